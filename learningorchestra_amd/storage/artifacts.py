@@ -1,0 +1,108 @@
+"""Artifact (binary) store — models and stage outputs.
+
+The reference persists model/stage binaries on shared Docker volumes keyed
+``/binaries/{service_type}/{filename}``, saving Keras models via
+``instance.save()`` and everything else with dill, and reading by trying dill
+first then keras (/root/reference/microservices/binary_executor_image/
+utils.py:195-233). The MI355X rebuild keeps the same path scheme under the
+single data root and the same save/load polymorphism, with torch-native
+modules saved as a ``state_dict`` + constructor spec (so checkpoints are
+portable and resumable) and arbitrary Python objects via dill.
+"""
+from __future__ import annotations
+
+import json
+import os
+from typing import Any, Optional
+
+import dill
+
+
+class ArtifactStore:
+    def __init__(self, root: Optional[str] = None):
+        if root is None:
+            from ..config import get_config
+            root = os.path.join(get_config().data_root, "binaries")
+        self._root = root
+        os.makedirs(root, exist_ok=True)
+
+    def _dir(self, service_type: str) -> str:
+        d = os.path.join(self._root, service_type.replace("/", "_"))
+        os.makedirs(d, exist_ok=True)
+        return d
+
+    def path(self, name: str, service_type: str) -> str:
+        return os.path.join(self._dir(service_type), name)
+
+    # -- save ---------------------------------------------------------------
+    def save(self, instance: Any, name: str, service_type: str) -> str:
+        """Persist ``instance``. Torch modules -> spec.json + state.pt;
+        everything else -> dill (mirrors utils.py:195-208 keras-then-dill)."""
+        base = self.path(name, service_type)
+        try:
+            import torch
+            if isinstance(instance, torch.nn.Module):
+                os.makedirs(base, exist_ok=True)
+                spec = getattr(instance, "lo_spec", None)
+                with open(os.path.join(base, "spec.json"), "w") as fh:
+                    json.dump({
+                        "format": "torch_module",
+                        "class_module": type(instance).__module__,
+                        "class_name": type(instance).__name__,
+                        "spec": spec,
+                    }, fh)
+                torch.save(instance.state_dict(), os.path.join(base, "state.pt"))
+                return base
+        except ImportError:
+            pass
+        with open(base + ".dill", "wb") as fh:
+            dill.dump(instance, fh)
+        return base + ".dill"
+
+    # -- load ---------------------------------------------------------------
+    def load(self, name: str, service_type: str) -> Any:
+        """Read back an artifact: dill first, then torch-module directory
+        (mirrors utils.py:210-221 dill-then-keras)."""
+        base = self.path(name, service_type)
+        if os.path.exists(base + ".dill"):
+            with open(base + ".dill", "rb") as fh:
+                return dill.load(fh)
+        if os.path.isdir(base) and os.path.exists(os.path.join(base, "spec.json")):
+            import importlib
+            import torch
+            with open(os.path.join(base, "spec.json")) as fh:
+                meta = json.load(fh)
+            module = importlib.import_module(meta["class_module"])
+            cls = getattr(module, meta["class_name"])
+            spec = meta.get("spec")
+            instance = cls(**spec) if isinstance(spec, dict) else cls()
+            state = torch.load(os.path.join(base, "state.pt"),
+                               map_location="cpu", weights_only=True)
+            instance.load_state_dict(state)
+            return instance
+        raise FileNotFoundError(f"no artifact '{name}' of type '{service_type}'")
+
+    def exists(self, name: str, service_type: str) -> bool:
+        base = self.path(name, service_type)
+        return os.path.exists(base + ".dill") or os.path.isdir(base)
+
+    def delete(self, name: str, service_type: str) -> None:
+        base = self.path(name, service_type)
+        if os.path.exists(base + ".dill"):
+            os.remove(base + ".dill")
+        elif os.path.isdir(base):
+            import shutil
+            shutil.rmtree(base)
+
+    # -- raw files (dataset/generic, database_api database.py:69-76) ---------
+    def save_raw(self, stream, name: str, service_type: str = "dataset/generic",
+                 chunk_size: int = 1 << 20) -> str:
+        path = self.path(name, service_type)
+        with open(path, "wb") as fh:
+            for chunk in stream:
+                if chunk:
+                    fh.write(chunk)
+        return path
+
+    def open_raw(self, name: str, service_type: str = "dataset/generic"):
+        return open(self.path(name, service_type), "rb")

@@ -1,0 +1,543 @@
+"""Embedded Mongo-compatible document store.
+
+The reference keeps *all* state — datasets as row-documents, metadata, results
+— in a 3-member MongoDB replica set (/root/reference/docker-compose.yml:42-90)
+accessed through a small per-service ``Database`` wrapper (e.g.
+/root/reference/microservices/database_api_image/utils.py:9-43). The MI355X
+rebuild is a single-node system, so the default backend is an embedded,
+thread-safe, optionally-persistent document store that speaks the same subset
+of the MongoDB API the reference uses:
+
+* ``insert_one`` / ``insert_many`` / ``find`` / ``find_one`` / ``update_one``
+  / ``delete_many`` / ``drop`` / ``estimated_document_count``
+* query operators ``$eq $ne $gt $gte $lt $lte $in $nin $exists $and $or $not``
+* ``sort`` / ``skip`` / ``limit`` cursors (utils.py:17-23 sorts by ``_id``)
+* ``aggregate`` with the ``$match`` / ``$group`` (+``$sum``/``$avg``/``$min``
+  /``$max``/``$count``) / ``$sort`` / ``$limit`` / ``$project`` stages — the
+  reference's histogram verb is a single ``$group`` pipeline
+  (/root/reference/microservices/histogram_image/histogram.py:31-32).
+
+If ``Config.mongo_uri`` is set and pymongo can reach a real server, `connect`
+returns a real pymongo database instead — the on-disk/metadata layout is the
+same either way (the "MongoDB-compatible" north star in BASELINE.json).
+
+Persistence: each collection snapshots to ``<root>/collections/<name>.jsonl``
+on ``flush()`` (and atexit). Loads are lazy at open.
+"""
+from __future__ import annotations
+
+import atexit
+import json
+import os
+import threading
+from typing import Any, Dict, Iterable, Iterator, List, Optional, Tuple
+
+_CMP_OPS = {"$eq", "$ne", "$gt", "$gte", "$lt", "$lte", "$in", "$nin", "$exists", "$not"}
+
+
+def _cmp(a: Any, b: Any, op: str) -> bool:
+    try:
+        if op == "$gt":
+            return a is not None and a > b
+        if op == "$gte":
+            return a is not None and a >= b
+        if op == "$lt":
+            return a is not None and a < b
+        if op == "$lte":
+            return a is not None and a <= b
+    except TypeError:
+        return False
+    raise ValueError(op)
+
+
+def _match_value(doc_value: Any, cond: Any, present: bool) -> bool:
+    if isinstance(cond, dict) and cond and all(k in _CMP_OPS for k in cond):
+        for op, operand in cond.items():
+            if op == "$eq":
+                ok = doc_value == operand
+            elif op == "$ne":
+                ok = doc_value != operand
+            elif op == "$in":
+                ok = doc_value in operand
+            elif op == "$nin":
+                ok = doc_value not in operand
+            elif op == "$exists":
+                ok = present == bool(operand)
+            elif op == "$not":
+                ok = not _match_value(doc_value, operand, present)
+            else:
+                ok = _cmp(doc_value, operand, op)
+            if not ok:
+                return False
+        return True
+    # plain equality (also matches element of an array field, like Mongo)
+    if isinstance(doc_value, list) and not isinstance(cond, list):
+        return cond in doc_value or doc_value == cond
+    return doc_value == cond
+
+
+def _get_path(doc: Dict[str, Any], path: str) -> Tuple[Any, bool]:
+    cur: Any = doc
+    for part in path.split("."):
+        if isinstance(cur, dict) and part in cur:
+            cur = cur[part]
+        else:
+            return None, False
+    return cur, True
+
+
+def match(doc: Dict[str, Any], query: Dict[str, Any]) -> bool:
+    """Evaluate a Mongo-style filter document against ``doc``."""
+    for key, cond in query.items():
+        if key == "$and":
+            if not all(match(doc, q) for q in cond):
+                return False
+        elif key == "$or":
+            if not any(match(doc, q) for q in cond):
+                return False
+        elif key == "$nor":
+            if any(match(doc, q) for q in cond):
+                return False
+        else:
+            value, present = _get_path(doc, key)
+            if not _match_value(value, cond, present):
+                return False
+    return True
+
+
+def _project(doc: Dict[str, Any], projection: Optional[Dict[str, int]]) -> Dict[str, Any]:
+    if not projection:
+        return dict(doc)
+    include = {k for k, v in projection.items() if v}
+    exclude = {k for k, v in projection.items() if not v}
+    if include:
+        out = {k: doc[k] for k in include if k in doc}
+        if "_id" in doc and projection.get("_id", 1):
+            out["_id"] = doc["_id"]
+        return out
+    return {k: v for k, v in doc.items() if k not in exclude}
+
+
+class Cursor:
+    """Chainable find() cursor: ``.sort(key, dir).skip(n).limit(n)``."""
+
+    def __init__(self, docs: List[Dict[str, Any]], projection: Optional[Dict[str, int]] = None):
+        self._docs = docs
+        self._projection = projection
+        self._skip = 0
+        self._limit: Optional[int] = None
+
+    def sort(self, key_or_list, direction: int = 1) -> "Cursor":
+        if isinstance(key_or_list, str):
+            keys = [(key_or_list, direction)]
+        else:
+            keys = list(key_or_list)
+        for key, dirn in reversed(keys):
+            self._docs.sort(key=lambda d, k=key: _sort_key(_get_path(d, k)[0]),
+                            reverse=(dirn < 0))
+        return self
+
+    def skip(self, n: int) -> "Cursor":
+        self._skip = n
+        return self
+
+    def limit(self, n: int) -> "Cursor":
+        self._limit = n
+        return self
+
+    def __iter__(self) -> Iterator[Dict[str, Any]]:
+        docs = self._docs[self._skip:]
+        if self._limit is not None:
+            docs = docs[: self._limit]
+        for d in docs:
+            yield _project(d, self._projection)
+
+    def __next__(self):  # pragma: no cover - iterator protocol convenience
+        return next(iter(self))
+
+    def count(self) -> int:
+        return len(list(iter(self)))
+
+
+def _sort_key(v: Any):
+    # total order across mixed types (None < numbers < strings < other)
+    if v is None:
+        return (0, 0)
+    if isinstance(v, bool):
+        return (1, int(v))
+    if isinstance(v, (int, float)):
+        return (1, v)
+    if isinstance(v, str):
+        return (2, v)
+    return (3, str(v))
+
+
+class InsertOneResult:
+    def __init__(self, inserted_id):
+        self.inserted_id = inserted_id
+        self.acknowledged = True
+
+
+class InsertManyResult:
+    def __init__(self, inserted_ids):
+        self.inserted_ids = inserted_ids
+        self.acknowledged = True
+
+
+class UpdateResult:
+    def __init__(self, matched: int, modified: int):
+        self.matched_count = matched
+        self.modified_count = modified
+        self.acknowledged = True
+
+
+class DeleteResult:
+    def __init__(self, deleted: int):
+        self.deleted_count = deleted
+        self.acknowledged = True
+
+
+class Collection:
+    def __init__(self, store: "DocumentStore", name: str):
+        self._store = store
+        self.name = name
+        self._docs: Dict[Any, Dict[str, Any]] = {}
+        self._order: List[Any] = []  # insertion order of _ids
+        self._lock = threading.RLock()
+        self._auto_id = 0
+        self._dirty = False
+
+    # -- write ------------------------------------------------------------
+    def insert_one(self, doc: Dict[str, Any]) -> InsertOneResult:
+        with self._lock:
+            doc = dict(doc)
+            if "_id" not in doc:
+                doc["_id"] = self._next_id()
+            if doc["_id"] in self._docs:
+                raise DuplicateKeyError(f"duplicate _id {doc['_id']} in {self.name}")
+            self._docs[doc["_id"]] = doc
+            self._order.append(doc["_id"])
+            self._dirty = True
+            return InsertOneResult(doc["_id"])
+
+    def insert_many(self, docs: Iterable[Dict[str, Any]]) -> InsertManyResult:
+        ids = []
+        with self._lock:
+            for doc in docs:
+                ids.append(self.insert_one(doc).inserted_id)
+        return InsertManyResult(ids)
+
+    def _next_id(self):
+        while self._auto_id in self._docs:
+            self._auto_id += 1
+        nid = self._auto_id
+        self._auto_id += 1
+        return nid
+
+    def update_one(self, flt: Dict[str, Any], update: Dict[str, Any],
+                   upsert: bool = False) -> UpdateResult:
+        with self._lock:
+            for _id in self._order:
+                doc = self._docs[_id]
+                if match(doc, flt):
+                    self._apply_update(doc, update)
+                    self._dirty = True
+                    return UpdateResult(1, 1)
+            if upsert:
+                base = {k: v for k, v in flt.items() if not k.startswith("$")
+                        and not isinstance(v, dict)}
+                self._apply_update(base, update)
+                self.insert_one(base)
+                return UpdateResult(0, 0)
+            return UpdateResult(0, 0)
+
+    def update_many(self, flt: Dict[str, Any], update: Dict[str, Any]) -> UpdateResult:
+        n = 0
+        with self._lock:
+            for _id in self._order:
+                doc = self._docs[_id]
+                if match(doc, flt):
+                    self._apply_update(doc, update)
+                    n += 1
+            if n:
+                self._dirty = True
+        return UpdateResult(n, n)
+
+    def replace_one(self, flt: Dict[str, Any], doc: Dict[str, Any],
+                    upsert: bool = False) -> UpdateResult:
+        with self._lock:
+            for _id in self._order:
+                old = self._docs[_id]
+                if match(old, flt):
+                    new = dict(doc)
+                    new["_id"] = _id
+                    self._docs[_id] = new
+                    self._dirty = True
+                    return UpdateResult(1, 1)
+            if upsert:
+                self.insert_one(dict(doc))
+                return UpdateResult(0, 0)
+            return UpdateResult(0, 0)
+
+    @staticmethod
+    def _apply_update(doc: Dict[str, Any], update: Dict[str, Any]) -> None:
+        for op, fields in update.items():
+            if op == "$set":
+                for k, v in fields.items():
+                    doc[k] = v
+            elif op == "$unset":
+                for k in fields:
+                    doc.pop(k, None)
+            elif op == "$inc":
+                for k, v in fields.items():
+                    doc[k] = doc.get(k, 0) + v
+            elif op == "$push":
+                for k, v in fields.items():
+                    doc.setdefault(k, []).append(v)
+            else:
+                raise ValueError(f"unsupported update operator {op}")
+
+    def delete_many(self, flt: Dict[str, Any]) -> DeleteResult:
+        with self._lock:
+            to_del = [i for i in self._order if match(self._docs[i], flt)]
+            for i in to_del:
+                del self._docs[i]
+            if to_del:
+                self._order = [i for i in self._order if i in self._docs]
+                self._dirty = True
+            return DeleteResult(len(to_del))
+
+    def delete_one(self, flt: Dict[str, Any]) -> DeleteResult:
+        with self._lock:
+            for i in self._order:
+                if match(self._docs[i], flt):
+                    del self._docs[i]
+                    self._order.remove(i)
+                    self._dirty = True
+                    return DeleteResult(1)
+            return DeleteResult(0)
+
+    def drop(self) -> None:
+        with self._lock:
+            self._docs.clear()
+            self._order.clear()
+            self._auto_id = 0
+            self._dirty = True
+        self._store._drop_collection(self.name)
+
+    # -- read -------------------------------------------------------------
+    def find(self, flt: Optional[Dict[str, Any]] = None,
+             projection: Optional[Dict[str, int]] = None) -> Cursor:
+        flt = flt or {}
+        with self._lock:
+            docs = [self._docs[i] for i in self._order if match(self._docs[i], flt)]
+        return Cursor(docs, projection)
+
+    def find_one(self, flt: Optional[Dict[str, Any]] = None,
+                 projection: Optional[Dict[str, int]] = None) -> Optional[Dict[str, Any]]:
+        for d in self.find(flt, projection).limit(1):
+            return d
+        return None
+
+    def count_documents(self, flt: Dict[str, Any]) -> int:
+        return self.find(flt).count()
+
+    def estimated_document_count(self) -> int:
+        with self._lock:
+            return len(self._docs)
+
+    # -- aggregate ---------------------------------------------------------
+    def aggregate(self, pipeline: List[Dict[str, Any]]) -> List[Dict[str, Any]]:
+        with self._lock:
+            docs = [dict(self._docs[i]) for i in self._order]
+        for stage in pipeline:
+            (op, spec), = stage.items()
+            if op == "$match":
+                docs = [d for d in docs if match(d, spec)]
+            elif op == "$group":
+                docs = _group(docs, spec)
+            elif op == "$sort":
+                for key, dirn in reversed(list(spec.items())):
+                    docs.sort(key=lambda d, k=key: _sort_key(_get_path(d, k)[0]),
+                              reverse=(dirn < 0))
+            elif op == "$limit":
+                docs = docs[:spec]
+            elif op == "$skip":
+                docs = docs[spec:]
+            elif op == "$project":
+                docs = [_project(d, spec) for d in docs]
+            elif op == "$count":
+                docs = [{spec: len(docs)}]
+            else:
+                raise ValueError(f"unsupported aggregate stage {op}")
+        return docs
+
+    # -- persistence --------------------------------------------------------
+    def _load_jsonl(self, path: str) -> None:
+        with self._lock, open(path, "r", encoding="utf-8") as fh:
+            for line in fh:
+                line = line.strip()
+                if not line:
+                    continue
+                doc = json.loads(line)
+                self._docs[doc["_id"]] = doc
+                self._order.append(doc["_id"])
+            self._dirty = False
+
+    def _save_jsonl(self, path: str) -> None:
+        with self._lock:
+            if not self._dirty:
+                return
+            tmp = path + ".tmp"
+            with open(tmp, "w", encoding="utf-8") as fh:
+                for i in self._order:
+                    fh.write(json.dumps(self._docs[i], default=_json_default) + "\n")
+            os.replace(tmp, path)
+            self._dirty = False
+
+
+def _json_default(o):
+    try:
+        import numpy as np
+        if isinstance(o, np.integer):
+            return int(o)
+        if isinstance(o, np.floating):
+            return float(o)
+        if isinstance(o, np.ndarray):
+            return o.tolist()
+    except Exception:
+        pass
+    return str(o)
+
+
+def _resolve_expr(doc: Dict[str, Any], expr: Any) -> Any:
+    if isinstance(expr, str) and expr.startswith("$"):
+        return _get_path(doc, expr[1:])[0]
+    return expr
+
+
+def _group(docs: List[Dict[str, Any]], spec: Dict[str, Any]) -> List[Dict[str, Any]]:
+    key_expr = spec["_id"]
+    accs = {k: v for k, v in spec.items() if k != "_id"}
+    groups: Dict[Any, Dict[str, Any]] = {}
+    order: List[Any] = []
+    for d in docs:
+        key = _resolve_expr(d, key_expr)
+        hkey = json.dumps(key, default=_json_default) if isinstance(key, (dict, list)) else key
+        if hkey not in groups:
+            groups[hkey] = {"_id": key}
+            for name, acc in accs.items():
+                (aop, _), = acc.items()
+                groups[hkey][name] = 0 if aop in ("$sum", "$count") else None
+                if aop == "$avg":
+                    groups[hkey]["__cnt_" + name] = 0
+            order.append(hkey)
+        g = groups[hkey]
+        for name, acc in accs.items():
+            (aop, aexpr), = acc.items()
+            val = _resolve_expr(d, aexpr)
+            if aop == "$sum":
+                g[name] += val if isinstance(val, (int, float)) and not isinstance(val, bool) else 0
+            elif aop == "$min":
+                g[name] = val if g[name] is None else min(g[name], val)
+            elif aop == "$max":
+                g[name] = val if g[name] is None else max(g[name], val)
+            elif aop == "$avg":
+                if isinstance(val, (int, float)):
+                    g[name] = (g[name] or 0) + val
+                    g["__cnt_" + name] += 1
+            elif aop == "$first":
+                if g[name] is None:
+                    g[name] = val
+            elif aop == "$push":
+                if g[name] is None or g[name] == 0:
+                    g[name] = []
+                g[name].append(val)
+            else:
+                raise ValueError(f"unsupported accumulator {aop}")
+    out = []
+    for hkey in order:
+        g = groups[hkey]
+        for name in list(g):
+            if name.startswith("__cnt_"):
+                tgt = name[len("__cnt_"):]
+                if g[name]:
+                    g[tgt] = g[tgt] / g[name]
+                del g[name]
+        out.append(g)
+    return out
+
+
+class DuplicateKeyError(Exception):
+    pass
+
+
+class DocumentStore:
+    """A database of named collections (the pymongo ``Database`` analog)."""
+
+    def __init__(self, root: Optional[str] = None):
+        self._root = root
+        self._collections: Dict[str, Collection] = {}
+        self._lock = threading.RLock()
+        if root:
+            os.makedirs(os.path.join(root, "collections"), exist_ok=True)
+            for fn in os.listdir(os.path.join(root, "collections")):
+                if fn.endswith(".jsonl"):
+                    name = fn[: -len(".jsonl")]
+                    col = Collection(self, name)
+                    col._load_jsonl(os.path.join(root, "collections", fn))
+                    self._collections[name] = col
+            atexit.register(self.flush)
+
+    def __getitem__(self, name: str) -> Collection:
+        with self._lock:
+            if name not in self._collections:
+                self._collections[name] = Collection(self, name)
+            return self._collections[name]
+
+    def __getattr__(self, name: str) -> Collection:
+        if name.startswith("_"):
+            raise AttributeError(name)
+        return self[name]
+
+    def list_collection_names(self) -> List[str]:
+        with self._lock:
+            return [n for n, c in self._collections.items()
+                    if c.estimated_document_count() > 0]
+
+    def drop_collection(self, name: str) -> None:
+        with self._lock:
+            if name in self._collections:
+                self._collections[name].drop()
+
+    def _drop_collection(self, name: str) -> None:
+        with self._lock:
+            self._collections.pop(name, None)
+            if self._root:
+                path = os.path.join(self._root, "collections", name + ".jsonl")
+                if os.path.exists(path):
+                    os.remove(path)
+
+    def flush(self) -> None:
+        if not self._root:
+            return
+        with self._lock:
+            cols = list(self._collections.items())
+        for name, col in cols:
+            col._save_jsonl(os.path.join(self._root, "collections", name + ".jsonl"))
+
+
+def connect(cfg=None) -> Any:
+    """Open the document database: real MongoDB if configured, embedded otherwise."""
+    from ..config import get_config
+    cfg = cfg or get_config()
+    if cfg.mongo_uri:
+        try:
+            import pymongo
+            client = pymongo.MongoClient(cfg.mongo_uri, serverSelectionTimeoutMS=2000)
+            client.admin.command("ping")
+            return client[cfg.database_name]
+        except Exception:
+            pass  # fall through to embedded store
+    return DocumentStore(os.path.join(cfg.data_root, cfg.database_name)
+                         if cfg.data_root else None)

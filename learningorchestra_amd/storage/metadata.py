@@ -1,0 +1,152 @@
+"""Metadata / catalog layer — the §2.2 protocol implemented once.
+
+The reference triplicates a ``Metadata`` class into every microservice
+(e.g. /root/reference/microservices/database_api_image/utils.py:46-77,
+binary_executor_image/utils.py:66-135). The contract it implements:
+
+* every dataset/artifact gets a collection whose document ``_id == 0`` is the
+  metadata document, created with ``finished: false`` *before* the async work
+  starts;
+* ``update_finished_flag`` flips it when the work (or its exception) lands;
+* executors append numbered *execution documents* (``_id = max+1``) recording
+  ``description``, parameters, and the ``exception`` (null on success);
+* lineage: every derived artifact records ``parentName``; the module/class of
+  the owning model is found by walking the parent chain until a ``model/*``
+  type (binary_executor_image/utils.py:257-276).
+"""
+from __future__ import annotations
+
+import time
+from typing import Any, Dict, List, Optional
+
+METADATA_ROW_ID = 0
+
+
+def _now() -> str:
+    return time.strftime("%Y-%m-%dT%H:%M:%S", time.localtime())
+
+
+class Metadata:
+    def __init__(self, database):
+        self._db = database
+
+    # -- creation -----------------------------------------------------------
+    def create_file(self, name: str, service_type: str, **extra) -> Dict[str, Any]:
+        """Create the `_id: 0` metadata document with finished=False.
+
+        Mirrors database_api_image/utils.py:50-63 (datasets carry url/fields)
+        and binary_executor_image/utils.py:79-97 (executions carry parentName,
+        module/class info) — extras land in the doc verbatim.
+        """
+        doc = {
+            "_id": METADATA_ROW_ID,
+            "datasetName": name,
+            "type": service_type,
+            "timeCreated": _now(),
+            "finished": False,
+            **extra,
+        }
+        col = self._db[name]
+        if col.find_one({"_id": METADATA_ROW_ID}) is not None:
+            col.replace_one({"_id": METADATA_ROW_ID}, doc)
+        else:
+            col.insert_one(doc)
+        return doc
+
+    def update_finished_flag(self, name: str, finished: bool = True,
+                             exception: Optional[str] = None) -> None:
+        update: Dict[str, Any] = {"finished": finished}
+        if exception is not None:
+            update["exception"] = exception
+        self._db[name].update_one({"_id": METADATA_ROW_ID}, {"$set": update})
+
+    def update_fields(self, name: str, **fields) -> None:
+        self._db[name].update_one({"_id": METADATA_ROW_ID}, {"$set": fields})
+
+    def update_file_headers(self, name: str, fields: List[str]) -> None:
+        self.update_fields(name, fields=fields)
+
+    # -- execution documents -------------------------------------------------
+    def create_execution_document(self, name: str, description: str,
+                                  parameters: Optional[Dict[str, Any]] = None,
+                                  exception: Optional[str] = None,
+                                  **extra) -> int:
+        """Append a versioned execution document at the next ``_id``
+        (binary_executor_image/utils.py:112-135 uses max(_id)+1)."""
+        col = self._db[name]
+        max_id = METADATA_ROW_ID
+        for doc in col.find({}, {"_id": 1}):
+            if isinstance(doc["_id"], int) and doc["_id"] > max_id:
+                max_id = doc["_id"]
+        doc = {
+            "_id": max_id + 1,
+            "description": description,
+            "executionParameters": parameters or {},
+            "exception": exception,
+            "timeCreated": _now(),
+            **extra,
+        }
+        col.insert_one(doc)
+        return doc["_id"]
+
+    # -- reads ---------------------------------------------------------------
+    def get_metadata(self, name: str) -> Optional[Dict[str, Any]]:
+        return self._db[name].find_one({"_id": METADATA_ROW_ID})
+
+    def is_finished(self, name: str) -> bool:
+        doc = self.get_metadata(name)
+        return bool(doc and doc.get("finished"))
+
+    def exists(self, name: str) -> bool:
+        return self.get_metadata(name) is not None
+
+    def get_type(self, name: str) -> Optional[str]:
+        doc = self.get_metadata(name)
+        return doc.get("type") if doc else None
+
+    def catalog(self, service_type: Optional[str] = None) -> List[Dict[str, Any]]:
+        """List all metadata documents, optionally filtered by type prefix
+        (the reference's ``read_files_descriptor``, database_api server.py:84-93)."""
+        out = []
+        for name in self._db.list_collection_names():
+            doc = self._db[name].find_one({"_id": METADATA_ROW_ID})
+            if doc is None:
+                continue
+            if service_type is None or doc.get("type") == service_type:
+                out.append(doc)
+        return out
+
+    # -- lineage --------------------------------------------------------------
+    def walk_to_model(self, name: str, max_depth: int = 64) -> Optional[Dict[str, Any]]:
+        """Walk the ``parentName`` chain until a ``model/*``-typed document
+        (binary_executor_image/utils.py:257-276) and return its metadata."""
+        seen = set()
+        cur = name
+        for _ in range(max_depth):
+            if cur in seen:
+                return None
+            seen.add(cur)
+            doc = self.get_metadata(cur)
+            if doc is None:
+                return None
+            if str(doc.get("type", "")).startswith("model/"):
+                return doc
+            parent = doc.get("parentName")
+            if not parent:
+                return None
+            cur = parent
+        return None
+
+    def lineage(self, name: str, max_depth: int = 64) -> List[Dict[str, Any]]:
+        """Full provenance chain root-last."""
+        chain = []
+        cur: Optional[str] = name
+        seen = set()
+        while cur and cur not in seen and len(chain) < max_depth:
+            seen.add(cur)
+            doc = self.get_metadata(cur)
+            if doc is None:
+                break
+            chain.append(doc)
+            cur = doc.get("parentName")
+        return chain
