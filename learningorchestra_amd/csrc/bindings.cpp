@@ -1,0 +1,265 @@
+// PyTorch bindings for the learningorchestra_amd gfx950 kernel library.
+// Thin argument-checking wrappers; all compute is in the .hip translation
+// units. Built in-tree (build_ext.py) as learningorchestra_amd/_lo_C.so.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <hip/hip_runtime.h>
+
+namespace lo {
+struct GemmArgs {
+  const void *A, *B;
+  void* C;
+  const float* bias;
+  long lda, ldb, ldc;
+  int M, N, K;
+  bool ta, tb;
+  int epi;
+  bool out_f32;
+  int splits;
+};
+bool gemm_dispatch(const GemmArgs& g, hipStream_t s);
+void launch_mfma_probe(const void* A, const void* B, float* D, hipStream_t s);
+void launch_im2col(const void* in, void* col, int B, int H, int W, int C,
+                   int KH, int KW, int SH, int SW, int PH, int PW,
+                   int OH, int OW, int Kpad, hipStream_t s);
+void launch_col2im(const void* dcol, void* dx, int B, int H, int W, int C,
+                   int KH, int KW, int PH, int PW, int OH, int OW, int Kpad,
+                   hipStream_t s);
+void launch_maxpool_fwd(const void* in, void* out, void* idx, int B, int H, int W,
+                        int C, int KH, int KW, int SH, int SW, int OH, int OW,
+                        hipStream_t s);
+void launch_maxpool_bwd(const void* dy, const void* idx, void* dx, int B, int H,
+                        int W, int C, int KH, int KW, int SH, int SW, int OH,
+                        int OW, hipStream_t s);
+void launch_relu_bwd(const void* dy, const void* y, void* dx, long n, hipStream_t s);
+void launch_sgd(void* master, const void* grad, void* mom, void* mirror, long n,
+                float lr, float mu, float wd, float gscale, hipStream_t s);
+void launch_adam(void* master, const void* grad, void* m1, void* m2, void* mirror,
+                 long n, float lr, float b1, float b2, float eps, float wd,
+                 float c1, float c2, float gscale, hipStream_t s);
+void launch_colsum(const void* dy, void* out, long M, int N, long ldy, hipStream_t s);
+void launch_argmax_rows(const void* x, void* out, long M, int C, int Cvalid,
+                        long ldx, hipStream_t s);
+void launch_accuracy_count(const void* pred, const void* label, void* out, long M,
+                           hipStream_t s);
+void launch_softmax_ce(const void* logits, const void* labels, void* dlogits,
+                       void* loss_sum, void* correct, long M, int C, int Cvalid,
+                       float gscale, hipStream_t s);
+}  // namespace lo
+
+namespace {
+
+hipStream_t stream() { return at::hip::getCurrentHIPStream().stream(); }
+
+void check_bf16(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+void check_f32(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == at::kFloat, name, " must be fp32");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// gemm: C = op_ta(A) @ op_tb(B) (+bias)(+relu). Returns false if no native
+// config covers the shape (python falls back to a library GEMM).
+bool gemm(at::Tensor A, at::Tensor B, at::Tensor C,
+          c10::optional<at::Tensor> bias, bool ta, bool tb, int64_t epi,
+          int64_t splits) {
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  TORCH_CHECK(C.is_cuda() && C.is_contiguous(), "C must be contiguous GPU");
+  const bool out_f32 = C.scalar_type() == at::kFloat;
+  TORCH_CHECK(out_f32 || C.scalar_type() == at::kBFloat16, "C must be bf16 or fp32");
+  const int M = (int)C.size(0), N = (int)C.size(1);
+  const int K = (int)(ta ? A.size(0) : A.size(1));
+  TORCH_CHECK((int)(ta ? A.size(1) : A.size(0)) == M, "A/M mismatch");
+  TORCH_CHECK((int)(tb ? B.size(1) : B.size(0)) == K, "B/K mismatch");
+  TORCH_CHECK((int)(tb ? B.size(0) : B.size(1)) == N, "B/N mismatch");
+  const float* bias_p = nullptr;
+  if (bias.has_value()) {
+    check_f32(*bias, "bias");
+    TORCH_CHECK(bias->numel() == N, "bias/N mismatch");
+    bias_p = bias->data_ptr<float>();
+  }
+  lo::GemmArgs g{A.data_ptr(), B.data_ptr(), C.data_ptr(), bias_p,
+                 A.size(1), B.size(1), C.size(1), M, N, K, ta, tb,
+                 (int)epi, out_f32, (int)splits};
+  return lo::gemm_dispatch(g, stream());
+}
+
+at::Tensor mfma_probe(at::Tensor A, at::Tensor B) {
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  TORCH_CHECK(A.size(0) == 16 && A.size(1) == 32 && B.size(0) == 32 && B.size(1) == 16);
+  auto D = at::zeros({16, 16}, A.options().dtype(at::kFloat));
+  lo::launch_mfma_probe(A.data_ptr(), B.data_ptr(), D.data_ptr<float>(), stream());
+  return D;
+}
+
+at::Tensor im2col(at::Tensor in, int64_t KH, int64_t KW, int64_t SH, int64_t SW,
+                  int64_t PH, int64_t PW, int64_t Kpad, at::Tensor col) {
+  check_bf16(in, "in");
+  check_bf16(col, "col");
+  const int B = (int)in.size(0), H = (int)in.size(1), W = (int)in.size(2),
+            C = (int)in.size(3);
+  const int OH = (H + 2 * (int)PH - (int)KH) / (int)SH + 1;
+  const int OW = (W + 2 * (int)PW - (int)KW) / (int)SW + 1;
+  TORCH_CHECK(col.size(0) == (long)B * OH * OW && col.size(1) == Kpad, "col shape");
+  TORCH_CHECK(Kpad >= KH * KW * C && Kpad % 8 == 0, "Kpad");
+  lo::launch_im2col(in.data_ptr(), col.data_ptr(), B, H, W, C, (int)KH, (int)KW,
+                    (int)SH, (int)SW, (int)PH, (int)PW, OH, OW, (int)Kpad, stream());
+  return col;
+}
+
+at::Tensor col2im(at::Tensor dcol, int64_t B, int64_t H, int64_t W, int64_t C,
+                  int64_t KH, int64_t KW, int64_t PH, int64_t PW, at::Tensor dx) {
+  check_bf16(dcol, "dcol");
+  check_bf16(dx, "dx");
+  const int OH = (int)(H + 2 * PH - KH) + 1, OW = (int)(W + 2 * PW - KW) + 1;
+  TORCH_CHECK(dcol.size(0) == B * OH * OW, "dcol rows");
+  lo::launch_col2im(dcol.data_ptr(), dx.data_ptr(), (int)B, (int)H, (int)W, (int)C,
+                    (int)KH, (int)KW, (int)PH, (int)PW, OH, OW,
+                    (int)dcol.size(1), stream());
+  return dx;
+}
+
+std::vector<at::Tensor> maxpool_fwd(at::Tensor in, int64_t KH, int64_t KW,
+                                    int64_t SH, int64_t SW) {
+  check_bf16(in, "in");
+  const int B = (int)in.size(0), H = (int)in.size(1), W = (int)in.size(2),
+            C = (int)in.size(3);
+  const int OH = (H - (int)KH) / (int)SH + 1, OW = (W - (int)KW) / (int)SW + 1;
+  auto out = at::empty({B, OH, OW, C}, in.options());
+  auto idx = at::empty({B, OH, OW, C}, in.options().dtype(at::kByte));
+  lo::launch_maxpool_fwd(in.data_ptr(), out.data_ptr(), idx.data_ptr(), B, H, W, C,
+                         (int)KH, (int)KW, (int)SH, (int)SW, OH, OW, stream());
+  return {out, idx};
+}
+
+at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor idx, int64_t H, int64_t W,
+                       int64_t KH, int64_t KW, int64_t SH, int64_t SW,
+                       at::Tensor dx) {
+  check_bf16(dy, "dy");
+  check_bf16(dx, "dx");
+  const int B = (int)dy.size(0), OH = (int)dy.size(1), OW = (int)dy.size(2),
+            C = (int)dy.size(3);
+  lo::launch_maxpool_bwd(dy.data_ptr(), idx.data_ptr(), dx.data_ptr(), B, (int)H,
+                         (int)W, C, (int)KH, (int)KW, (int)SH, (int)SW, OH, OW,
+                         stream());
+  return dx;
+}
+
+at::Tensor relu_bwd(at::Tensor dy, at::Tensor y, at::Tensor dx) {
+  check_bf16(dy, "dy");
+  check_bf16(y, "y");
+  check_bf16(dx, "dx");
+  TORCH_CHECK(dy.numel() == y.numel() && dy.numel() == dx.numel());
+  TORCH_CHECK(dy.numel() % 8 == 0, "relu_bwd needs 8-aligned numel");
+  lo::launch_relu_bwd(dy.data_ptr(), y.data_ptr(), dx.data_ptr(), dy.numel(),
+                      stream());
+  return dx;
+}
+
+void sgd_step(at::Tensor master, at::Tensor grad, at::Tensor mom, at::Tensor mirror,
+              double lr, double mu, double wd, double gscale) {
+  check_f32(master, "master");
+  check_f32(grad, "grad");
+  check_f32(mom, "mom");
+  check_bf16(mirror, "mirror");
+  TORCH_CHECK(master.numel() == grad.numel() && master.numel() == mom.numel() &&
+              master.numel() == mirror.numel());
+  TORCH_CHECK(master.numel() % 4 == 0, "arena must be 4-aligned");
+  lo::launch_sgd(master.data_ptr(), grad.data_ptr(), mom.data_ptr(),
+                 mirror.data_ptr(), master.numel(), (float)lr, (float)mu,
+                 (float)wd, (float)gscale, stream());
+}
+
+void adam_step(at::Tensor master, at::Tensor grad, at::Tensor m1, at::Tensor m2,
+               at::Tensor mirror, double lr, double b1, double b2, double eps,
+               double wd, double c1, double c2, double gscale) {
+  check_f32(master, "master");
+  check_f32(grad, "grad");
+  check_f32(m1, "m1");
+  check_f32(m2, "m2");
+  check_bf16(mirror, "mirror");
+  TORCH_CHECK(master.numel() % 4 == 0, "arena must be 4-aligned");
+  lo::launch_adam(master.data_ptr(), grad.data_ptr(), m1.data_ptr(), m2.data_ptr(),
+                  mirror.data_ptr(), master.numel(), (float)lr, (float)b1,
+                  (float)b2, (float)eps, (float)wd, (float)c1, (float)c2,
+                  (float)gscale, stream());
+}
+
+at::Tensor colsum(at::Tensor dy, at::Tensor out) {
+  check_bf16(dy, "dy");
+  check_f32(out, "out");
+  TORCH_CHECK(dy.dim() == 2 && out.numel() == dy.size(1));
+  lo::launch_colsum(dy.data_ptr(), out.data_ptr(), dy.size(0), (int)dy.size(1),
+                    dy.size(1), stream());
+  return out;
+}
+
+at::Tensor argmax_rows(at::Tensor x, int64_t cvalid) {
+  check_bf16(x, "x");
+  TORCH_CHECK(x.dim() == 2);
+  auto out = at::empty({x.size(0)}, x.options().dtype(at::kInt));
+  lo::launch_argmax_rows(x.data_ptr(), out.data_ptr(), x.size(0), (int)x.size(1),
+                         (int)cvalid, x.size(1), stream());
+  return out;
+}
+
+at::Tensor accuracy_count(at::Tensor pred, at::Tensor label) {
+  TORCH_CHECK(pred.scalar_type() == at::kInt && label.scalar_type() == at::kLong);
+  auto out = at::zeros({1}, pred.options());
+  lo::launch_accuracy_count(pred.data_ptr(), label.data_ptr(), out.data_ptr(),
+                            pred.numel(), stream());
+  return out;
+}
+
+// fused softmax-CE: writes dlogits; returns nothing (loss_sum/correct are
+// caller-provided accumulators so one graph-captured step reuses them)
+void softmax_ce(at::Tensor logits, at::Tensor labels, at::Tensor dlogits,
+                c10::optional<at::Tensor> loss_sum, c10::optional<at::Tensor> correct,
+                int64_t cvalid, double gscale) {
+  check_bf16(logits, "logits");
+  check_bf16(dlogits, "dlogits");
+  TORCH_CHECK(labels.scalar_type() == at::kLong && labels.is_cuda());
+  TORCH_CHECK(logits.dim() == 2 && logits.sizes() == dlogits.sizes());
+  TORCH_CHECK(cvalid <= logits.size(1));
+  TORCH_CHECK(cvalid <= 32 || logits.size(1) % 8 == 0,
+              "wave path needs 8-aligned C");
+  void* lp = nullptr;
+  void* cp = nullptr;
+  if (loss_sum.has_value()) { check_f32(*loss_sum, "loss_sum"); lp = loss_sum->data_ptr(); }
+  if (correct.has_value()) { cp = correct->data_ptr(); }
+  lo::launch_softmax_ce(logits.data_ptr(), labels.data_ptr(), dlogits.data_ptr(),
+                        lp, cp, logits.size(0), (int)logits.size(1), (int)cvalid,
+                        (float)gscale, stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gemm", &gemm, "bf16 MFMA GEMM (gfx950)",
+        py::arg("A"), py::arg("B"), py::arg("C"), py::arg("bias") = py::none(),
+        py::arg("ta") = false, py::arg("tb") = false, py::arg("epi") = 0,
+        py::arg("splits") = 1);
+  m.def("mfma_probe", &mfma_probe);
+  m.def("im2col", &im2col);
+  m.def("col2im", &col2im);
+  m.def("maxpool_fwd", &maxpool_fwd);
+  m.def("maxpool_bwd", &maxpool_bwd);
+  m.def("relu_bwd", &relu_bwd);
+  m.def("sgd_step", &sgd_step);
+  m.def("adam_step", &adam_step);
+  m.def("colsum", &colsum);
+  m.def("argmax_rows", &argmax_rows);
+  m.def("accuracy_count", &accuracy_count);
+  m.def("softmax_ce", &softmax_ce,
+        py::arg("logits"), py::arg("labels"), py::arg("dlogits"),
+        py::arg("loss_sum") = py::none(), py::arg("correct") = py::none(),
+        py::arg("cvalid") = 0, py::arg("gscale") = 1.0);
+}

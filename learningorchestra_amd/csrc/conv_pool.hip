@@ -1,0 +1,272 @@
+// Conv2d support kernels for gfx950 — NHWC layout throughout.
+//
+// The CNN path (SURVEY.md §2.9 "Conv2d fwd/bwd -> implicit-GEMM conv") is
+// im2col + the MFMA GEMM of gemm.hip: NHWC makes every im2col row
+// k-contiguous (kh,kw,c) and the GEMM output [B*OH*OW, outC] directly the
+// next layer's NHWC activation — no transposes anywhere in the hot loop.
+//
+// * im2col:  in[B,H,W,C] -> col[B*OH*OW, Kpad] (Kpad >= KH*KW*C, %8==0, the
+//            pad tail is pre-zeroed by the allocator and never written)
+// * col2im:  gather-style dcol -> dX (no atomics; each input element sums its
+//            covering patches)
+// * maxpool2d fwd/bwd with u8 argmax indices (non-overlap needs no atomics)
+
+#include "lo_common.h"
+
+namespace lo {
+
+// --------------------------------------------------------------- im2col ----
+template <bool VEC8>
+__global__ void im2col_kernel(const bf16* __restrict__ in, bf16* __restrict__ col,
+                              int B, int H, int W, int C, int KH, int KW,
+                              int SH, int SW, int PH, int PW,
+                              int OH, int OW, int Kpad) {
+  const int CV = VEC8 ? C / 8 : C;            // channel units per position
+  const long total = (long)B * OH * OW * KH * KW * CV;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int cu = i % CV;
+    long r = i / CV;
+    const int kw = r % KW; r /= KW;
+    const int kh = r % KH; r /= KH;
+    const int ow = r % OW; r /= OW;
+    const int oh = r % OH; r /= OH;
+    const int b = r;
+    const int h = oh * SH - PH + kh, w = ow * SW - PW + kw;
+    const long row = ((long)b * OH + oh) * OW + ow;
+    const long kidx = ((long)kh * KW + kw) * C + cu * (VEC8 ? 8 : 1);
+    bf16* dst = col + row * Kpad + kidx;
+    const bool inside = (h >= 0 && h < H && w >= 0 && w < W);
+    if (VEC8) {
+      bf16x8 v = {};
+      if (inside)
+        v = *(const bf16x8*)(in + (((long)b * H + h) * W + w) * C + cu * 8);
+      *(bf16x8*)dst = v;
+    } else {
+      bf16 v = bf16(0.f);
+      if (inside) v = in[(((long)b * H + h) * W + w) * C + cu];
+      *dst = v;
+    }
+  }
+}
+
+void launch_im2col(const void* in, void* col, int B, int H, int W, int C,
+                   int KH, int KW, int SH, int SW, int PH, int PW,
+                   int OH, int OW, int Kpad, hipStream_t s) {
+  const bool vec = (C % 8 == 0);
+  const long total = (long)B * OH * OW * KH * KW * (vec ? C / 8 : C);
+  const int block = 256;
+  const int grid = (int)min((total + block - 1) / block, (long)2048);
+  if (vec)
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(im2col_kernel<true>), dim3(grid), dim3(block), 0, s,
+                       (const bf16*)in, (bf16*)col, B, H, W, C, KH, KW, SH, SW,
+                       PH, PW, OH, OW, Kpad);
+  else
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(im2col_kernel<false>), dim3(grid), dim3(block), 0, s,
+                       (const bf16*)in, (bf16*)col, B, H, W, C, KH, KW, SH, SW,
+                       PH, PW, OH, OW, Kpad);
+}
+
+// --------------------------------------------------------------- col2im ----
+// stride-1 gather: dX[b,h,w,c] = sum over (kh,kw) with oh=h+PH-kh in [0,OH),
+// ow=w+PW-kw in [0,OW) of dcol[row(b,oh,ow)][(kh*KW+kw)*C + c]
+template <bool VEC8>
+__global__ void col2im_kernel(const bf16* __restrict__ dcol, bf16* __restrict__ dx,
+                              int B, int H, int W, int C, int KH, int KW,
+                              int PH, int PW, int OH, int OW, int Kpad) {
+  const int CV = VEC8 ? C / 8 : C;
+  const long total = (long)B * H * W * CV;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int cu = i % CV;
+    long r = i / CV;
+    const int w = r % W; r /= W;
+    const int h = r % H; r /= H;
+    const int b = r;
+    float acc[VEC8 ? 8 : 1] = {};
+    for (int kh = 0; kh < KH; ++kh) {
+      const int oh = h + PH - kh;
+      if (oh < 0 || oh >= OH) continue;
+      for (int kw = 0; kw < KW; ++kw) {
+        const int ow = w + PW - kw;
+        if (ow < 0 || ow >= OW) continue;
+        const long row = ((long)b * OH + oh) * OW + ow;
+        const long kidx = ((long)kh * KW + kw) * C + cu * (VEC8 ? 8 : 1);
+        if (VEC8) {
+          bf16x8 v = *(const bf16x8*)(dcol + row * Kpad + kidx);
+          #pragma unroll
+          for (int j = 0; j < 8; ++j) acc[j] += tofloat(v[j]);
+        } else {
+          acc[0] += tofloat(dcol[row * Kpad + kidx]);
+        }
+      }
+    }
+    if (VEC8) {
+      bf16x8 o;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) o[j] = tobf16(acc[j]);
+      *(bf16x8*)(dx + (((long)b * H + h) * W + w) * C + cu * 8) = o;
+    } else {
+      dx[(((long)b * H + h) * W + w) * C + cu] = tobf16(acc[0]);
+    }
+  }
+}
+
+void launch_col2im(const void* dcol, void* dx, int B, int H, int W, int C,
+                   int KH, int KW, int PH, int PW, int OH, int OW, int Kpad,
+                   hipStream_t s) {
+  const bool vec = (C % 8 == 0);
+  const long total = (long)B * H * W * (vec ? C / 8 : C);
+  const int block = 256;
+  const int grid = (int)min((total + block - 1) / block, (long)2048);
+  if (vec)
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(col2im_kernel<true>), dim3(grid), dim3(block), 0, s,
+                       (const bf16*)dcol, (bf16*)dx, B, H, W, C, KH, KW, PH, PW, OH, OW, Kpad);
+  else
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(col2im_kernel<false>), dim3(grid), dim3(block), 0, s,
+                       (const bf16*)dcol, (bf16*)dx, B, H, W, C, KH, KW, PH, PW, OH, OW, Kpad);
+}
+
+// ------------------------------------------------------------- maxpool -----
+template <bool VEC8>
+__global__ void maxpool_fwd_kernel(const bf16* __restrict__ in, bf16* __restrict__ out,
+                                   unsigned char* __restrict__ idx,
+                                   int B, int H, int W, int C, int KH, int KW,
+                                   int SH, int SW, int OH, int OW) {
+  const int CV = VEC8 ? C / 8 : C;
+  const long total = (long)B * OH * OW * CV;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int cu = i % CV;
+    long r = i / CV;
+    const int ow = r % OW; r /= OW;
+    const int oh = r % OH; r /= OH;
+    const int b = r;
+    const int NE = VEC8 ? 8 : 1;
+    float best[NE];
+    unsigned char bidx[NE];
+    #pragma unroll
+    for (int j = 0; j < NE; ++j) { best[j] = -3.0e38f; bidx[j] = 0; }
+    for (int kh = 0; kh < KH; ++kh) {
+      const int h = oh * SH + kh;
+      if (h >= H) continue;
+      for (int kw = 0; kw < KW; ++kw) {
+        const int w = ow * SW + kw;
+        if (w >= W) continue;
+        const long base = (((long)b * H + h) * W + w) * C + cu * NE;
+        if (VEC8) {
+          bf16x8 v = *(const bf16x8*)(in + base);
+          #pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const float f = tofloat(v[j]);
+            if (f > best[j]) { best[j] = f; bidx[j] = (unsigned char)(kh * KW + kw); }
+          }
+        } else {
+          const float f = tofloat(in[base]);
+          if (f > best[0]) { best[0] = f; bidx[0] = (unsigned char)(kh * KW + kw); }
+        }
+      }
+    }
+    const long obase = (((long)b * OH + oh) * OW + ow) * C + cu * NE;
+    if (VEC8) {
+      bf16x8 o;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) o[j] = tobf16(best[j]);
+      *(bf16x8*)(out + obase) = o;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) idx[obase + j] = bidx[j];
+    } else {
+      out[obase] = tobf16(best[0]);
+      idx[obase] = bidx[0];
+    }
+  }
+}
+
+// backward: per input element, sum dY of covering outputs whose argmax is it
+// (with SH==KH/SW==KW pooling each input has <=1 cover -> no atomics needed)
+template <bool VEC8>
+__global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
+                                   const unsigned char* __restrict__ idx,
+                                   bf16* __restrict__ dx,
+                                   int B, int H, int W, int C, int KH, int KW,
+                                   int SH, int SW, int OH, int OW) {
+  const int CV = VEC8 ? C / 8 : C;
+  const long total = (long)B * H * W * CV;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int cu = i % CV;
+    long r = i / CV;
+    const int w = r % W; r /= W;
+    const int h = r % H; r /= H;
+    const int b = r;
+    const int NE = VEC8 ? 8 : 1;
+    float acc[NE];
+    #pragma unroll
+    for (int j = 0; j < NE; ++j) acc[j] = 0.f;
+    const int oh_lo = max(0, (h - KH + SH) / SH), oh_hi = min(OH - 1, h / SH);
+    const int ow_lo = max(0, (w - KW + SW) / SW), ow_hi = min(OW - 1, w / SW);
+    for (int oh = oh_lo; oh <= oh_hi; ++oh) {
+      const int kh = h - oh * SH;
+      if (kh < 0 || kh >= KH) continue;
+      for (int ow = ow_lo; ow <= ow_hi; ++ow) {
+        const int kw = w - ow * SW;
+        if (kw < 0 || kw >= KW) continue;
+        const long obase = (((long)b * OH + oh) * OW + ow) * C + cu * NE;
+        const unsigned char want = (unsigned char)(kh * KW + kw);
+        if (VEC8) {
+          bf16x8 g = *(const bf16x8*)(dy + obase);
+          #pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (idx[obase + j] == want) acc[j] += tofloat(g[j]);
+        } else {
+          if (idx[obase] == want) acc[0] += tofloat(dy[obase]);
+        }
+      }
+    }
+    const long ibase = (((long)b * H + h) * W + w) * C + cu * NE;
+    if (VEC8) {
+      bf16x8 o;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) o[j] = tobf16(acc[j]);
+      *(bf16x8*)(dx + ibase) = o;
+    } else {
+      dx[ibase] = tobf16(acc[0]);
+    }
+  }
+}
+
+void launch_maxpool_fwd(const void* in, void* out, void* idx, int B, int H, int W,
+                        int C, int KH, int KW, int SH, int SW, int OH, int OW,
+                        hipStream_t s) {
+  const bool vec = (C % 8 == 0);
+  const long total = (long)B * OH * OW * (vec ? C / 8 : C);
+  const int block = 256;
+  const int grid = (int)min((total + block - 1) / block, (long)2048);
+  if (vec)
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(maxpool_fwd_kernel<true>), dim3(grid), dim3(block), 0, s,
+                       (const bf16*)in, (bf16*)out, (unsigned char*)idx,
+                       B, H, W, C, KH, KW, SH, SW, OH, OW);
+  else
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(maxpool_fwd_kernel<false>), dim3(grid), dim3(block), 0, s,
+                       (const bf16*)in, (bf16*)out, (unsigned char*)idx,
+                       B, H, W, C, KH, KW, SH, SW, OH, OW);
+}
+
+void launch_maxpool_bwd(const void* dy, const void* idx, void* dx, int B, int H,
+                        int W, int C, int KH, int KW, int SH, int SW, int OH,
+                        int OW, hipStream_t s) {
+  const bool vec = (C % 8 == 0);
+  const long total = (long)B * H * W * (vec ? C / 8 : C);
+  const int block = 256;
+  const int grid = (int)min((total + block - 1) / block, (long)2048);
+  if (vec)
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(maxpool_bwd_kernel<true>), dim3(grid), dim3(block), 0, s,
+                       (const bf16*)dy, (const unsigned char*)idx, (bf16*)dx,
+                       B, H, W, C, KH, KW, SH, SW, OH, OW);
+  else
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(maxpool_bwd_kernel<false>), dim3(grid), dim3(block), 0, s,
+                       (const bf16*)dy, (const unsigned char*)idx, (bf16*)dx,
+                       B, H, W, C, KH, KW, SH, SW, OH, OW);
+}
+
+}  // namespace lo

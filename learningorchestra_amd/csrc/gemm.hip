@@ -1,0 +1,275 @@
+// MFMA bf16 GEMM for gfx950 (MI355X) — the framework's core compute kernel.
+//
+// Computes C[M,N] = A' @ B' (+bias, +ReLU), fp32 accumulate, where
+//   A' = A  ([M,K] row-major)        if !TA, else A^T (A is [K,M] row-major)
+//   B' = B  ([K,N] row-major)        if !TB, else B^T (B is [N,K] row-major)
+// The (TA=true) path + fp32 atomic output is the split-K weight-gradient GEMM.
+//
+// Replaces what the reference delegated to Spark MLlib / TF black boxes
+// ("Dense GEMM fwd/bwd" -> bf16 MFMA GEMM). Design per the
+// CDNA4 guide §5: v_mfma_f32_16x16x32_bf16 tiles, LDS staging with the
+// bank-conflict XOR swizzle (guide §6 G4), XCD-aware block remap (T1).
+// Layouts are normalized in LDS to As[BM][BK] / Bs[BN][BK] (k-contiguous
+// rows) so every MFMA fragment read is one swizzled ds_read_b128.
+
+#include "lo_common.h"
+
+namespace lo {
+
+// mfma_f32_16x16x32_bf16 operand maps (verified by tests/test_gpu_ops.py's
+// probe): lane l holds A[row=l%16][k=(l/16)*8 + j], j=0..7 (contiguous k),
+// B[k][col] mirrored; C/D: col=lane&15, row=(lane>>4)*4+reg (guide §3).
+
+template <int BM, int BN, int BK, int WM, int WN, bool TA, bool TB,
+          int EPI, bool OUT_F32, bool ATOMIC>
+__global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
+    const bf16* __restrict__ A, long lda, const bf16* __restrict__ B, long ldb,
+    void* __restrict__ Cv, long ldc, const float* __restrict__ bias,
+    int M, int N, int K, int kStart, int kChunk) {
+  constexpr int T = WM * WN * 64;
+  constexpr int WTM = BM / WM;          // wave tile rows
+  constexpr int WTN = BN / WN;          // wave tile cols
+  constexpr int MFRAG = WTM / 16;
+  constexpr int NFRAG = WTN / 16;
+  constexpr int BKB = BK * 2;           // LDS row bytes
+  constexpr int SWZ = (BKB >= 128) ? 7 : (BKB >= 64 ? 3 : 1);
+  static_assert(BM % (WM * 16) == 0 && BN % (WN * 16) == 0, "wave tiling");
+  static_assert(BK % 32 == 0, "BK multiple of MFMA K");
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* smA = smem;                     // BM*BKB bytes
+  char* smB = smem + BM * BKB;          // BN*BKB bytes
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid / WN, wc = wid % WN;
+
+  const int nb = (N + BN - 1) / BN;
+  const int mb = (M + BM - 1) / BM;
+  int bid = xcd_swizzle(blockIdx.x, mb * nb);
+  const int bm = bid % mb, bn = bid / mb;   // consecutive blocks share B panel
+  const int m0 = bm * BM, n0 = bn * BN;
+
+  const int kBegin = kStart + blockIdx.z * kChunk;
+  const int kEnd = min(K, kBegin + kChunk);
+
+  f32x4 acc[MFRAG][NFRAG] = {};
+
+  const int fr = lane & 15;
+  const int fkb = (lane >> 4) * 8;      // fragment k base within MFMA step
+
+  for (int k0 = kBegin; k0 < kEnd; k0 += BK) {
+    // ---- stage A tile -> As[BM][BK] ------------------------------------
+    if (!TA) {
+      constexpr int CH = BM * BK / 8;
+      for (int c = tid; c < CH; c += T) {
+        const int row = c / (BK / 8), kc = c % (BK / 8);
+        const int gm = m0 + row, gk = k0 + kc * 8;
+        bf16x8 v = {};
+        if (gm < M && gk < K) v = *(const bf16x8*)(A + (long)gm * lda + gk);
+        *(bf16x8*)(smA + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = v;
+      }
+    } else {
+      // A is [K,M]: read 8 contiguous m at one k (coalesced), scatter to LDS
+      constexpr int CH = BK * BM / 8;
+      for (int c = tid; c < CH; c += T) {
+        const int krow = c / (BM / 8), mc = c % (BM / 8);
+        const int gk = k0 + krow, gm0 = m0 + mc * 8;
+        bf16x8 v = {};
+        if (gk < K) {
+          if (gm0 + 8 <= M) v = *(const bf16x8*)(A + (long)gk * lda + gm0);
+          else if (gm0 < M)
+            for (int j = 0; j < 8 && gm0 + j < M; ++j) v[j] = A[(long)gk * lda + gm0 + j];
+        }
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int row = mc * 8 + j;
+          *(bf16*)(smA + row * BKB + ((krow * 2) ^ ((row & SWZ) << 4))) = v[j];
+        }
+      }
+    }
+    // ---- stage B tile -> Bs[BN][BK] ------------------------------------
+    if (TB) {
+      constexpr int CH = BN * BK / 8;
+      for (int c = tid; c < CH; c += T) {
+        const int row = c / (BK / 8), kc = c % (BK / 8);
+        const int gn = n0 + row, gk = k0 + kc * 8;
+        bf16x8 v = {};
+        if (gn < N && gk < K) v = *(const bf16x8*)(B + (long)gn * ldb + gk);
+        *(bf16x8*)(smB + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = v;
+      }
+    } else {
+      constexpr int CH = BK * BN / 8;
+      for (int c = tid; c < CH; c += T) {
+        const int krow = c / (BN / 8), nc = c % (BN / 8);
+        const int gk = k0 + krow, gn0 = n0 + nc * 8;
+        bf16x8 v = {};
+        if (gk < K) {
+          if (gn0 + 8 <= N) v = *(const bf16x8*)(B + (long)gk * ldb + gn0);
+          else if (gn0 < N)
+            for (int j = 0; j < 8 && gn0 + j < N; ++j) v[j] = B[(long)gk * ldb + gn0 + j];
+        }
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int row = nc * 8 + j;
+          *(bf16*)(smB + row * BKB + ((krow * 2) ^ ((row & SWZ) << 4))) = v[j];
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- MFMA over the tile --------------------------------------------
+    #pragma unroll
+    for (int kk = 0; kk < BK / 32; ++kk) {
+      bf16x8 af[MFRAG], bf[NFRAG];
+      #pragma unroll
+      for (int mi = 0; mi < MFRAG; ++mi) {
+        const int row = wr * WTM + mi * 16 + fr;
+        af[mi] = *(const bf16x8*)(smA + row * BKB +
+                                  (((kk * 32 + fkb) * 2) ^ ((row & SWZ) << 4)));
+      }
+      #pragma unroll
+      for (int ni = 0; ni < NFRAG; ++ni) {
+        const int row = wc * WTN + ni * 16 + fr;
+        bf[ni] = *(const bf16x8*)(smB + row * BKB +
+                                  (((kk * 32 + fkb) * 2) ^ ((row & SWZ) << 4)));
+      }
+      #pragma unroll
+      for (int mi = 0; mi < MFRAG; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < NFRAG; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: bias / ReLU / store (bf16 | f32 | f32-atomic) -----------
+  float* Cf = (float*)Cv;
+  bf16* Cb = (bf16*)Cv;
+  #pragma unroll
+  for (int ni = 0; ni < NFRAG; ++ni) {
+    const int col = n0 + wc * WTN + ni * 16 + (lane & 15);
+    if (col >= N) continue;
+    const float bv = bias ? bias[col] : 0.f;
+    #pragma unroll
+    for (int mi = 0; mi < MFRAG; ++mi) {
+      const int row0 = m0 + wr * WTM + mi * 16 + (lane >> 4) * 4;
+      #pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int row = row0 + j;
+        if (row >= M) continue;
+        float v = acc[mi][ni][j] + bv;
+        if (EPI == 1) v = fmaxf(v, 0.f);
+        if (ATOMIC) atomicAdd(Cf + (long)row * ldc + col, v);
+        else if (OUT_F32) Cf[(long)row * ldc + col] = v;
+        else Cb[(long)row * ldc + col] = tobf16(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// MFMA operand-layout probe: computes D = A[16,32] @ B[32,16] with the
+// contiguous-k fragment map so the GPU test can verify the map empirically
+// (guide §3: "Always A=I-check with ASYMMETRIC B").
+__global__ void mfma_probe_kernel(const bf16* A, const bf16* B, float* D) {
+  const int lane = threadIdx.x & 63;
+  bf16x8 a, b;
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a[j] = A[(lane & 15) * 32 + (lane >> 4) * 8 + j];
+    b[j] = B[((lane >> 4) * 8 + j) * 16 + (lane & 15)];
+  }
+  f32x4 c = {};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+  #pragma unroll
+  for (int j = 0; j < 4; ++j)
+    D[((lane >> 4) * 4 + j) * 16 + (lane & 15)] = c[j];
+}
+
+// ---------------------------------------------------------------------------
+// host-side dispatch
+
+struct GemmArgs {
+  const void *A, *B;
+  void* C;
+  const float* bias;
+  long lda, ldb, ldc;
+  int M, N, K;
+  bool ta, tb;    // operand transposes (see file header)
+  int epi;        // 0 none, 1 relu
+  bool out_f32;
+  int splits;     // >1 => atomic split-K accumulate into fp32 C (zeroed by caller)
+};
+
+template <int BM, int BN, int BK, int WM, int WN, bool TA, bool TB,
+          int EPI, bool OUT_F32, bool ATOMIC>
+static void launch_cfg(const GemmArgs& g, hipStream_t s) {
+  const int mb = cdiv(g.M, BM), nb = cdiv(g.N, BN);
+  int kChunk;
+  if (g.splits > 1) {
+    const int kb = cdiv(g.K, BK);
+    kChunk = cdiv(kb, g.splits) * BK;
+  } else {
+    kChunk = cdiv(g.K, BK) * BK;
+  }
+  const int zs = g.splits > 1 ? cdiv(g.K, kChunk) : 1;
+  dim3 grid(mb * nb, 1, zs), block(WM * WN * 64);
+  size_t lds = (size_t)(BM + BN) * BK * 2;
+  hipLaunchKernelGGL(HIP_KERNEL_NAME(
+      gemm_kernel<BM, BN, BK, WM, WN, TA, TB, EPI, OUT_F32, ATOMIC>),
+      grid, block, lds, s,
+      (const bf16*)g.A, g.lda, (const bf16*)g.B, g.ldb, g.C, g.ldc, g.bias,
+      g.M, g.N, g.K, 0, kChunk);
+}
+
+// Per-(TA,TB) tile-size selection. Instantiates only the combos the engine
+// uses (fwd = N,T; dX = N,N; dW = T,N split-K) plus (T,T) for completeness.
+template <bool TA, bool TB>
+static bool dispatch_tiles(const GemmArgs& g, hipStream_t s) {
+  #define LO_EPI_CASES(BM_, BN_, WM_, WN_)                                     \
+    do {                                                                       \
+      if (g.out_f32) {                                                         \
+        if (g.epi == 1) launch_cfg<BM_, BN_, 64, WM_, WN_, TA, TB, 1, true, false>(g, s);  \
+        else            launch_cfg<BM_, BN_, 64, WM_, WN_, TA, TB, 0, true, false>(g, s);  \
+      } else {                                                                 \
+        if (g.epi == 1) launch_cfg<BM_, BN_, 64, WM_, WN_, TA, TB, 1, false, false>(g, s); \
+        else            launch_cfg<BM_, BN_, 64, WM_, WN_, TA, TB, 0, false, false>(g, s); \
+      }                                                                        \
+      return true;                                                             \
+    } while (0)
+
+  if (g.N > 64 && g.M >= 4096) LO_EPI_CASES(128, 128, 2, 2);
+  if (g.N > 64) LO_EPI_CASES(64, 128, 2, 2);     // small-M wide-N (dX tails)
+  if (g.N > 32) LO_EPI_CASES(128, 64, 2, 2);
+  if (g.N > 16) LO_EPI_CASES(256, 32, 4, 1);
+  LO_EPI_CASES(128, 16, 4, 1);
+  #undef LO_EPI_CASES
+}
+
+// Returns true if a native config covered the shape; false => caller falls
+// back to a library GEMM (rocBLAS via torch) for the cold path.
+bool gemm_dispatch(const GemmArgs& g, hipStream_t s) {
+  if (g.K % 8 != 0) return false;
+  if (g.splits > 1) {
+    // split-K atomic accumulate (weight gradients): dW = A^T @ B
+    if (!g.out_f32 || g.epi != 0) return false;
+    if (!g.ta && g.tb) launch_cfg<32, 64, 64, 1, 4, false, true, 0, true, true>(g, s);
+    else if (g.ta && !g.tb) launch_cfg<32, 64, 64, 1, 4, true, false, 0, true, true>(g, s);
+    else return false;
+    return true;
+  }
+  if (!g.ta && g.tb) return dispatch_tiles<false, true>(g, s);
+  if (!g.ta && !g.tb) return dispatch_tiles<false, false>(g, s);
+  if (g.ta && !g.tb) return dispatch_tiles<true, false>(g, s);
+  return dispatch_tiles<true, true>(g, s);
+}
+
+void launch_mfma_probe(const void* A, const void* B, float* D, hipStream_t s) {
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, s,
+                     (const bf16*)A, (const bf16*)B, D);
+}
+
+}  // namespace lo

@@ -1,0 +1,253 @@
+"""Explicit-backward NHWC layers over the kernel library.
+
+This is the in-process training engine that replaces the reference's
+delegation to TF/keras ``fit`` (SURVEY §3.3 hot loop): a fixed layer graph
+with hand-managed forward/backward, all activations in persistent
+preallocated buffers (hipGraph-capture friendly — zero allocations in the
+steady-state step), parameters/grads in the flat ParamArena.
+
+Gradient-scale convention: the loss head folds 1/(global_batch) into dlogits
+(softmax_ce gscale), so every weight grad lands already-averaged and the DDP
+all-reduce is a plain SUM.
+"""
+from __future__ import annotations
+
+import math
+from typing import List, Optional, Tuple
+
+import torch
+
+from ..ops import functional as F
+from .arena import ParamArena
+
+
+def _pad8(n: int) -> int:
+    return (n + 7) // 8 * 8
+
+
+class Layer:
+    def build(self, arena: ParamArena) -> None:  # register params
+        pass
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        raise NotImplementedError
+
+    def backward(self, dy: torch.Tensor) -> Optional[torch.Tensor]:
+        raise NotImplementedError
+
+
+class Conv2dNHWC(Layer):
+    """conv = im2col + MFMA GEMM (+fused bias/ReLU epilogue)."""
+
+    def __init__(self, name: str, in_c: int, out_c: int, kh: int, kw: int,
+                 stride: int = 1, pad: int = 0, relu: bool = True,
+                 first: bool = False):
+        self.name = name
+        self.in_c, self.out_c = in_c, out_c
+        self.kh, self.kw, self.stride, self.pad = kh, kw, stride, pad
+        self.relu = relu
+        self.first = first  # input layer: skip dX
+        self.kdim = kh * kw * in_c
+        self.kpad = _pad8(self.kdim)
+        self.arena: Optional[ParamArena] = None
+        self._bufs = {}
+
+    def build(self, arena: ParamArena) -> None:
+        self.arena = arena
+        std = math.sqrt(2.0 / self.kdim)  # he-init over true fan-in
+        kdim, kpad, out_c = self.kdim, self.kpad, self.out_c
+
+        def init_w(shape):
+            g = torch.Generator(device="cpu").manual_seed(
+                abs(hash(self.name)) % (2 ** 31))
+            t = torch.zeros(shape, dtype=torch.float32)
+            t[:, :kdim] = torch.randn((out_c, kdim), generator=g) * std
+            return t
+
+        arena.add(self.name + ".w", (self.out_c, self.kpad), init_w)
+        arena.add(self.name + ".b", (self.out_c,), torch.zeros(self.out_c))
+
+    def _alloc(self, B: int, H: int, W: int, dev, dtype):
+        OH = (H + 2 * self.pad - self.kh) // self.stride + 1
+        OW = (W + 2 * self.pad - self.kw) // self.stride + 1
+        key = (B, H, W)
+        if self._bufs.get("key") != key:
+            M = B * OH * OW
+            self._bufs = {
+                "key": key, "B": B, "H": H, "W": W, "OH": OH, "OW": OW,
+                # col pad tail stays zero forever (im2col never writes it)
+                "col": torch.zeros((M, self.kpad), device=dev, dtype=dtype),
+                "y": torch.empty((M, self.out_c), device=dev, dtype=dtype),
+            }
+            if not self.first:
+                self._bufs["dcol"] = torch.empty((M, self.kpad), device=dev, dtype=dtype)
+                self._bufs["dx"] = torch.empty((B, H, W, self.in_c), device=dev, dtype=dtype)
+        return self._bufs
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        B, H, W, C = x.shape
+        assert C == self.in_c
+        bufs = self._alloc(B, H, W, x.device, x.dtype)
+        F.im2col(x, self.kh, self.kw, self.stride, self.stride, self.pad,
+                 self.pad, self.kpad, out=bufs["col"])
+        F.gemm(bufs["col"], self.arena.p(self.name + ".w"), tb=True,
+               bias=self.arena.pf(self.name + ".b"), relu=self.relu,
+               out=bufs["y"])
+        return bufs["y"].view(B, bufs["OH"], bufs["OW"], self.out_c)
+
+    def backward(self, dy: torch.Tensor) -> Optional[torch.Tensor]:
+        bufs = self._bufs
+        M = bufs["col"].shape[0]
+        dy2 = dy.reshape(M, self.out_c)
+        if self.relu:
+            F.relu_bwd(dy2, bufs["y"], out=dy2)
+        # weight grad: dW[outC, kpad] = dY^T @ col, split-K when M is deep
+        gw = self.arena.g(self.name + ".w")
+        splits = _splitk_heuristic(self.out_c, self.kpad, M)
+        F.gemm(dy2, bufs["col"], ta=True, out=gw, splits=splits)
+        F.colsum(dy2, out=self.arena.g(self.name + ".b"))
+        if self.first:
+            return None
+        assert self.stride == 1, "col2im gather path is stride-1"
+        F.gemm(dy2, self.arena.p(self.name + ".w"), out=bufs["dcol"])
+        F.col2im(bufs["dcol"], bufs["B"], bufs["H"], bufs["W"], self.in_c,
+                 self.kh, self.kw, self.pad, self.pad, out=bufs["dx"])
+        return bufs["dx"]
+
+
+class MaxPool2dNHWC(Layer):
+    def __init__(self, k: int = 2, stride: Optional[int] = None):
+        self.k = k
+        self.stride = stride or k
+        self._bufs = {}
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        self._bufs["shape"] = x.shape
+        out, idx = F.maxpool2d(x, self.k, self.k, self.stride, self.stride)
+        self._bufs["idx"] = idx
+        if "dx" not in self._bufs or self._bufs["dx"].shape != x.shape:
+            self._bufs["dx"] = torch.empty_like(x)
+        return out
+
+    def backward(self, dy: torch.Tensor) -> torch.Tensor:
+        B, H, W, C = self._bufs["shape"]
+        return F.maxpool2d_bwd(dy, self._bufs["idx"], H, W, self.k, self.k,
+                               self.stride, self.stride, out=self._bufs["dx"])
+
+
+class Flatten(Layer):
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        self._shape = x.shape
+        return x.reshape(x.shape[0], -1)
+
+    def backward(self, dy: torch.Tensor) -> torch.Tensor:
+        return dy.reshape(self._shape)
+
+
+class Linear(Layer):
+    def __init__(self, name: str, in_f: int, out_f: int, relu: bool = False):
+        self.name = name
+        self.in_f, self.out_f = in_f, out_f
+        self.relu = relu
+        self.arena: Optional[ParamArena] = None
+        self._bufs = {}
+
+    def build(self, arena: ParamArena) -> None:
+        self.arena = arena
+        std = math.sqrt(2.0 / self.in_f)
+        g = torch.Generator(device="cpu").manual_seed(abs(hash(self.name)) % (2 ** 31))
+        arena.add(self.name + ".w", (self.out_f, self.in_f),
+                  torch.randn((self.out_f, self.in_f), generator=g) * std)
+        arena.add(self.name + ".b", (self.out_f,), torch.zeros(self.out_f))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        M = x.shape[0]
+        if self._bufs.get("M") != M:
+            self._bufs = {"M": M,
+                          "y": torch.empty((M, self.out_f), device=x.device, dtype=x.dtype),
+                          "dx": torch.empty((M, self.in_f), device=x.device, dtype=x.dtype)}
+        self._bufs["x"] = x
+        F.gemm(x, self.arena.p(self.name + ".w"), tb=True,
+               bias=self.arena.pf(self.name + ".b"), relu=self.relu,
+               out=self._bufs["y"])
+        return self._bufs["y"]
+
+    def backward(self, dy: torch.Tensor) -> torch.Tensor:
+        bufs = self._bufs
+        if self.relu:
+            F.relu_bwd(dy, bufs["y"], out=dy)
+        gw = self.arena.g(self.name + ".w")
+        splits = _splitk_heuristic(self.out_f, self.in_f, bufs["M"])
+        F.gemm(dy, bufs["x"], ta=True, out=gw, splits=splits)
+        F.colsum(dy, out=self.arena.g(self.name + ".b"))
+        F.gemm(dy, self.arena.p(self.name + ".w"), out=bufs["dx"])
+        return bufs["dx"]
+
+
+def _splitk_heuristic(m: int, n: int, k: int) -> int:
+    """Split-K factor for the dW GEMM C[m,n] with reduction depth k: fill the
+    256-CU chip (~512 blocks of 32x64 tiles) without shredding k."""
+    base_blocks = ((m + 31) // 32) * ((n + 63) // 64)
+    if base_blocks >= 256 or k < 4096:
+        return 1
+    return max(1, min(512 // base_blocks, k // 1024))
+
+
+class SequentialClassifier:
+    """A fixed feed-forward classifier on the explicit-backward engine, with
+    the padded-class softmax-CE head (padded logits masked in the kernel)."""
+
+    def __init__(self, layers: List[Layer], num_classes: int, device="cpu",
+                 seed: int = 0):
+        self.layers = layers
+        self.num_classes = num_classes
+        self.arena = ParamArena(device)
+        for lay in layers:
+            lay.build(self.arena)
+        self.arena.finalize(seed)
+        dev = torch.device(device)
+        self.loss_sum = torch.zeros(1, dtype=torch.float32, device=dev)
+        self.correct = torch.zeros(1, dtype=torch.int32, device=dev)
+        self._dlogits = None
+
+    @property
+    def cpad(self) -> int:
+        # classifier head output width (padded to 8/16 for aligned GEMMs)
+        return self.layers[-1].out_f
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        for lay in self.layers:
+            x = lay.forward(x)
+        return x
+
+    def train_step(self, x: torch.Tensor, y: torch.Tensor,
+                   gscale: Optional[float] = None) -> Tuple[torch.Tensor, torch.Tensor]:
+        """One fused fwd+bwd. Returns (loss_sum, correct) device tensors;
+        optimizer step is separate (Trainer composes allreduce between)."""
+        if gscale is None:
+            gscale = 1.0 / x.shape[0]
+        logits = self.forward(x)
+        if self._dlogits is None or self._dlogits.shape != logits.shape:
+            self._dlogits = torch.empty_like(logits)
+        self.loss_sum.zero_()
+        self.correct.zero_()
+        F.softmax_ce(logits, y, self._dlogits, self.loss_sum, self.correct,
+                     cvalid=self.num_classes, gscale=gscale)
+        dy = self._dlogits
+        for lay in reversed(self.layers):
+            dy = lay.backward(dy)
+            if dy is None:
+                break
+        return self.loss_sum, self.correct
+
+    @torch.no_grad()
+    def predict(self, x: torch.Tensor) -> torch.Tensor:
+        logits = self.forward(x)
+        return F.argmax_rows(logits, self.num_classes)
+
+    # -- checkpoint/resume ------------------------------------
+    def state_dict(self):
+        return self.arena.state_dict()
+
+    def load_state_dict(self, sd):
+        self.arena.load_state_dict(sd)

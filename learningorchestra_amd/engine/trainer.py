@@ -1,0 +1,94 @@
+"""Training driver: composes model step + RCCL all-reduce + fused optimizer,
+with optional hipGraph capture of the whole step (launch-bound small models
+like the MNIST CNN are exactly the "capture launch-bound inner loops in
+hipGraphs" case)."""
+from __future__ import annotations
+
+import time
+from typing import Callable, Optional, Tuple
+
+import torch
+
+from ..parallel import all_reduce_grads, get_world_size, is_distributed
+from .arena import SGD, Adam
+from .layers import SequentialClassifier
+
+
+class Trainer:
+    def __init__(self, model: SequentialClassifier, optimizer, device="cpu",
+                 use_graph: bool = False):
+        self.model = model
+        self.opt = optimizer
+        self.device = torch.device(device)
+        self.use_graph = use_graph and self.device.type == "cuda"
+        self._graph = None
+        self._static_x = None
+        self._static_y = None
+
+    # full step on given tensors (eager)
+    def _step_body(self, x: torch.Tensor, y: torch.Tensor) -> None:
+        world = get_world_size()
+        gscale = 1.0 / (x.shape[0] * world)
+        self.model.train_step(x, y, gscale=gscale)
+        all_reduce_grads(self.model.arena.grad)
+        self.opt.step()
+
+    def step(self, x: torch.Tensor, y: torch.Tensor) -> Tuple[float, float]:
+        """One training step. Returns (mean loss, accuracy) — NOTE: these
+        device->host reads sync; use step_async inside timed loops."""
+        self.step_async(x, y)
+        b = x.shape[0]
+        return (self.model.loss_sum.item() / b,
+                self.model.correct.item() / b)
+
+    def step_async(self, x: torch.Tensor, y: torch.Tensor) -> None:
+        if self.use_graph:
+            if self._graph is None:
+                self._capture(x, y)
+            else:
+                self._static_x.copy_(x, non_blocking=True)
+                self._static_y.copy_(y, non_blocking=True)
+                self._graph.replay()
+            return
+        self._step_body(x, y)
+
+    def _capture(self, x: torch.Tensor, y: torch.Tensor) -> None:
+        # warm up eagerly (allocates every persistent buffer), then capture
+        self._static_x = x.clone()
+        self._static_y = y.clone()
+        for _ in range(3):
+            self._step_body(self._static_x, self._static_y)
+        torch.cuda.synchronize()
+        self._graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self._graph):
+            self._step_body(self._static_x, self._static_y)
+        torch.cuda.synchronize()
+
+    def train(self, data_iter, steps: int, log_every: int = 0,
+              log_fn: Optional[Callable[[str], None]] = None) -> dict:
+        t0 = time.perf_counter()
+        samples = 0
+        last_loss, last_acc = float("nan"), float("nan")
+        for i in range(steps):
+            x, y = next(data_iter)
+            if log_every and (i + 1) % log_every == 0:
+                last_loss, last_acc = self.step(x, y)
+                if log_fn:
+                    log_fn(f"step {i+1}/{steps} loss={last_loss:.4f} acc={last_acc:.3f}")
+            else:
+                self.step_async(x, y)
+            samples += x.shape[0]
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        return {"steps": steps, "seconds": dt, "samples": samples,
+                "samples_per_sec": samples / dt if dt > 0 else float("inf"),
+                "loss": last_loss, "accuracy": last_acc}
+
+
+def make_sgd(model: SequentialClassifier, lr=0.05, momentum=0.9, wd=0.0) -> SGD:
+    return SGD(model.arena, lr=lr, momentum=momentum, weight_decay=wd)
+
+
+def make_adam(model: SequentialClassifier, lr=1e-3) -> Adam:
+    return Adam(model.arena, lr=lr)

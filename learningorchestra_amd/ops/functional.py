@@ -1,0 +1,259 @@
+"""Op-level functional API: HIP kernels on GPU, plain-PyTorch fp32 reference
+on CPU.
+
+Dispatch rule (the build contract): a CUDA/HIP tensor MUST run the in-tree
+gfx950 kernel — ``require_ext()`` raises rather than silently falling back to
+eager PyTorch; CPU tensors run the reference implementation (which is also
+what GPU numerics tests compare against). Shapes the native GEMM doesn't
+cover fall back to torch.matmul (= rocBLAS, a plain library GEMM) — only the
+cold path ever takes that branch.
+
+Activation layout is NHWC everywhere (SURVEY §2.9 + conv_pool.hip header).
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from ._ext import has_ext, require_ext
+
+
+def _is_gpu(t: torch.Tensor) -> bool:
+    return t.is_cuda
+
+
+# --------------------------------------------------------------------- GEMM
+def gemm(A: torch.Tensor, B: torch.Tensor, *, ta: bool = False, tb: bool = False,
+         bias: Optional[torch.Tensor] = None, relu: bool = False,
+         out: Optional[torch.Tensor] = None, out_dtype: Optional[torch.dtype] = None,
+         splits: int = 1) -> torch.Tensor:
+    """C[M,N] = op(A) @ op(B) (+bias) (+relu). bf16 inputs, fp32 accumulate.
+
+    op(A) = A.T if ta (A stored [K,M]); op(B) = B.T if tb (B stored [N,K]).
+    splits > 1 uses the split-K fp32-atomic path (out must be fp32, zeroed
+    here).
+    """
+    M = A.shape[1] if ta else A.shape[0]
+    N = B.shape[0] if tb else B.shape[1]
+    dtype = out_dtype or (torch.float32 if splits > 1 else A.dtype)
+    if out is None:
+        out = torch.empty((M, N), device=A.device, dtype=dtype)
+    if _is_gpu(A):
+        lo = require_ext()
+        if splits > 1:
+            out.zero_()
+        ok = lo.gemm(A, B, out, bias, ta, tb, 1 if relu else 0, splits)
+        if ok:
+            return out
+        # cold-path shapes: plain library GEMM (rocBLAS via torch.matmul)
+    a = (A.transpose(0, 1) if ta else A).float()
+    b = (B.transpose(0, 1) if tb else B).float()
+    c = a @ b
+    if bias is not None:
+        c = c + bias.float()
+    if relu:
+        c = torch.relu(c)
+    out.copy_(c.to(out.dtype))
+    return out
+
+
+# ------------------------------------------------------------------- im2col
+def im2col(x: torch.Tensor, kh: int, kw: int, sh: int, sw: int, ph: int, pw: int,
+           kpad: int, out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """NHWC x [B,H,W,C] -> col [B*OH*OW, kpad] with row layout (kh, kw, c),
+    zero-padded to kpad columns."""
+    B, H, W, C = x.shape
+    OH = (H + 2 * ph - kh) // sh + 1
+    OW = (W + 2 * pw - kw) // sw + 1
+    K = kh * kw * C
+    assert kpad >= K and kpad % 8 == 0
+    if out is None:
+        out = torch.zeros((B * OH * OW, kpad), device=x.device, dtype=x.dtype)
+    if _is_gpu(x):
+        lo = require_ext()
+        lo.im2col(x, kh, kw, sh, sw, ph, pw, kpad, out)
+        return out
+    # reference: unfold on NCHW then reorder (c,kh,kw) -> (kh,kw,c)
+    xn = x.permute(0, 3, 1, 2).float()  # NCHW
+    cols = torch.nn.functional.unfold(xn, (kh, kw), padding=(ph, pw),
+                                      stride=(sh, sw))  # [B, C*kh*kw, L]
+    cols = cols.reshape(B, C, kh * kw, -1).permute(0, 3, 2, 1)  # [B,L,khkw,C]
+    cols = cols.reshape(B * OH * OW, K)
+    out[:, :K] = cols.to(out.dtype)
+    out[:, K:] = 0
+    return out
+
+
+def col2im(dcol: torch.Tensor, B: int, H: int, W: int, C: int, kh: int, kw: int,
+           ph: int, pw: int, out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Stride-1 col2im (gather): dcol [B*OH*OW, kpad] -> dx [B,H,W,C]."""
+    OH = H + 2 * ph - kh + 1
+    OW = W + 2 * pw - kw + 1
+    if out is None:
+        out = torch.empty((B, H, W, C), device=dcol.device, dtype=dcol.dtype)
+    if _is_gpu(dcol):
+        lo = require_ext()
+        lo.col2im(dcol, B, H, W, C, kh, kw, ph, pw, out)
+        return out
+    K = kh * kw * C
+    cols = dcol[:, :K].float().reshape(B, OH * OW, kh * kw, C)
+    cols = cols.permute(0, 3, 2, 1).reshape(B, C * kh * kw, OH * OW)
+    xn = torch.nn.functional.fold(cols, (H, W), (kh, kw), padding=(ph, pw))
+    out.copy_(xn.permute(0, 2, 3, 1).to(out.dtype))
+    return out
+
+
+# ------------------------------------------------------------------ maxpool
+def maxpool2d(x: torch.Tensor, kh: int, kw: int, sh: int, sw: int
+              ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """NHWC maxpool; returns (out, idx u8 of kh*KW+kw argmax)."""
+    if _is_gpu(x):
+        lo = require_ext()
+        out, idx = lo.maxpool_fwd(x, kh, kw, sh, sw)
+        return out, idx
+    B, H, W, C = x.shape
+    OH = (H - kh) // sh + 1
+    OW = (W - kw) // sw + 1
+    xn = x.permute(0, 3, 1, 2).float()
+    out_n, ind = torch.nn.functional.max_pool2d(xn, (kh, kw), (sh, sw),
+                                                return_indices=True)
+    out = out_n.permute(0, 2, 3, 1).to(x.dtype)
+    # flat NCHW index -> (kh,kw) offset index
+    hh = (ind // W)
+    ww = (ind % W)
+    oh = torch.arange(OH, device=x.device).view(1, 1, OH, 1)
+    ow = torch.arange(OW, device=x.device).view(1, 1, 1, OW)
+    rel = (hh - oh * sh) * kw + (ww - ow * sw)
+    idx = rel.permute(0, 2, 3, 1).to(torch.uint8).contiguous()
+    return out, idx
+
+
+def maxpool2d_bwd(dy: torch.Tensor, idx: torch.Tensor, H: int, W: int,
+                  kh: int, kw: int, sh: int, sw: int,
+                  out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    B, OH, OW, C = dy.shape
+    if out is None:
+        out = torch.empty((B, H, W, C), device=dy.device, dtype=dy.dtype)
+    if _is_gpu(dy):
+        lo = require_ext()
+        lo.maxpool_bwd(dy, idx, H, W, kh, kw, sh, sw, out)
+        return out
+    dx = torch.zeros((B, H, W, C), dtype=torch.float32)
+    rel = idx.long()
+    khh, kww = rel // kw, rel % kw
+    oh = torch.arange(OH).view(1, OH, 1, 1)
+    ow = torch.arange(OW).view(1, 1, OW, 1)
+    hsrc = oh * sh + khh
+    wsrc = ow * sw + kww
+    b = torch.arange(B).view(B, 1, 1, 1).expand_as(rel)
+    c = torch.arange(C).view(1, 1, 1, C).expand_as(rel)
+    dx.index_put_((b.reshape(-1), hsrc.reshape(-1), wsrc.reshape(-1),
+                   c.reshape(-1)), dy.float().reshape(-1), accumulate=True)
+    out.copy_(dx.to(out.dtype))
+    return out
+
+
+# -------------------------------------------------------------- activations
+def relu_bwd(dy: torch.Tensor, y: torch.Tensor,
+             out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if out is None:
+        out = torch.empty_like(dy)
+    if _is_gpu(dy):
+        lo = require_ext()
+        lo.relu_bwd(dy.contiguous().view(-1), y.contiguous().view(-1),
+                    out.view(-1))
+        return out
+    out.copy_((dy.float() * (y.float() > 0)).to(out.dtype))
+    return out
+
+
+# ---------------------------------------------------------------- optimizer
+def sgd_step(master: torch.Tensor, grad: torch.Tensor, mom: torch.Tensor,
+             mirror: torch.Tensor, lr: float, mu: float, wd: float,
+             gscale: float = 1.0) -> None:
+    if _is_gpu(master):
+        lo = require_ext()
+        lo.sgd_step(master, grad, mom, mirror, lr, mu, wd, gscale)
+        return
+    g = grad * gscale + wd * master
+    mom.mul_(mu).add_(g)
+    master.add_(mom, alpha=-lr)
+    mirror.copy_(master.to(mirror.dtype))
+
+
+def adam_step(master: torch.Tensor, grad: torch.Tensor, m1: torch.Tensor,
+              m2: torch.Tensor, mirror: torch.Tensor, lr: float, b1: float,
+              b2: float, eps: float, wd: float, step: int,
+              gscale: float = 1.0) -> None:
+    c1 = 1.0 / (1.0 - b1 ** step)
+    c2 = 1.0 / (1.0 - b2 ** step)
+    if _is_gpu(master):
+        lo = require_ext()
+        lo.adam_step(master, grad, m1, m2, mirror, lr, b1, b2, eps, wd, c1, c2,
+                     gscale)
+        return
+    g = grad * gscale + wd * master
+    m1.mul_(b1).add_(g, alpha=1 - b1)
+    m2.mul_(b2).addcmul_(g, g, value=1 - b2)
+    master.addcdiv_(m1 * c1, (m2 * c2).sqrt() + eps, value=-lr)
+    mirror.copy_(master.to(mirror.dtype))
+
+
+# --------------------------------------------------------------- reductions
+def colsum(dy: torch.Tensor, out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if out is None:
+        out = torch.empty(dy.shape[1], device=dy.device, dtype=torch.float32)
+    if _is_gpu(dy):
+        lo = require_ext()
+        lo.colsum(dy, out)
+        return out
+    out.copy_(dy.float().sum(0))
+    return out
+
+
+def argmax_rows(x: torch.Tensor, cvalid: Optional[int] = None) -> torch.Tensor:
+    cvalid = cvalid or x.shape[1]
+    if _is_gpu(x):
+        lo = require_ext()
+        return lo.argmax_rows(x, cvalid)
+    return x[:, :cvalid].float().argmax(1).to(torch.int32)
+
+
+def accuracy_count(pred: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+    if _is_gpu(pred):
+        lo = require_ext()
+        return lo.accuracy_count(pred, labels)
+    return (pred.long() == labels).sum().to(torch.int32).reshape(1)
+
+
+# --------------------------------------------------------------- softmax-CE
+def softmax_ce(logits: torch.Tensor, labels: torch.Tensor,
+               dlogits: Optional[torch.Tensor] = None,
+               loss_sum: Optional[torch.Tensor] = None,
+               correct: Optional[torch.Tensor] = None,
+               cvalid: Optional[int] = None, gscale: float = 1.0):
+    """Fused fwd+bwd: writes dlogits; accumulates loss_sum (fp32[1]) and
+    correct (int32[1]) in place. Returns (dlogits, loss_sum, correct)."""
+    M, C = logits.shape
+    cvalid = cvalid or C
+    if dlogits is None:
+        dlogits = torch.empty_like(logits)
+    if loss_sum is None:
+        loss_sum = torch.zeros(1, device=logits.device, dtype=torch.float32)
+    if correct is None:
+        correct = torch.zeros(1, device=logits.device, dtype=torch.int32)
+    if _is_gpu(logits):
+        lo = require_ext()
+        lo.softmax_ce(logits, labels, dlogits, loss_sum, correct, cvalid, gscale)
+        return dlogits, loss_sum, correct
+    lg = logits[:, :cvalid].float()
+    p = torch.softmax(lg, dim=1)
+    loss = torch.nn.functional.cross_entropy(lg, labels, reduction="sum")
+    onehot = torch.nn.functional.one_hot(labels, cvalid).float()
+    d = (p - onehot) * gscale
+    dlogits.zero_()
+    dlogits[:, :cvalid] = d.to(dlogits.dtype)
+    loss_sum += loss
+    correct += (lg.argmax(1) == labels).sum().to(torch.int32)
+    return dlogits, loss_sum, correct

@@ -1,0 +1,60 @@
+"""Data parallelism: RCCL all-reduce over xGMI (one process per GPU).
+
+Replaces the reference's Spark worker fan-out. On ROCm the
+torch.distributed "nccl" backend IS RCCL; on CPU test hosts "gloo" runs the
+same code path (multi-process CPU tests, world_size 2).
+
+The gradient "bucket" is the ParamArena's single flat fp32 grad tensor, so
+one step needs exactly ONE all-reduce — sized in the hundreds of MB for the
+big models, which is the right granularity for the 7-link point-to-point xGMI
+fabric (per-link-bound rings want few, large transfers — SURVEY §5.8).
+"""
+from __future__ import annotations
+
+import os
+from datetime import timedelta
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+
+def is_distributed() -> bool:
+    return dist.is_available() and dist.is_initialized()
+
+
+def get_rank() -> int:
+    return dist.get_rank() if is_distributed() else 0
+
+
+def get_world_size() -> int:
+    return dist.get_world_size() if is_distributed() else 1
+
+
+def init_distributed(backend: Optional[str] = None, timeout_s: int = 300) -> int:
+    """Initialize from torchrun env (RANK/WORLD_SIZE/LOCAL_RANK). Returns
+    local rank. No-op (rank 0) when not launched distributed."""
+    if "RANK" not in os.environ or int(os.environ.get("WORLD_SIZE", "1")) == 1:
+        return 0
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if not dist.is_initialized():
+        if backend == "nccl":
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend=backend,
+                                timeout=timedelta(seconds=timeout_s))
+    return local_rank
+
+
+def all_reduce_grads(grad_flat: torch.Tensor, async_op: bool = False):
+    """SUM all-reduce of the flat grad arena (grads carry 1/(B*world) so sum
+    = global-batch mean)."""
+    if not is_distributed():
+        return None
+    return dist.all_reduce(grad_flat, op=dist.ReduceOp.SUM, async_op=async_op)
+
+
+def barrier() -> None:
+    if is_distributed():
+        dist.barrier()

@@ -1,0 +1,146 @@
+"""Engine tests on CPU: arena layout, layer gradients vs torch autograd,
+end-to-end MNIST-CNN training step + convergence on a separable toy problem."""
+import math
+
+import pytest
+import torch
+
+from learningorchestra_amd.engine.arena import ParamArena, SGD
+from learningorchestra_amd.engine.layers import (Conv2dNHWC, Flatten, Linear,
+                                                 MaxPool2dNHWC,
+                                                 SequentialClassifier)
+from learningorchestra_amd.engine.trainer import Trainer, make_sgd
+from learningorchestra_amd.models.mnist_cnn import build_mnist_cnn
+from learningorchestra_amd.data.synthetic import mnist_batch
+
+
+def test_arena_views_and_step():
+    a = ParamArena("cpu")
+    a.add("w", (3, 5), 0.1)
+    a.add("b", (7,), torch.arange(7.0))
+    a.finalize(seed=1)
+    assert a.numel == 16 + 8  # 8-padded
+    assert a.p("w").shape == (3, 5) and a.p("w").dtype == torch.bfloat16
+    assert torch.allclose(a.pf("b"), torch.arange(7.0))
+    a.g("w").fill_(1.0)
+    opt = SGD(a, lr=0.5, momentum=0.0)
+    before = a.pf("w").clone()
+    opt.step()
+    assert torch.allclose(a.pf("w"), before - 0.5)
+    # mirror tracks master
+    assert torch.allclose(a.p("w").float(), a.pf("w"), atol=0.01)
+
+
+def _rel_close(a, b, tol=0.05):
+    a, b = a.float().flatten(), b.float().flatten()
+    return ((a - b).norm() / (b.norm() + 1e-8)).item() < tol
+
+def _torch_ref_linear(x, w, b, relu):
+    y = x.float() @ w.float().t() + b.float()
+    return torch.relu(y) if relu else y
+
+
+def test_linear_grads_match_autograd():
+    torch.manual_seed(0)
+    M, K, N = 32, 24, 16
+    lay = Linear("l", K, N, relu=True)
+    arena = ParamArena("cpu")
+    lay.build(arena)
+    arena.finalize()
+    x = torch.randn(M, K).bfloat16()
+    y = lay.forward(x)
+
+    xw = x.float().requires_grad_(True)
+    w = arena.pf("l.w").detach().clone().requires_grad_(True)
+    b = arena.pf("l.b").detach().clone().requires_grad_(True)
+    yr = _torch_ref_linear(xw, w, b, True)
+    assert torch.allclose(y.float(), yr, atol=0.15, rtol=0.05)
+
+    dy = torch.randn(M, N)
+    yr.backward(dy)
+    dx = lay.backward(dy.bfloat16().clone())
+    assert _rel_close(arena.g("l.w"), w.grad)
+    assert _rel_close(arena.g("l.b"), b.grad)
+    assert _rel_close(dx, xw.grad)
+
+
+def test_conv_grads_match_autograd():
+    torch.manual_seed(0)
+    B, H, W, Cin, Cout, k = 2, 10, 10, 8, 8, 3
+    lay = Conv2dNHWC("c", Cin, Cout, k, k, relu=False, first=False)
+    arena = ParamArena("cpu")
+    lay.build(arena)
+    arena.finalize()
+    x = torch.randn(B, H, W, Cin).bfloat16()
+    y = lay.forward(x)
+    OH = H - k + 1
+    assert y.shape == (B, OH, OH, Cout)
+
+    # torch reference in NCHW
+    xr = x.float().permute(0, 3, 1, 2).requires_grad_(True)
+    wflat = arena.pf("c.w").detach().clone()  # [Cout, kpad]
+    kdim = k * k * Cin
+    # our weight row layout is (kh, kw, c)
+    wr = (wflat[:, :kdim].reshape(Cout, k, k, Cin).permute(0, 3, 1, 2)
+          .contiguous().requires_grad_(True))
+    br = arena.pf("c.b").detach().clone().requires_grad_(True)
+    yr = torch.nn.functional.conv2d(xr, wr, br)
+    assert torch.allclose(y.float().permute(0, 3, 1, 2), yr, atol=0.3, rtol=0.1)
+
+    dy = torch.randn(B, OH, OH, Cout)
+    yr.backward(dy.permute(0, 3, 1, 2))
+    dx = lay.backward(dy.bfloat16().clone())
+    gw = arena.g("c.w")[:, :kdim].reshape(Cout, k, k, Cin).permute(0, 3, 1, 2)
+    assert _rel_close(gw, wr.grad)
+    assert _rel_close(arena.g("c.b"), br.grad)
+    assert _rel_close(dx.float().permute(0, 3, 1, 2), xr.grad)
+    # padded weight-grad columns stay zero (kpad trick invariant)
+    if arena.g("c.w").shape[1] > kdim:
+        assert arena.g("c.w")[:, kdim:].abs().max() == 0
+
+
+def test_mnist_cnn_step_and_shapes():
+    model = build_mnist_cnn("cpu", seed=0)
+    x, y = mnist_batch(32, dtype=torch.bfloat16, seed=0)
+    loss, correct = model.train_step(x, y)
+    assert loss.item() > 0 and 0 <= correct.item() <= 32
+    # loss near log(10) at random init
+    assert abs(loss.item() / 32 - math.log(10)) < 1.0
+    preds = model.predict(x)
+    assert preds.shape == (32,) and preds.max() < 10
+
+
+def test_training_reduces_loss():
+    """Tiny separable problem: the label is encoded in the image mean —
+    a few SGD steps must cut the loss substantially."""
+    torch.manual_seed(0)
+    model = build_mnist_cnn("cpu", seed=1, channels=(8, 8), fc_width=32)
+    opt = make_sgd(model, lr=0.1, momentum=0.9)
+    trainer = Trainer(model, opt, device="cpu")
+
+    def batch():
+        y = torch.randint(0, 10, (64,))
+        x = (y.float().view(-1, 1, 1, 1) / 10.0 +
+             0.05 * torch.randn(64, 28, 28, 1)).bfloat16()
+        return x, y
+
+    x0, y0 = batch()
+    first, _ = trainer.step(x0, y0)
+    for _ in range(80):
+        trainer.step_async(*batch())
+    last, acc = trainer.step(*batch())
+    assert last < first * 0.75, (first, last)
+    assert acc > 0.2
+
+
+def test_checkpoint_resume():
+    model = build_mnist_cnn("cpu", seed=2, channels=(4, 4), fc_width=16)
+    sd = model.state_dict()
+    x, y = mnist_batch(8, dtype=torch.bfloat16, seed=3)
+    logits_before = model.forward(x).float().clone()
+    # perturb, then restore
+    model.arena.master.add_(1.0)
+    model.arena.mirror.copy_(model.arena.master.bfloat16())
+    assert not torch.allclose(model.forward(x).float(), logits_before)
+    model.load_state_dict(sd)
+    assert torch.allclose(model.forward(x).float(), logits_before)
