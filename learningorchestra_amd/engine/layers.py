@@ -13,6 +13,7 @@ all-reduce is a plain SUM.
 from __future__ import annotations
 
 import math
+import zlib
 from typing import List, Optional, Tuple
 
 import torch
@@ -59,7 +60,7 @@ class Conv2dNHWC(Layer):
 
         def init_w(shape):
             g = torch.Generator(device="cpu").manual_seed(
-                abs(hash(self.name)) % (2 ** 31))
+                zlib.crc32(self.name.encode()) % (2 ** 31))
             t = torch.zeros(shape, dtype=torch.float32)
             t[:, :kdim] = torch.randn((out_c, kdim), generator=g) * std
             return t
@@ -155,7 +156,7 @@ class Linear(Layer):
     def build(self, arena: ParamArena) -> None:
         self.arena = arena
         std = math.sqrt(2.0 / self.in_f)
-        g = torch.Generator(device="cpu").manual_seed(abs(hash(self.name)) % (2 ** 31))
+        g = torch.Generator(device="cpu").manual_seed(zlib.crc32(self.name.encode()) % (2 ** 31))
         arena.add(self.name + ".w", (self.out_f, self.in_f),
                   torch.randn((self.out_f, self.in_f), generator=g) * std)
         arena.add(self.name + ".b", (self.out_f,), torch.zeros(self.out_f))
