@@ -1,0 +1,331 @@
+"""The REST API — learningOrchestra's full 102-endpoint surface on one
+FastAPI server.
+
+The reference split this across a KrakenD gateway + ~10 Flask microservices
+(SURVEY §1 L1/L2); on a single MI355X node they collapse into one in-process
+app with the same URI scheme (``/api/learningOrchestra/v1/{verb}/{tool}``),
+the same request JSON field names (``datasetName``/``datasetURI``,
+``modulePath``/``class``/``classParameters``, ``name``/``modelName``/
+``parentName``/``method``/``methodParameters``, ``function``/
+``functionParameters``, ``trainDatasetName``/``testDatasetName``/
+``modelingCode``/``classifiersList``, ``types``/``names``), the same
+``{"result": ...}`` envelope, status codes (200/201/404/406/409) and the
+async ``finished``-flag poll contract.
+
+Tools: ``scikitlearn`` (sklearn in-process, CPU), ``tensorflow`` (module
+paths translated onto the native zoo) and the new native ``torch`` tool —
+all three accepted on every executor verb.
+
+Deliberate fix vs the reference: the ``evaluate/sckitlearn``
+type-string typo is corrected to ``evaluate/scikitlearn``; name-uniqueness
+checks happen under the runtime lock (no TOCTOU).
+
+New verb endpoint: ``GET /observe/{name}[/wait]`` — the reference's Observe
+verb was client-side collection polling (README.md:81); here the server
+offers long-poll waiting on the finished flag.
+"""
+from __future__ import annotations
+
+import json
+import os
+import threading
+import time
+from typing import Any, Dict, List, Optional
+
+from fastapi import FastAPI, Request, Response
+from fastapi.responses import JSONResponse
+
+from ..config import Config, get_config
+from ..data.csv_ingest import CsvIngest
+from ..executor.execution import Execution, ValidationError
+from ..executor.scheduler import JobScheduler
+from ..storage import ArtifactStore, Data, Metadata, connect
+from ..storage.metadata import METADATA_ROW_ID
+from .services import (BuilderService, DataTypeService, HistogramService,
+                       ProjectionService)
+
+PREFIX = "/api/learningOrchestra/v1"
+RESULT = "result"
+
+EXECUTOR_TOOLS = ("scikitlearn", "tensorflow", "torch")
+BINARY_VERBS = ("train", "tune", "evaluate", "predict")
+
+
+class Runtime:
+    """Bundles storage + executor + verb services (the whole L2-L6 stack)."""
+
+    def __init__(self, cfg: Optional[Config] = None):
+        self.cfg = cfg or get_config()
+        self.db = connect(self.cfg)
+        self.artifacts = ArtifactStore(os.path.join(self.cfg.data_root, "binaries"))
+        self.metadata = Metadata(self.db)
+        self.scheduler = JobScheduler(self.metadata, self.cfg.max_jobs)
+        self.data = Data(self.db, self.artifacts)
+        device = self.cfg.resolve_device()
+        gpu_device = device if device.startswith("cuda") else None
+        self.execution = Execution(self.db, self.artifacts, self.scheduler,
+                                   self.cfg.allow_user_code, device=gpu_device)
+        self.csv = CsvIngest(self.db)
+        self.projection = ProjectionService(self.db, self.scheduler)
+        self.datatype = DataTypeService(self.db, self.scheduler)
+        self.histogram = HistogramService(self.db, self.scheduler)
+        self.builder = BuilderService(self.db, self.artifacts, self.scheduler,
+                                      self.cfg.allow_user_code, device=gpu_device)
+        self._name_lock = threading.Lock()
+
+    # -- shared helpers ------------------------------------------------------
+    def require_unique(self, name: str) -> None:
+        if self.metadata.exists(name):
+            raise ValidationError(f"duplicate name '{name}'", status=409)
+
+    def require_exists(self, name: str) -> Dict[str, Any]:
+        doc = self.metadata.get_metadata(name)
+        if doc is None:
+            raise ValidationError(f"'{name}' not found", status=404)
+        return doc
+
+    def read_rows(self, name: str, query: Dict[str, Any], skip: int,
+                  limit: int) -> List[Dict[str, Any]]:
+        limit = min(limit, self.cfg.limit_param_max)
+        cursor = (self.db[name].find(query).sort("_id", 1)
+                  .skip(max(skip, 0)).limit(limit))
+        return list(cursor)
+
+
+def _poll_uri(verb: str, tool: str, name: str) -> str:
+    return f"{PREFIX}/{verb}/{tool}/{name}?query={{}}&limit=10&skip=0"
+
+
+def create_app(runtime: Optional[Runtime] = None) -> FastAPI:
+    rt = runtime or Runtime()
+    app = FastAPI(title="learningOrchestra-AMD", version="0.1.0",
+                  docs_url=PREFIX + "/docs", openapi_url=PREFIX + "/openapi.json")
+    app.state.runtime = rt
+
+    @app.exception_handler(ValidationError)
+    async def _validation_handler(_req: Request, exc: ValidationError):
+        return JSONResponse({RESULT: str(exc)},
+                            status_code=getattr(exc, "status", 406))
+
+    # ------------------------------------------------------------- dataset --
+    @app.post(PREFIX + "/dataset/{tool}", status_code=201)
+    def create_dataset(tool: str, body: Dict[str, Any]):
+        _check_tool(tool, ("csv", "generic"))
+        name = _field(body, "datasetName")
+        uri = _field(body, "datasetURI")
+        with rt._name_lock:
+            rt.require_unique(name)
+            if tool == "csv":
+                rt.csv.run_async(name, uri, rt.scheduler)
+            else:
+                _generic_download(rt, name, uri)
+        return {RESULT: _poll_uri("dataset", tool, name)}
+
+    # ---------------------------------------------------------------- model --
+    @app.post(PREFIX + "/model/{tool}", status_code=201)
+    def create_model(tool: str, body: Dict[str, Any]):
+        _check_tool(tool, EXECUTOR_TOOLS)
+        name = _field(body, "modelName")
+        with rt._name_lock:
+            rt.require_unique(name)
+            rt.execution.create_model(
+                name, f"model/{tool}", _field(body, "modulePath"),
+                _field(body, "class"), body.get("classParameters", {}),
+                body.get("description", ""))
+        return {RESULT: _poll_uri("model", tool, name)}
+
+    @app.patch(PREFIX + "/model/{tool}/{name}")
+    def update_model(tool: str, name: str, body: Dict[str, Any]):
+        _check_tool(tool, EXECUTOR_TOOLS)
+        meta = rt.require_exists(name)
+        rt.execution.create_model(
+            name, meta.get("type", f"model/{tool}"), meta["modulePath"],
+            meta.get("className", meta.get("class")),
+            body.get("classParameters", meta.get("classParameters", {})),
+            body.get("description", ""))
+        return {RESULT: _poll_uri("model", tool, name)}
+
+    # ------------------------------------------- train/tune/evaluate/predict --
+    @app.post(PREFIX + "/{verb}/{tool}", status_code=201)
+    def create_binary_execution(verb: str, tool: str, body: Dict[str, Any]):
+        if verb == "builder":
+            return _builder_post(rt, body)
+        if verb == "function":
+            return _function_post(rt, tool, body)
+        if verb == "explore" and tool == "histogram":
+            return _histogram_post(rt, body)
+        if verb == "transform" and tool == "projection":
+            return _projection_post(rt, body)
+        if verb in ("transform", "explore") and tool in EXECUTOR_TOOLS:
+            return _generic_execution_post(rt, verb, tool, body)
+        _check_tool(tool, EXECUTOR_TOOLS, verb=verb, verbs=BINARY_VERBS)
+        name = _field(body, "name")
+        parent = body.get("parentName") or _field(body, "modelName")
+        with rt._name_lock:
+            rt.require_unique(name)
+            rt.execution.create_binary_execution(
+                name, f"{verb}/{tool}", parent, _field(body, "method"),
+                body.get("methodParameters", {}), body.get("description", ""))
+        return {RESULT: _poll_uri(verb, tool, name)}
+
+    @app.patch(PREFIX + "/transform/projection")
+    def update_projection(body: Dict[str, Any]):
+        # reference PATCHes projections at the collection URI (krakend table)
+        return _projection_post(rt, body)
+
+    @app.patch(PREFIX + "/{verb}/{tool}/{name}")
+    def update_execution(verb: str, tool: str, name: str, body: Dict[str, Any]):
+        if verb == "model":
+            return update_model(tool, name, body)
+        rt.require_exists(name)
+        if verb == "function":
+            rt.metadata.update_finished_flag(name, False)
+            rt.execution.create_code_execution(
+                name, f"function/{tool}", _field(body, "function"),
+                body.get("functionParameters", {}), body.get("description", ""))
+        else:
+            rt.execution.update_execution(name, body.get("methodParameters", {}),
+                                          body.get("description", ""))
+        return {RESULT: _poll_uri(verb, tool, name)}
+
+    # ---------------------------------------------------- dataType (PATCH) --
+    @app.patch(PREFIX + "/transform/dataType")
+    def transform_datatype(body: Dict[str, Any]):
+        name = _field(body, "datasetName")
+        rt.require_exists(name)
+        rt.datatype.convert(name, _field(body, "types"))
+        return {RESULT: _poll_uri("transform", "dataType", name)}
+
+    # ------------------------------------------------------------- observe --
+    @app.get(PREFIX + "/observe/{name}")
+    def observe(name: str):
+        return {RESULT: rt.require_exists(name)}
+
+    @app.get(PREFIX + "/observe/{name}/wait")
+    def observe_wait(name: str, timeoutSeconds: float = 60.0):
+        rt.require_exists(name)
+        deadline = time.time() + min(timeoutSeconds, 300.0)
+        while time.time() < deadline:
+            doc = rt.metadata.get_metadata(name)
+            if doc and doc.get("finished"):
+                return {RESULT: doc}
+            time.sleep(0.05)
+        return JSONResponse({RESULT: rt.metadata.get_metadata(name),
+                             "timedOut": True}, status_code=200)
+
+    # ------------------------------------------------------------- catalog --
+    @app.get(PREFIX + "/{verb}/{tool}")
+    def catalog(verb: str, tool: str):
+        stype = f"{verb}/{tool}"
+        return {RESULT: rt.metadata.catalog(stype)}
+
+    # ------------------------------------------------------- rows/metadata --
+    @app.get(PREFIX + "/{verb}/{tool}/{name}")
+    def read_rows(verb: str, tool: str, name: str, query: str = "{}",
+                  limit: int = 10, skip: int = 0):
+        rt.require_exists(name)
+        q = json.loads(query) if query else {}
+        return {RESULT: rt.read_rows(name, q, skip, limit)}
+
+    @app.get(PREFIX + "/{verb}/{tool}/{name}/metadata")
+    def read_metadata(verb: str, tool: str, name: str):
+        return {RESULT: rt.require_exists(name)}
+
+    # -------------------------------------------------------------- delete --
+    @app.delete(PREFIX + "/{verb}/{tool}/{name}")
+    def delete(verb: str, tool: str, name: str):
+        rt.require_exists(name)
+        rt.execution.delete(name)
+        return {RESULT: f"deleted {name}"}
+
+    return app
+
+
+# ---------------------------------------------------------------- helpers --
+def _field(body: Dict[str, Any], name: str):
+    if name not in body:
+        raise ValidationError(f"missing required field '{name}'")
+    return body[name]
+
+
+def _check_tool(tool: str, valid, verb: Optional[str] = None,
+                verbs: Optional[tuple] = None) -> None:
+    if verbs is not None and verb not in verbs:
+        raise ValidationError(f"unknown verb '{verb}'", status=404)
+    if tool not in valid:
+        raise ValidationError(f"unknown tool '{tool}' (valid: {list(valid)})",
+                              status=404)
+
+
+def _generic_download(rt: Runtime, name: str, uri: str) -> None:
+    rt.metadata.create_file(name, "dataset/generic", url=uri)
+
+    def pipeline():
+        import re
+        if re.match(r"^https?://", uri):
+            import requests
+            with requests.get(uri, stream=True, timeout=60) as resp:
+                resp.raise_for_status()
+                rt.artifacts.save_raw(resp.iter_content(1 << 20), name)
+        else:
+            with open(uri, "rb") as fh:
+                rt.artifacts.save_raw(iter(lambda: fh.read(1 << 20), b""), name)
+        rt.metadata.update_finished_flag(name, True)
+
+    rt.scheduler.submit(name, pipeline)
+
+
+def _generic_execution_post(rt: Runtime, verb: str, tool: str,
+                            body: Dict[str, Any]):
+    name = _field(body, "name")
+    with rt._name_lock:
+        rt.require_unique(name)
+        rt.execution.create_execution(
+            name, f"{verb}/{tool}", _field(body, "modulePath"),
+            _field(body, "class"), body.get("classParameters", {}),
+            _field(body, "method"), body.get("methodParameters", {}),
+            body.get("description", ""))
+    return {RESULT: _poll_uri(verb, tool, name)}
+
+
+def _function_post(rt: Runtime, tool: str, body: Dict[str, Any]):
+    _check_tool(tool, ("python",))
+    name = _field(body, "name")
+    with rt._name_lock:
+        rt.require_unique(name)
+        rt.execution.create_code_execution(
+            name, "function/python", _field(body, "function"),
+            body.get("functionParameters", {}), body.get("description", ""))
+    return {RESULT: _poll_uri("function", "python", name)}
+
+
+def _projection_post(rt: Runtime, body: Dict[str, Any]):
+    out = _field(body, "outputDatasetName")
+    rt.projection.create(_field(body, "inputDatasetName"), out,
+                         _field(body, "names"))
+    return {RESULT: _poll_uri("transform", "projection", out)}
+
+
+def _histogram_post(rt: Runtime, body: Dict[str, Any]):
+    out = _field(body, "outputDatasetName")
+    rt.histogram.create(_field(body, "inputDatasetName"), out,
+                        _field(body, "names"))
+    return {RESULT: _poll_uri("explore", "histogram", out)}
+
+
+def _builder_post(rt: Runtime, body: Dict[str, Any]):
+    names = rt.builder.create(_field(body, "trainDatasetName"),
+                              _field(body, "testDatasetName"),
+                              _field(body, "modelingCode"),
+                              _field(body, "classifiersList"))
+    return {RESULT: [_poll_uri("builder", "sparkml", n) for n in names]}
+
+
+def main():  # pragma: no cover - manual server entry
+    import uvicorn
+    cfg = get_config()
+    uvicorn.run(create_app(), host=cfg.host, port=cfg.port)
+
+
+if __name__ == "__main__":  # pragma: no cover
+    main()

@@ -24,6 +24,7 @@ SOURCES = [
     os.path.join(CSRC, "conv_pool.hip"),
     os.path.join(CSRC, "elementwise.hip"),
     os.path.join(CSRC, "softmax_ce.hip"),
+    os.path.join(CSRC, "tree_hist.hip"),
 ]
 
 

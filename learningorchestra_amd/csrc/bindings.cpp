@@ -46,6 +46,9 @@ void launch_accuracy_count(const void* pred, const void* label, void* out, long 
 void launch_softmax_ce(const void* logits, const void* labels, void* dlogits,
                        void* loss_sum, void* correct, long M, int C, int Cvalid,
                        float gscale, hipStream_t s);
+void launch_tree_hist(const void* binned, const void* node_of, const void* grad,
+                      const void* hess, void* hist, long N, int F, int n_nodes,
+                      int B, hipStream_t s);
 }  // namespace lo
 
 namespace {
@@ -240,6 +243,23 @@ void softmax_ce(at::Tensor logits, at::Tensor labels, at::Tensor dlogits,
                         (float)gscale, stream());
 }
 
+void tree_hist(at::Tensor binned, at::Tensor node_of, at::Tensor grad,
+               at::Tensor hess, at::Tensor hist, int64_t n_nodes, int64_t n_bins) {
+  TORCH_CHECK(binned.is_cuda() && binned.scalar_type() == at::kByte &&
+              binned.is_contiguous(), "binned must be contiguous u8 GPU");
+  TORCH_CHECK(node_of.scalar_type() == at::kInt && node_of.is_contiguous());
+  check_f32(grad, "grad");
+  check_f32(hess, "hess");
+  check_f32(hist, "hist");
+  const long N = binned.size(0);
+  const int F = (int)binned.size(1);
+  TORCH_CHECK(node_of.numel() == N && grad.numel() == N && hess.numel() == N);
+  TORCH_CHECK(hist.numel() == n_nodes * F * n_bins * 2, "hist size");
+  lo::launch_tree_hist(binned.data_ptr(), node_of.data_ptr(), grad.data_ptr(),
+                       hess.data_ptr(), hist.data_ptr(), N, F, (int)n_nodes,
+                       (int)n_bins, stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -258,6 +278,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum", &colsum);
   m.def("argmax_rows", &argmax_rows);
   m.def("accuracy_count", &accuracy_count);
+  m.def("tree_hist", &tree_hist);
   m.def("softmax_ce", &softmax_ce,
         py::arg("logits"), py::arg("labels"), py::arg("dlogits"),
         py::arg("loss_sum") = py::none(), py::arg("correct") = py::none(),

@@ -33,7 +33,12 @@ from .scheduler import JobScheduler
 
 class ValidationError(ValueError):
     """4xx-able validation failure (reference: per-endpoint validator chains,
-    e.g. database_executor_image/server.py:201-278)."""
+    e.g. database_executor_image/server.py:201-278). ``status`` maps to the
+    HTTP code (406 default; 404 not-found; 409 duplicate)."""
+
+    def __init__(self, message: str, status: int = 406):
+        super().__init__(message)
+        self.status = status
 
 
 class ReflectiveRuntime:

@@ -1,0 +1,273 @@
+"""Histogram-based decision-tree ensembles — the MLlib-parity tree family
+(reference builder.py:55-61 drives Spark MLlib DecisionTree/RandomForest/GBT;
+SURVEY §2.9: "Tree histogram build + split (RF/GB/DT) -> per-feature
+histogram build, split-gain reduce").
+
+Design (LightGBM/XGBoost-style, one GPU):
+  1. quantize features to <=255 uint8 bins (quantile grid from a sample);
+  2. grow trees level-order; per level build per-node (grad,hess) histograms
+     over [node, feature, bin] — the hot op (tree_hist HIP kernel on GPU,
+     torch scatter_add reference on CPU);
+  3. split gain G_L^2/(H_L+lam) + G_R^2/(H_R+lam) - G^2/(H+lam) via tensor
+     cumsums (small), partition samples by the chosen (feature, threshold).
+
+GBT: logistic loss (grad = p - y, hess = p(1-p)).
+RF:  squared-loss trees on bootstrap + feature subsample, averaged.
+DT:  a single deeper tree.
+"""
+from __future__ import annotations
+
+import math
+from typing import List, Optional, Tuple
+
+import torch
+
+from ..ops._ext import has_ext, require_ext
+
+MAX_BINS = 255
+
+
+def quantize(X: torch.Tensor, n_bins: int = MAX_BINS,
+             sample: int = 100_000) -> Tuple[torch.Tensor, torch.Tensor]:
+    """X [N,F] float -> (binned uint8 [N,F], bin_edges [F, n_bins-1])."""
+    N, F = X.shape
+    idx = torch.randperm(N, device=X.device)[: min(N, sample)]
+    qs = torch.linspace(0, 1, n_bins + 1, device=X.device)[1:-1]
+    edges = torch.quantile(X[idx].float(), qs, dim=0).T.contiguous()  # [F, n_bins-1]
+    binned = torch.searchsorted(edges, X.T.contiguous().float()).T  # [N,F] in [0,n_bins-1]
+    return binned.to(torch.uint8).contiguous(), edges
+
+
+def build_histograms(binned: torch.Tensor, node_of: torch.Tensor,
+                     grad: torch.Tensor, hess: torch.Tensor,
+                     n_nodes: int, n_bins: int = MAX_BINS + 1) -> torch.Tensor:
+    """-> hist [n_nodes, F, n_bins, 2] (sum grad, sum hess) for samples with
+    node_of >= 0. GPU: tree_hist HIP kernel; CPU: torch index_add reference."""
+    N, F = binned.shape
+    if binned.is_cuda:
+        lo = require_ext()
+        hist = torch.zeros(n_nodes * F * n_bins * 2, device=binned.device,
+                           dtype=torch.float32)
+        lo.tree_hist(binned, node_of, grad, hess, hist, n_nodes, n_bins)
+        return hist.view(n_nodes, F, n_bins, 2)
+    hist = torch.zeros(n_nodes * F * n_bins, 2, dtype=torch.float32)
+    active = node_of >= 0
+    nb = node_of[active].long()
+    bb = binned[active].long()                      # [n, F]
+    flat = (nb.unsqueeze(1) * F + torch.arange(F)) * n_bins + bb  # [n, F]
+    src = torch.stack([grad[active], hess[active]], dim=1)  # [n, 2]
+    hist.index_add_(0, flat.reshape(-1),
+                    src.repeat_interleave(F, dim=0))
+    return hist.view(n_nodes, F, n_bins, 2)
+
+
+def best_splits(hist: torch.Tensor, lam: float = 1.0, min_child_hess: float = 1e-3
+                ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """hist [nodes, F, B, 2] -> (gain [nodes], feature [nodes], bin [nodes]).
+    Vectorized cumsum split-gain reduce (small tensor — stays in torch)."""
+    g = hist[..., 0]
+    h = hist[..., 1]
+    GL = g.cumsum(-1)
+    HL = h.cumsum(-1)
+    G = GL[..., -1:].expand_as(GL)
+    H = HL[..., -1:].expand_as(HL)
+    GR, HR = G - GL, H - HL
+    gain = GL.square() / (HL + lam) + GR.square() / (HR + lam) - G.square() / (H + lam)
+    valid = (HL > min_child_hess) & (HR > min_child_hess)
+    gain = torch.where(valid, gain, torch.full_like(gain, -1e30))
+    flat = gain.flatten(1)                      # [nodes, F*B]
+    best = flat.argmax(1)
+    B = hist.shape[2]
+    return flat.gather(1, best.unsqueeze(1)).squeeze(1), best // B, best % B
+
+
+class Tree:
+    __slots__ = ("feature", "threshold_bin", "left", "right", "value", "edges")
+
+    def __init__(self, n_nodes: int, device):
+        self.feature = torch.full((n_nodes,), -1, dtype=torch.int64, device=device)
+        self.threshold_bin = torch.zeros(n_nodes, dtype=torch.int64, device=device)
+        self.value = torch.zeros(n_nodes, dtype=torch.float32, device=device)
+
+    def predict_binned(self, binned: torch.Tensor) -> torch.Tensor:
+        """Vectorized level-order traversal on the binned matrix."""
+        N = binned.shape[0]
+        node = torch.zeros(N, dtype=torch.int64, device=binned.device)
+        depth = int(math.log2(self.feature.shape[0] + 1))
+        for _ in range(depth):
+            f = self.feature[node]
+            leaf = f < 0
+            fsafe = f.clamp(min=0)
+            b = binned.gather(1, fsafe.unsqueeze(1)).squeeze(1).long()
+            go_left = b <= self.threshold_bin[node]
+            nxt = torch.where(go_left, 2 * node + 1, 2 * node + 2)
+            node = torch.where(leaf, node, nxt)
+        return self.value[node]
+
+
+class TreeLearner:
+    """Grows one tree level-order on (grad, hess)."""
+
+    def __init__(self, max_depth: int = 6, lam: float = 1.0,
+                 min_gain: float = 1e-6, lr: float = 1.0):
+        self.max_depth = max_depth
+        self.lam = lam
+        self.min_gain = min_gain
+        self.lr = lr
+
+    def fit(self, binned: torch.Tensor, grad: torch.Tensor, hess: torch.Tensor,
+            sample_mask: Optional[torch.Tensor] = None) -> Tree:
+        N, F = binned.shape
+        device = binned.device
+        total_nodes = 2 ** (self.max_depth + 1) - 1
+        tree = Tree(total_nodes, device)
+        node_of = torch.zeros(N, dtype=torch.int32, device=device)
+        if sample_mask is not None:
+            node_of = torch.where(sample_mask, node_of,
+                                  torch.full_like(node_of, -1))
+        level_start = 0
+        for depth in range(self.max_depth + 1):
+            level_nodes = 2 ** depth
+            rel_node = node_of - level_start
+            hist = build_histograms(binned, rel_node, grad, hess, level_nodes)
+            # each sample lands once per feature; per-node totals from feature 0
+            Gn = hist[:, 0, :, 0].sum(-1)
+            Hn = hist[:, 0, :, 1].sum(-1)
+            values = -Gn / (Hn + self.lam) * self.lr
+            tree.value[level_start:level_start + level_nodes] = values
+            if depth == self.max_depth:
+                break
+            gain, feat, tbin = best_splits(hist, self.lam)
+            do_split = gain > self.min_gain
+            abs_nodes = torch.arange(level_nodes, device=device) + level_start
+            tree.feature[abs_nodes] = torch.where(do_split, feat,
+                                                  torch.full_like(feat, -1))
+            tree.threshold_bin[abs_nodes] = tbin
+            # partition: samples in splitting nodes move to children
+            nrel = rel_node.long().clamp(min=0)
+            f_of = feat[nrel]
+            t_of = tbin[nrel]
+            split_of = do_split[nrel]
+            b = binned.gather(1, f_of.unsqueeze(1)).squeeze(1).long()
+            go_left = b <= t_of
+            parent_abs = node_of.long()
+            child = torch.where(go_left, 2 * parent_abs + 1, 2 * parent_abs + 2)
+            new_node = torch.where(split_of & (node_of >= 0), child, -torch.ones_like(child))
+            node_of = new_node.to(torch.int32)
+            level_start += level_nodes
+        return tree
+
+
+class _TreeEnsembleBase:
+    def __init__(self, n_trees: int, max_depth: int, lr: float, device=None,
+                 seed: int = 0, subsample: float = 1.0, colsample: float = 1.0):
+        self.n_trees = n_trees
+        self.max_depth = max_depth
+        self.lr = lr
+        self.device = device or ("cuda" if torch.cuda.is_available() else "cpu")
+        self.seed = seed
+        self.subsample = subsample
+        self.colsample = colsample
+        self.trees: List[Tree] = []
+        self.edges: Optional[torch.Tensor] = None
+        self.base_score = 0.0
+
+    def _to_device_tensor(self, X) -> torch.Tensor:
+        if not isinstance(X, torch.Tensor):
+            X = torch.as_tensor(X, dtype=torch.float32)
+        return X.to(self.device, torch.float32)
+
+    def _bin(self, X: torch.Tensor) -> torch.Tensor:
+        binned = torch.searchsorted(self.edges, X.T.contiguous()).T
+        return binned.to(torch.uint8).contiguous()
+
+    def _raw_predict(self, binned: torch.Tensor) -> torch.Tensor:
+        out = torch.full((binned.shape[0],), self.base_score,
+                         device=binned.device)
+        for t in self.trees:
+            out += t.predict_binned(binned)
+        return out
+
+
+class GBTClassifier(_TreeEnsembleBase):
+    """Gradient-boosted trees, binary logistic loss (MLlib GBTClassifier
+    parity; reference builder.py:58)."""
+
+    def __init__(self, n_trees: int = 50, max_depth: int = 5, lr: float = 0.2,
+                 **kw):
+        super().__init__(n_trees, max_depth, lr, **kw)
+
+    def fit(self, X, y) -> "GBTClassifier":
+        X = self._to_device_tensor(X)
+        y = self._to_device_tensor(y).clamp(0, 1)
+        binned, self.edges = quantize(X)
+        p0 = y.mean().clamp(1e-4, 1 - 1e-4)
+        self.base_score = float(torch.log(p0 / (1 - p0)))
+        raw = torch.full_like(y, self.base_score)
+        learner = TreeLearner(self.max_depth, lr=self.lr)
+        g = torch.Generator(device="cpu").manual_seed(self.seed)
+        for _ in range(self.n_trees):
+            p = torch.sigmoid(raw)
+            grad = p - y
+            hess = p * (1 - p)
+            mask = None
+            if self.subsample < 1.0:
+                mask = (torch.rand(y.shape[0], generator=g) <
+                        self.subsample).to(self.device)
+            tree = learner.fit(binned, grad, hess, mask)
+            self.trees.append(tree)
+            raw += tree.predict_binned(binned)
+        return self
+
+    def predict_proba(self, X):
+        X = self._to_device_tensor(X)
+        p1 = torch.sigmoid(self._raw_predict(self._bin(X)))
+        return torch.stack([1 - p1, p1], 1).cpu().numpy()
+
+    def predict(self, X):
+        return self.predict_proba(X)[:, 1] > 0.5
+
+
+class RandomForestClassifier(_TreeEnsembleBase):
+    """Bagged squared-loss trees on bootstrap samples (MLlib
+    RandomForestClassifier parity; reference builder.py:59)."""
+
+    def __init__(self, n_trees: int = 20, max_depth: int = 8, **kw):
+        super().__init__(n_trees, max_depth, lr=1.0, subsample=0.8, **kw)
+
+    def fit(self, X, y) -> "RandomForestClassifier":
+        X = self._to_device_tensor(X)
+        y = self._to_device_tensor(y).clamp(0, 1)
+        binned, self.edges = quantize(X)
+        self.base_score = float(y.mean())
+        learner = TreeLearner(self.max_depth, lr=1.0)
+        g = torch.Generator(device="cpu").manual_seed(self.seed)
+        grad0 = (self.base_score - y)
+        hess = torch.ones_like(y)
+        for _ in range(self.n_trees):
+            mask = (torch.rand(y.shape[0], generator=g) < self.subsample
+                    ).to(self.device)
+            tree = learner.fit(binned, grad0, hess, mask)
+            self.trees.append(tree)
+        return self
+
+    def predict_proba(self, X):
+        X = self._to_device_tensor(X)
+        raw = torch.full((X.shape[0],), 0.0, device=self.device)
+        binned = self._bin(X)
+        for t in self.trees:
+            raw += t.predict_binned(binned)
+        p1 = (self.base_score + raw / max(len(self.trees), 1)).clamp(0, 1)
+        return torch.stack([1 - p1, p1], 1).cpu().numpy()
+
+    def predict(self, X):
+        return self.predict_proba(X)[:, 1] > 0.5
+
+
+class DecisionTreeClassifier(RandomForestClassifier):
+    """Single deep tree (MLlib DecisionTreeClassifier parity)."""
+
+    def __init__(self, max_depth: int = 10, **kw):
+        kw.pop("n_trees", None)
+        super().__init__(n_trees=1, max_depth=max_depth, **kw)
+        self.subsample = 1.0

@@ -1,0 +1,221 @@
+"""REST surface tests: the reference's URI scheme, field names, envelope,
+status codes and poll contract, driven end-to-end (the Titanic plumbing
+config from BASELINE.json runs entirely through HTTP here)."""
+import json
+import time
+
+import pytest
+from fastapi.testclient import TestClient
+
+from learningorchestra_amd.api.app import PREFIX, Runtime, create_app
+from learningorchestra_amd.data.synthetic import titanic_csv
+
+
+@pytest.fixture()
+def client(tmp_config, tmp_path):
+    rt = Runtime(tmp_config)
+    app = create_app(rt)
+    with TestClient(app) as c:
+        c.rt = rt
+        yield c
+
+
+def wait_finished(client, name, timeout=30.0):
+    r = client.get(f"{PREFIX}/observe/{name}/wait",
+                   params={"timeoutSeconds": timeout})
+    assert r.status_code == 200
+    doc = r.json()["result"]
+    assert doc is not None and doc.get("finished"), doc
+    return doc
+
+
+def ingest_titanic(client, tmp_path, name="titanic"):
+    p = tmp_path / "titanic.csv"
+    p.write_text(titanic_csv(rows=200))
+    r = client.post(f"{PREFIX}/dataset/csv",
+                    json={"datasetName": name, "datasetURI": str(p)})
+    assert r.status_code == 201
+    assert r.json()["result"].startswith(f"{PREFIX}/dataset/csv/{name}")
+    return wait_finished(client, name)
+
+
+def test_dataset_csv_contract(client, tmp_path):
+    meta = ingest_titanic(client, tmp_path)
+    assert meta["type"] == "dataset/csv" and "PassengerId" in meta["fields"]
+    # catalog
+    r = client.get(f"{PREFIX}/dataset/csv")
+    assert [d["datasetName"] for d in r.json()["result"]] == ["titanic"]
+    # paged rows: first doc is the _id:0 metadata document (reference
+    # database.read_file semantics)
+    r = client.get(f"{PREFIX}/dataset/csv/titanic",
+                   params={"query": "{}", "limit": 3, "skip": 0})
+    rows = r.json()["result"]
+    assert rows[0]["_id"] == 0 and rows[1]["_id"] == 1
+    # filtered query
+    r = client.get(f"{PREFIX}/dataset/csv/titanic",
+                   params={"query": json.dumps({"Sex": "male"}), "limit": 5})
+    assert all(x["Sex"] == "male" for x in r.json()["result"])
+    # duplicate -> 409
+    r = client.post(f"{PREFIX}/dataset/csv",
+                    json={"datasetName": "titanic", "datasetURI": "x.csv"})
+    assert r.status_code == 409
+    # missing -> 404
+    assert client.get(f"{PREFIX}/dataset/csv/nope").status_code == 404
+
+
+def test_projection_and_datatype_and_histogram(client, tmp_path):
+    ingest_titanic(client, tmp_path)
+    r = client.post(f"{PREFIX}/transform/projection",
+                    json={"inputDatasetName": "titanic",
+                          "outputDatasetName": "titanic_p",
+                          "names": ["Sex", "Age", "Survived"]})
+    assert r.status_code == 201
+    wait_finished(client, "titanic_p")
+    r = client.get(f"{PREFIX}/transform/projection/titanic_p",
+                   params={"limit": 2, "skip": 1})
+    row = r.json()["result"][0]
+    assert set(row) == {"_id", "Sex", "Age", "Survived"}
+    # invalid field -> 406
+    r = client.post(f"{PREFIX}/transform/projection",
+                    json={"inputDatasetName": "titanic",
+                          "outputDatasetName": "bad", "names": ["NoSuch"]})
+    assert r.status_code == 406
+
+    # dataType: stringify then re-numberify Age
+    r = client.patch(f"{PREFIX}/transform/dataType",
+                     json={"datasetName": "titanic_p",
+                           "types": {"Age": "string"}})
+    assert r.status_code == 200
+    wait_finished(client, "titanic_p")
+    row = client.get(f"{PREFIX}/dataset/csv/titanic_p",
+                     params={"limit": 2, "skip": 1}).json()["result"][0]
+    assert row["Age"] is None or isinstance(row["Age"], str)
+    client.patch(f"{PREFIX}/transform/dataType",
+                 json={"datasetName": "titanic_p", "types": {"Age": "number"}})
+    wait_finished(client, "titanic_p")
+
+    # histogram
+    r = client.post(f"{PREFIX}/explore/histogram",
+                    json={"inputDatasetName": "titanic",
+                          "outputDatasetName": "hist_sex", "names": ["Sex"]})
+    assert r.status_code == 201
+    wait_finished(client, "hist_sex")
+    rows = client.get(f"{PREFIX}/explore/histogram/hist_sex",
+                      params={"limit": 10}).json()["result"]
+    hist = next(x for x in rows if x.get("field") == "Sex")
+    assert set(hist["values"]) == {"male", "female"}
+    assert sum(hist["values"].values()) == 200
+
+
+def test_model_train_predict_evaluate_sklearn(client, tmp_path):
+    """The reference's canonical sklearn chain over REST: model -> train ->
+    predict -> evaluate, with $dataset parameter resolution."""
+    ingest_titanic(client, tmp_path)
+    client.post(f"{PREFIX}/transform/projection",
+                json={"inputDatasetName": "titanic", "outputDatasetName": "feat",
+                      "names": ["Pclass", "SibSp", "Parch", "Fare", "Survived"]})
+    wait_finished(client, "feat")
+
+    r = client.post(f"{PREFIX}/model/scikitlearn",
+                    json={"modelName": "lr_model",
+                          "modulePath": "sklearn.linear_model",
+                          "class": "LogisticRegression",
+                          "classParameters": {"max_iter": 200}})
+    assert r.status_code == 201
+    wait_finished(client, "lr_model")
+
+    r = client.post(f"{PREFIX}/train/scikitlearn",
+                    json={"name": "lr_trained", "modelName": "lr_model",
+                          "parentName": "lr_model", "method": "fit",
+                          "methodParameters": {"X": "#[[0.],[1.],[2.],[3.]]",
+                                               "y": "#[0,0,1,1]"}})
+    assert r.status_code == 201
+    doc = wait_finished(client, "lr_trained")
+    assert doc.get("exception") in (None, "")
+
+    r = client.post(f"{PREFIX}/predict/scikitlearn",
+                    json={"name": "lr_pred", "modelName": "lr_model",
+                          "parentName": "lr_trained", "method": "predict",
+                          "methodParameters": {"X": "#[[0],[3]]"}})
+    assert r.status_code == 201
+    doc = wait_finished(client, "lr_pred")
+    assert doc.get("exception") in (None, "")
+    # evaluate verb (score) on the same chain
+    r = client.post(f"{PREFIX}/evaluate/scikitlearn",
+                    json={"name": "lr_eval", "modelName": "lr_model",
+                          "parentName": "lr_trained", "method": "score",
+                          "methodParameters": {"X": "#[[0],[3]]", "y": "#[0,1]"}})
+    assert r.status_code == 201
+    wait_finished(client, "lr_eval")
+    # lineage: predict walks back to the model (binary_executor utils.py:257)
+    assert client.rt.metadata.walk_to_model("lr_pred")["datasetName"] == "lr_model"
+
+
+def test_function_python_verb(client):
+    r = client.post(f"{PREFIX}/function/python",
+                    json={"name": "fn1", "function": "print('hi')\nresponse = a * 2",
+                          "functionParameters": {"a": 21}})
+    assert r.status_code == 201
+    wait_finished(client, "fn1")
+    rows = client.get(f"{PREFIX}/function/python/fn1",
+                      params={"limit": 10}).json()["result"]
+    exec_doc = next(x for x in rows if x["_id"] == 1)
+    assert exec_doc["functionMessage"] == "hi\n"
+    assert client.rt.artifacts.load("fn1", "function/python") == 42
+
+
+def test_builder_sparkml_verb(client, tmp_path):
+    """builder/sparkml: modelingCode + native {lr,nb} classifiers end-to-end
+    (dt/rf/gb covered in tree tests; two classifiers keep this test fast)."""
+    ingest_titanic(client, tmp_path)
+    code = (
+        "import pandas as pd\n"
+        "def prep(df):\n"
+        "    out = df[['Pclass','SibSp','Parch','Fare']].copy()\n"
+        "    out['sex_n'] = (df['Sex'] == 'female').astype(float)\n"
+        "    out['label'] = df['Survived'].astype(float)\n"
+        "    return out.fillna(0.0)\n"
+        "features_training = prep(training_df)\n"
+        "features_evaluation = prep(testing_df)\n"
+        "features_testing = prep(testing_df).drop(columns=['label'])\n")
+    r = client.post(f"{PREFIX}/builder/sparkml",
+                    json={"trainDatasetName": "titanic",
+                          "testDatasetName": "titanic",
+                          "modelingCode": code,
+                          "classifiersList": ["lr", "nb"]})
+    assert r.status_code == 201
+    uris = r.json()["result"]
+    assert len(uris) == 2
+    for c in ("lr", "nb"):
+        doc = wait_finished(client, f"titanic{c}", timeout=120)
+        assert doc.get("exception") in (None, ""), doc
+        assert doc["accuracy"] > 0.5 and doc["fitTime"] > 0
+        rows = client.get(f"{PREFIX}/builder/sparkml/titanic{c}",
+                          params={"limit": 5, "skip": 1}).json()["result"]
+        assert all("prediction" in x for x in rows)
+    # invalid classifier -> 406
+    r = client.post(f"{PREFIX}/builder/sparkml",
+                    json={"trainDatasetName": "titanic",
+                          "testDatasetName": "titanic",
+                          "modelingCode": code, "classifiersList": ["svm"]})
+    assert r.status_code == 406
+
+
+def test_torch_tool_model_verb(client):
+    """The native 'torch' tool: instantiate an engine model via the model
+    verb (tensorflow module paths route here too)."""
+    r = client.post(f"{PREFIX}/model/torch",
+                    json={"modelName": "cnn0",
+                          "modulePath": "learningorchestra_amd.models.zoo",
+                          "class": "MnistCNN",
+                          "classParameters": {"seed": 1}})
+    assert r.status_code == 201
+    doc = wait_finished(client, "cnn0")
+    assert doc.get("exception") in (None, "")
+
+
+def test_delete_and_404(client, tmp_path):
+    ingest_titanic(client, tmp_path)
+    assert client.delete(f"{PREFIX}/dataset/csv/titanic").status_code == 200
+    assert client.get(f"{PREFIX}/dataset/csv/titanic").status_code == 404
+    assert client.delete(f"{PREFIX}/dataset/csv/titanic").status_code == 404
