@@ -30,7 +30,10 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=50)
     ap.add_argument("--warmup", type=int, default=10)
-    ap.add_argument("--batch", type=int, default=8192, help="per-GPU batch")
+    ap.add_argument("--batch", type=int, default=0, help="per-GPU batch (0 = model default)")
+    ap.add_argument("--model", default="mnist-cnn",
+                    choices=["mnist-cnn", "textcnn"],
+                    help="flagship = mnist-cnn (BASELINE.json headline)")
     ap.add_argument("--no-graph", action="store_true")
     ap.add_argument("--lr", type=float, default=0.05)
     args = ap.parse_args()
@@ -40,7 +43,8 @@ def main() -> None:
                                                 init_distributed)
     from learningorchestra_amd.engine.trainer import Trainer, make_sgd
     from learningorchestra_amd.models.mnist_cnn import build_mnist_cnn
-    from learningorchestra_amd.data.synthetic import mnist_batch
+    from learningorchestra_amd.models.textcnn import build_textcnn
+    from learningorchestra_amd.data.synthetic import imdb_batch, mnist_batch
 
     local_rank = init_distributed()
     world = get_world_size()
@@ -50,15 +54,28 @@ def main() -> None:
     if use_gpu:
         torch.cuda.set_device(local_rank)
 
-    model = build_mnist_cnn(device, seed=0)
+    if args.model == "mnist-cnn":
+        batch = args.batch or 8192
+        model = build_mnist_cnn(device, seed=0)
+        model_desc = "MNIST-CNN (LeNet-style, conv32-conv64-fc256)"
+        extra_cfg = {"image": "28x28x1"}
+    else:
+        batch = args.batch or 2048
+        model = build_textcnn(device, seed=0)
+        model_desc = "TextCNN (IMDb sentiment, emb128, filters 128 x k3/4/5)"
+        extra_cfg = {"seq_len": 256, "vocab": 20000}
     # graph capture: single-rank only (RCCL collectives stay outside graphs
     # until validated under capture)
     use_graph = use_gpu and world == 1 and not args.no_graph
     trainer = Trainer(model, make_sgd(model, lr=args.lr), device=device,
                       use_graph=use_graph)
 
-    x, y = mnist_batch(args.batch, device=device, dtype=torch.bfloat16,
-                       seed=1234 + rank)
+    if args.model == "mnist-cnn":
+        x, y = mnist_batch(batch, device=device, dtype=torch.bfloat16,
+                           seed=1234 + rank)
+    else:
+        x, y = imdb_batch(batch, seq_len=256, vocab=20000, seed=1234 + rank)
+        x, y = x.to(device), y.to(device)
 
     for _ in range(args.warmup):
         trainer.step_async(x, y)
@@ -82,8 +99,8 @@ def main() -> None:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 
-    loss = model.loss_sum.item() / args.batch
-    samples = args.batch * world * args.steps
+    loss = model.loss_sum.item() / batch
+    samples = batch * world * args.steps
     value = samples / elapsed
     if rank == 0:
         print(json.dumps({
@@ -99,10 +116,10 @@ def main() -> None:
             "vs_baseline": None,
             "dtype": "bf16",
             "data": "synthetic",
-            "config": {"model": "MNIST-CNN (LeNet-style, conv32-conv64-fc256)",
-                       "global_batch": args.batch * world,
-                       "per_gpu_batch": args.batch,
-                       "image": "28x28x1",
+            "config": {"model": model_desc,
+                       "global_batch": batch * world,
+                       "per_gpu_batch": batch,
+                       **extra_cfg,
                        "parallelism": f"dp{world}",
                        "graph_capture": use_graph,
                        "final_loss": round(loss, 4)},

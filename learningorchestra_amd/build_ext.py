@@ -25,6 +25,7 @@ SOURCES = [
     os.path.join(CSRC, "elementwise.hip"),
     os.path.join(CSRC, "softmax_ce.hip"),
     os.path.join(CSRC, "tree_hist.hip"),
+    os.path.join(CSRC, "embedding.hip"),
 ]
 
 

@@ -49,6 +49,10 @@ void launch_softmax_ce(const void* logits, const void* labels, void* dlogits,
 void launch_tree_hist(const void* binned, const void* node_of, const void* grad,
                       const void* hess, void* hist, long N, int F, int n_nodes,
                       int B, hipStream_t s);
+void launch_embedding_fwd(const void* ids, const void* table, void* out,
+                          long n, int dim, hipStream_t s);
+void launch_embedding_bwd(const void* ids, const void* dy, void* gtable,
+                          long n, int dim, hipStream_t s);
 }  // namespace lo
 
 namespace {
@@ -260,6 +264,28 @@ void tree_hist(at::Tensor binned, at::Tensor node_of, at::Tensor grad,
                        (int)n_bins, stream());
 }
 
+at::Tensor embedding_fwd(at::Tensor ids, at::Tensor table, at::Tensor out) {
+  TORCH_CHECK(ids.scalar_type() == at::kLong && ids.is_cuda() && ids.is_contiguous());
+  check_bf16(table, "table");
+  check_bf16(out, "out");
+  const int dim = (int)table.size(1);
+  TORCH_CHECK(dim % 8 == 0, "embedding dim must be 8-aligned");
+  TORCH_CHECK(out.numel() == ids.numel() * dim, "out shape");
+  lo::launch_embedding_fwd(ids.data_ptr(), table.data_ptr(), out.data_ptr(),
+                           ids.numel(), dim, stream());
+  return out;
+}
+
+void embedding_bwd(at::Tensor ids, at::Tensor dy, at::Tensor gtable) {
+  TORCH_CHECK(ids.scalar_type() == at::kLong && ids.is_cuda() && ids.is_contiguous());
+  check_bf16(dy, "dy");
+  check_f32(gtable, "gtable");
+  const int dim = (int)gtable.size(1);
+  TORCH_CHECK(dy.numel() == ids.numel() * dim, "dy shape");
+  lo::launch_embedding_bwd(ids.data_ptr(), dy.data_ptr(), gtable.data_ptr(),
+                           ids.numel(), dim, stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -279,6 +305,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("argmax_rows", &argmax_rows);
   m.def("accuracy_count", &accuracy_count);
   m.def("tree_hist", &tree_hist);
+  m.def("embedding_fwd", &embedding_fwd);
+  m.def("embedding_bwd", &embedding_bwd);
   m.def("softmax_ce", &softmax_ce,
         py::arg("logits"), py::arg("labels"), py::arg("dlogits"),
         py::arg("loss_sum") = py::none(), py::arg("correct") = py::none(),

@@ -117,14 +117,24 @@ class Conv2dNHWC(Layer):
 
 
 class MaxPool2dNHWC(Layer):
-    def __init__(self, k: int = 2, stride: Optional[int] = None):
-        self.k = k
-        self.stride = stride or k
+    def __init__(self, k: int = 2, stride: Optional[int] = None,
+                 kw: Optional[int] = None, sw: Optional[int] = None):
+        self.kh = k
+        self.kw = kw if kw is not None else k
+        self.sh = stride or k
+        self.sw = sw if sw is not None else (stride or self.kw)
         self._bufs = {}
+
+    # aliases used by models that retune pooling per sequence length
+    def set_window(self, kh: int, kw: int, sh: Optional[int] = None,
+                   sw: Optional[int] = None) -> None:
+        self.kh, self.kw = kh, kw
+        self.sh = sh if sh is not None else kh
+        self.sw = sw if sw is not None else kw
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         self._bufs["shape"] = x.shape
-        out, idx = F.maxpool2d(x, self.k, self.k, self.stride, self.stride)
+        out, idx = F.maxpool2d(x, self.kh, self.kw, self.sh, self.sw)
         self._bufs["idx"] = idx
         if "dx" not in self._bufs or self._bufs["dx"].shape != x.shape:
             self._bufs["dx"] = torch.empty_like(x)
@@ -132,8 +142,8 @@ class MaxPool2dNHWC(Layer):
 
     def backward(self, dy: torch.Tensor) -> torch.Tensor:
         B, H, W, C = self._bufs["shape"]
-        return F.maxpool2d_bwd(dy, self._bufs["idx"], H, W, self.k, self.k,
-                               self.stride, self.stride, out=self._bufs["dx"])
+        return F.maxpool2d_bwd(dy, self._bufs["idx"], H, W, self.kh, self.kw,
+                               self.sh, self.sw, out=self._bufs["dx"])
 
 
 class Flatten(Layer):
@@ -252,3 +262,40 @@ class SequentialClassifier:
 
     def load_state_dict(self, sd):
         self.arena.load_state_dict(sd)
+
+
+class Embedding(Layer):
+    """Token embedding gather; backward scatter-adds into the grad arena
+    (embedding.hip; fp32 atomics)."""
+
+    def __init__(self, name: str, vocab: int, dim: int):
+        assert dim % 8 == 0, "embedding dim must be 8-aligned"
+        self.name = name
+        self.vocab, self.dim = vocab, dim
+        self.arena: Optional[ParamArena] = None
+        self._ids = None
+        self._out = None
+
+    def build(self, arena: ParamArena) -> None:
+        self.arena = arena
+        g = torch.Generator(device="cpu").manual_seed(
+            zlib.crc32(self.name.encode()) % (2 ** 31))
+        arena.add(self.name + ".w", (self.vocab, self.dim),
+                  torch.randn((self.vocab, self.dim), generator=g) * 0.05)
+
+    def forward(self, ids: torch.Tensor) -> torch.Tensor:
+        from ..ops import functional as F_
+        self._ids = ids
+        if self._out is None or self._out.shape[:ids.dim()] != ids.shape:
+            self._out = torch.empty(*ids.shape, self.dim,
+                                    device=self.arena.mirror.device,
+                                    dtype=torch.bfloat16)
+        F_.embedding(ids, self.arena.p(self.name + ".w"), out=self._out)
+        return self._out
+
+    def backward(self, dy: torch.Tensor) -> None:
+        from ..ops import functional as F_
+        g = self.arena.g(self.name + ".w")
+        g.zero_()  # scatter-add accumulates; arena grads are per-step
+        F_.embedding_bwd(self._ids, dy, g)
+        return None

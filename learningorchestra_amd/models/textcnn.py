@@ -1,0 +1,112 @@
+"""Text-CNN for IMDb-style sentiment — BASELINE.json config 3
+("IMDb text-CNN sentiment train, data-parallel RCCL all-reduce on 8xMI355X").
+
+Kim-2014-style architecture on the explicit-backward engine: embedding
+gather (embedding.hip) -> parallel 1-D convolutions (kernel sizes 3/4/5 as
+NHWC convs with W=1 over the sequence axis, im2col + MFMA GEMM with fused
+bias+ReLU) -> max-over-time pooling (maxpool kernel, KH = T-k+1) -> concat ->
+padded-head linear + fused softmax-CE. Every hot op is a gfx950 HIP kernel;
+branch concat/split and the branch-grad sum are device memcpy/add glue.
+"""
+from __future__ import annotations
+
+from typing import Optional, Sequence, Tuple
+
+import torch
+
+from ..engine.arena import ParamArena
+from ..engine.layers import Conv2dNHWC, Embedding, Linear, MaxPool2dNHWC
+from ..ops import functional as F
+
+
+class TextCNN:
+    def __init__(self, vocab: int = 20000, emb_dim: int = 128,
+                 filters: int = 128, kernel_sizes: Sequence[int] = (3, 4, 5),
+                 num_classes: int = 2, device="cpu", seed: int = 0):
+        self.vocab, self.emb_dim, self.filters = vocab, emb_dim, filters
+        self.kernel_sizes = tuple(kernel_sizes)
+        self.num_classes = num_classes
+        self.cpad = max(8, (num_classes + 7) // 8 * 8)
+        self.device = torch.device(device)
+
+        self.arena = ParamArena(device)
+        self.emb = Embedding("emb", vocab, emb_dim)
+        self.convs = [Conv2dNHWC(f"conv{k}", emb_dim, filters, k, 1, relu=True)
+                      for k in self.kernel_sizes]
+        self.pools = [MaxPool2dNHWC(1) for _ in self.kernel_sizes]  # k set per fwd
+        self.fc = Linear("fc", filters * len(self.kernel_sizes), self.cpad)
+        for lay in [self.emb, *self.convs, self.fc]:
+            lay.build(self.arena)
+        self.arena.finalize(seed)
+
+        self.loss_sum = torch.zeros(1, dtype=torch.float32, device=self.device)
+        self.correct = torch.zeros(1, dtype=torch.int32, device=self.device)
+        self._bufs = {}
+
+    def _alloc(self, B: int, S: int):
+        key = (B, S)
+        if self._bufs.get("key") != key:
+            nb = len(self.kernel_sizes)
+            self._bufs = {
+                "key": key,
+                "cat": torch.empty((B, self.filters * nb), device=self.device,
+                                   dtype=torch.bfloat16),
+                "dxe": torch.empty((B, S, 1, self.emb_dim), device=self.device,
+                                   dtype=torch.bfloat16),
+                "dlogits": torch.empty((B, self.cpad), device=self.device,
+                                       dtype=torch.bfloat16),
+            }
+        return self._bufs
+
+    def forward(self, ids: torch.Tensor) -> torch.Tensor:
+        B, S = ids.shape
+        bufs = self._alloc(B, S)
+        e = self.emb.forward(ids).view(B, S, 1, self.emb_dim)
+        Fn = self.filters
+        for i, (k, conv, pool) in enumerate(zip(self.kernel_sizes, self.convs,
+                                                self.pools)):
+            y = conv.forward(e)                       # [B, S-k+1, 1, F]
+            pool.set_window(y.shape[1], 1)            # max-over-time
+            p = pool.forward(y)                       # [B, 1, 1, F]
+            bufs["cat"][:, i * Fn:(i + 1) * Fn] = p.view(B, Fn)
+        return self.fc.forward(bufs["cat"])
+
+    def train_step(self, ids: torch.Tensor, y: torch.Tensor,
+                   gscale: Optional[float] = None) -> Tuple[torch.Tensor, torch.Tensor]:
+        B, S = ids.shape
+        if gscale is None:
+            gscale = 1.0 / B
+        logits = self.forward(ids)
+        bufs = self._bufs
+        self.loss_sum.zero_()
+        self.correct.zero_()
+        F.softmax_ce(logits, y, bufs["dlogits"], self.loss_sum, self.correct,
+                     cvalid=self.num_classes, gscale=gscale)
+        dcat = self.fc.backward(bufs["dlogits"])      # [B, 3F]
+        Fn = self.filters
+        dxe = bufs["dxe"]
+        for i, (k, conv, pool) in enumerate(zip(self.kernel_sizes, self.convs,
+                                                self.pools)):
+            dpool = dcat[:, i * Fn:(i + 1) * Fn].contiguous().view(B, 1, 1, Fn)
+            dconv = pool.backward(dpool)              # [B, S-k+1, 1, F]
+            dbranch = conv.backward(dconv)            # [B, S, 1, emb]
+            if i == 0:
+                dxe.copy_(dbranch)
+            else:
+                dxe.add_(dbranch)
+        self.emb.backward(dxe.view(B, S, self.emb_dim))
+        return self.loss_sum, self.correct
+
+    @torch.no_grad()
+    def predict(self, ids: torch.Tensor) -> torch.Tensor:
+        return F.argmax_rows(self.forward(ids), self.num_classes)
+
+    def state_dict(self):
+        return self.arena.state_dict()
+
+    def load_state_dict(self, sd):
+        self.arena.load_state_dict(sd)
+
+
+def build_textcnn(device="cpu", seed: int = 0, **kw) -> TextCNN:
+    return TextCNN(device=device, seed=seed, **kw)

@@ -257,3 +257,30 @@ def softmax_ce(logits: torch.Tensor, labels: torch.Tensor,
     loss_sum += loss
     correct += (lg.argmax(1) == labels).sum().to(torch.int32)
     return dlogits, loss_sum, correct
+
+
+# ---------------------------------------------------------------- embedding
+def embedding(ids: torch.Tensor, table: torch.Tensor,
+              out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Gather rows: out[..., :] = table[ids[...], :] (bf16)."""
+    dim = table.shape[1]
+    if out is None:
+        out = torch.empty(*ids.shape, dim, device=table.device, dtype=table.dtype)
+    if _is_gpu(table):
+        lo = require_ext()
+        lo.embedding_fwd(ids.contiguous(), table, out.view(-1, dim))
+        return out
+    out.copy_(table[ids])
+    return out
+
+
+def embedding_bwd(ids: torch.Tensor, dy: torch.Tensor,
+                  gtable: torch.Tensor) -> None:
+    """Scatter-add: gtable[ids[...], :] += dy (fp32 accumulate). Caller zeros
+    gtable (the grad arena is fully rewritten each step)."""
+    dim = gtable.shape[1]
+    if _is_gpu(gtable):
+        lo = require_ext()
+        lo.embedding_bwd(ids.contiguous(), dy.contiguous().view(-1, dim), gtable)
+        return
+    gtable.index_add_(0, ids.reshape(-1), dy.float().reshape(-1, dim))
