@@ -32,7 +32,7 @@ def main() -> None:
     ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--batch", type=int, default=0, help="per-GPU batch (0 = model default)")
     ap.add_argument("--model", default="mnist-cnn",
-                    choices=["mnist-cnn", "textcnn"],
+                    choices=["mnist-cnn", "textcnn", "resnet50"],
                     help="flagship = mnist-cnn (BASELINE.json headline)")
     ap.add_argument("--no-graph", action="store_true")
     ap.add_argument("--lr", type=float, default=0.05)
@@ -44,7 +44,8 @@ def main() -> None:
     from learningorchestra_amd.engine.trainer import Trainer, make_sgd
     from learningorchestra_amd.models.mnist_cnn import build_mnist_cnn
     from learningorchestra_amd.models.textcnn import build_textcnn
-    from learningorchestra_amd.data.synthetic import imdb_batch, mnist_batch
+    from learningorchestra_amd.data.synthetic import (imagenet_batch,
+                                                      imdb_batch, mnist_batch)
 
     local_rank = init_distributed()
     world = get_world_size()
@@ -59,11 +60,17 @@ def main() -> None:
         model = build_mnist_cnn(device, seed=0)
         model_desc = "MNIST-CNN (LeNet-style, conv32-conv64-fc256)"
         extra_cfg = {"image": "28x28x1"}
-    else:
+    elif args.model == "textcnn":
         batch = args.batch or 2048
         model = build_textcnn(device, seed=0)
         model_desc = "TextCNN (IMDb sentiment, emb128, filters 128 x k3/4/5)"
         extra_cfg = {"seq_len": 256, "vocab": 20000}
+    else:
+        from learningorchestra_amd.models.resnet import build_resnet50
+        batch = args.batch or 128
+        model = build_resnet50(device, seed=0)
+        model_desc = "ResNet-50 (bottleneck v1, 1000 classes)"
+        extra_cfg = {"image": "224x224x3"}
     # graph capture: single-rank only (RCCL collectives stay outside graphs
     # until validated under capture)
     use_graph = use_gpu and world == 1 and not args.no_graph
@@ -73,9 +80,12 @@ def main() -> None:
     if args.model == "mnist-cnn":
         x, y = mnist_batch(batch, device=device, dtype=torch.bfloat16,
                            seed=1234 + rank)
-    else:
+    elif args.model == "textcnn":
         x, y = imdb_batch(batch, seq_len=256, vocab=20000, seed=1234 + rank)
         x, y = x.to(device), y.to(device)
+    else:
+        x, y = imagenet_batch(batch, device=device, dtype=torch.bfloat16,
+                              seed=1234 + rank)
 
     for _ in range(args.warmup):
         trainer.step_async(x, y)

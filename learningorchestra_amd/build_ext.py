@@ -26,6 +26,7 @@ SOURCES = [
     os.path.join(CSRC, "softmax_ce.hip"),
     os.path.join(CSRC, "tree_hist.hip"),
     os.path.join(CSRC, "embedding.hip"),
+    os.path.join(CSRC, "batchnorm.hip"),
 ]
 
 

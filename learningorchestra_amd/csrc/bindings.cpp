@@ -24,14 +24,32 @@ void launch_im2col(const void* in, void* col, int B, int H, int W, int C,
                    int KH, int KW, int SH, int SW, int PH, int PW,
                    int OH, int OW, int Kpad, hipStream_t s);
 void launch_col2im(const void* dcol, void* dx, int B, int H, int W, int C,
-                   int KH, int KW, int PH, int PW, int OH, int OW, int Kpad,
-                   hipStream_t s);
+                   int KH, int KW, int SH, int SW, int PH, int PW, int OH,
+                   int OW, int Kpad, hipStream_t s);
 void launch_maxpool_fwd(const void* in, void* out, void* idx, int B, int H, int W,
-                        int C, int KH, int KW, int SH, int SW, int OH, int OW,
-                        hipStream_t s);
+                        int C, int KH, int KW, int SH, int SW, int PH, int PW,
+                        int OH, int OW, hipStream_t s);
 void launch_maxpool_bwd(const void* dy, const void* idx, void* dx, int B, int H,
-                        int W, int C, int KH, int KW, int SH, int SW, int OH,
-                        int OW, hipStream_t s);
+                        int W, int C, int KH, int KW, int SH, int SW, int PH,
+                        int PW, int OH, int OW, hipStream_t s);
+void launch_bn_stats(const void* x, void* sum, void* sumsq, long M, int C,
+                     hipStream_t s);
+void launch_bn_fwd(const void* x, void* y, const void* mean, const void* invstd,
+                   const void* gamma, const void* beta, long M, int C, int relu,
+                   hipStream_t s);
+void launch_bn_bwd_reduce(const void* dy, const void* y, const void* x,
+                          const void* mean, const void* invstd, void* dbeta,
+                          void* dgamma, long M, int C, int relu, hipStream_t s);
+void launch_bn_bwd_dx(const void* dy, const void* y, const void* x, void* dx,
+                      const void* mean, const void* invstd, const void* gamma,
+                      const void* dbeta, const void* dgamma, long M, int C,
+                      int relu, hipStream_t s);
+void launch_add_relu(const void* a, const void* b, void* z, long n, int relu,
+                     hipStream_t s);
+void launch_avgpool_global(const void* x, void* out, int B, int HW, int C,
+                           hipStream_t s);
+void launch_avgpool_global_bwd(const void* dy, void* dx, int B, int HW, int C,
+                               hipStream_t s);
 void launch_relu_bwd(const void* dy, const void* y, void* dx, long n, hipStream_t s);
 void launch_sgd(void* master, const void* grad, void* mom, void* mirror, long n,
                 float lr, float mu, float wd, float gscale, hipStream_t s);
@@ -123,40 +141,117 @@ at::Tensor im2col(at::Tensor in, int64_t KH, int64_t KW, int64_t SH, int64_t SW,
 }
 
 at::Tensor col2im(at::Tensor dcol, int64_t B, int64_t H, int64_t W, int64_t C,
-                  int64_t KH, int64_t KW, int64_t PH, int64_t PW, at::Tensor dx) {
+                  int64_t KH, int64_t KW, int64_t SH, int64_t SW, int64_t PH,
+                  int64_t PW, at::Tensor dx) {
   check_bf16(dcol, "dcol");
   check_bf16(dx, "dx");
-  const int OH = (int)(H + 2 * PH - KH) + 1, OW = (int)(W + 2 * PW - KW) + 1;
+  const int OH = (int)((H + 2 * PH - KH) / SH) + 1;
+  const int OW = (int)((W + 2 * PW - KW) / SW) + 1;
   TORCH_CHECK(dcol.size(0) == B * OH * OW, "dcol rows");
   lo::launch_col2im(dcol.data_ptr(), dx.data_ptr(), (int)B, (int)H, (int)W, (int)C,
-                    (int)KH, (int)KW, (int)PH, (int)PW, OH, OW,
+                    (int)KH, (int)KW, (int)SH, (int)SW, (int)PH, (int)PW, OH, OW,
                     (int)dcol.size(1), stream());
   return dx;
 }
 
 std::vector<at::Tensor> maxpool_fwd(at::Tensor in, int64_t KH, int64_t KW,
-                                    int64_t SH, int64_t SW) {
+                                    int64_t SH, int64_t SW, int64_t PH, int64_t PW) {
   check_bf16(in, "in");
   const int B = (int)in.size(0), H = (int)in.size(1), W = (int)in.size(2),
             C = (int)in.size(3);
-  const int OH = (H - (int)KH) / (int)SH + 1, OW = (W - (int)KW) / (int)SW + 1;
+  const int OH = (H + 2 * (int)PH - (int)KH) / (int)SH + 1;
+  const int OW = (W + 2 * (int)PW - (int)KW) / (int)SW + 1;
   auto out = at::empty({B, OH, OW, C}, in.options());
   auto idx = at::empty({B, OH, OW, C}, in.options().dtype(at::kByte));
   lo::launch_maxpool_fwd(in.data_ptr(), out.data_ptr(), idx.data_ptr(), B, H, W, C,
-                         (int)KH, (int)KW, (int)SH, (int)SW, OH, OW, stream());
+                         (int)KH, (int)KW, (int)SH, (int)SW, (int)PH, (int)PW,
+                         OH, OW, stream());
   return {out, idx};
 }
 
 at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor idx, int64_t H, int64_t W,
                        int64_t KH, int64_t KW, int64_t SH, int64_t SW,
-                       at::Tensor dx) {
+                       int64_t PH, int64_t PW, at::Tensor dx) {
   check_bf16(dy, "dy");
   check_bf16(dx, "dx");
   const int B = (int)dy.size(0), OH = (int)dy.size(1), OW = (int)dy.size(2),
             C = (int)dy.size(3);
   lo::launch_maxpool_bwd(dy.data_ptr(), idx.data_ptr(), dx.data_ptr(), B, (int)H,
-                         (int)W, C, (int)KH, (int)KW, (int)SH, (int)SW, OH, OW,
-                         stream());
+                         (int)W, C, (int)KH, (int)KW, (int)SH, (int)SW,
+                         (int)PH, (int)PW, OH, OW, stream());
+  return dx;
+}
+
+// ----------------------------------------------------------- batchnorm ----
+void bn_stats(at::Tensor x, at::Tensor sum, at::Tensor sumsq) {
+  check_bf16(x, "x");
+  check_f32(sum, "sum");
+  check_f32(sumsq, "sumsq");
+  const int C = (int)x.size(-1);
+  TORCH_CHECK(C % 8 == 0 && sum.numel() == C && sumsq.numel() == C);
+  lo::launch_bn_stats(x.data_ptr(), sum.data_ptr(), sumsq.data_ptr(),
+                      x.numel() / C, C, stream());
+}
+
+void bn_fwd(at::Tensor x, at::Tensor y, at::Tensor mean, at::Tensor invstd,
+            at::Tensor gamma, at::Tensor beta, bool relu) {
+  check_bf16(x, "x");
+  check_bf16(y, "y");
+  const int C = (int)x.size(-1);
+  lo::launch_bn_fwd(x.data_ptr(), y.data_ptr(), mean.data_ptr(),
+                    invstd.data_ptr(), gamma.data_ptr(), beta.data_ptr(),
+                    x.numel() / C, C, relu ? 1 : 0, stream());
+}
+
+void bn_bwd_reduce(at::Tensor dy, at::Tensor y, at::Tensor x, at::Tensor mean,
+                   at::Tensor invstd, at::Tensor dbeta, at::Tensor dgamma,
+                   bool relu) {
+  check_bf16(dy, "dy");
+  const int C = (int)x.size(-1);
+  lo::launch_bn_bwd_reduce(dy.data_ptr(), y.data_ptr(), x.data_ptr(),
+                           mean.data_ptr(), invstd.data_ptr(), dbeta.data_ptr(),
+                           dgamma.data_ptr(), x.numel() / C, C, relu ? 1 : 0,
+                           stream());
+}
+
+void bn_bwd_dx(at::Tensor dy, at::Tensor y, at::Tensor x, at::Tensor dx,
+               at::Tensor mean, at::Tensor invstd, at::Tensor gamma,
+               at::Tensor dbeta, at::Tensor dgamma, bool relu) {
+  check_bf16(dx, "dx");
+  const int C = (int)x.size(-1);
+  lo::launch_bn_bwd_dx(dy.data_ptr(), y.data_ptr(), x.data_ptr(), dx.data_ptr(),
+                       mean.data_ptr(), invstd.data_ptr(), gamma.data_ptr(),
+                       dbeta.data_ptr(), dgamma.data_ptr(), x.numel() / C, C,
+                       relu ? 1 : 0, stream());
+}
+
+void add_relu(at::Tensor a, at::Tensor b, at::Tensor z, bool relu) {
+  check_bf16(a, "a");
+  check_bf16(b, "b");
+  check_bf16(z, "z");
+  TORCH_CHECK(a.numel() == b.numel() && a.numel() == z.numel() &&
+              a.numel() % 8 == 0);
+  lo::launch_add_relu(a.data_ptr(), b.data_ptr(), z.data_ptr(), a.numel(),
+                      relu ? 1 : 0, stream());
+}
+
+at::Tensor avgpool_global(at::Tensor x, at::Tensor out) {
+  check_bf16(x, "x");
+  check_bf16(out, "out");
+  const int B = (int)x.size(0);
+  const int C = (int)x.size(-1);
+  const int HW = (int)(x.numel() / ((long)B * C));
+  lo::launch_avgpool_global(x.data_ptr(), out.data_ptr(), B, HW, C, stream());
+  return out;
+}
+
+at::Tensor avgpool_global_bwd(at::Tensor dy, at::Tensor dx) {
+  check_bf16(dy, "dy");
+  check_bf16(dx, "dx");
+  const int B = (int)dy.size(0);
+  const int C = (int)dy.size(-1);
+  const int HW = (int)(dx.numel() / ((long)B * C));
+  lo::launch_avgpool_global_bwd(dy.data_ptr(), dx.data_ptr(), B, HW, C, stream());
   return dx;
 }
 
@@ -306,6 +401,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("accuracy_count", &accuracy_count);
   m.def("tree_hist", &tree_hist);
   m.def("embedding_fwd", &embedding_fwd);
+  m.def("bn_stats", &bn_stats);
+  m.def("bn_fwd", &bn_fwd);
+  m.def("bn_bwd_reduce", &bn_bwd_reduce);
+  m.def("bn_bwd_dx", &bn_bwd_dx);
+  m.def("add_relu", &add_relu);
+  m.def("avgpool_global", &avgpool_global);
+  m.def("avgpool_global_bwd", &avgpool_global_bwd);
   m.def("embedding_bwd", &embedding_bwd);
   m.def("softmax_ce", &softmax_ce,
         py::arg("logits"), py::arg("labels"), py::arg("dlogits"),

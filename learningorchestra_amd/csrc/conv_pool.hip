@@ -68,12 +68,14 @@ void launch_im2col(const void* in, void* col, int B, int H, int W, int C,
 }
 
 // --------------------------------------------------------------- col2im ----
-// stride-1 gather: dX[b,h,w,c] = sum over (kh,kw) with oh=h+PH-kh in [0,OH),
-// ow=w+PW-kw in [0,OW) of dcol[row(b,oh,ow)][(kh*KW+kw)*C + c]
+// gather form for any stride: dX[b,h,w,c] = sum over (kh,kw) with
+// oh = (h+PH-kh)/SH integral in [0,OH) (likewise ow) of
+// dcol[row(b,oh,ow)][(kh*KW+kw)*C + c] — no atomics.
 template <bool VEC8>
 __global__ void col2im_kernel(const bf16* __restrict__ dcol, bf16* __restrict__ dx,
                               int B, int H, int W, int C, int KH, int KW,
-                              int PH, int PW, int OH, int OW, int Kpad) {
+                              int SH, int SW, int PH, int PW, int OH, int OW,
+                              int Kpad) {
   const int CV = VEC8 ? C / 8 : C;
   const long total = (long)B * H * W * CV;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
@@ -85,11 +87,15 @@ __global__ void col2im_kernel(const bf16* __restrict__ dcol, bf16* __restrict__ 
     const int b = r;
     float acc[VEC8 ? 8 : 1] = {};
     for (int kh = 0; kh < KH; ++kh) {
-      const int oh = h + PH - kh;
-      if (oh < 0 || oh >= OH) continue;
+      const int ohs = h + PH - kh;
+      if (ohs < 0 || ohs % SH != 0) continue;
+      const int oh = ohs / SH;
+      if (oh >= OH) continue;
       for (int kw = 0; kw < KW; ++kw) {
-        const int ow = w + PW - kw;
-        if (ow < 0 || ow >= OW) continue;
+        const int ows = w + PW - kw;
+        if (ows < 0 || ows % SW != 0) continue;
+        const int ow = ows / SW;
+        if (ow >= OW) continue;
         const long row = ((long)b * OH + oh) * OW + ow;
         const long kidx = ((long)kh * KW + kw) * C + cu * (VEC8 ? 8 : 1);
         if (VEC8) {
@@ -113,18 +119,18 @@ __global__ void col2im_kernel(const bf16* __restrict__ dcol, bf16* __restrict__ 
 }
 
 void launch_col2im(const void* dcol, void* dx, int B, int H, int W, int C,
-                   int KH, int KW, int PH, int PW, int OH, int OW, int Kpad,
-                   hipStream_t s) {
+                   int KH, int KW, int SH, int SW, int PH, int PW, int OH,
+                   int OW, int Kpad, hipStream_t s) {
   const bool vec = (C % 8 == 0);
   const long total = (long)B * H * W * (vec ? C / 8 : C);
   const int block = 256;
   const int grid = (int)min((total + block - 1) / block, (long)2048);
   if (vec)
     hipLaunchKernelGGL(HIP_KERNEL_NAME(col2im_kernel<true>), dim3(grid), dim3(block), 0, s,
-                       (const bf16*)dcol, (bf16*)dx, B, H, W, C, KH, KW, PH, PW, OH, OW, Kpad);
+                       (const bf16*)dcol, (bf16*)dx, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW, Kpad);
   else
     hipLaunchKernelGGL(HIP_KERNEL_NAME(col2im_kernel<false>), dim3(grid), dim3(block), 0, s,
-                       (const bf16*)dcol, (bf16*)dx, B, H, W, C, KH, KW, PH, PW, OH, OW, Kpad);
+                       (const bf16*)dcol, (bf16*)dx, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW, Kpad);
 }
 
 // ------------------------------------------------------------- maxpool -----
@@ -132,7 +138,7 @@ template <bool VEC8>
 __global__ void maxpool_fwd_kernel(const bf16* __restrict__ in, bf16* __restrict__ out,
                                    unsigned char* __restrict__ idx,
                                    int B, int H, int W, int C, int KH, int KW,
-                                   int SH, int SW, int OH, int OW) {
+                                   int SH, int SW, int PH, int PW, int OH, int OW) {
   const int CV = VEC8 ? C / 8 : C;
   const long total = (long)B * OH * OW * CV;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
@@ -148,11 +154,11 @@ __global__ void maxpool_fwd_kernel(const bf16* __restrict__ in, bf16* __restrict
     #pragma unroll
     for (int j = 0; j < NE; ++j) { best[j] = -3.0e38f; bidx[j] = 0; }
     for (int kh = 0; kh < KH; ++kh) {
-      const int h = oh * SH + kh;
-      if (h >= H) continue;
+      const int h = oh * SH + kh - PH;
+      if (h < 0 || h >= H) continue;
       for (int kw = 0; kw < KW; ++kw) {
-        const int w = ow * SW + kw;
-        if (w >= W) continue;
+        const int w = ow * SW + kw - PW;
+        if (w < 0 || w >= W) continue;
         const long base = (((long)b * H + h) * W + w) * C + cu * NE;
         if (VEC8) {
           bf16x8 v = *(const bf16x8*)(in + base);
@@ -189,7 +195,7 @@ __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
                                    const unsigned char* __restrict__ idx,
                                    bf16* __restrict__ dx,
                                    int B, int H, int W, int C, int KH, int KW,
-                                   int SH, int SW, int OH, int OW) {
+                                   int SH, int SW, int PH, int PW, int OH, int OW) {
   const int CV = VEC8 ? C / 8 : C;
   const long total = (long)B * H * W * CV;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
@@ -203,13 +209,14 @@ __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
     float acc[NE];
     #pragma unroll
     for (int j = 0; j < NE; ++j) acc[j] = 0.f;
-    const int oh_lo = max(0, (h - KH + SH) / SH), oh_hi = min(OH - 1, h / SH);
-    const int ow_lo = max(0, (w - KW + SW) / SW), ow_hi = min(OW - 1, w / SW);
+    const int hp = h + PH, wp = w + PW;
+    const int oh_lo = max(0, (hp - KH + SH) / SH), oh_hi = min(OH - 1, hp / SH);
+    const int ow_lo = max(0, (wp - KW + SW) / SW), ow_hi = min(OW - 1, wp / SW);
     for (int oh = oh_lo; oh <= oh_hi; ++oh) {
-      const int kh = h - oh * SH;
+      const int kh = hp - oh * SH;
       if (kh < 0 || kh >= KH) continue;
       for (int ow = ow_lo; ow <= ow_hi; ++ow) {
-        const int kw = w - ow * SW;
+        const int kw = wp - ow * SW;
         if (kw < 0 || kw >= KW) continue;
         const long obase = (((long)b * OH + oh) * OW + ow) * C + cu * NE;
         const unsigned char want = (unsigned char)(kh * KW + kw);
@@ -236,8 +243,8 @@ __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
 }
 
 void launch_maxpool_fwd(const void* in, void* out, void* idx, int B, int H, int W,
-                        int C, int KH, int KW, int SH, int SW, int OH, int OW,
-                        hipStream_t s) {
+                        int C, int KH, int KW, int SH, int SW, int PH, int PW,
+                        int OH, int OW, hipStream_t s) {
   const bool vec = (C % 8 == 0);
   const long total = (long)B * OH * OW * (vec ? C / 8 : C);
   const int block = 256;
@@ -245,16 +252,16 @@ void launch_maxpool_fwd(const void* in, void* out, void* idx, int B, int H, int 
   if (vec)
     hipLaunchKernelGGL(HIP_KERNEL_NAME(maxpool_fwd_kernel<true>), dim3(grid), dim3(block), 0, s,
                        (const bf16*)in, (bf16*)out, (unsigned char*)idx,
-                       B, H, W, C, KH, KW, SH, SW, OH, OW);
+                       B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
   else
     hipLaunchKernelGGL(HIP_KERNEL_NAME(maxpool_fwd_kernel<false>), dim3(grid), dim3(block), 0, s,
                        (const bf16*)in, (bf16*)out, (unsigned char*)idx,
-                       B, H, W, C, KH, KW, SH, SW, OH, OW);
+                       B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
 }
 
 void launch_maxpool_bwd(const void* dy, const void* idx, void* dx, int B, int H,
-                        int W, int C, int KH, int KW, int SH, int SW, int OH,
-                        int OW, hipStream_t s) {
+                        int W, int C, int KH, int KW, int SH, int SW, int PH,
+                        int PW, int OH, int OW, hipStream_t s) {
   const bool vec = (C % 8 == 0);
   const long total = (long)B * H * W * (vec ? C / 8 : C);
   const int block = 256;
@@ -262,11 +269,11 @@ void launch_maxpool_bwd(const void* dy, const void* idx, void* dx, int B, int H,
   if (vec)
     hipLaunchKernelGGL(HIP_KERNEL_NAME(maxpool_bwd_kernel<true>), dim3(grid), dim3(block), 0, s,
                        (const bf16*)dy, (const unsigned char*)idx, (bf16*)dx,
-                       B, H, W, C, KH, KW, SH, SW, OH, OW);
+                       B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
   else
     hipLaunchKernelGGL(HIP_KERNEL_NAME(maxpool_bwd_kernel<false>), dim3(grid), dim3(block), 0, s,
                        (const bf16*)dy, (const unsigned char*)idx, (bf16*)dx,
-                       B, H, W, C, KH, KW, SH, SW, OH, OW);
+                       B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
 }
 
 }  // namespace lo

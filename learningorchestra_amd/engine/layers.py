@@ -42,14 +42,18 @@ class Conv2dNHWC(Layer):
 
     def __init__(self, name: str, in_c: int, out_c: int, kh: int, kw: int,
                  stride: int = 1, pad: int = 0, relu: bool = True,
-                 first: bool = False):
+                 first: bool = False, bias: bool = True):
         self.name = name
+        self.bias = bias
         self.in_c, self.out_c = in_c, out_c
         self.kh, self.kw, self.stride, self.pad = kh, kw, stride, pad
         self.relu = relu
         self.first = first  # input layer: skip dX
         self.kdim = kh * kw * in_c
         self.kpad = _pad8(self.kdim)
+        # 1x1 stride-1 convs skip im2col entirely (col IS the input)
+        self._is_1x1 = (kh == 1 and kw == 1 and stride == 1 and pad == 0
+                        and self.kpad == in_c)
         self.arena: Optional[ParamArena] = None
         self._bufs = {}
 
@@ -66,7 +70,8 @@ class Conv2dNHWC(Layer):
             return t
 
         arena.add(self.name + ".w", (self.out_c, self.kpad), init_w)
-        arena.add(self.name + ".b", (self.out_c,), torch.zeros(self.out_c))
+        if self.bias:
+            arena.add(self.name + ".b", (self.out_c,), torch.zeros(self.out_c))
 
     def _alloc(self, B: int, H: int, W: int, dev, dtype):
         OH = (H + 2 * self.pad - self.kh) // self.stride + 1
@@ -76,24 +81,31 @@ class Conv2dNHWC(Layer):
             M = B * OH * OW
             self._bufs = {
                 "key": key, "B": B, "H": H, "W": W, "OH": OH, "OW": OW,
-                # col pad tail stays zero forever (im2col never writes it)
-                "col": torch.zeros((M, self.kpad), device=dev, dtype=dtype),
                 "y": torch.empty((M, self.out_c), device=dev, dtype=dtype),
             }
+            if not self._is_1x1:
+                # col pad tail stays zero forever (im2col never writes it)
+                self._bufs["col"] = torch.zeros((M, self.kpad), device=dev, dtype=dtype)
             if not self.first:
-                self._bufs["dcol"] = torch.empty((M, self.kpad), device=dev, dtype=dtype)
                 self._bufs["dx"] = torch.empty((B, H, W, self.in_c), device=dev, dtype=dtype)
+                if not self._is_1x1:
+                    self._bufs["dcol"] = torch.empty((M, self.kpad), device=dev, dtype=dtype)
         return self._bufs
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         B, H, W, C = x.shape
         assert C == self.in_c
         bufs = self._alloc(B, H, W, x.device, x.dtype)
-        F.im2col(x, self.kh, self.kw, self.stride, self.stride, self.pad,
-                 self.pad, self.kpad, out=bufs["col"])
-        F.gemm(bufs["col"], self.arena.p(self.name + ".w"), tb=True,
-               bias=self.arena.pf(self.name + ".b"), relu=self.relu,
-               out=bufs["y"])
+        if self._is_1x1:
+            col = x.view(B * H * W, C)
+            bufs["col"] = col
+        else:
+            F.im2col(x, self.kh, self.kw, self.stride, self.stride, self.pad,
+                     self.pad, self.kpad, out=bufs["col"])
+            col = bufs["col"]
+        F.gemm(col, self.arena.p(self.name + ".w"), tb=True,
+               bias=self.arena.pf(self.name + ".b") if self.bias else None,
+               relu=self.relu, out=bufs["y"])
         return bufs["y"].view(B, bufs["OH"], bufs["OW"], self.out_c)
 
     def backward(self, dy: torch.Tensor) -> Optional[torch.Tensor]:
@@ -106,23 +118,31 @@ class Conv2dNHWC(Layer):
         gw = self.arena.g(self.name + ".w")
         splits = _splitk_heuristic(self.out_c, self.kpad, M)
         F.gemm(dy2, bufs["col"], ta=True, out=gw, splits=splits)
-        F.colsum(dy2, out=self.arena.g(self.name + ".b"))
+        if self.bias:
+            F.colsum(dy2, out=self.arena.g(self.name + ".b"))
         if self.first:
             return None
-        assert self.stride == 1, "col2im gather path is stride-1"
+        if self._is_1x1:
+            # 1x1/s1 conv: col IS x, so dcol IS dx
+            dx = bufs["dx"]
+            F.gemm(dy2, self.arena.p(self.name + ".w"), out=dx.view(M, self.in_c))
+            return dx
         F.gemm(dy2, self.arena.p(self.name + ".w"), out=bufs["dcol"])
         F.col2im(bufs["dcol"], bufs["B"], bufs["H"], bufs["W"], self.in_c,
-                 self.kh, self.kw, self.pad, self.pad, out=bufs["dx"])
+                 self.kh, self.kw, self.stride, self.stride, self.pad,
+                 self.pad, out=bufs["dx"])
         return bufs["dx"]
 
 
 class MaxPool2dNHWC(Layer):
     def __init__(self, k: int = 2, stride: Optional[int] = None,
-                 kw: Optional[int] = None, sw: Optional[int] = None):
+                 kw: Optional[int] = None, sw: Optional[int] = None,
+                 pad: int = 0):
         self.kh = k
         self.kw = kw if kw is not None else k
         self.sh = stride or k
         self.sw = sw if sw is not None else (stride or self.kw)
+        self.ph = self.pw = pad
         self._bufs = {}
 
     # aliases used by models that retune pooling per sequence length
@@ -134,7 +154,8 @@ class MaxPool2dNHWC(Layer):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         self._bufs["shape"] = x.shape
-        out, idx = F.maxpool2d(x, self.kh, self.kw, self.sh, self.sw)
+        out, idx = F.maxpool2d(x, self.kh, self.kw, self.sh, self.sw,
+                               self.ph, self.pw)
         self._bufs["idx"] = idx
         if "dx" not in self._bufs or self._bufs["dx"].shape != x.shape:
             self._bufs["dx"] = torch.empty_like(x)
@@ -143,7 +164,8 @@ class MaxPool2dNHWC(Layer):
     def backward(self, dy: torch.Tensor) -> torch.Tensor:
         B, H, W, C = self._bufs["shape"]
         return F.maxpool2d_bwd(dy, self._bufs["idx"], H, W, self.kh, self.kw,
-                               self.sh, self.sw, out=self._bufs["dx"])
+                               self.sh, self.sw, self.ph, self.pw,
+                               out=self._bufs["dx"])
 
 
 class Flatten(Layer):
@@ -299,3 +321,98 @@ class Embedding(Layer):
         g.zero_()  # scatter-add accumulates; arena grads are per-step
         F_.embedding_bwd(self._ids, dy, g)
         return None
+
+
+class BatchNormReLU(Layer):
+    '''Per-channel BN (+fused ReLU) over NHWC activations (batchnorm.hip).
+    Running stats live on the layer (non-trainable state, included in the
+    model's extra-state checkpoint); gamma/beta in the arena.'''
+
+    def __init__(self, name: str, channels: int, relu: bool = True,
+                 eps: float = 1e-5, momentum: float = 0.1):
+        assert channels % 8 == 0
+        self.name = name
+        self.c = channels
+        self.relu = relu
+        self.eps = eps
+        self.momentum = momentum
+        self.training = True
+        self.arena: Optional[ParamArena] = None
+        self._bufs = {}
+
+    def build(self, arena: ParamArena) -> None:
+        self.arena = arena
+        arena.add(self.name + ".g", (self.c,), torch.ones(self.c))
+        arena.add(self.name + ".b", (self.c,), torch.zeros(self.c))
+
+    def _alloc(self, shape, dev):
+        if self._bufs.get("shape") != shape:
+            M = 1
+            for d in shape[:-1]:
+                M *= d
+            self._bufs = {
+                "shape": shape, "M": M,
+                "y": torch.empty((M, self.c), device=dev, dtype=torch.bfloat16),
+                "dx": torch.empty((M, self.c), device=dev, dtype=torch.bfloat16),
+                "mean": torch.zeros(self.c, device=dev),
+                "invstd": torch.ones(self.c, device=dev),
+                "scratch": torch.zeros((2, self.c), device=dev),
+                "running_mean": torch.zeros(self.c, device=dev),
+                "running_var": torch.ones(self.c, device=dev),
+            }
+        return self._bufs
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        bufs = self._alloc(tuple(x.shape), x.device)
+        x2 = x.reshape(bufs["M"], self.c)
+        self._x2 = x2
+        gamma = self.arena.pf(self.name + ".g")
+        beta = self.arena.pf(self.name + ".b")
+        if self.training:
+            F.bn_fwd_train(x2, gamma, beta, self.eps, bufs["y"], bufs["mean"],
+                           bufs["invstd"], bufs["scratch"], self.relu)
+            m = self.momentum
+            bufs["running_mean"].mul_(1 - m).add_(bufs["mean"], alpha=m)
+            var = bufs["invstd"].square().reciprocal() - self.eps
+            bufs["running_var"].mul_(1 - m).add_(var, alpha=m)
+        else:
+            F.bn_fwd_eval(x2, gamma, beta, bufs["running_mean"],
+                          bufs["running_var"], self.eps, bufs["y"], self.relu)
+        return bufs["y"].view(x.shape)
+
+    def backward(self, dy: torch.Tensor) -> torch.Tensor:
+        bufs = self._bufs
+        dy2 = dy.reshape(bufs["M"], self.c)
+        F.bn_bwd(dy2, bufs["y"], self._x2, bufs["mean"], bufs["invstd"],
+                 self.arena.pf(self.name + ".g"),
+                 self.arena.g(self.name + ".g"), self.arena.g(self.name + ".b"),
+                 bufs["dx"], self.relu)
+        return bufs["dx"].view(dy.shape)
+
+    # checkpointable non-arena state
+    def extra_state(self):
+        if not self._bufs:
+            return {}
+        return {self.name + ".running_mean": self._bufs["running_mean"].cpu(),
+                self.name + ".running_var": self._bufs["running_var"].cpu()}
+
+    def load_extra_state(self, sd):
+        if self._bufs and self.name + ".running_mean" in sd:
+            self._bufs["running_mean"].copy_(sd[self.name + ".running_mean"])
+            self._bufs["running_var"].copy_(sd[self.name + ".running_var"])
+
+
+class AvgPoolGlobal(Layer):
+    '''Global average pool [B,H,W,C] -> [B,C] (batchnorm.hip kernels).'''
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        self._shape = x.shape
+        B, H, W, C = x.shape
+        if not hasattr(self, "_out") or self._out.shape[0] != B:
+            self._out = torch.empty((B, C), device=x.device, dtype=x.dtype)
+            self._dx = torch.empty_like(x)
+        return F.avgpool_global(x, out=self._out)
+
+    def backward(self, dy: torch.Tensor) -> torch.Tensor:
+        B, H, W, C = self._shape
+        return F.avgpool_global_bwd(dy, H, W, out=self._dx)

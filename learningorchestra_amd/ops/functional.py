@@ -86,37 +86,41 @@ def im2col(x: torch.Tensor, kh: int, kw: int, sh: int, sw: int, ph: int, pw: int
 
 
 def col2im(dcol: torch.Tensor, B: int, H: int, W: int, C: int, kh: int, kw: int,
-           ph: int, pw: int, out: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """Stride-1 col2im (gather): dcol [B*OH*OW, kpad] -> dx [B,H,W,C]."""
-    OH = H + 2 * ph - kh + 1
-    OW = W + 2 * pw - kw + 1
+           sh: int, sw: int, ph: int, pw: int,
+           out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Gather-form col2im for any stride: dcol [B*OH*OW, kpad] -> dx [B,H,W,C]."""
+    OH = (H + 2 * ph - kh) // sh + 1
+    OW = (W + 2 * pw - kw) // sw + 1
     if out is None:
         out = torch.empty((B, H, W, C), device=dcol.device, dtype=dcol.dtype)
     if _is_gpu(dcol):
         lo = require_ext()
-        lo.col2im(dcol, B, H, W, C, kh, kw, ph, pw, out)
+        lo.col2im(dcol, B, H, W, C, kh, kw, sh, sw, ph, pw, out)
         return out
     K = kh * kw * C
     cols = dcol[:, :K].float().reshape(B, OH * OW, kh * kw, C)
     cols = cols.permute(0, 3, 2, 1).reshape(B, C * kh * kw, OH * OW)
-    xn = torch.nn.functional.fold(cols, (H, W), (kh, kw), padding=(ph, pw))
+    xn = torch.nn.functional.fold(cols, (H, W), (kh, kw), padding=(ph, pw),
+                                  stride=(sh, sw))
     out.copy_(xn.permute(0, 2, 3, 1).to(out.dtype))
     return out
 
 
 # ------------------------------------------------------------------ maxpool
-def maxpool2d(x: torch.Tensor, kh: int, kw: int, sh: int, sw: int
-              ) -> Tuple[torch.Tensor, torch.Tensor]:
-    """NHWC maxpool; returns (out, idx u8 of kh*KW+kw argmax)."""
+def maxpool2d(x: torch.Tensor, kh: int, kw: int, sh: int, sw: int,
+              ph: int = 0, pw: int = 0) -> Tuple[torch.Tensor, torch.Tensor]:
+    """NHWC maxpool (zero-pad treated as -inf); returns (out, idx u8 of
+    kh*KW+kw argmax)."""
     if _is_gpu(x):
         lo = require_ext()
-        out, idx = lo.maxpool_fwd(x, kh, kw, sh, sw)
+        out, idx = lo.maxpool_fwd(x, kh, kw, sh, sw, ph, pw)
         return out, idx
     B, H, W, C = x.shape
-    OH = (H - kh) // sh + 1
-    OW = (W - kw) // sw + 1
+    OH = (H + 2 * ph - kh) // sh + 1
+    OW = (W + 2 * pw - kw) // sw + 1
     xn = x.permute(0, 3, 1, 2).float()
     out_n, ind = torch.nn.functional.max_pool2d(xn, (kh, kw), (sh, sw),
+                                                padding=(ph, pw),
                                                 return_indices=True)
     out = out_n.permute(0, 2, 3, 1).to(x.dtype)
     # flat NCHW index -> (kh,kw) offset index
@@ -124,28 +128,28 @@ def maxpool2d(x: torch.Tensor, kh: int, kw: int, sh: int, sw: int
     ww = (ind % W)
     oh = torch.arange(OH, device=x.device).view(1, 1, OH, 1)
     ow = torch.arange(OW, device=x.device).view(1, 1, 1, OW)
-    rel = (hh - oh * sh) * kw + (ww - ow * sw)
+    rel = (hh - (oh * sh - ph)) * kw + (ww - (ow * sw - pw))
     idx = rel.permute(0, 2, 3, 1).to(torch.uint8).contiguous()
     return out, idx
 
 
 def maxpool2d_bwd(dy: torch.Tensor, idx: torch.Tensor, H: int, W: int,
-                  kh: int, kw: int, sh: int, sw: int,
+                  kh: int, kw: int, sh: int, sw: int, ph: int = 0, pw: int = 0,
                   out: Optional[torch.Tensor] = None) -> torch.Tensor:
     B, OH, OW, C = dy.shape
     if out is None:
         out = torch.empty((B, H, W, C), device=dy.device, dtype=dy.dtype)
     if _is_gpu(dy):
         lo = require_ext()
-        lo.maxpool_bwd(dy, idx, H, W, kh, kw, sh, sw, out)
+        lo.maxpool_bwd(dy, idx, H, W, kh, kw, sh, sw, ph, pw, out)
         return out
     dx = torch.zeros((B, H, W, C), dtype=torch.float32)
     rel = idx.long()
     khh, kww = rel // kw, rel % kw
     oh = torch.arange(OH).view(1, OH, 1, 1)
     ow = torch.arange(OW).view(1, 1, OW, 1)
-    hsrc = oh * sh + khh
-    wsrc = ow * sw + kww
+    hsrc = oh * sh + khh - ph
+    wsrc = ow * sw + kww - pw
     b = torch.arange(B).view(B, 1, 1, 1).expand_as(rel)
     c = torch.arange(C).view(1, 1, 1, C).expand_as(rel)
     dx.index_put_((b.reshape(-1), hsrc.reshape(-1), wsrc.reshape(-1),
@@ -284,3 +288,111 @@ def embedding_bwd(ids: torch.Tensor, dy: torch.Tensor,
         lo.embedding_bwd(ids.contiguous(), dy.contiguous().view(-1, dim), gtable)
         return
     gtable.index_add_(0, ids.reshape(-1), dy.float().reshape(-1, dim))
+
+
+# --------------------------------------------------------------- batchnorm
+def bn_fwd_train(x2d: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
+                 eps: float, out: torch.Tensor, mean: torch.Tensor,
+                 invstd: torch.Tensor, scratch: torch.Tensor,
+                 relu: bool = True) -> None:
+    """Training-mode BN over [M, C] (+fused ReLU). Writes out (bf16), mean,
+    invstd (fp32 [C]); ``scratch`` is a [2, C] fp32 workspace (sum/sumsq)."""
+    M, C = x2d.shape
+    if _is_gpu(x2d):
+        lo = require_ext()
+        lo.bn_stats(x2d, scratch[0], scratch[1])
+        mean.copy_(scratch[0] / M)
+        invstd.copy_((scratch[1] / M - mean.square()).clamp_(min=0)
+                     .add_(eps).rsqrt_())
+        lo.bn_fwd(x2d, out, mean, invstd, gamma, beta, relu)
+        return
+    xf = x2d.float()
+    mean.copy_(xf.mean(0))
+    invstd.copy_((xf.var(0, unbiased=False) + eps).rsqrt())
+    y = (xf - mean) * invstd * gamma + beta
+    if relu:
+        y = torch.relu(y)
+    out.copy_(y.to(out.dtype))
+
+
+def bn_fwd_eval(x2d: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
+                running_mean: torch.Tensor, running_var: torch.Tensor,
+                eps: float, out: torch.Tensor, relu: bool = True) -> None:
+    invstd = (running_var + eps).rsqrt()
+    if _is_gpu(x2d):
+        lo = require_ext()
+        lo.bn_fwd(x2d, out, running_mean.contiguous(), invstd.contiguous(),
+                  gamma, beta, relu)
+        return
+    y = (x2d.float() - running_mean) * invstd * gamma + beta
+    if relu:
+        y = torch.relu(y)
+    out.copy_(y.to(out.dtype))
+
+
+def bn_bwd(dy2d: torch.Tensor, y2d: torch.Tensor, x2d: torch.Tensor,
+           mean: torch.Tensor, invstd: torch.Tensor, gamma: torch.Tensor,
+           dgamma: torch.Tensor, dbeta: torch.Tensor, dx: torch.Tensor,
+           relu: bool = True) -> None:
+    """BN backward (ReLU-fused variant masks dy by y>0). Writes dgamma,
+    dbeta (fp32 [C], overwritten) and dx (bf16 [M,C])."""
+    M, C = x2d.shape
+    if _is_gpu(x2d):
+        lo = require_ext()
+        lo.bn_bwd_reduce(dy2d, y2d, x2d, mean, invstd, dbeta, dgamma, relu)
+        lo.bn_bwd_dx(dy2d, y2d, x2d, dx, mean, invstd, gamma, dbeta, dgamma,
+                     relu)
+        return
+    g = dy2d.float()
+    if relu:
+        g = g * (y2d.float() > 0)
+    xhat = (x2d.float() - mean) * invstd
+    dbeta.copy_(g.sum(0))
+    dgamma.copy_((g * xhat).sum(0))
+    d = gamma * invstd * (g - dbeta / M - xhat * dgamma / M)
+    dx.copy_(d.to(dx.dtype))
+
+
+# ------------------------------------------------------------ residual add
+def add_relu(a: torch.Tensor, b: torch.Tensor, out: Optional[torch.Tensor] = None,
+             relu: bool = True) -> torch.Tensor:
+    if out is None:
+        out = torch.empty_like(a)
+    if _is_gpu(a):
+        lo = require_ext()
+        lo.add_relu(a.view(-1), b.view(-1), out.view(-1), relu)
+        return out
+    z = a.float() + b.float()
+    if relu:
+        z = torch.relu(z)
+    out.copy_(z.to(out.dtype))
+    return out
+
+
+# ------------------------------------------------------ global average pool
+def avgpool_global(x: torch.Tensor, out: Optional[torch.Tensor] = None
+                   ) -> torch.Tensor:
+    """[B,H,W,C] -> [B,C] mean over H*W."""
+    B, H, W, C = x.shape
+    if out is None:
+        out = torch.empty((B, C), device=x.device, dtype=x.dtype)
+    if _is_gpu(x):
+        lo = require_ext()
+        lo.avgpool_global(x, out)
+        return out
+    out.copy_(x.float().mean((1, 2)).to(out.dtype))
+    return out
+
+
+def avgpool_global_bwd(dy: torch.Tensor, H: int, W: int,
+                       out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    B, C = dy.shape
+    if out is None:
+        out = torch.empty((B, H, W, C), device=dy.device, dtype=dy.dtype)
+    if _is_gpu(dy):
+        lo = require_ext()
+        lo.avgpool_global_bwd(dy, out)
+        return out
+    out.copy_((dy.float() / (H * W)).view(B, 1, 1, C)
+              .expand(B, H, W, C).to(out.dtype))
+    return out

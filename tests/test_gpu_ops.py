@@ -98,8 +98,8 @@ def test_im2col_col2im_vs_reference(lo):
     col_gpu = F.im2col(x.cuda(), k, k, 1, 1, 0, 0, kpad).cpu()
     assert torch.equal(col_ref.float(), col_gpu.float())
     d = torch.randn(col_ref.shape).bfloat16()
-    dx_ref = F.col2im(d, B, H, W, C, k, k, 0, 0)
-    dx_gpu = F.col2im(d.cuda(), B, H, W, C, k, k, 0, 0).cpu()
+    dx_ref = F.col2im(d, B, H, W, C, k, k, 1, 1, 0, 0)
+    dx_gpu = F.col2im(d.cuda(), B, H, W, C, k, k, 1, 1, 0, 0).cpu()
     assert _rel(dx_gpu, dx_ref) < 1e-2
 
 

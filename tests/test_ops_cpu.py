@@ -43,7 +43,7 @@ def test_im2col_col2im_adjoint():
     OH = H + 2 - k + 1
     assert col.shape == (B * OH * OH, kpad)
     d = torch.randn(col.shape).bfloat16()
-    dx = F.col2im(d, B, H, W, C, k, k, 1, 1)
+    dx = F.col2im(d, B, H, W, C, k, k, 1, 1, 1, 1)
     lhs = (col.float() * d.float()).sum()
     rhs = (x.float() * dx.float()).sum()
     assert abs(lhs - rhs) / (abs(lhs) + 1e-6) < 0.05
