@@ -20,10 +20,12 @@ namespace lo {
 
 // generic dual-column reduction: each thread owns an 8-col chunk; works for
 // any C % 8 == 0 (C/8 <= blockDim keeps every lane busy; larger C loops).
-template <typename F1, typename F2>
+// accum(r, c0, acc1[8], acc2[8]): vector-load the 8-column chunk at (r, c0)
+// and add into the accumulators.
+template <typename ACCUM>
 __device__ __forceinline__ void colreduce2(long M, int C, int ldx,
                                            float* out1, float* out2,
-                                           F1 val1, F2 val2) {
+                                           ACCUM accum) {
   extern __shared__ __attribute__((aligned(16))) float lacc[];  // [2][C]
   for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) lacc[i] = 0.f;
   __syncthreads();
@@ -40,14 +42,8 @@ __device__ __forceinline__ void colreduce2(long M, int C, int ldx,
       #pragma unroll
       for (int j = 0; j < 8; ++j) { acc1[j] = 0.f; acc2[j] = 0.f; }
       const long rStride = (long)gridDim.x * rowsPerBlock;
-      for (long r = (long)blockIdx.x * rowsPerBlock + tr; r < M; r += rStride) {
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int c = tc * 8 + j;
-          acc1[j] += val1(r, c);
-          acc2[j] += val2(r, c);
-        }
-      }
+      for (long r = (long)blockIdx.x * rowsPerBlock + tr; r < M; r += rStride)
+        accum(r, tc * 8, acc1, acc2);
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
         atomicAdd(lacc + tc * 8 + j, acc1[j]);
@@ -65,8 +61,15 @@ __device__ __forceinline__ void colreduce2(long M, int C, int ldx,
 __global__ void bn_stats_kernel(const bf16* __restrict__ x, float* __restrict__ sum,
                                 float* __restrict__ sumsq, long M, int C) {
   colreduce2(M, C, C, sum, sumsq,
-             [&](long r, int c) { return tofloat(x[r * C + c]); },
-             [&](long r, int c) { float v = tofloat(x[r * C + c]); return v * v; });
+             [&](long r, int c0, float (&a1)[8], float (&a2)[8]) {
+               bf16x8 v = *(const bf16x8*)(x + r * C + c0);
+               #pragma unroll
+               for (int j = 0; j < 8; ++j) {
+                 const float f = tofloat(v[j]);
+                 a1[j] += f;
+                 a2[j] += f * f;
+               }
+             });
 }
 
 __global__ void bn_fwd_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
@@ -103,15 +106,19 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ dy,
                                      float* __restrict__ dgamma,
                                      long M, int C, int relu) {
   colreduce2(M, C, C, dbeta, dgamma,
-             [&](long r, int c) {
-               float g = tofloat(dy[r * C + c]);
-               if (relu && tofloat(y[r * C + c]) <= 0.f) g = 0.f;
-               return g;
-             },
-             [&](long r, int c) {
-               float g = tofloat(dy[r * C + c]);
-               if (relu && tofloat(y[r * C + c]) <= 0.f) g = 0.f;
-               return g * (tofloat(x[r * C + c]) - mean[c]) * invstd[c];
+             [&](long r, int c0, float (&a1)[8], float (&a2)[8]) {
+               bf16x8 gv = *(const bf16x8*)(dy + r * C + c0);
+               bf16x8 xv = *(const bf16x8*)(x + r * C + c0);
+               bf16x8 yv;
+               if (relu) yv = *(const bf16x8*)(y + r * C + c0);
+               #pragma unroll
+               for (int j = 0; j < 8; ++j) {
+                 float g = tofloat(gv[j]);
+                 if (relu && tofloat(yv[j]) <= 0.f) g = 0.f;
+                 const int c = c0 + j;
+                 a1[j] += g;
+                 a2[j] += g * (tofloat(xv[j]) - mean[c]) * invstd[c];
+               }
              });
 }
 
@@ -133,12 +140,14 @@ __global__ void bn_bwd_dx_kernel(const bf16* __restrict__ dy,
     const int c0 = (int)(i % (C / 8)) * 8;
     bf16x8 gv = *(const bf16x8*)(dy + r * C + c0);
     bf16x8 xv = *(const bf16x8*)(x + r * C + c0);
+    bf16x8 yv;
+    if (relu) yv = *(const bf16x8*)(y + r * C + c0);
     bf16x8 o;
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const int c = c0 + j;
       float g = tofloat(gv[j]);
-      if (relu && tofloat(y[r * C + c0 + j]) <= 0.f) g = 0.f;
+      if (relu && tofloat(yv[j]) <= 0.f) g = 0.f;
       const float xhat = (tofloat(xv[j]) - mean[c]) * invstd[c];
       const float d = gamma[c] * invstd[c] *
                       (g - dbeta[c] * invM - xhat * dgamma[c] * invM);

@@ -71,21 +71,31 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
         *(bf16x8*)(smA + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = v;
       }
     } else {
-      // A is [K,M]: read 8 contiguous m at one k (coalesced), scatter to LDS
-      constexpr int CH = BK * BM / 8;
+      // A is [K,M]: each thread loads an 8x8 (k x m) tile with coalesced
+      // vector loads, transposes in registers, writes 8 m-rows as b128
+      constexpr int CH = BK * BM / 64;
       for (int c = tid; c < CH; c += T) {
-        const int krow = c / (BM / 8), mc = c % (BM / 8);
-        const int gk = k0 + krow, gm0 = m0 + mc * 8;
-        bf16x8 v = {};
-        if (gk < K) {
-          if (gm0 + 8 <= M) v = *(const bf16x8*)(A + (long)gk * lda + gm0);
-          else if (gm0 < M)
-            for (int j = 0; j < 8 && gm0 + j < M; ++j) v[j] = A[(long)gk * lda + gm0 + j];
+        const int kc = c / (BM / 8), mc = c % (BM / 8);
+        const int gk0 = k0 + kc * 8, gm0 = m0 + mc * 8;
+        bf16x8 v[8];
+        #pragma unroll
+        for (int kk = 0; kk < 8; ++kk) {
+          const int gk = gk0 + kk;
+          bf16x8 t = {};
+          if (gk < K) {
+            if (gm0 + 8 <= M) t = *(const bf16x8*)(A + (long)gk * lda + gm0);
+            else if (gm0 < M)
+              for (int j = 0; j < 8 && gm0 + j < M; ++j) t[j] = A[(long)gk * lda + gm0 + j];
+          }
+          v[kk] = t;
         }
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
+          bf16x8 w;
+          #pragma unroll
+          for (int kk = 0; kk < 8; ++kk) w[kk] = v[kk][j];
           const int row = mc * 8 + j;
-          *(bf16*)(smA + row * BKB + ((krow * 2) ^ ((row & SWZ) << 4))) = v[j];
+          *(bf16x8*)(smA + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = w;
         }
       }
     }
@@ -100,20 +110,29 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
         *(bf16x8*)(smB + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = v;
       }
     } else {
-      constexpr int CH = BK * BN / 8;
+      constexpr int CH = BK * BN / 64;
       for (int c = tid; c < CH; c += T) {
-        const int krow = c / (BN / 8), nc = c % (BN / 8);
-        const int gk = k0 + krow, gn0 = n0 + nc * 8;
-        bf16x8 v = {};
-        if (gk < K) {
-          if (gn0 + 8 <= N) v = *(const bf16x8*)(B + (long)gk * ldb + gn0);
-          else if (gn0 < N)
-            for (int j = 0; j < 8 && gn0 + j < N; ++j) v[j] = B[(long)gk * ldb + gn0 + j];
+        const int kc = c / (BN / 8), nc = c % (BN / 8);
+        const int gk0 = k0 + kc * 8, gn0 = n0 + nc * 8;
+        bf16x8 v[8];
+        #pragma unroll
+        for (int kk = 0; kk < 8; ++kk) {
+          const int gk = gk0 + kk;
+          bf16x8 t = {};
+          if (gk < K) {
+            if (gn0 + 8 <= N) t = *(const bf16x8*)(B + (long)gk * ldb + gn0);
+            else if (gn0 < N)
+              for (int j = 0; j < 8 && gn0 + j < N; ++j) t[j] = B[(long)gk * ldb + gn0 + j];
+          }
+          v[kk] = t;
         }
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
+          bf16x8 w;
+          #pragma unroll
+          for (int kk = 0; kk < 8; ++kk) w[kk] = v[kk][j];
           const int row = nc * 8 + j;
-          *(bf16*)(smB + row * BKB + ((krow * 2) ^ ((row & SWZ) << 4))) = v[j];
+          *(bf16x8*)(smB + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = w;
         }
       }
     }
