@@ -71,15 +71,22 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
         *(bf16x8*)(smA + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = v;
       }
     } else {
-      // A is [K,M]: each thread loads an 8x8 (k x m) tile with coalesced
-      // vector loads, transposes in registers, writes 8 m-rows as b128
-      constexpr int CH = BK * BM / 64;
+      // A is [K,M]: each thread loads a (KRA x 8) k-by-m tile with coalesced
+      // vector loads, transposes in registers, writes 8 m-row pieces of KRA
+      // contiguous k (2*KRA-byte LDS writes stay inside one 16-B granule, so
+      // the XOR swizzle is preserved). KRA adapts so all T threads stay busy
+      // on small tiles.
+      constexpr int CHV = BK * BM / 8;  // 8-wide vector loads in the tile
+      constexpr int KRA = (CHV >= T * 8) ? 8 : (CHV >= T * 4 ? 4
+                          : (CHV >= T * 2 ? 2 : 1));
+      constexpr int CH = CHV / KRA;
+      typedef bf16 bfv __attribute__((ext_vector_type(KRA)));
       for (int c = tid; c < CH; c += T) {
         const int kc = c / (BM / 8), mc = c % (BM / 8);
-        const int gk0 = k0 + kc * 8, gm0 = m0 + mc * 8;
-        bf16x8 v[8];
+        const int gk0 = k0 + kc * KRA, gm0 = m0 + mc * 8;
+        bf16x8 v[KRA];
         #pragma unroll
-        for (int kk = 0; kk < 8; ++kk) {
+        for (int kk = 0; kk < KRA; ++kk) {
           const int gk = gk0 + kk;
           bf16x8 t = {};
           if (gk < K) {
@@ -91,11 +98,11 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
         }
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          bf16x8 w;
+          bfv w;
           #pragma unroll
-          for (int kk = 0; kk < 8; ++kk) w[kk] = v[kk][j];
+          for (int kk = 0; kk < KRA; ++kk) w[kk] = v[kk][j];
           const int row = mc * 8 + j;
-          *(bf16x8*)(smA + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = w;
+          *(bfv*)(smA + row * BKB + ((kc * KRA * 2) ^ ((row & SWZ) << 4))) = w;
         }
       }
     }
@@ -110,13 +117,17 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
         *(bf16x8*)(smB + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = v;
       }
     } else {
-      constexpr int CH = BK * BN / 64;
+      constexpr int CHV = BK * BN / 8;
+      constexpr int KRB = (CHV >= T * 8) ? 8 : (CHV >= T * 4 ? 4
+                          : (CHV >= T * 2 ? 2 : 1));
+      constexpr int CH = CHV / KRB;
+      typedef bf16 bfvb __attribute__((ext_vector_type(KRB)));
       for (int c = tid; c < CH; c += T) {
         const int kc = c / (BN / 8), nc = c % (BN / 8);
-        const int gk0 = k0 + kc * 8, gn0 = n0 + nc * 8;
-        bf16x8 v[8];
+        const int gk0 = k0 + kc * KRB, gn0 = n0 + nc * 8;
+        bf16x8 v[KRB];
         #pragma unroll
-        for (int kk = 0; kk < 8; ++kk) {
+        for (int kk = 0; kk < KRB; ++kk) {
           const int gk = gk0 + kk;
           bf16x8 t = {};
           if (gk < K) {
@@ -128,11 +139,11 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
         }
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          bf16x8 w;
+          bfvb w;
           #pragma unroll
-          for (int kk = 0; kk < 8; ++kk) w[kk] = v[kk][j];
+          for (int kk = 0; kk < KRB; ++kk) w[kk] = v[kk][j];
           const int row = nc * 8 + j;
-          *(bf16x8*)(smB + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = w;
+          *(bfvb*)(smB + row * BKB + ((kc * KRB * 2) ^ ((row & SWZ) << 4))) = w;
         }
       }
     }
