@@ -223,7 +223,13 @@ def create_app(runtime: Optional[Runtime] = None) -> FastAPI:
     @app.get(PREFIX + "/{verb}/{tool}/{name}")
     def read_rows(verb: str, tool: str, name: str, query: str = "{}",
                   limit: int = 10, skip: int = 0):
-        rt.require_exists(name)
+        meta = rt.require_exists(name)
+        # explore plots: serve the rendered PNG (reference database_executor
+        # server.py:151-166 returned the image for explore GETs)
+        png = rt.artifacts.path(name, meta.get("type", f"{verb}/{tool}")) + ".png"
+        if os.path.exists(png):
+            with open(png, "rb") as fh:
+                return Response(fh.read(), media_type="image/png")
         q = json.loads(query) if query else {}
         return {RESULT: rt.read_rows(name, q, skip, limit)}
 

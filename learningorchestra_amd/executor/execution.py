@@ -276,8 +276,12 @@ class Execution:
     # ---------------------------------------------------------------- helpers
     def _store_result(self, name: str, service_type: str, result: Any) -> None:
         """Explore results that are DataFrame-like go to the document store as
-        row-documents (so the poll/GET contract serves them); everything else
+        row-documents (so the poll/GET contract serves them); matplotlib
+        figures render to PNG (the reference's seaborn-plot-to-PNG explore
+        storage, database_executor_image/utils.py:300-309); everything else
         to the artifact store (database_executor_image/server.py:52-58)."""
+        if _render_figure_png(result, self._artifacts.path(name, service_type)):
+            return
         try:
             import pandas as pd
             if isinstance(result, pd.DataFrame):
@@ -303,6 +307,25 @@ class Execution:
             self._artifacts.delete(name, stype)
         except Exception:
             pass
+
+
+def _render_figure_png(result: Any, base_path: str) -> bool:
+    """If ``result`` is a matplotlib Figure/Axes, save PNG and return True."""
+    try:
+        import matplotlib.figure
+        fig = None
+        if isinstance(result, matplotlib.figure.Figure):
+            fig = result
+        elif hasattr(result, "get_figure"):
+            fig = result.get_figure()
+        if fig is None:
+            return False
+        fig.savefig(base_path + ".png", format="png", bbox_inches="tight")
+        import matplotlib.pyplot as plt
+        plt.close(fig)
+        return True
+    except ImportError:
+        return False
 
 
 def _safe_params(params: Optional[Dict[str, Any]]) -> Dict[str, Any]:

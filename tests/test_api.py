@@ -219,3 +219,22 @@ def test_delete_and_404(client, tmp_path):
     assert client.delete(f"{PREFIX}/dataset/csv/titanic").status_code == 200
     assert client.get(f"{PREFIX}/dataset/csv/titanic").status_code == 404
     assert client.delete(f"{PREFIX}/dataset/csv/titanic").status_code == 404
+
+
+def test_explore_plot_png(client, tmp_path):
+    """explore/{tool} with the native Plot class renders a PNG served on GET
+    (reference seaborn-scatterplot-to-PNG path)."""
+    ingest_titanic(client, tmp_path)
+    r = client.post(f"{PREFIX}/explore/torch",
+                    json={"name": "plot1",
+                          "modulePath": "learningorchestra_amd.models.explore",
+                          "class": "Plot", "classParameters": {},
+                          "method": "scatter",
+                          "methodParameters": {"data": "$titanic",
+                                               "x": "Age", "y": "Fare"}})
+    assert r.status_code == 201
+    wait_finished(client, "plot1")
+    resp = client.get(f"{PREFIX}/explore/torch/plot1")
+    assert resp.status_code == 200
+    assert resp.headers["content-type"] == "image/png"
+    assert resp.content[:8] == b"\x89PNG\r\n\x1a\n"
