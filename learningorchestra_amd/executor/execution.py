@@ -23,6 +23,7 @@ import importlib
 import inspect
 import io
 import sys
+import time
 import traceback
 from contextlib import redirect_stdout
 from typing import Any, Dict, Optional, Tuple
@@ -139,12 +140,14 @@ class Execution:
                                    description=description)
 
         def pipeline():
+            t0 = time.time()
             treated = self.parameters().treat(class_parameters)
             instance = cls(**treated)
             self._artifacts.save(instance, name, service_type)
             self._metadata.create_execution_document(
                 name, description or f"instantiate {class_name}",
-                {"classParameters": _safe_params(class_parameters)})
+                {"classParameters": _safe_params(class_parameters)},
+                durationSeconds=round(time.time() - t0, 4))
             self._metadata.update_finished_flag(name, True)
 
         self._scheduler.submit(name, pipeline)
@@ -163,6 +166,7 @@ class Execution:
                                    description=description)
 
         def pipeline():
+            t0 = time.time()
             params = self.parameters()
             instance = cls(**params.treat(class_parameters))
             method = getattr(instance, method_name)
@@ -171,7 +175,8 @@ class Execution:
                                else instance)
             self._metadata.create_execution_document(
                 name, description or f"{class_name}.{method_name}",
-                {"methodParameters": _safe_params(method_parameters)})
+                {"methodParameters": _safe_params(method_parameters)},
+                durationSeconds=round(time.time() - t0, 4))
             self._metadata.update_finished_flag(name, True)
 
         self._scheduler.submit(name, pipeline, device=self._device)
@@ -189,6 +194,7 @@ class Execution:
                                    method=method_name, description=description)
 
         def pipeline():
+            t0 = time.time()
             parent_type = parent_meta.get("type", "")
             instance = self._artifacts.load(parent_name, parent_type)
             self._rt.validate_method(instance, method_name)
@@ -203,7 +209,8 @@ class Execution:
                 self._store_result(name, service_type, result)
             self._metadata.create_execution_document(
                 name, description or f"{method_name} on {parent_name}",
-                {"methodParameters": _safe_params(method_parameters)})
+                {"methodParameters": _safe_params(method_parameters)},
+                durationSeconds=round(time.time() - t0, 4))
             self._metadata.update_finished_flag(name, True)
 
         self._scheduler.submit(name, pipeline, device=self._device)
