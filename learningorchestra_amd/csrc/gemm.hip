@@ -37,8 +37,10 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
   static_assert(BK % 32 == 0, "BK multiple of MFMA K");
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* smA = smem;                     // BM*BKB bytes
-  char* smB = smem + BM * BKB;          // BN*BKB bytes
+  // two (A,B) tile buffers: stage t+1 while computing t (guide T3 2-phase);
+  // computed as offsets (a pointer array of LDS addrspacecasts is rejected
+  // by the backend as a static initializer)
+  constexpr int BUFB = (BM + BN) * BKB;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -59,16 +61,42 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
   const int fr = lane & 15;
   const int fkb = (lane >> 4) * 8;      // fragment k base within MFMA step
 
-  for (int k0 = kBegin; k0 < kEnd; k0 += BK) {
+  constexpr int NW = T / 64;
+  constexpr int ROWS_PER_SEG = 1024 / BKB;        // LDS rows per 1-KiB wave DMA
+  constexpr int KCH = BKB / 16;                   // 16-B chunks per LDS row
+  const bool a_rows_in = !TA && (m0 + BM <= M);
+  const bool b_rows_in = TB && (n0 + BN <= N);
+
+  auto stage_tile = [&](int buf, int k0) {
+    char* smA = smem + buf * BUFB;
+    char* smB = smem + buf * BUFB + BM * BKB;
+    const bool k_in = (k0 + BK <= K);
     // ---- stage A tile -> As[BM][BK] ------------------------------------
     if (!TA) {
-      constexpr int CH = BM * BK / 8;
-      for (int c = tid; c < CH; c += T) {
-        const int row = c / (BK / 8), kc = c % (BK / 8);
-        const int gm = m0 + row, gk = k0 + kc * 8;
-        bf16x8 v = {};
-        if (gm < M && gk < K) v = *(const bf16x8*)(A + (long)gm * lda + gk);
-        *(bf16x8*)(smA + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = v;
+      if (a_rows_in && k_in && ROWS_PER_SEG > 0) {
+        // interior tiles: LDS-DMA (global_load_lds, 16 B/lane — guide §5
+        // ladder step 3). glds writes lane-linear, so the bank-conflict XOR
+        // swizzle moves to the per-lane SOURCE address (rule 21); the
+        // ds_read side keeps the same XOR (involution).
+        #pragma unroll
+        for (int seg = wid; seg < BM / ROWS_PER_SEG; seg += NW) {
+          const int row = seg * ROWS_PER_SEG + lane / KCH;
+          const int kc = (lane % KCH) ^ (row & SWZ);
+          auto* gsrc = (const __attribute__((address_space(1))) unsigned int*)
+              (const char*)(A + (long)(m0 + row) * lda + k0 + kc * 8);
+          auto* ldst = (__attribute__((address_space(3))) unsigned int*)
+              (__attribute__((address_space(3))) char*)(smA + seg * 1024);
+          __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 0);
+        }
+      } else {
+        constexpr int CH = BM * BK / 8;
+        for (int c = tid; c < CH; c += T) {
+          const int row = c / (BK / 8), kc = c % (BK / 8);
+          const int gm = m0 + row, gk = k0 + kc * 8;
+          bf16x8 v = {};
+          if (gm < M && gk < K) v = *(const bf16x8*)(A + (long)gm * lda + gk);
+          *(bf16x8*)(smA + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = v;
+        }
       }
     } else {
       // A is [K,M]: each thread loads a (KRA x 8) k-by-m tile with coalesced
@@ -108,13 +136,26 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
     }
     // ---- stage B tile -> Bs[BN][BK] ------------------------------------
     if (TB) {
-      constexpr int CH = BN * BK / 8;
-      for (int c = tid; c < CH; c += T) {
-        const int row = c / (BK / 8), kc = c % (BK / 8);
-        const int gn = n0 + row, gk = k0 + kc * 8;
-        bf16x8 v = {};
-        if (gn < N && gk < K) v = *(const bf16x8*)(B + (long)gn * ldb + gk);
-        *(bf16x8*)(smB + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = v;
+      if (b_rows_in && k_in && ROWS_PER_SEG > 0) {
+        #pragma unroll
+        for (int seg = wid; seg < BN / ROWS_PER_SEG; seg += NW) {
+          const int row = seg * ROWS_PER_SEG + lane / KCH;
+          const int kc = (lane % KCH) ^ (row & SWZ);
+          auto* gsrc = (const __attribute__((address_space(1))) unsigned int*)
+              (const char*)(B + (long)(n0 + row) * ldb + k0 + kc * 8);
+          auto* ldst = (__attribute__((address_space(3))) unsigned int*)
+              (__attribute__((address_space(3))) char*)(smB + seg * 1024);
+          __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 0);
+        }
+      } else {
+        constexpr int CH = BN * BK / 8;
+        for (int c = tid; c < CH; c += T) {
+          const int row = c / (BK / 8), kc = c % (BK / 8);
+          const int gn = n0 + row, gk = k0 + kc * 8;
+          bf16x8 v = {};
+          if (gn < N && gk < K) v = *(const bf16x8*)(B + (long)gn * ldb + gk);
+          *(bf16x8*)(smB + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = v;
+        }
       }
     } else {
       constexpr int CHV = BK * BN / 8;
@@ -147,9 +188,11 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
         }
       }
     }
-    __syncthreads();
+  };
 
-    // ---- MFMA over the tile --------------------------------------------
+  auto compute_tile = [&](int buf) {
+    const char* smA = smem + buf * BUFB;
+    const char* smB = smem + buf * BUFB + BM * BKB;
     #pragma unroll
     for (int kk = 0; kk < BK / 32; ++kk) {
       bf16x8 af[MFRAG], bf[NFRAG];
@@ -172,7 +215,18 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     }
+  };
+
+  // 2-phase pipeline: issue tile t+1's loads BEFORE computing tile t, one
+  // barrier per tile (guide §5.5 T3 minimum 2-phase recipe).
+  int cur = 0;
+  stage_tile(0, kBegin);
+  __syncthreads();
+  for (int k0 = kBegin; k0 < kEnd; k0 += BK) {
+    if (k0 + BK < kEnd) stage_tile(cur ^ 1, k0 + BK);
+    compute_tile(cur);
     __syncthreads();
+    cur ^= 1;
   }
 
   // ---- epilogue: bias / ReLU / store (bf16 | f32 | f32-atomic) -----------
@@ -247,7 +301,7 @@ static void launch_cfg(const GemmArgs& g, hipStream_t s) {
   }
   const int zs = g.splits > 1 ? cdiv(g.K, kChunk) : 1;
   dim3 grid(mb * nb, 1, zs), block(WM * WN * 64);
-  size_t lds = (size_t)(BM + BN) * BK * 2;
+  size_t lds = (size_t)(BM + BN) * BK * 2 * 2;  // double-buffered
   hipLaunchKernelGGL(HIP_KERNEL_NAME(
       gemm_kernel<BM, BN, BK, WM, WN, TA, TB, EPI, OUT_F32, ATOMIC>),
       grid, block, lds, s,
