@@ -30,6 +30,11 @@ class Layer:
     def build(self, arena: ParamArena) -> None:  # register params
         pass
 
+    def post_opt_step(self) -> None:
+        """Refresh derived compute state after an optimizer step (e.g. the
+        transposed-weight mirror that keeps dX GEMMs on the fast k-contiguous
+        path)."""
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         raise NotImplementedError
 
@@ -92,6 +97,16 @@ class Conv2dNHWC(Layer):
                     self._bufs["dcol"] = torch.empty((M, self.kpad), device=dev, dtype=dtype)
         return self._bufs
 
+    def _wt(self) -> torch.Tensor:
+        # transposed mirror [kpad, out_c] (see Linear._wt)
+        if getattr(self, "_wt_buf", None) is None:
+            self._wt_buf = self.arena.p(self.name + ".w").t().contiguous()
+        return self._wt_buf
+
+    def post_opt_step(self) -> None:
+        if getattr(self, "_wt_buf", None) is not None:
+            self._wt_buf.copy_(self.arena.p(self.name + ".w").t())
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         B, H, W, C = x.shape
         assert C == self.in_c
@@ -125,9 +140,9 @@ class Conv2dNHWC(Layer):
         if self._is_1x1:
             # 1x1/s1 conv: col IS x, so dcol IS dx
             dx = bufs["dx"]
-            F.gemm(dy2, self.arena.p(self.name + ".w"), out=dx.view(M, self.in_c))
+            F.gemm(dy2, self._wt(), tb=True, out=dx.view(M, self.in_c))
             return dx
-        F.gemm(dy2, self.arena.p(self.name + ".w"), out=bufs["dcol"])
+        F.gemm(dy2, self._wt(), tb=True, out=bufs["dcol"])
         F.col2im(bufs["dcol"], bufs["B"], bufs["H"], bufs["W"], self.in_c,
                  self.kh, self.kw, self.stride, self.stride, self.pad,
                  self.pad, out=bufs["dx"])
@@ -205,6 +220,17 @@ class Linear(Layer):
                out=self._bufs["y"])
         return self._bufs["y"]
 
+    def _wt(self) -> torch.Tensor:
+        # transposed mirror [in_f, out_f]: dX = dY @ W runs as the fast (F,T)
+        # GEMM (k-contiguous B) instead of the transpose-staged (F,F) path
+        if getattr(self, "_wt_buf", None) is None:
+            self._wt_buf = self.arena.p(self.name + ".w").t().contiguous()
+        return self._wt_buf
+
+    def post_opt_step(self) -> None:
+        if getattr(self, "_wt_buf", None) is not None:
+            self._wt_buf.copy_(self.arena.p(self.name + ".w").t())
+
     def backward(self, dy: torch.Tensor) -> torch.Tensor:
         bufs = self._bufs
         if self.relu:
@@ -213,7 +239,7 @@ class Linear(Layer):
         splits = _splitk_heuristic(self.out_f, self.in_f, bufs["M"])
         F.gemm(dy, bufs["x"], ta=True, out=gw, splits=splits)
         F.colsum(dy, out=self.arena.g(self.name + ".b"))
-        F.gemm(dy, self.arena.p(self.name + ".w"), out=bufs["dx"])
+        F.gemm(dy, self._wt(), tb=True, out=bufs["dx"])
         return bufs["dx"]
 
 
@@ -273,6 +299,10 @@ class SequentialClassifier:
                 break
         return self.loss_sum, self.correct
 
+    def post_opt_step(self) -> None:
+        for lay in self.layers:
+            lay.post_opt_step()
+
     @torch.no_grad()
     def predict(self, x: torch.Tensor) -> torch.Tensor:
         logits = self.forward(x)
@@ -284,6 +314,7 @@ class SequentialClassifier:
 
     def load_state_dict(self, sd):
         self.arena.load_state_dict(sd)
+        self.post_opt_step()  # refresh transposed weight mirrors
 
 
 class Embedding(Layer):

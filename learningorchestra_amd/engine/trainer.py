@@ -32,6 +32,9 @@ class Trainer:
         self.model.train_step(x, y, gscale=gscale)
         all_reduce_grads(self.model.arena.grad)
         self.opt.step()
+        post = getattr(self.model, "post_opt_step", None)
+        if post is not None:
+            post()
 
     def step(self, x: torch.Tensor, y: torch.Tensor) -> Tuple[float, float]:
         """One training step. Returns (mean loss, accuracy) — NOTE: these

@@ -160,6 +160,13 @@ class ResNet:
         self.stem_conv.backward(dy)
         return self.loss_sum, self.correct
 
+    def post_opt_step(self) -> None:
+        self.stem_conv.post_opt_step()
+        for blk in self.blocks:
+            for lay in blk.layers():
+                lay.post_opt_step()
+        self.fc.post_opt_step()
+
     @torch.no_grad()
     def predict(self, x: torch.Tensor) -> torch.Tensor:
         return F.argmax_rows(self.forward(x), self.num_classes)
@@ -176,6 +183,7 @@ class ResNet:
                                     if k in self.arena._offsets})
         for bn in self._bn_layers():
             bn.load_extra_state(sd)
+        self.post_opt_step()
 
 
 def build_resnet50(device="cpu", seed: int = 0, num_classes: int = NUM_CLASSES,

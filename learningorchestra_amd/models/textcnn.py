@@ -97,6 +97,10 @@ class TextCNN:
         self.emb.backward(dxe.view(B, S, self.emb_dim))
         return self.loss_sum, self.correct
 
+    def post_opt_step(self) -> None:
+        for lay in [*self.convs, self.fc]:
+            lay.post_opt_step()
+
     @torch.no_grad()
     def predict(self, ids: torch.Tensor) -> torch.Tensor:
         return F.argmax_rows(self.forward(ids), self.num_classes)
@@ -106,6 +110,7 @@ class TextCNN:
 
     def load_state_dict(self, sd):
         self.arena.load_state_dict(sd)
+        self.post_opt_step()
 
 
 def build_textcnn(device="cpu", seed: int = 0, **kw) -> TextCNN:
