@@ -4,6 +4,7 @@ like the MNIST CNN are exactly the "capture launch-bound inner loops in
 hipGraphs" case)."""
 from __future__ import annotations
 
+import os
 import time
 from typing import Callable, Optional, Tuple
 
@@ -66,6 +67,24 @@ class Trainer:
         with torch.cuda.graph(self._graph):
             self._step_body(self._static_x, self._static_y)
         torch.cuda.synchronize()
+
+    # -- in-training checkpoint/resume (the reference had no
+    # epoch-level resume; fit was atomic) --------------------------------
+    def save_checkpoint(self, path: str, step: int = 0) -> None:
+        state = {"model": self.model.state_dict(), "step": step}
+        opt_sd = getattr(self.opt, "state_dict", None)
+        if opt_sd is not None:
+            state["optimizer"] = opt_sd()
+        tmp = path + ".tmp"
+        torch.save(state, tmp)
+        os.replace(tmp, path)
+
+    def load_checkpoint(self, path: str) -> int:
+        state = torch.load(path, map_location="cpu", weights_only=True)
+        self.model.load_state_dict(state["model"])
+        if "optimizer" in state and hasattr(self.opt, "load_state_dict"):
+            self.opt.load_state_dict(state["optimizer"])
+        return int(state.get("step", 0))
 
     def train(self, data_iter, steps: int, log_every: int = 0,
               log_fn: Optional[Callable[[str], None]] = None) -> dict:

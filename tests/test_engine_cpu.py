@@ -144,3 +144,36 @@ def test_checkpoint_resume():
     assert not torch.allclose(model.forward(x).float(), logits_before)
     model.load_state_dict(sd)
     assert torch.allclose(model.forward(x).float(), logits_before)
+
+
+def test_trainer_checkpoint_resume(tmp_path):
+    from learningorchestra_amd.engine.trainer import Trainer, make_sgd
+    model = build_mnist_cnn("cpu", seed=5, channels=(4, 4), fc_width=16)
+    tr = Trainer(model, make_sgd(model, lr=0.05), device="cpu")
+    x, y = mnist_batch(16, dtype=torch.bfloat16, seed=1)
+    tr.step(x, y)
+    tr.save_checkpoint(str(tmp_path / "ck.pt"), step=7)
+    ref_master = model.arena.master.clone()
+    ref_mom = tr.opt.mom.clone()
+    tr.step(x, y)  # diverge
+    assert not torch.equal(model.arena.master, ref_master)
+    step = tr.load_checkpoint(str(tmp_path / "ck.pt"))
+    assert step == 7
+    assert torch.equal(model.arena.master, ref_master)
+    assert torch.equal(tr.opt.mom, ref_mom)
+
+
+def test_grid_search_tuning():
+    from learningorchestra_amd.models.tuning import GridSearch
+    import numpy as np
+    rng = np.random.RandomState(0)
+    X = rng.randn(200, 4).astype("float32")
+    y = (X[:, 0] > 0).astype("int64")
+    gs = GridSearch("learningorchestra_amd.models.tabular",
+                    "LogisticRegressionClassifier",
+                    {"lr": [0.01, 0.3]},
+                    fixedParameters={"epochs": 15, "device": "cpu"})
+    gs.fit(X, y)
+    assert len(gs.results_) == 2
+    assert gs.best_score_ > 0.8
+    assert gs.predict(X[:10]).shape == (10,)
