@@ -16,6 +16,45 @@
 namespace lo {
 
 // --------------------------------------------------------------- im2col ----
+// small-kpad fast path (e.g. C=1 5x5 convs, kpad<=64): one thread assembles
+// a FULL col row in registers (the input tile is tiny and L1/L2-hot) and
+// vector-stores it — the generic per-element path is scalar-store-bound at
+// C==1 (was 12% of the MNIST step).
+template <int KPAD>
+__global__ void im2col_row_kernel(const bf16* __restrict__ in, bf16* __restrict__ col,
+                                  int B, int H, int W, int C, int KH, int KW,
+                                  int SH, int SW, int PH, int PW,
+                                  int OH, int OW) {
+  const long total = (long)B * OH * OW;
+  const int K = KH * KW * C;
+  for (long rrow = (long)blockIdx.x * blockDim.x + threadIdx.x; rrow < total;
+       rrow += (long)gridDim.x * blockDim.x) {
+    long r = rrow;
+    const int ow = r % OW; r /= OW;
+    const int oh = r % OH; r /= OH;
+    const int b = r;
+    bf16 vals[KPAD];
+    #pragma unroll
+    for (int i = 0; i < KPAD; ++i) vals[i] = bf16(0.f);
+    int k = 0;
+    for (int kh = 0; kh < KH; ++kh) {
+      const int h = oh * SH - PH + kh;
+      for (int kw = 0; kw < KW; ++kw) {
+        const int w = ow * SW - PW + kw;
+        if (h >= 0 && h < H && w >= 0 && w < W)
+          for (int c = 0; c < C; ++c)
+            vals[k + c] = in[(((long)b * H + h) * W + w) * C + c];
+        k += C;
+      }
+    }
+    (void)K;
+    bf16* dst = col + rrow * KPAD;
+    #pragma unroll
+    for (int i = 0; i < KPAD / 8; ++i)
+      *(bf16x8*)(dst + i * 8) = *(bf16x8*)(vals + i * 8);
+  }
+}
+
 template <bool VEC8>
 __global__ void im2col_kernel(const bf16* __restrict__ in, bf16* __restrict__ col,
                               int B, int H, int W, int C, int KH, int KW,
@@ -53,6 +92,36 @@ __global__ void im2col_kernel(const bf16* __restrict__ in, bf16* __restrict__ co
 void launch_im2col(const void* in, void* col, int B, int H, int W, int C,
                    int KH, int KW, int SH, int SW, int PH, int PW,
                    int OH, int OW, int Kpad, hipStream_t s) {
+  if (Kpad <= 64) {
+    const long rows = (long)B * OH * OW;
+    const int block = 256;
+    const int grid = (int)min((rows + block - 1) / block, (long)4096);
+    if (Kpad == 32)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(im2col_row_kernel<32>), dim3(grid), dim3(block), 0, s,
+                         (const bf16*)in, (bf16*)col, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
+    else if (Kpad == 64)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(im2col_row_kernel<64>), dim3(grid), dim3(block), 0, s,
+                         (const bf16*)in, (bf16*)col, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
+    else if (Kpad == 16)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(im2col_row_kernel<16>), dim3(grid), dim3(block), 0, s,
+                         (const bf16*)in, (bf16*)col, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
+    else if (Kpad == 8)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(im2col_row_kernel<8>), dim3(grid), dim3(block), 0, s,
+                         (const bf16*)in, (bf16*)col, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
+    else if (Kpad == 40)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(im2col_row_kernel<40>), dim3(grid), dim3(block), 0, s,
+                         (const bf16*)in, (bf16*)col, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
+    else if (Kpad == 48)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(im2col_row_kernel<48>), dim3(grid), dim3(block), 0, s,
+                         (const bf16*)in, (bf16*)col, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
+    else if (Kpad == 56)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(im2col_row_kernel<56>), dim3(grid), dim3(block), 0, s,
+                         (const bf16*)in, (bf16*)col, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
+    else
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(im2col_row_kernel<24>), dim3(grid), dim3(block), 0, s,
+                         (const bf16*)in, (bf16*)col, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
+    return;
+  }
   const bool vec = (C % 8 == 0);
   const long total = (long)B * OH * OW * KH * KW * (vec ? C / 8 : C);
   const int block = 256;

@@ -313,24 +313,33 @@ static void launch_cfg(const GemmArgs& g, hipStream_t s) {
 // uses (fwd = N,T; dX = N,N; dW = T,N split-K) plus (T,T) for completeness.
 template <bool TA, bool TB>
 static bool dispatch_tiles(const GemmArgs& g, hipStream_t s) {
-  #define LO_EPI_CASES(BM_, BN_, WM_, WN_)                                     \
+  #define LO_EPI_CASES_BK(BM_, BN_, BK_, WM_, WN_)                             \
     do {                                                                       \
       if (g.out_f32) {                                                         \
-        if (g.epi == 1) launch_cfg<BM_, BN_, 64, WM_, WN_, TA, TB, 1, true, false>(g, s);  \
-        else            launch_cfg<BM_, BN_, 64, WM_, WN_, TA, TB, 0, true, false>(g, s);  \
+        if (g.epi == 1) launch_cfg<BM_, BN_, BK_, WM_, WN_, TA, TB, 1, true, false>(g, s);  \
+        else            launch_cfg<BM_, BN_, BK_, WM_, WN_, TA, TB, 0, true, false>(g, s);  \
       } else {                                                                 \
-        if (g.epi == 1) launch_cfg<BM_, BN_, 64, WM_, WN_, TA, TB, 1, false, false>(g, s); \
-        else            launch_cfg<BM_, BN_, 64, WM_, WN_, TA, TB, 0, false, false>(g, s); \
+        if (g.epi == 1) launch_cfg<BM_, BN_, BK_, WM_, WN_, TA, TB, 1, false, false>(g, s); \
+        else            launch_cfg<BM_, BN_, BK_, WM_, WN_, TA, TB, 0, false, false>(g, s); \
       }                                                                        \
       return true;                                                             \
     } while (0)
 
-  if (g.N > 64 && g.M >= 4096) LO_EPI_CASES(128, 128, 2, 2);
-  if (g.N > 64) LO_EPI_CASES(64, 128, 2, 2);     // small-M wide-N (dX tails)
-  if (g.N > 32) LO_EPI_CASES(128, 64, 2, 2);
-  if (g.N > 16) LO_EPI_CASES(256, 32, 4, 1);
-  LO_EPI_CASES(128, 16, 4, 1);
-  #undef LO_EPI_CASES
+  if (g.K <= 32) {
+    // shallow reductions (e.g. C=1 5x5 conv, kpad 32): half-depth tiles so
+    // the LDS image and glds path are fully used
+    if (g.N > 64 && g.M >= 4096) LO_EPI_CASES_BK(128, 128, 32, 2, 2);
+    if (g.N > 64) LO_EPI_CASES_BK(64, 128, 32, 2, 2);
+    if (g.N > 32) LO_EPI_CASES_BK(128, 64, 32, 2, 2);
+    if (g.N > 16) LO_EPI_CASES_BK(256, 32, 32, 4, 1);
+    LO_EPI_CASES_BK(128, 16, 32, 4, 1);
+  }
+  if (g.N > 64 && g.M >= 4096) LO_EPI_CASES_BK(128, 128, 64, 2, 2);
+  if (g.N > 64) LO_EPI_CASES_BK(64, 128, 64, 2, 2);   // small-M wide-N (dX tails)
+  if (g.N > 32) LO_EPI_CASES_BK(128, 64, 64, 2, 2);
+  if (g.N > 16) LO_EPI_CASES_BK(256, 32, 64, 4, 1);
+  LO_EPI_CASES_BK(128, 16, 64, 4, 1);
+  #undef LO_EPI_CASES_BK
 }
 
 // Returns true if a native config covered the shape; false => caller falls
