@@ -347,19 +347,13 @@ static bool dispatch_tiles(const GemmArgs& g, hipStream_t s) {
 bool gemm_dispatch(const GemmArgs& g, hipStream_t s) {
   if (g.K % 8 != 0) return false;
   if (g.splits > 1) {
-    // split-K atomic accumulate (weight gradients): dW = A^T @ B. Wide-N
-    // shapes take a 64x128 tile (deeper register-transpose staging: KRA=2,
-    // KRB=4); narrow ones the 32x64 tile.
+    // split-K atomic accumulate (weight gradients): dW = A^T @ B
+    // (32x64 tiles: measured better than 64x128 — block count beats
+    // staging depth for these shapes)
     if (!g.out_f32 || g.epi != 0) return false;
-    if (!g.ta && g.tb) {
-      if (g.N >= 96) launch_cfg<64, 128, 64, 2, 2, false, true, 0, true, true>(g, s);
-      else launch_cfg<32, 64, 64, 1, 4, false, true, 0, true, true>(g, s);
-    } else if (g.ta && !g.tb) {
-      if (g.N >= 96) launch_cfg<64, 128, 64, 2, 2, true, false, 0, true, true>(g, s);
-      else launch_cfg<32, 64, 64, 1, 4, true, false, 0, true, true>(g, s);
-    } else {
-      return false;
-    }
+    if (!g.ta && g.tb) launch_cfg<32, 64, 64, 1, 4, false, true, 0, true, true>(g, s);
+    else if (g.ta && !g.tb) launch_cfg<32, 64, 64, 1, 4, true, false, 0, true, true>(g, s);
+    else return false;
     return true;
   }
   if (!g.ta && g.tb) return dispatch_tiles<false, true>(g, s);
