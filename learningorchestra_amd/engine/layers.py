@@ -249,7 +249,7 @@ def _splitk_heuristic(m: int, n: int, k: int) -> int:
     base_blocks = ((m + 31) // 32) * ((n + 63) // 64)
     if base_blocks >= 256 or k < 4096:
         return 1
-    return max(1, min(512 // base_blocks, k // 1024))
+    return max(1, min(2048 // base_blocks, k // 1024))
 
 
 class SequentialClassifier:
