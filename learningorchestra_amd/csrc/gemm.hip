@@ -229,26 +229,69 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
     cur ^= 1;
   }
 
-  // ---- epilogue: bias / ReLU / store (bf16 | f32 | f32-atomic) -----------
+  // ---- epilogue ----------------------------------------------------------
+  // MFMA C fragments are column-fragmented (lane l holds rows r0..r0+3 of ONE
+  // column), so direct stores are 2-4 B scalars. Non-atomic outputs stage the
+  // C tile through LDS (padded rows) and store 16 B vectors instead; the
+  // split-K path keeps per-element fp32 atomics.
   float* Cf = (float*)Cv;
   bf16* Cb = (bf16*)Cv;
-  #pragma unroll
-  for (int ni = 0; ni < NFRAG; ++ni) {
-    const int col = n0 + wc * WTN + ni * 16 + (lane & 15);
-    if (col >= N) continue;
-    const float bv = bias ? bias[col] : 0.f;
+  if (ATOMIC) {
     #pragma unroll
-    for (int mi = 0; mi < MFRAG; ++mi) {
-      const int row0 = m0 + wr * WTM + mi * 16 + (lane >> 4) * 4;
+    for (int ni = 0; ni < NFRAG; ++ni) {
+      const int col = n0 + wc * WTN + ni * 16 + (lane & 15);
+      if (col >= N) continue;
       #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const int row = row0 + j;
-        if (row >= M) continue;
-        float v = acc[mi][ni][j] + bv;
-        if (EPI == 1) v = fmaxf(v, 0.f);
-        if (ATOMIC) atomicAdd(Cf + (long)row * ldc + col, v);
-        else if (OUT_F32) Cf[(long)row * ldc + col] = v;
-        else Cb[(long)row * ldc + col] = tobf16(v);
+      for (int mi = 0; mi < MFRAG; ++mi) {
+        const int row0 = m0 + wr * WTM + mi * 16 + (lane >> 4) * 4;
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int row = row0 + j;
+          if (row >= M) continue;
+          atomicAdd(Cf + (long)row * ldc + col, acc[mi][ni][j]);
+        }
+      }
+    }
+    return;
+  }
+  constexpr int EB = OUT_F32 ? 4 : 2;          // bytes per element in LDS
+  constexpr int CROWB = BN * EB + 16;          // padded LDS row
+  char* ct = smem;                              // staging buffers are done
+  {
+    #pragma unroll
+    for (int ni = 0; ni < NFRAG; ++ni) {
+      const int colL = wc * WTN + ni * 16 + (lane & 15);
+      const int col = n0 + colL;
+      const float bv = (bias && col < N) ? bias[col] : 0.f;
+      #pragma unroll
+      for (int mi = 0; mi < MFRAG; ++mi) {
+        const int rowL0 = wr * WTM + mi * 16 + (lane >> 4) * 4;
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float v = acc[mi][ni][j] + bv;
+          if (EPI == 1) v = fmaxf(v, 0.f);
+          if (OUT_F32) *(float*)(ct + (rowL0 + j) * CROWB + colL * 4) = v;
+          else *(bf16*)(ct + (rowL0 + j) * CROWB + colL * 2) = tobf16(v);
+        }
+      }
+    }
+  }
+  __syncthreads();
+  constexpr int VE = 16 / EB;                  // elements per 16-B store
+  constexpr int NCH = BN / VE;
+  for (int c = tid; c < BM * NCH; c += T) {
+    const int rowL = c / NCH, ch = c % NCH;
+    const int gr = m0 + rowL;
+    if (gr >= M) continue;
+    const int gc = n0 + ch * VE;
+    const char* src = ct + rowL * CROWB + ch * 16;
+    if (gc + VE <= N) {
+      if (OUT_F32) *(f32x4*)(Cf + (long)gr * ldc + gc) = *(const f32x4*)src;
+      else *(bf16x8*)(Cb + (long)gr * ldc + gc) = *(const bf16x8*)src;
+    } else if (gc < N) {
+      for (int e = 0; e < VE && gc + e < N; ++e) {
+        if (OUT_F32) Cf[(long)gr * ldc + gc + e] = ((const float*)src)[e];
+        else Cb[(long)gr * ldc + gc + e] = ((const bf16*)src)[e];
       }
     }
   }
@@ -301,7 +344,9 @@ static void launch_cfg(const GemmArgs& g, hipStream_t s) {
   }
   const int zs = g.splits > 1 ? cdiv(g.K, kChunk) : 1;
   dim3 grid(mb * nb, 1, zs), block(WM * WN * 64);
-  size_t lds = (size_t)(BM + BN) * BK * 2 * 2;  // double-buffered
+  size_t lds_stage = (size_t)(BM + BN) * BK * 2 * 2;  // double-buffered
+  size_t lds_epi = (size_t)BM * (BN * (g.out_f32 ? 4 : 2) + 16);
+  size_t lds = lds_stage > lds_epi ? lds_stage : lds_epi;
   hipLaunchKernelGGL(HIP_KERNEL_NAME(
       gemm_kernel<BM, BN, BK, WM, WN, TA, TB, EPI, OUT_F32, ATOMIC>),
       grid, block, lds, s,
