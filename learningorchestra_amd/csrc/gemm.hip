@@ -16,6 +16,8 @@
 
 namespace lo {
 
+typedef short s16x4 __attribute__((ext_vector_type(4)));
+
 // mfma_f32_16x16x32_bf16 operand maps (verified by tests/test_gpu_ops.py's
 // probe): lane l holds A[row=l%16][k=(l/16)*8 + j], j=0..7 (contiguous k),
 // B[k][col] mirrored; C/D: col=lane&15, row=(lane>>4)*4+reg (guide §3).
@@ -99,39 +101,23 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
         }
       }
     } else {
-      // A is [K,M]: each thread loads a (KRA x 8) k-by-m tile with coalesced
-      // vector loads, transposes in registers, writes 8 m-row pieces of KRA
-      // contiguous k (2*KRA-byte LDS writes stay inside one 16-B granule, so
-      // the XOR swizzle is preserved). KRA adapts so all T threads stay busy
-      // on small tiles.
-      constexpr int CHV = BK * BM / 8;  // 8-wide vector loads in the tile
-      constexpr int KRA = (CHV >= T * 8) ? 8 : (CHV >= T * 4 ? 4
-                          : (CHV >= T * 2 ? 2 : 1));
-      constexpr int CH = CHV / KRA;
-      typedef bf16 bfv __attribute__((ext_vector_type(KRA)));
+      // A is [K,M]: stage NATURALLY (no transpose, no swizzle) as 16-wide
+      // m-subtiles — image elem (k, m) at (m/16)*(BK*16) + k*16 + m%16 — and
+      // let ds_read_b64_tr_b16 do the transpose at fragment-read time (each
+      // 16-lane group reads a contiguous 128-B region and receives its
+      // [4 k][16 m] transpose; verified by the on-GPU probe).
+      constexpr int CH = BK * BM / 8;
       for (int c = tid; c < CH; c += T) {
-        const int kc = c / (BM / 8), mc = c % (BM / 8);
-        const int gk0 = k0 + kc * KRA, gm0 = m0 + mc * 8;
-        bf16x8 v[KRA];
-        #pragma unroll
-        for (int kk = 0; kk < KRA; ++kk) {
-          const int gk = gk0 + kk;
-          bf16x8 t = {};
-          if (gk < K) {
-            if (gm0 + 8 <= M) t = *(const bf16x8*)(A + (long)gk * lda + gm0);
-            else if (gm0 < M)
-              for (int j = 0; j < 8 && gm0 + j < M; ++j) t[j] = A[(long)gk * lda + gm0 + j];
-          }
-          v[kk] = t;
+        const int k = c / (BM / 8), mc = c % (BM / 8);
+        const int gk = k0 + k, gm0 = m0 + mc * 8;
+        bf16x8 v = {};
+        if (gk < K) {
+          if (gm0 + 8 <= M) v = *(const bf16x8*)(A + (long)gk * lda + gm0);
+          else if (gm0 < M)
+            for (int j = 0; j < 8 && gm0 + j < M; ++j) v[j] = A[(long)gk * lda + gm0 + j];
         }
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          bfv w;
-          #pragma unroll
-          for (int kk = 0; kk < KRA; ++kk) w[kk] = v[kk][j];
-          const int row = mc * 8 + j;
-          *(bfv*)(smA + row * BKB + ((kc * KRA * 2) ^ ((row & SWZ) << 4))) = w;
-        }
+        const int msub = (mc * 8) / 16, mrem = (mc * 8) % 16;
+        *(bf16x8*)(smA + (msub * BK * 16 + k * 16 + mrem) * 2) = v;
       }
     }
     // ---- stage B tile -> Bs[BN][BK] ------------------------------------
@@ -158,34 +144,19 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
         }
       }
     } else {
-      constexpr int CHV = BK * BN / 8;
-      constexpr int KRB = (CHV >= T * 8) ? 8 : (CHV >= T * 4 ? 4
-                          : (CHV >= T * 2 ? 2 : 1));
-      constexpr int CH = CHV / KRB;
-      typedef bf16 bfvb __attribute__((ext_vector_type(KRB)));
+      // B is [K,N]: natural k-major 16-wide n-subtile image + tr16 reads
+      constexpr int CH = BK * BN / 8;
       for (int c = tid; c < CH; c += T) {
-        const int kc = c / (BN / 8), nc = c % (BN / 8);
-        const int gk0 = k0 + kc * KRB, gn0 = n0 + nc * 8;
-        bf16x8 v[KRB];
-        #pragma unroll
-        for (int kk = 0; kk < KRB; ++kk) {
-          const int gk = gk0 + kk;
-          bf16x8 t = {};
-          if (gk < K) {
-            if (gn0 + 8 <= N) t = *(const bf16x8*)(B + (long)gk * ldb + gn0);
-            else if (gn0 < N)
-              for (int j = 0; j < 8 && gn0 + j < N; ++j) t[j] = B[(long)gk * ldb + gn0 + j];
-          }
-          v[kk] = t;
+        const int k = c / (BN / 8), nc = c % (BN / 8);
+        const int gk = k0 + k, gn0 = n0 + nc * 8;
+        bf16x8 v = {};
+        if (gk < K) {
+          if (gn0 + 8 <= N) v = *(const bf16x8*)(B + (long)gk * ldb + gn0);
+          else if (gn0 < N)
+            for (int j = 0; j < 8 && gn0 + j < N; ++j) v[j] = B[(long)gk * ldb + gn0 + j];
         }
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          bfvb w;
-          #pragma unroll
-          for (int kk = 0; kk < KRB; ++kk) w[kk] = v[kk][j];
-          const int row = nc * 8 + j;
-          *(bfvb*)(smB + row * BKB + ((kc * KRB * 2) ^ ((row & SWZ) << 4))) = w;
-        }
+        const int nsub = (nc * 8) / 16, nrem = (nc * 8) % 16;
+        *(bf16x8*)(smB + (nsub * BK * 16 + k * 16 + nrem) * 2) = v;
       }
     }
   };
@@ -198,15 +169,47 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
       bf16x8 af[MFRAG], bf[NFRAG];
       #pragma unroll
       for (int mi = 0; mi < MFRAG; ++mi) {
-        const int row = wr * WTM + mi * 16 + fr;
-        af[mi] = *(const bf16x8*)(smA + row * BKB +
-                                  (((kk * 32 + fkb) * 2) ^ ((row & SWZ) << 4)));
+        if (!TA) {
+          const int row = wr * WTM + mi * 16 + fr;
+          af[mi] = *(const bf16x8*)(smA + row * BKB +
+                                    (((kk * 32 + fkb) * 2) ^ ((row & SWZ) << 4)));
+        } else {
+          // hardware-transpose read from the k-major image: two b64_tr_b16
+          // reads give k = kbase..+4 and +4..+8 of this lane's m column
+          const int msub = (wr * WTM + mi * 16) / 16;
+          const char* base = smA + (msub * BK * 16 + (kk * 32 + fkb) * 16) * 2
+                             + (lane & 15) * 8;
+          s16x4 lo4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (__attribute__((address_space(3))) s16x4*)base);
+          s16x4 hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (__attribute__((address_space(3))) s16x4*)(base + 128));
+          #pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            ((short*)&af[mi])[j] = lo4[j];
+            ((short*)&af[mi])[j + 4] = hi4[j];
+          }
+        }
       }
       #pragma unroll
       for (int ni = 0; ni < NFRAG; ++ni) {
-        const int row = wc * WTN + ni * 16 + fr;
-        bf[ni] = *(const bf16x8*)(smB + row * BKB +
-                                  (((kk * 32 + fkb) * 2) ^ ((row & SWZ) << 4)));
+        if (TB) {
+          const int row = wc * WTN + ni * 16 + fr;
+          bf[ni] = *(const bf16x8*)(smB + row * BKB +
+                                    (((kk * 32 + fkb) * 2) ^ ((row & SWZ) << 4)));
+        } else {
+          const int nsub = (wc * WTN + ni * 16) / 16;
+          const char* base = smB + (nsub * BK * 16 + (kk * 32 + fkb) * 16) * 2
+                             + (lane & 15) * 8;
+          s16x4 lo4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (__attribute__((address_space(3))) s16x4*)base);
+          s16x4 hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (__attribute__((address_space(3))) s16x4*)(base + 128));
+          #pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            ((short*)&bf[ni])[j] = lo4[j];
+            ((short*)&bf[ni])[j + 4] = hi4[j];
+          }
+        }
       }
       #pragma unroll
       for (int mi = 0; mi < MFRAG; ++mi)
