@@ -18,6 +18,17 @@ namespace lo {
 
 typedef short s16x4 __attribute__((ext_vector_type(4)));
 
+// tr16 image region permutation: k-block kb (4 k-rows each, 8 per 32-k MFMA
+// step) is stored at slot (kb/2) + (kb%2)*4 within its step, so one
+// ds_read_b64_tr_b16 instruction's four 16-lane groups (k = g*8 + j, i.e.
+// kb = {0,2,4,6} then {1,3,5,7}) read CONSECUTIVE 128-B regions — the
+// conflict-free pattern; the naive k-order is a 4-way bank conflict.
+LO_DEVICE int tr16_slot(int k) {
+  const int step = k >> 5;              // 32-k MFMA step
+  const int kb = (k & 31) >> 2;         // k-block within step
+  return step * 8 + (kb >> 1) + (kb & 1) * 4;
+}
+
 // mfma_f32_16x16x32_bf16 operand maps (verified by tests/test_gpu_ops.py's
 // probe): lane l holds A[row=l%16][k=(l/16)*8 + j], j=0..7 (contiguous k),
 // B[k][col] mirrored; C/D: col=lane&15, row=(lane>>4)*4+reg (guide §3).
@@ -117,7 +128,8 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
             for (int j = 0; j < 8 && gm0 + j < M; ++j) v[j] = A[(long)gk * lda + gm0 + j];
         }
         const int msub = (mc * 8) / 16, mrem = (mc * 8) % 16;
-        *(bf16x8*)(smA + (msub * BK * 16 + k * 16 + mrem) * 2) = v;
+        *(bf16x8*)(smA + (msub * BK * 16 + tr16_slot(k) * 64 + (k & 3) * 16
+                          + mrem) * 2) = v;
       }
     }
     // ---- stage B tile -> Bs[BN][BK] ------------------------------------
@@ -156,7 +168,8 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
             for (int j = 0; j < 8 && gn0 + j < N; ++j) v[j] = B[(long)gk * ldb + gn0 + j];
         }
         const int nsub = (nc * 8) / 16, nrem = (nc * 8) % 16;
-        *(bf16x8*)(smB + (nsub * BK * 16 + k * 16 + nrem) * 2) = v;
+        *(bf16x8*)(smB + (nsub * BK * 16 + tr16_slot(k) * 64 + (k & 3) * 16
+                          + nrem) * 2) = v;
       }
     }
   };
@@ -177,12 +190,12 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
           // hardware-transpose read from the k-major image: two b64_tr_b16
           // reads give k = kbase..+4 and +4..+8 of this lane's m column
           const int msub = (wr * WTM + mi * 16) / 16;
-          const char* base = smA + (msub * BK * 16 + (kk * 32 + fkb) * 16) * 2
-                             + (lane & 15) * 8;
+          const char* base = smA + (msub * BK * 16) * 2
+                             + (kk * 8 + (lane >> 4)) * 128 + (lane & 15) * 8;
           s16x4 lo4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
               (__attribute__((address_space(3))) s16x4*)base);
           s16x4 hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-              (__attribute__((address_space(3))) s16x4*)(base + 128));
+              (__attribute__((address_space(3))) s16x4*)(base + 512));
           #pragma unroll
           for (int j = 0; j < 4; ++j) {
             ((short*)&af[mi])[j] = lo4[j];
@@ -198,12 +211,12 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
                                     (((kk * 32 + fkb) * 2) ^ ((row & SWZ) << 4)));
         } else {
           const int nsub = (wc * WTN + ni * 16) / 16;
-          const char* base = smB + (nsub * BK * 16 + (kk * 32 + fkb) * 16) * 2
-                             + (lane & 15) * 8;
+          const char* base = smB + (nsub * BK * 16) * 2
+                             + (kk * 8 + (lane >> 4)) * 128 + (lane & 15) * 8;
           s16x4 lo4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
               (__attribute__((address_space(3))) s16x4*)base);
           s16x4 hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-              (__attribute__((address_space(3))) s16x4*)(base + 128));
+              (__attribute__((address_space(3))) s16x4*)(base + 512));
           #pragma unroll
           for (int j = 0; j < 4; ++j) {
             ((short*)&bf[ni])[j] = lo4[j];

@@ -247,7 +247,7 @@ def _splitk_heuristic(m: int, n: int, k: int) -> int:
     """Split-K factor for the dW GEMM C[m,n] with reduction depth k: fill the
     256-CU chip (~512 blocks of 32x64 tiles) without shredding k."""
     base_blocks = ((m + 31) // 32) * ((n + 63) // 64)
-    if base_blocks >= 256 or k < 4096:
+    if base_blocks >= 2048 or k < 2048:
         return 1
     return max(1, min(2048 // base_blocks, k // 1024))
 
