@@ -7,6 +7,11 @@
 #include <hip/hip_runtime.h>
 
 namespace lo {
+struct ConvGeom {
+  int B, H, W, C;
+  int KH, KW, SH, SW, PH, PW;
+  int OH, OW;
+};
 struct GemmArgs {
   const void *A, *B;
   void* C;
@@ -17,6 +22,8 @@ struct GemmArgs {
   int epi;
   bool out_f32;
   int splits;
+  int gather;
+  ConvGeom geom;
 };
 bool gemm_dispatch(const GemmArgs& g, hipStream_t s);
 void launch_mfma_probe(const void* A, const void* B, float* D, hipStream_t s);
@@ -112,7 +119,55 @@ bool gemm(at::Tensor A, at::Tensor B, at::Tensor C,
   }
   lo::GemmArgs g{A.data_ptr(), B.data_ptr(), C.data_ptr(), bias_p,
                  A.size(1), B.size(1), C.size(1), M, N, K, ta, tb,
-                 (int)epi, out_f32, (int)splits};
+                 (int)epi, out_f32, (int)splits, 0, {}};
+  return lo::gemm_dispatch(g, stream());
+}
+
+// implicit-GEMM conv forward: y2d[B*OH*OW, outC] = im2col(x) @ W^T (+bias,
+// +relu) without materializing col. W is [outC, kpad].
+bool gemm_conv_fwd(at::Tensor x, at::Tensor Wt, at::Tensor C,
+                   c10::optional<at::Tensor> bias, bool relu,
+                   int64_t KH, int64_t KW, int64_t SH, int64_t SW,
+                   int64_t PH, int64_t PW) {
+  check_bf16(x, "x");
+  check_bf16(Wt, "W");
+  check_bf16(C, "C");
+  const int B = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2),
+            Ci = (int)x.size(3);
+  const int OH = (int)((H + 2 * PH - KH) / SH) + 1;
+  const int OW = (int)((W + 2 * PW - KW) / SW) + 1;
+  const int M = (int)C.size(0), N = (int)C.size(1);
+  const int K = (int)Wt.size(1);  // kpad
+  TORCH_CHECK(M == (long)B * OH * OW && N == Wt.size(0), "conv shapes");
+  const float* bias_p = nullptr;
+  if (bias.has_value()) { check_f32(*bias, "bias"); bias_p = bias->data_ptr<float>(); }
+  lo::ConvGeom geom{B, H, W, Ci, (int)KH, (int)KW, (int)SH, (int)SW,
+                    (int)PH, (int)PW, OH, OW};
+  lo::GemmArgs g{x.data_ptr(), Wt.data_ptr(), C.data_ptr(), bias_p,
+                 K, Wt.size(1), C.size(1), M, N, K, false, true,
+                 relu ? 1 : 0, false, 1, 1, geom};
+  return lo::gemm_dispatch(g, stream());
+}
+
+// implicit-GEMM conv dW: gw[outC, kpad] = dY^T @ im2col(x), fp32 split-K
+bool gemm_conv_dw(at::Tensor dy2, at::Tensor x, at::Tensor gw, int64_t splits,
+                  int64_t KH, int64_t KW, int64_t SH, int64_t SW,
+                  int64_t PH, int64_t PW) {
+  check_bf16(dy2, "dy2");
+  check_bf16(x, "x");
+  check_f32(gw, "gw");
+  const int B = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2),
+            Ci = (int)x.size(3);
+  const int OH = (int)((H + 2 * PH - KH) / SH) + 1;
+  const int OW = (int)((W + 2 * PW - KW) / SW) + 1;
+  const int M = (int)gw.size(0), N = (int)gw.size(1);   // outC, kpad
+  const long K = dy2.size(0);                           // B*OH*OW
+  TORCH_CHECK(K == (long)B * OH * OW && M == dy2.size(1), "conv dW shapes");
+  lo::ConvGeom geom{B, H, W, Ci, (int)KH, (int)KW, (int)SH, (int)SW,
+                    (int)PH, (int)PW, OH, OW};
+  lo::GemmArgs g{dy2.data_ptr(), x.data_ptr(), gw.data_ptr(), nullptr,
+                 dy2.size(1), N, gw.size(1), M, N, (int)K, true, false,
+                 0, true, (int)std::max<int64_t>(splits, 2), 2, geom};
   return lo::gemm_dispatch(g, stream());
 }
 
@@ -388,6 +443,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("A"), py::arg("B"), py::arg("C"), py::arg("bias") = py::none(),
         py::arg("ta") = false, py::arg("tb") = false, py::arg("epi") = 0,
         py::arg("splits") = 1);
+  m.def("gemm_conv_fwd", &gemm_conv_fwd);
+  m.def("gemm_conv_dw", &gemm_conv_dw);
   m.def("mfma_probe", &mfma_probe);
   m.def("im2col", &im2col);
   m.def("col2im", &col2im);

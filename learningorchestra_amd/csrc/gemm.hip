@@ -29,16 +29,60 @@ LO_DEVICE int tr16_slot(int k) {
   return step * 8 + (kb >> 1) + (kb & 1) * 4;
 }
 
+// Implicit-GEMM conv: geometry for gathering im2col rows inside the GEMM
+// staging (SURVEY §2.9 implicit conv; never materializes the col matrix).
+struct ConvGeom {
+  int B, H, W, C;        // NHWC input
+  int KH, KW, SH, SW, PH, PW;
+  int OH, OW;
+};
+
+// gather 8 consecutive col elements of row `m` starting at column `k`
+// (C % 8 == 0: the run lies in one (kh,kw) channel segment; C < 8: scalar)
+LO_DEVICE bf16x8 conv_gather8(const bf16* __restrict__ x, const ConvGeom& g,
+                              long m, int k) {
+  bf16x8 v = {};
+  const int ow = (int)(m % g.OW);
+  const long t = m / g.OW;
+  const int oh = (int)(t % g.OH);
+  const int b = (int)(t / g.OH);
+  const int kk = g.KH * g.KW * g.C;
+  if (g.C % 8 == 0) {
+    if (k >= kk) return v;
+    const int c = k % g.C;
+    const int p = k / g.C;
+    const int kw = p % g.KW, kh = p / g.KW;
+    const int h = oh * g.SH - g.PH + kh, w = ow * g.SW - g.PW + kw;
+    if (h >= 0 && h < g.H && w >= 0 && w < g.W)
+      v = *(const bf16x8*)(x + (((long)b * g.H + h) * g.W + w) * g.C + c);
+    return v;
+  }
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    const int ke = k + e;
+    if (ke >= kk) break;
+    const int c = ke % g.C;
+    const int p = ke / g.C;
+    const int kw = p % g.KW, kh = p / g.KW;
+    const int h = oh * g.SH - g.PH + kh, w = ow * g.SW - g.PW + kw;
+    if (h >= 0 && h < g.H && w >= 0 && w < g.W)
+      v[e] = x[(((long)b * g.H + h) * g.W + w) * g.C + c];
+  }
+  return v;
+}
+
 // mfma_f32_16x16x32_bf16 operand maps (verified by tests/test_gpu_ops.py's
 // probe): lane l holds A[row=l%16][k=(l/16)*8 + j], j=0..7 (contiguous k),
 // B[k][col] mirrored; C/D: col=lane&15, row=(lane>>4)*4+reg (guide §3).
 
+// GATHER: 0 = plain, 1 = A is an im2col view of `A` as NHWC input (fwd
+// conv), 2 = B is an im2col view of `B` (dW with tr16 k-major image).
 template <int BM, int BN, int BK, int WM, int WN, bool TA, bool TB,
-          int EPI, bool OUT_F32, bool ATOMIC>
+          int EPI, bool OUT_F32, bool ATOMIC, int GATHER = 0>
 __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
     const bf16* __restrict__ A, long lda, const bf16* __restrict__ B, long ldb,
     void* __restrict__ Cv, long ldc, const float* __restrict__ bias,
-    int M, int N, int K, int kStart, int kChunk) {
+    int M, int N, int K, int kStart, int kChunk, ConvGeom geom) {
   constexpr int T = WM * WN * 64;
   constexpr int WTM = BM / WM;          // wave tile rows
   constexpr int WTN = BN / WN;          // wave tile cols
@@ -85,7 +129,19 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
     char* smB = smem + buf * BUFB + BM * BKB;
     const bool k_in = (k0 + BK <= K);
     // ---- stage A tile -> As[BM][BK] ------------------------------------
-    if (!TA) {
+    if (GATHER == 1) {
+      // implicit im2col gather of A rows (no col matrix): plain-load path,
+      // swizzled like the register path below
+      constexpr int CH = BM * BK / 8;
+      for (int c = tid; c < CH; c += T) {
+        const int row = c / (BK / 8), kc = c % (BK / 8);
+        const long gm = m0 + row;
+        const int gk = k0 + kc * 8;
+        bf16x8 v = {};
+        if (gm < M && gk < K) v = conv_gather8(A, geom, gm, gk);
+        *(bf16x8*)(smA + row * BKB + ((kc * 16) ^ ((row & SWZ) << 4))) = v;
+      }
+    } else if (!TA) {
       if (a_rows_in && k_in && ROWS_PER_SEG > 0) {
         // interior tiles: LDS-DMA (global_load_lds, 16 B/lane — guide §5
         // ladder step 3). glds writes lane-linear, so the bank-conflict XOR
@@ -156,13 +212,17 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
         }
       }
     } else {
-      // B is [K,N]: natural k-major 16-wide n-subtile image + tr16 reads
+      // B is [K,N]: natural k-major 16-wide n-subtile image + tr16 reads;
+      // GATHER==2 sources the row from the implicit im2col of the NHWC input
+      // (dW never re-reads a col matrix)
       constexpr int CH = BK * BN / 8;
       for (int c = tid; c < CH; c += T) {
         const int k = c / (BN / 8), nc = c % (BN / 8);
         const int gk = k0 + k, gn0 = n0 + nc * 8;
         bf16x8 v = {};
-        if (gk < K) {
+        if (GATHER == 2) {
+          if (gk < K && gn0 < N) v = conv_gather8(B, geom, (long)gk, gn0);
+        } else if (gk < K) {
           if (gn0 + 8 <= N) v = *(const bf16x8*)(B + (long)gk * ldb + gn0);
           else if (gn0 < N)
             for (int j = 0; j < 8 && gn0 + j < N; ++j) v[j] = B[(long)gk * ldb + gn0 + j];
@@ -345,10 +405,12 @@ struct GemmArgs {
   int epi;        // 0 none, 1 relu
   bool out_f32;
   int splits;     // >1 => atomic split-K accumulate into fp32 C (zeroed by caller)
+  int gather;     // 0 plain; 1 A = implicit im2col of NHWC input; 2 B likewise
+  ConvGeom geom;
 };
 
 template <int BM, int BN, int BK, int WM, int WN, bool TA, bool TB,
-          int EPI, bool OUT_F32, bool ATOMIC>
+          int EPI, bool OUT_F32, bool ATOMIC, int GATHER = 0>
 static void launch_cfg(const GemmArgs& g, hipStream_t s) {
   const int mb = cdiv(g.M, BM), nb = cdiv(g.N, BN);
   int kChunk;
@@ -364,10 +426,10 @@ static void launch_cfg(const GemmArgs& g, hipStream_t s) {
   size_t lds_epi = (size_t)BM * (BN * (g.out_f32 ? 4 : 2) + 16);
   size_t lds = lds_stage > lds_epi ? lds_stage : lds_epi;
   hipLaunchKernelGGL(HIP_KERNEL_NAME(
-      gemm_kernel<BM, BN, BK, WM, WN, TA, TB, EPI, OUT_F32, ATOMIC>),
+      gemm_kernel<BM, BN, BK, WM, WN, TA, TB, EPI, OUT_F32, ATOMIC, GATHER>),
       grid, block, lds, s,
       (const bf16*)g.A, g.lda, (const bf16*)g.B, g.ldb, g.C, g.ldc, g.bias,
-      g.M, g.N, g.K, 0, kChunk);
+      g.M, g.N, g.K, 0, kChunk, g.geom);
 }
 
 // Per-(TA,TB) tile-size selection. Instantiates only the combos the engine
@@ -403,10 +465,40 @@ static bool dispatch_tiles(const GemmArgs& g, hipStream_t s) {
   #undef LO_EPI_CASES_BK
 }
 
+// implicit-conv forward: C[M=B*OH*OW, N=outC] = im2col(x) @ W^T (+bias/relu)
+static bool dispatch_conv_fwd(const GemmArgs& g, hipStream_t s) {
+  #define LO_CONV_FWD(BM_, BN_, BK_, WM_, WN_)                                 \
+    do {                                                                       \
+      if (g.epi == 1) launch_cfg<BM_, BN_, BK_, WM_, WN_, false, true, 1, false, false, 1>(g, s); \
+      else            launch_cfg<BM_, BN_, BK_, WM_, WN_, false, true, 0, false, false, 1>(g, s); \
+      return true;                                                             \
+    } while (0)
+  if (g.out_f32) return false;
+  if (g.K <= 32) {
+    if (g.N > 32) LO_CONV_FWD(128, 64, 32, 2, 2);
+    if (g.N > 16) LO_CONV_FWD(256, 32, 32, 4, 1);
+    LO_CONV_FWD(128, 16, 32, 4, 1);
+  }
+  if (g.N > 64) LO_CONV_FWD(128, 128, 64, 2, 2);
+  if (g.N > 32) LO_CONV_FWD(128, 64, 64, 2, 2);
+  if (g.N > 16) LO_CONV_FWD(256, 32, 64, 4, 1);
+  LO_CONV_FWD(128, 16, 64, 4, 1);
+  #undef LO_CONV_FWD
+}
+
+// implicit-conv dW: C[M=outC, N=kpad] = dY^T @ im2col(x), split-K atomics
+static bool dispatch_conv_dw(const GemmArgs& g, hipStream_t s) {
+  if (!g.out_f32 || g.epi != 0 || !g.ta || g.tb) return false;
+  launch_cfg<32, 64, 64, 1, 4, true, false, 0, true, true, 2>(g, s);
+  return true;
+}
+
 // Returns true if a native config covered the shape; false => caller falls
 // back to a library GEMM (rocBLAS via torch) for the cold path.
 bool gemm_dispatch(const GemmArgs& g, hipStream_t s) {
   if (g.K % 8 != 0) return false;
+  if (g.gather == 1) return dispatch_conv_fwd(g, s);
+  if (g.gather == 2) return dispatch_conv_dw(g, s);
   if (g.splits > 1) {
     // split-K atomic accumulate (weight gradients): dW = A^T @ B
     // (32x64 tiles: measured better than 64x128 — block count beats

@@ -88,8 +88,9 @@ class Conv2dNHWC(Layer):
                 "key": key, "B": B, "H": H, "W": W, "OH": OH, "OW": OW,
                 "y": torch.empty((M, self.out_c), device=dev, dtype=dtype),
             }
-            if not self._is_1x1:
-                # col pad tail stays zero forever (im2col never writes it)
+            if not self._is_1x1 and not dev.type == "cuda":
+                # CPU reference path materializes col (GPU gathers it inside
+                # the GEMM staging — implicit conv)
                 self._bufs["col"] = torch.zeros((M, self.kpad), device=dev, dtype=dtype)
             if not self.first:
                 self._bufs["dx"] = torch.empty((B, H, W, self.in_c), device=dev, dtype=dtype)
@@ -111,28 +112,40 @@ class Conv2dNHWC(Layer):
         B, H, W, C = x.shape
         assert C == self.in_c
         bufs = self._alloc(B, H, W, x.device, x.dtype)
+        self._x = x
+        bias = self.arena.pf(self.name + ".b") if self.bias else None
         if self._is_1x1:
-            col = x.view(B * H * W, C)
-            bufs["col"] = col
+            bufs["col"] = x.view(B * H * W, C)
+            F.gemm(bufs["col"], self.arena.p(self.name + ".w"), tb=True,
+                   bias=bias, relu=self.relu, out=bufs["y"])
+        elif x.is_cuda:
+            # implicit conv: im2col gathered inside the GEMM staging
+            F.conv2d_fwd_implicit(x, self.arena.p(self.name + ".w"), self.kh,
+                                  self.kw, self.stride, self.stride, self.pad,
+                                  self.pad, bias=bias, relu=self.relu,
+                                  out=bufs["y"])
         else:
             F.im2col(x, self.kh, self.kw, self.stride, self.stride, self.pad,
                      self.pad, self.kpad, out=bufs["col"])
-            col = bufs["col"]
-        F.gemm(col, self.arena.p(self.name + ".w"), tb=True,
-               bias=self.arena.pf(self.name + ".b") if self.bias else None,
-               relu=self.relu, out=bufs["y"])
+            F.gemm(bufs["col"], self.arena.p(self.name + ".w"), tb=True,
+                   bias=bias, relu=self.relu, out=bufs["y"])
         return bufs["y"].view(B, bufs["OH"], bufs["OW"], self.out_c)
 
     def backward(self, dy: torch.Tensor) -> Optional[torch.Tensor]:
         bufs = self._bufs
-        M = bufs["col"].shape[0]
+        M = bufs["y"].shape[0]
         dy2 = dy.reshape(M, self.out_c)
         if self.relu:
             F.relu_bwd(dy2, bufs["y"], out=dy2)
         # weight grad: dW[outC, kpad] = dY^T @ col, split-K when M is deep
         gw = self.arena.g(self.name + ".w")
         splits = _splitk_heuristic(self.out_c, self.kpad, M)
-        F.gemm(dy2, bufs["col"], ta=True, out=gw, splits=splits)
+        if dy2.is_cuda and not self._is_1x1:
+            F.conv2d_dw_implicit(dy2, self._x, gw, self.kh, self.kw,
+                                 self.stride, self.stride, self.pad, self.pad,
+                                 splits)
+        else:
+            F.gemm(dy2, bufs["col"], ta=True, out=gw, splits=splits)
         if self.bias:
             F.colsum(dy2, out=self.arena.g(self.name + ".b"))
         if self.first:

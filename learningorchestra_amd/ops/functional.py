@@ -396,3 +396,38 @@ def avgpool_global_bwd(dy: torch.Tensor, H: int, W: int,
     out.copy_((dy.float() / (H * W)).view(B, 1, 1, C)
               .expand(B, H, W, C).to(out.dtype))
     return out
+
+
+# ------------------------------------------------------- implicit-GEMM conv
+def conv2d_fwd_implicit(x: torch.Tensor, w: torch.Tensor, kh: int, kw: int,
+                        sh: int, sw: int, ph: int, pw: int,
+                        bias: Optional[torch.Tensor] = None, relu: bool = False,
+                        out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """GPU-only fused conv forward: y2d[B*OH*OW, outC] = im2col(x) @ w^T
+    gathered inside the GEMM staging (no col matrix). w is [outC, kpad]."""
+    assert x.is_cuda
+    B, H, W, C = x.shape
+    OH = (H + 2 * ph - kh) // sh + 1
+    OW = (W + 2 * pw - kw) // sw + 1
+    if out is None:
+        out = torch.empty((B * OH * OW, w.shape[0]), device=x.device,
+                          dtype=x.dtype)
+    lo = require_ext()
+    ok = lo.gemm_conv_fwd(x, w, out, bias, relu, kh, kw, sh, sw, ph, pw)
+    if not ok:
+        raise RuntimeError("gemm_conv_fwd: no config for this shape")
+    return out
+
+
+def conv2d_dw_implicit(dy2: torch.Tensor, x: torch.Tensor, gw: torch.Tensor,
+                       kh: int, kw: int, sh: int, sw: int, ph: int, pw: int,
+                       splits: int) -> torch.Tensor:
+    """GPU-only fused conv weight grad: gw[outC, kpad] = dY^T @ im2col(x),
+    split-K fp32 atomics (gw zeroed here)."""
+    assert x.is_cuda
+    lo = require_ext()
+    gw.zero_()
+    ok = lo.gemm_conv_dw(dy2, x, gw, max(splits, 2), kh, kw, sh, sw, ph, pw)
+    if not ok:
+        raise RuntimeError("gemm_conv_dw: no config for this shape")
+    return gw

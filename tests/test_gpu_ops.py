@@ -197,3 +197,43 @@ def test_relu_bwd(lo):
     ref = F.relu_bwd(dy.clone(), y)
     got = F.relu_bwd(dy.cuda(), y.cuda()).cpu()
     assert torch.equal(ref.float(), got.float())
+
+
+def test_implicit_conv_fwd_matches_im2col(lo):
+    """Implicit-GEMM conv (gather inside staging) vs the materialized
+    im2col + GEMM path — bit-comparable (same math order)."""
+    from learningorchestra_amd.ops import functional as F
+    torch.manual_seed(11)
+    for (B, H, W, C, outC, k, s, p) in [(4, 28, 28, 1, 32, 5, 1, 0),
+                                        (2, 14, 14, 32, 64, 5, 1, 0),
+                                        (2, 16, 16, 32, 64, 3, 2, 1),
+                                        (2, 33, 33, 8, 24, 3, 1, 1)]:
+        kdim = k * k * C
+        kpad = (kdim + 7) // 8 * 8
+        x = torch.randn(B, H, W, C).bfloat16().cuda()
+        w = torch.zeros(outC, kpad).bfloat16().cuda()
+        w[:, :kdim] = torch.randn(outC, kdim).bfloat16()
+        bias = torch.randn(outC).cuda()
+        col = F.im2col(x, k, k, s, s, p, p, kpad)
+        ref = F.gemm(col, w, tb=True, bias=bias, relu=True)
+        got = F.conv2d_fwd_implicit(x, w, k, k, s, s, p, p, bias=bias, relu=True)
+        rel = ((got.float() - ref.float()).norm() / (ref.float().norm() + 1e-8)).item()
+        assert rel < 1e-3, (B, H, W, C, outC, k, s, p, rel)
+
+
+def test_implicit_conv_dw_matches_im2col(lo):
+    from learningorchestra_amd.ops import functional as F
+    torch.manual_seed(12)
+    for (B, H, W, C, outC, k, s, p) in [(4, 28, 28, 1, 32, 5, 1, 0),
+                                        (2, 16, 16, 32, 64, 3, 2, 1)]:
+        kdim = k * k * C
+        kpad = (kdim + 7) // 8 * 8
+        OH = (H + 2 * p - k) // s + 1
+        x = torch.randn(B, H, W, C).bfloat16().cuda()
+        dy2 = torch.randn(B * OH * OH, outC).bfloat16().cuda()
+        col = F.im2col(x, k, k, s, s, p, p, kpad)
+        ref = F.gemm(dy2, col, ta=True, splits=4, out_dtype=torch.float32)
+        gw = torch.empty(outC, kpad, device="cuda", dtype=torch.float32)
+        F.conv2d_dw_implicit(dy2, x, gw, k, k, s, s, p, p, splits=4)
+        rel = ((gw - ref).norm() / (ref.norm() + 1e-8)).item()
+        assert rel < 1e-3, (B, H, W, C, outC, k, s, p, rel)
