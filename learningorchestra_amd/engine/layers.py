@@ -47,9 +47,14 @@ class Conv2dNHWC(Layer):
 
     def __init__(self, name: str, in_c: int, out_c: int, kh: int, kw: int,
                  stride: int = 1, pad: int = 0, relu: bool = True,
-                 first: bool = False, bias: bool = True):
+                 first: bool = False, bias: bool = True,
+                 implicit: bool = False):
         self.name = name
         self.bias = bias
+        # implicit-GEMM conv (col gathered inside GEMM staging): numerically
+        # identical, saves the col buffer entirely, but measured slightly
+        # slower than glds-staged materialized col on MI355X — opt-in.
+        self.implicit = implicit
         self.in_c, self.out_c = in_c, out_c
         self.kh, self.kw, self.stride, self.pad = kh, kw, stride, pad
         self.relu = relu
@@ -88,7 +93,7 @@ class Conv2dNHWC(Layer):
                 "key": key, "B": B, "H": H, "W": W, "OH": OH, "OW": OW,
                 "y": torch.empty((M, self.out_c), device=dev, dtype=dtype),
             }
-            if not self._is_1x1 and not dev.type == "cuda":
+            if not self._is_1x1 and not (dev.type == "cuda" and self.implicit):
                 # CPU reference path materializes col (GPU gathers it inside
                 # the GEMM staging — implicit conv)
                 self._bufs["col"] = torch.zeros((M, self.kpad), device=dev, dtype=dtype)
@@ -118,7 +123,7 @@ class Conv2dNHWC(Layer):
             bufs["col"] = x.view(B * H * W, C)
             F.gemm(bufs["col"], self.arena.p(self.name + ".w"), tb=True,
                    bias=bias, relu=self.relu, out=bufs["y"])
-        elif x.is_cuda:
+        elif x.is_cuda and self.implicit:
             # implicit conv: im2col gathered inside the GEMM staging
             F.conv2d_fwd_implicit(x, self.arena.p(self.name + ".w"), self.kh,
                                   self.kw, self.stride, self.stride, self.pad,
@@ -140,7 +145,7 @@ class Conv2dNHWC(Layer):
         # weight grad: dW[outC, kpad] = dY^T @ col, split-K when M is deep
         gw = self.arena.g(self.name + ".w")
         splits = _splitk_heuristic(self.out_c, self.kpad, M)
-        if dy2.is_cuda and not self._is_1x1:
+        if dy2.is_cuda and self.implicit and not self._is_1x1:
             F.conv2d_dw_implicit(dy2, self._x, gw, self.kh, self.kw,
                                  self.stride, self.stride, self.pad, self.pad,
                                  splits)

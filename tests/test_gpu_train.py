@@ -39,10 +39,10 @@ def test_training_converges_gpu():
         return x, y
 
     first, _ = trainer.step(*batch())
-    for _ in range(100):
+    for _ in range(200):
         trainer.step_async(*batch())
     last, acc = trainer.step(*batch())
-    assert last < first * 0.75, (first, last)
+    assert last < first * 0.7, (first, last)
 
 
 def test_graph_capture_step():
