@@ -196,6 +196,22 @@ def create_app(runtime: Optional[Runtime] = None) -> FastAPI:
         rt.datatype.convert(name, _field(body, "types"))
         return {RESULT: _poll_uri("transform", "dataType", name)}
 
+    # ------------------------------------------------------------- metrics --
+    @app.get(PREFIX + "/metrics")
+    def metrics():
+        import torch
+        cats = {}
+        for doc in rt.metadata.catalog():
+            t = doc.get("type", "?")
+            cats[t] = cats.get(t, 0) + 1
+        gpu = {"available": torch.cuda.is_available()}
+        if gpu["available"]:
+            gpu["device"] = torch.cuda.get_device_name(0)
+            gpu["mem_allocated"] = torch.cuda.memory_allocated()
+        return {RESULT: {"artifactsByType": cats,
+                         "collections": len(rt.db.list_collection_names()),
+                         "device": rt.cfg.resolve_device(), "gpu": gpu}}
+
     # ------------------------------------------------------------- observe --
     @app.get(PREFIX + "/observe/{name}")
     def observe(name: str):

@@ -106,3 +106,68 @@ class MnistCNN:
 
     def load_state_dict(self, sd):
         self.model.load_state_dict(sd)
+
+
+class TextCNNClassifier:
+    """Keras-surface wrapper over the TextCNN engine model (IMDb config)."""
+
+    def __init__(self, vocab: int = 20000, emb_dim: int = 128,
+                 filters: int = 128, kernel_sizes=(3, 4, 5),
+                 num_classes: int = 2, seed: int = 0, lr: float = 0.05,
+                 batch_size: int = 512, device: Optional[str] = None):
+        from .textcnn import build_textcnn
+        self.device = _default_device(device)
+        self.model = build_textcnn(self.device, seed=seed, vocab=vocab,
+                                   emb_dim=emb_dim, filters=filters,
+                                   kernel_sizes=tuple(kernel_sizes),
+                                   num_classes=num_classes)
+        self.trainer = Trainer(self.model, make_sgd(self.model, lr=lr),
+                               device=self.device)
+        self.batch_size = batch_size
+
+    def _to_ids(self, x) -> torch.Tensor:
+        if hasattr(x, "to_numpy"):
+            x = x.to_numpy()
+        return torch.as_tensor(np.asarray(x), dtype=torch.long,
+                               device=self.device)
+
+    def fit(self, x=None, y=None, epochs: int = 1,
+            batch_size: Optional[int] = None, verbose: int = 0) -> dict:
+        ids = self._to_ids(x)
+        yt = torch.as_tensor(np.asarray(y), dtype=torch.long,
+                             device=self.device)
+        bs = min(batch_size or self.batch_size, ids.shape[0])
+        losses = []
+        for _ in range(int(epochs)):
+            perm = torch.randperm(ids.shape[0], device=self.device)
+            tot = seen = 0.0
+            for i in range(0, ids.shape[0] - bs + 1, bs):
+                sel = perm[i:i + bs]
+                loss, _ = self.trainer.step(ids[sel].contiguous(),
+                                            yt[sel].contiguous())
+                tot += loss * bs
+                seen += bs
+            losses.append(tot / max(seen, 1))
+        return {"loss": losses}
+
+    @torch.no_grad()
+    def predict(self, x, batch_size: int = 4096) -> np.ndarray:
+        ids = self._to_ids(x)
+        outs = []
+        for i in range(0, ids.shape[0], batch_size):
+            outs.append(self.model.predict(ids[i:i + batch_size].contiguous())
+                        .cpu().numpy())
+        return np.concatenate(outs) if outs else np.empty(0, dtype=np.int32)
+
+    @torch.no_grad()
+    def evaluate(self, x, y, batch_size: int = 4096) -> dict:
+        preds = self.predict(x, batch_size)
+        yt = np.asarray(y)
+        return {"accuracy": float((preds == yt.astype(preds.dtype)).mean()),
+                "n": int(len(preds))}
+
+    def state_dict(self):
+        return self.model.state_dict()
+
+    def load_state_dict(self, sd):
+        self.model.load_state_dict(sd)
