@@ -61,8 +61,8 @@ void launch_relu_bwd(const void* dy, const void* y, void* dx, long n, hipStream_
 void launch_sgd(void* master, const void* grad, void* mom, void* mirror, long n,
                 float lr, float mu, float wd, float gscale, hipStream_t s);
 void launch_adam(void* master, const void* grad, void* m1, void* m2, void* mirror,
-                 long n, float lr, float b1, float b2, float eps, float wd,
-                 float c1, float c2, float gscale, hipStream_t s);
+                 const void* step_dev, long n, float lr, float b1, float b2,
+                 float eps, float wd, float gscale, hipStream_t s);
 void launch_colsum(const void* dy, void* out, long M, int N, long ldy, hipStream_t s);
 void launch_argmax_rows(const void* x, void* out, long M, int C, int Cvalid,
                         long ldx, hipStream_t s);
@@ -336,17 +336,19 @@ void sgd_step(at::Tensor master, at::Tensor grad, at::Tensor mom, at::Tensor mir
 }
 
 void adam_step(at::Tensor master, at::Tensor grad, at::Tensor m1, at::Tensor m2,
-               at::Tensor mirror, double lr, double b1, double b2, double eps,
-               double wd, double c1, double c2, double gscale) {
+               at::Tensor mirror, at::Tensor step_dev, double lr, double b1,
+               double b2, double eps, double wd, double gscale) {
   check_f32(master, "master");
   check_f32(grad, "grad");
   check_f32(m1, "m1");
   check_f32(m2, "m2");
   check_bf16(mirror, "mirror");
+  TORCH_CHECK(step_dev.scalar_type() == at::kInt && step_dev.is_cuda(),
+              "step_dev must be int32 on GPU");
   TORCH_CHECK(master.numel() % 4 == 0, "arena must be 4-aligned");
   lo::launch_adam(master.data_ptr(), grad.data_ptr(), m1.data_ptr(), m2.data_ptr(),
-                  mirror.data_ptr(), master.numel(), (float)lr, (float)b1,
-                  (float)b2, (float)eps, (float)wd, (float)c1, (float)c2,
+                  mirror.data_ptr(), step_dev.data_ptr(), master.numel(),
+                  (float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
                   (float)gscale, stream());
 }
 

@@ -75,11 +75,17 @@ void launch_sgd(void* master, const void* grad, void* mom, void* mirror, long n,
 }
 
 // ---------------------------------------------------------- adam_step ------
+// step_dev: device step counter (graph-capture-safe bias correction — the
+// count is read on DEVICE, so replaying a captured step keeps correcting)
 __global__ void adam_kernel(float* __restrict__ master, const float* __restrict__ grad,
                             float* __restrict__ m1, float* __restrict__ m2,
-                            bf16* __restrict__ mirror, long n4, float lr,
+                            bf16* __restrict__ mirror,
+                            const int* __restrict__ step_dev, long n4, float lr,
                             float b1, float b2, float eps, float wd,
-                            float c1, float c2, float gscale) {
+                            float gscale) {
+  const float t = (float)*step_dev;
+  const float c1 = 1.f / (1.f - __powf(b1, t));
+  const float c2 = 1.f / (1.f - __powf(b2, t));
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
        i += (long)gridDim.x * blockDim.x) {
     f32x4 p = *(const f32x4*)(master + i * 4);
@@ -105,14 +111,15 @@ __global__ void adam_kernel(float* __restrict__ master, const float* __restrict_
 }
 
 void launch_adam(void* master, const void* grad, void* m1, void* m2, void* mirror,
-                 long n, float lr, float b1, float b2, float eps, float wd,
-                 float c1, float c2, float gscale, hipStream_t s) {
+                 const void* step_dev, long n, float lr, float b1, float b2,
+                 float eps, float wd, float gscale, hipStream_t s) {
   const long n4 = n / 4;
   const int block = 256;
   const int grid = (int)min((n4 + block - 1) / block, (long)2048);
   hipLaunchKernelGGL(adam_kernel, dim3(grid), dim3(block), 0, s,
                      (float*)master, (const float*)grad, (float*)m1, (float*)m2,
-                     (bf16*)mirror, n4, lr, b1, b2, eps, wd, c1, c2, gscale);
+                     (bf16*)mirror, (const int*)step_dev, n4, lr, b1, b2, eps,
+                     wd, gscale);
 }
 
 // ------------------------------------------------------------- colsum ------

@@ -129,10 +129,18 @@ class Adam:
         self.lr, self.betas, self.eps, self.weight_decay = lr, betas, eps, weight_decay
         self.m1 = torch.zeros_like(arena.master)
         self.m2 = torch.zeros_like(arena.master)
-        self.t = 0
+        if arena.master.is_cuda:
+            # device step counter: the increment and the bias correction both
+            # happen on-device, so a hipGraph-captured step stays correct
+            self.t = torch.zeros(1, dtype=torch.int32, device=arena.master.device)
+        else:
+            self.t = 0
 
     def step(self, gscale: float = 1.0) -> None:
-        self.t += 1
+        if torch.is_tensor(self.t):
+            self.t += 1
+        else:
+            self.t += 1
         a = self.arena
         F.adam_step(a.master, a.grad, self.m1, self.m2, a.mirror, self.lr,
                     self.betas[0], self.betas[1], self.eps, self.weight_decay,

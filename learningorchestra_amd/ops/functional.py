@@ -188,15 +188,21 @@ def sgd_step(master: torch.Tensor, grad: torch.Tensor, mom: torch.Tensor,
 
 def adam_step(master: torch.Tensor, grad: torch.Tensor, m1: torch.Tensor,
               m2: torch.Tensor, mirror: torch.Tensor, lr: float, b1: float,
-              b2: float, eps: float, wd: float, step: int,
+              b2: float, eps: float, wd: float, step,
               gscale: float = 1.0) -> None:
-    c1 = 1.0 / (1.0 - b1 ** step)
-    c2 = 1.0 / (1.0 - b2 ** step)
+    ``step``: int on CPU, or an int32 device scalar tensor on GPU (read
+    on device — hipGraph-capture-safe bias correction).
     if _is_gpu(master):
         lo = require_ext()
-        lo.adam_step(master, grad, m1, m2, mirror, lr, b1, b2, eps, wd, c1, c2,
+        if not torch.is_tensor(step):
+            step = torch.tensor([int(step)], dtype=torch.int32,
+                                device=master.device)
+        lo.adam_step(master, grad, m1, m2, mirror, step, lr, b1, b2, eps, wd,
                      gscale)
         return
+    t = int(step.item()) if torch.is_tensor(step) else int(step)
+    c1 = 1.0 / (1.0 - b1 ** t)
+    c2 = 1.0 / (1.0 - b2 ** t)
     g = grad * gscale + wd * master
     m1.mul_(b1).add_(g, alpha=1 - b1)
     m2.mul_(b2).addcmul_(g, g, value=1 - b2)
