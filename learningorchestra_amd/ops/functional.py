@@ -190,8 +190,8 @@ def adam_step(master: torch.Tensor, grad: torch.Tensor, m1: torch.Tensor,
               m2: torch.Tensor, mirror: torch.Tensor, lr: float, b1: float,
               b2: float, eps: float, wd: float, step,
               gscale: float = 1.0) -> None:
-    ``step``: int on CPU, or an int32 device scalar tensor on GPU (read
-    on device — hipGraph-capture-safe bias correction).
+    """``step``: int on CPU, or an int32 device scalar tensor on GPU (read
+    on device -- hipGraph-capture-safe bias correction)."""
     if _is_gpu(master):
         lo = require_ext()
         if not torch.is_tensor(step):
