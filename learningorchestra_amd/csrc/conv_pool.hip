@@ -29,10 +29,10 @@ __global__ void im2col_row_kernel(const bf16* __restrict__ in, bf16* __restrict_
   const int K = KH * KW * C;
   for (long rrow = (long)blockIdx.x * blockDim.x + threadIdx.x; rrow < total;
        rrow += (long)gridDim.x * blockDim.x) {
-    long r = rrow;
-    const int ow = r % OW; r /= OW;
-    const int oh = r % OH; r /= OH;
-    const int b = r;
+    unsigned int r32 = (unsigned int)rrow;
+    const int ow = r32 % OW; r32 /= OW;
+    const int oh = r32 % OH; r32 /= OH;
+    const int b = r32;
     bf16 vals[KPAD];
     #pragma unroll
     for (int i = 0; i < KPAD; ++i) vals[i] = bf16(0.f);
@@ -64,13 +64,13 @@ __global__ void im2col_kernel(const bf16* __restrict__ in, bf16* __restrict__ co
   const long total = (long)B * OH * OW * KH * KW * CV;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
-    int cu = i % CV;
-    long r = i / CV;
-    const int kw = r % KW; r /= KW;
-    const int kh = r % KH; r /= KH;
-    const int ow = r % OW; r /= OW;
-    const int oh = r % OH; r /= OH;
-    const int b = r;
+    unsigned int r32 = (unsigned int)i;   // totals < 2^31 by launch contract
+    const int cu = r32 % CV; r32 /= CV;
+    const int kw = r32 % KW; r32 /= KW;
+    const int kh = r32 % KH; r32 /= KH;
+    const int ow = r32 % OW; r32 /= OW;
+    const int oh = r32 % OH; r32 /= OH;
+    const int b = r32;
     const int h = oh * SH - PH + kh, w = ow * SW - PW + kw;
     const long row = ((long)b * OH + oh) * OW + ow;
     const long kidx = ((long)kh * KW + kw) * C + cu * (VEC8 ? 8 : 1);
@@ -149,11 +149,11 @@ __global__ void col2im_kernel(const bf16* __restrict__ dcol, bf16* __restrict__ 
   const long total = (long)B * H * W * CV;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
-    int cu = i % CV;
-    long r = i / CV;
-    const int w = r % W; r /= W;
-    const int h = r % H; r /= H;
-    const int b = r;
+    unsigned int r32 = (unsigned int)i;
+    const int cu = r32 % CV; r32 /= CV;
+    const int w = r32 % W; r32 /= W;
+    const int h = r32 % H; r32 /= H;
+    const int b = r32;
     float acc[VEC8 ? 8 : 1] = {};
     for (int kh = 0; kh < KH; ++kh) {
       const int ohs = h + PH - kh;
@@ -212,11 +212,11 @@ __global__ void maxpool_fwd_kernel(const bf16* __restrict__ in, bf16* __restrict
   const long total = (long)B * OH * OW * CV;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
-    int cu = i % CV;
-    long r = i / CV;
-    const int ow = r % OW; r /= OW;
-    const int oh = r % OH; r /= OH;
-    const int b = r;
+    unsigned int r32 = (unsigned int)i;
+    const int cu = r32 % CV; r32 /= CV;
+    const int ow = r32 % OW; r32 /= OW;
+    const int oh = r32 % OH; r32 /= OH;
+    const int b = r32;
     const int NE = VEC8 ? 8 : 1;
     float best[NE];
     unsigned char bidx[NE];
@@ -269,11 +269,11 @@ __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
   const long total = (long)B * H * W * CV;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
-    int cu = i % CV;
-    long r = i / CV;
-    const int w = r % W; r /= W;
-    const int h = r % H; r /= H;
-    const int b = r;
+    unsigned int r32 = (unsigned int)i;
+    const int cu = r32 % CV; r32 /= CV;
+    const int w = r32 % W; r32 /= W;
+    const int h = r32 % H; r32 /= H;
+    const int b = r32;
     const int NE = VEC8 ? 8 : 1;
     float acc[NE];
     #pragma unroll
