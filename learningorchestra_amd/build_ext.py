@@ -21,6 +21,7 @@ EXT_NAME = "_lo_C"
 SOURCES = [
     os.path.join(CSRC, "bindings.cpp"),
     os.path.join(CSRC, "gemm.hip"),
+    os.path.join(CSRC, "gemm_8phase.hip"),
     os.path.join(CSRC, "conv_pool.hip"),
     os.path.join(CSRC, "elementwise.hip"),
     os.path.join(CSRC, "softmax_ce.hip"),
