@@ -96,30 +96,34 @@ __global__ __launch_bounds__(512) void gemm256_kernel(
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   asm volatile("s_barrier" ::: "memory");
 
-  // ---- main loop: 4 phases per K-tile ------------------------------------
+  // ---- main loop: 4 phases per K-tile. The A quadrant for phase q+1 is
+  // ds_read right after phase q's MFMA cluster (same buffer, stable until
+  // the NEXT tile's stages), so its lgkmcnt wait spans two barriers and the
+  // staging section — the per-phase LDS latency is hidden (PMC showed
+  // WAIT_INST_LDS ~= busy cycles with reads issued in their own phase).
   for (int s = 0; s < ntiles; ++s) {
     const int buf = s & 1;
-    #pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      // phase reads: A quadrant q (4 x ds_read_b128); q0 also B (8)
-      bf16x8 afr[2][2];
+    bf16x8 afr[2][2][2];                    // [phase parity][mi-pair][kk]
+    auto read_quad = [&](int q, int pp) {
       #pragma unroll
       for (int j = 0; j < 2; ++j) {
-        const int mi = q * 2 + j;
-        const int row = wr * 128 + mi * 16 + fr;
+        const int row = wr * 128 + (q * 2 + j) * 16 + fr;
         #pragma unroll
         for (int kk = 0; kk < 2; ++kk)
-          afr[j][kk] = lds_frag(buf, 0, row, kk);
+          afr[pp][j][kk] = lds_frag(buf, 0, row, kk);
       }
-      if (q == 0) {
-        #pragma unroll
-        for (int ni = 0; ni < NFRAG; ++ni) {
-          const int row = wc * 64 + ni * 16 + fr;
-          #pragma unroll
-          for (int kk = 0; kk < 2; ++kk)
-            bfr[ni][kk] = lds_frag(buf, BM * BKB, row, kk);
-        }
-      }
+    };
+    // tile prologue (q0 section): B frags + quadrant 0
+    #pragma unroll
+    for (int ni = 0; ni < NFRAG; ++ni) {
+      const int row = wc * 64 + ni * 16 + fr;
+      #pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        bfr[ni][kk] = lds_frag(buf, BM * BKB, row, kk);
+    }
+    read_quad(0, 0);
+    #pragma unroll
+    for (int q = 0; q < 4; ++q) {
       // phase staging (schedule in the header comment)
       if (q == 1) stage_A(s + 1, 0);
       else if (q == 2) stage_A(s + 1, 1);
@@ -137,8 +141,9 @@ __global__ __launch_bounds__(512) void gemm256_kernel(
           #pragma unroll
           for (int kk = 0; kk < 2; ++kk)
             acc[q * 2 + j][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afr[j][kk], bfr[ni][kk], acc[q * 2 + j][ni], 0, 0, 0);
+                afr[q & 1][j][kk], bfr[ni][kk], acc[q * 2 + j][ni], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
+      if (q < 3) read_quad(q + 1, (q + 1) & 1);
       asm volatile("s_barrier" ::: "memory");
     }
   }
