@@ -6,7 +6,7 @@
 // cluster (guide T5).
 //
 // The staging schedule was derived with an explicit region-lifetime + vmcnt
-// FIFO simulator (tools/sched_sim; 156 valid schedules, this is the
+// FIFO simulator (tools/sched_sim.py; 156 valid schedules, this is the
 // latest-staging one):
 //   tile s computes in phases q0..q3 (globally 4s+1..4s+4), buffer s%2
 //   q0: read B frags (8 ds_read_b128) + A quadrant 0

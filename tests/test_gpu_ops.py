@@ -237,3 +237,20 @@ def test_implicit_conv_dw_matches_im2col(lo):
         F.conv2d_dw_implicit(dy2, x, gw, k, k, s, s, p, p, splits=4)
         rel = ((gw - ref).norm() / (ref.norm() + 1e-8)).item()
         assert rel < 1e-3, (B, H, W, C, outC, k, s, p, rel)
+
+
+def test_gemm_8phase_race_screen(lo):
+    """The 256^2 8-phase kernel has hand-placed barriers/counted vmcnt (a
+    sync-structure template): multi-trial multi-shape refcheck per the
+    guide's two-lane discipline."""
+    from learningorchestra_amd.ops import functional as F
+    torch.manual_seed(123)
+    for trial in range(3):
+        for Msz, Nsz, Ksz in [(4096, 4096, 512), (4096, 4096, 384),
+                              (8192, 256, 2048)]:
+            A = torch.randn(Msz, Ksz, device="cuda").bfloat16()
+            B = torch.randn(Nsz, Ksz, device="cuda").bfloat16()
+            out = F.gemm(A, B, tb=True)
+            ref = A.float() @ B.float().t()
+            rel = ((out.float() - ref).norm() / ref.norm()).item()
+            assert rel < 2e-2, (trial, Msz, Nsz, Ksz, rel)
