@@ -460,16 +460,19 @@ static bool dispatch_tiles(const GemmArgs& g, hipStream_t s) {
       return true;                                                             \
     } while (0)
 
+  // 128x128 only when its grid still fills the chip; else halve BM for 2x
+  // the blocks (e.g. fc1 fwd 8192x256 was 128 blocks at 128^2)
+  const bool big_grid = ((long)cdiv(g.M, 128) * cdiv(g.N, 128)) >= 256;
   if (g.K <= 32) {
     // shallow reductions (e.g. C=1 5x5 conv, kpad 32): half-depth tiles so
     // the LDS image and glds path are fully used
-    if (g.N > 64 && g.M >= 4096) LO_EPI_CASES_BK(128, 128, 32, 2, 2);
+    if (g.N > 64 && big_grid) LO_EPI_CASES_BK(128, 128, 32, 2, 2);
     if (g.N > 64) LO_EPI_CASES_BK(64, 128, 32, 2, 2);
     if (g.N > 32) LO_EPI_CASES_BK(128, 64, 32, 2, 2);
     if (g.N > 16) LO_EPI_CASES_BK(256, 32, 32, 4, 1);
     LO_EPI_CASES_BK(128, 16, 32, 4, 1);
   }
-  if (g.N > 64 && g.M >= 4096) LO_EPI_CASES_BK(128, 128, 64, 2, 2);
+  if (g.N > 64 && big_grid) LO_EPI_CASES_BK(128, 128, 64, 2, 2);
   if (g.N > 64) LO_EPI_CASES_BK(64, 128, 64, 2, 2);   // small-M wide-N (dX tails)
   if (g.N > 32) LO_EPI_CASES_BK(128, 64, 64, 2, 2);
   if (g.N > 16) LO_EPI_CASES_BK(256, 32, 64, 4, 1);

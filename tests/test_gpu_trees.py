@@ -45,3 +45,33 @@ def test_gbt_gpu_end_to_end():
     acc = (clf.predict(X[:20000].numpy()).astype(int)
            == y[:20000].numpy().astype(int)).mean()
     assert acc > 0.8, acc
+
+
+def test_builder_native_classifiers_gpu(tmp_path):
+    """builder/sparkml with the native GPU tabular engines end-to-end via the
+    in-process client (LR on the engine + GBT histogram trees on cuda)."""
+    from learning_orchestra_client import BuilderSparkMl, Context, DatasetCsv
+    from learningorchestra_amd.api.app import Runtime
+    from learningorchestra_amd.config import Config
+    from learningorchestra_amd.data.synthetic import titanic_csv
+    cfg = Config(data_root=str(tmp_path), mongo_uri="")
+    ctx = Context.in_process(Runtime(cfg))
+    p = tmp_path / "t.csv"
+    p.write_text(titanic_csv(rows=400))
+    DatasetCsv(ctx).insert_sync("titanic", str(p))
+    code = (
+        "import pandas as pd\n"
+        "def prep(df):\n"
+        "    out = df[['Pclass','SibSp','Parch','Fare']].copy()\n"
+        "    out['sex_n'] = (df['Sex'] == 'female').astype(float)\n"
+        "    out['label'] = df['Survived'].astype(float)\n"
+        "    return out.fillna(0.0)\n"
+        "features_training = prep(training_df)\n"
+        "features_evaluation = prep(testing_df)\n"
+        "features_testing = prep(testing_df).drop(columns=['label'])\n")
+    b = BuilderSparkMl(ctx)
+    b.build("titanic", "titanic", code, ["lr", "gb"])
+    for c in ("lr", "gb"):
+        doc = b.wait(f"titanic{c}", timeout=240)
+        assert doc.get("exception") in (None, ""), doc
+        assert doc["accuracy"] > 0.55, (c, doc["accuracy"])
