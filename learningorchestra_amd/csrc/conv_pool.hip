@@ -311,9 +311,80 @@ __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
   }
 }
 
+// global max-over-time (OH==OW==1, C%8==0): one block per image, lanes
+// cooperate over the window with an LDS tree — the generic kernel serializes
+// the whole window per thread (254 iterations in the TextCNN pool).
+__global__ void maxpool_global_kernel(const bf16* __restrict__ in,
+                                      bf16* __restrict__ out,
+                                      unsigned char* __restrict__ idx,
+                                      int HW, int C) {
+  extern __shared__ __attribute__((aligned(16))) float lmx[];  // [C][2] max+idx
+  const int b = blockIdx.x;
+  const int nch = C / 8;
+  const int lanesPerChunk = max(1, (int)blockDim.x / nch);
+  const int tc = (int)threadIdx.x % nch;
+  const int tl = (int)threadIdx.x / nch;
+  float best[8];
+  int bidx[8];
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) { best[j] = -3.0e38f; bidx[j] = 0; }
+  if (tl < lanesPerChunk) {
+    for (int p = tl; p < HW; p += lanesPerChunk) {
+      bf16x8 v = *(const bf16x8*)(in + ((long)b * HW + p) * C + tc * 8);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f = tofloat(v[j]);
+        if (f > best[j]) { best[j] = f; bidx[j] = p; }
+      }
+    }
+  }
+  // LDS argmax reduce per element: atomicMax on a monotone unsigned key
+  // (sign-flip transform; raw float bits mis-order negatives), then winners
+  // record their window index in a second pass.
+  auto fkey = [](float f) -> unsigned {
+    const unsigned u = __float_as_uint(f);
+    return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+  };
+  unsigned* lk = (unsigned*)lmx;            // [C] keys then [C] idx
+  for (int i = threadIdx.x; i < C; i += blockDim.x) {
+    lk[i] = 0;
+    lk[C + i] = 0xffffffffu;
+  }
+  __syncthreads();
+  if (tl < lanesPerChunk) {
+    #pragma unroll
+    for (int j = 0; j < 8; ++j)
+      atomicMax(lk + tc * 8 + j, fkey(best[j]));
+  }
+  __syncthreads();
+  if (tl < lanesPerChunk) {
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = tc * 8 + j;
+      if (fkey(best[j]) == lk[c])
+        atomicMin(lk + C + c, (unsigned)bidx[j]);  // first-index tie-break
+                                            // (matches the reference argmax)
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < C; i += blockDim.x) {
+    const unsigned u = lk[i];
+    const unsigned raw = (u & 0x80000000u) ? (u & 0x7fffffffu) : ~u;
+    out[(long)b * C + i] = tobf16(__uint_as_float(raw));
+    idx[(long)b * C + i] = (unsigned char)min(lk[C + i], 255u);
+  }
+}
+
 void launch_maxpool_fwd(const void* in, void* out, void* idx, int B, int H, int W,
                         int C, int KH, int KW, int SH, int SW, int PH, int PW,
                         int OH, int OW, hipStream_t s) {
+  if (OH == 1 && OW == 1 && PH == 0 && PW == 0 && KH == H && KW == W &&
+      C % 8 == 0 && C <= 1024 && (long)H * W <= 255) {
+    hipLaunchKernelGGL(maxpool_global_kernel, dim3(B), dim3(256),
+                       C * 2 * sizeof(float), s, (const bf16*)in, (bf16*)out,
+                       (unsigned char*)idx, H * W, C);
+    return;
+  }
   const bool vec = (C % 8 == 0);
   const long total = (long)B * OH * OW * (vec ? C / 8 : C);
   const int block = 256;
