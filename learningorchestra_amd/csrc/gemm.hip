@@ -520,12 +520,10 @@ bool gemm_dispatch(const GemmArgs& g, hipStream_t s) {
     // staging depth for these shapes)
     if (!g.out_f32 || g.epi != 0) return false;
     if (!g.ta && g.tb) {
-      if (g.M >= 128) launch_cfg<128, 64, 64, 2, 2, false, true, 0, true, true>(g, s);
-      else if (g.M >= 64) launch_cfg<64, 64, 64, 2, 2, false, true, 0, true, true>(g, s);
+      if (g.M >= 64) launch_cfg<64, 64, 64, 2, 2, false, true, 0, true, true>(g, s);
       else launch_cfg<32, 64, 64, 1, 4, false, true, 0, true, true>(g, s);
     } else if (g.ta && !g.tb) {
-      if (g.M >= 128) launch_cfg<128, 64, 64, 2, 2, true, false, 0, true, true>(g, s);
-      else if (g.M >= 64) launch_cfg<64, 64, 64, 2, 2, true, false, 0, true, true>(g, s);
+      if (g.M >= 64) launch_cfg<64, 64, 64, 2, 2, true, false, 0, true, true>(g, s);
       else launch_cfg<32, 64, 64, 1, 4, true, false, 0, true, true>(g, s);
     } else {
       return false;

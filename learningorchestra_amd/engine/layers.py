@@ -264,7 +264,7 @@ class Linear(Layer):
 def _splitk_heuristic(m: int, n: int, k: int) -> int:
     """Split-K factor for the dW GEMM C[m,n] with reduction depth k: fill the
     256-CU chip (~512 blocks of 32x64 tiles) without shredding k."""
-    bm = 128 if m >= 128 else (64 if m >= 64 else 32)  # mirror split-K tiles
+    bm = 64 if m >= 64 else 32   # mirror the split-K tile selection
     base_blocks = ((m + bm - 1) // bm) * ((n + 63) // 64)
     if base_blocks >= 2048 or k < 2048:
         return 1
