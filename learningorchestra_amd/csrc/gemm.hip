@@ -124,7 +124,17 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
   const bool a_rows_in = !TA && (m0 + BM <= M);
   const bool b_rows_in = TB && (n0 + BN <= N);
 
-  auto stage_tile = [&](int buf, int k0) {
+  // T14 async-STAGE split for the register-staged (tr16) operand paths:
+  // issue the next tile's global loads BEFORE the compute phase (HBM latency
+  // hides under the MFMAs) and do the LDS writes after it — writing
+  // immediately after the loads would force the vmcnt wait at the ds_write
+  // and expose the full latency (guide §6 G15; this was the split-K dW
+  // kernel's bound).
+  constexpr int CHA_T = TA ? (BK * BM / 8 + T - 1) / T : 1;
+  constexpr int CHB_T = (!TB) ? (BK * BN / 8 + T - 1) / T : 1;
+  bf16x8 stgA[CHA_T], stgB[CHB_T];
+
+  auto stage_next = [&](int buf, int k0) {
     char* smA = smem + buf * BUFB;
     char* smB = smem + buf * BUFB + BM * BKB;
     const bool k_in = (k0 + BK <= K);
@@ -168,24 +178,24 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
         }
       }
     } else {
-      // A is [K,M]: stage NATURALLY (no transpose, no swizzle) as 16-wide
-      // m-subtiles — image elem (k, m) at (m/16)*(BK*16) + k*16 + m%16 — and
-      // let ds_read_b64_tr_b16 do the transpose at fragment-read time (each
-      // 16-lane group reads a contiguous 128-B region and receives its
-      // [4 k][16 m] transpose; verified by the on-GPU probe).
+      // A is [K,M]: load NATURALLY (no transpose) into registers; the LDS
+      // write happens in stage_finish; fragments use ds_read_b64_tr_b16
+      // against the k-major 16-wide m-subtile image (probe-verified).
       constexpr int CH = BK * BM / 8;
-      for (int c = tid; c < CH; c += T) {
-        const int k = c / (BM / 8), mc = c % (BM / 8);
-        const int gk = k0 + k, gm0 = m0 + mc * 8;
+      #pragma unroll
+      for (int i = 0; i < CHA_T; ++i) {
+        const int c = tid + i * T;
         bf16x8 v = {};
-        if (gk < K) {
-          if (gm0 + 8 <= M) v = *(const bf16x8*)(A + (long)gk * lda + gm0);
-          else if (gm0 < M)
-            for (int j = 0; j < 8 && gm0 + j < M; ++j) v[j] = A[(long)gk * lda + gm0 + j];
+        if (c < CH) {
+          const int k = c / (BM / 8), mc = c % (BM / 8);
+          const int gk = k0 + k, gm0 = m0 + mc * 8;
+          if (gk < K) {
+            if (gm0 + 8 <= M) v = *(const bf16x8*)(A + (long)gk * lda + gm0);
+            else if (gm0 < M)
+              for (int j = 0; j < 8 && gm0 + j < M; ++j) v[j] = A[(long)gk * lda + gm0 + j];
+          }
         }
-        const int msub = (mc * 8) / 16, mrem = (mc * 8) % 16;
-        *(bf16x8*)(smA + (msub * BK * 16 + tr16_slot(k) * 64 + (k & 3) * 16
-                          + mrem) * 2) = v;
+        stgA[i] = v;
       }
     }
     // ---- stage B tile -> Bs[BN][BK] ------------------------------------
@@ -212,24 +222,57 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
         }
       }
     } else {
-      // B is [K,N]: natural k-major 16-wide n-subtile image + tr16 reads;
-      // GATHER==2 sources the row from the implicit im2col of the NHWC input
+      // B is [K,N]: natural k-major load into registers (write deferred);
+      // GATHER==2 gathers the implicit im2col row of the NHWC input instead
       // (dW never re-reads a col matrix)
       constexpr int CH = BK * BN / 8;
-      for (int c = tid; c < CH; c += T) {
-        const int k = c / (BN / 8), nc = c % (BN / 8);
-        const int gk = k0 + k, gn0 = n0 + nc * 8;
+      #pragma unroll
+      for (int i = 0; i < CHB_T; ++i) {
+        const int c = tid + i * T;
         bf16x8 v = {};
-        if (GATHER == 2) {
-          if (gk < K && gn0 < N) v = conv_gather8(B, geom, (long)gk, gn0);
-        } else if (gk < K) {
-          if (gn0 + 8 <= N) v = *(const bf16x8*)(B + (long)gk * ldb + gn0);
-          else if (gn0 < N)
-            for (int j = 0; j < 8 && gn0 + j < N; ++j) v[j] = B[(long)gk * ldb + gn0 + j];
+        if (c < CH) {
+          const int k = c / (BN / 8), nc = c % (BN / 8);
+          const int gk = k0 + k, gn0 = n0 + nc * 8;
+          if (GATHER == 2) {
+            if (gk < K && gn0 < N) v = conv_gather8(B, geom, (long)gk, gn0);
+          } else if (gk < K) {
+            if (gn0 + 8 <= N) v = *(const bf16x8*)(B + (long)gk * ldb + gn0);
+            else if (gn0 < N)
+              for (int j = 0; j < 8 && gn0 + j < N; ++j) v[j] = B[(long)gk * ldb + gn0 + j];
+          }
         }
-        const int nsub = (nc * 8) / 16, nrem = (nc * 8) % 16;
-        *(bf16x8*)(smB + (nsub * BK * 16 + tr16_slot(k) * 64 + (k & 3) * 16
-                          + nrem) * 2) = v;
+        stgB[i] = v;
+      }
+    }
+  };
+
+  auto stage_finish = [&](int buf) {
+    char* smA = smem + buf * BUFB;
+    char* smB = smem + buf * BUFB + BM * BKB;
+    if (TA) {
+      constexpr int CH = BK * BM / 8;
+      #pragma unroll
+      for (int i = 0; i < CHA_T; ++i) {
+        const int c = tid + i * T;
+        if (c < CH) {
+          const int k = c / (BM / 8), mc = c % (BM / 8);
+          const int msub = (mc * 8) / 16, mrem = (mc * 8) % 16;
+          *(bf16x8*)(smA + (msub * BK * 16 + tr16_slot(k) * 64 + (k & 3) * 16
+                            + mrem) * 2) = stgA[i];
+        }
+      }
+    }
+    if (!TB) {
+      constexpr int CH = BK * BN / 8;
+      #pragma unroll
+      for (int i = 0; i < CHB_T; ++i) {
+        const int c = tid + i * T;
+        if (c < CH) {
+          const int k = c / (BN / 8), nc = c % (BN / 8);
+          const int nsub = (nc * 8) / 16, nrem = (nc * 8) % 16;
+          *(bf16x8*)(smB + (nsub * BK * 16 + tr16_slot(k) * 64 + (k & 3) * 16
+                            + nrem) * 2) = stgB[i];
+        }
       }
     }
   };
@@ -293,14 +336,18 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
     }
   };
 
-  // 2-phase pipeline: issue tile t+1's loads BEFORE computing tile t, one
-  // barrier per tile (guide §5.5 T3 minimum 2-phase recipe).
+  // 2-phase pipeline: issue tile t+1's loads BEFORE computing tile t, LDS
+  // writes for the register paths AFTER it, one barrier per tile (guide
+  // §5.5 T3 minimum 2-phase + G15 async-STAGE split).
   int cur = 0;
-  stage_tile(0, kBegin);
+  stage_next(0, kBegin);
+  stage_finish(0);
   __syncthreads();
   for (int k0 = kBegin; k0 < kEnd; k0 += BK) {
-    if (k0 + BK < kEnd) stage_tile(cur ^ 1, k0 + BK);
+    const bool more = (k0 + BK < kEnd);
+    if (more) stage_next(cur ^ 1, k0 + BK);
     compute_tile(cur);
+    if (more) stage_finish(cur ^ 1);
     __syncthreads();
     cur ^= 1;
   }
