@@ -82,7 +82,8 @@ template <int BM, int BN, int BK, int WM, int WN, bool TA, bool TB,
 __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
     const bf16* __restrict__ A, long lda, const bf16* __restrict__ B, long ldb,
     void* __restrict__ Cv, long ldc, const float* __restrict__ bias,
-    int M, int N, int K, int kStart, int kChunk, ConvGeom geom) {
+    int M, int N, int K, int kStart, int kChunk, ConvGeom geom,
+    float* __restrict__ stats_sum, float* __restrict__ stats_sumsq) {
   constexpr int T = WM * WN * 64;
   constexpr int WTM = BM / WM;          // wave tile rows
   constexpr int WTN = BN / WN;          // wave tile cols
@@ -418,6 +419,35 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
       }
     }
   }
+  // fused BatchNorm statistics over the staged tile (bf16-out convs feeding
+  // BN): per-column partial sums + one global atomic per column per block —
+  // saves the separate bn_stats pass over the activation.
+  if (!OUT_F32 && stats_sum != nullptr) {
+    const int tc = tid % (BN / 8);
+    const int tr = tid / (BN / 8);
+    const int rowsPerBlock = T / (BN / 8);
+    const int gc0 = n0 + tc * 8;
+    if (gc0 < N) {
+      float s1[8] = {}, s2[8] = {};
+      for (int rowL = tr; rowL < BM; rowL += rowsPerBlock) {
+        if (m0 + rowL >= M) break;
+        bf16x8 v = *(const bf16x8*)(ct + rowL * CROWB + tc * 16);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float f = tofloat(v[j]);
+          s1[j] += f;
+          s2[j] += f * f;
+        }
+      }
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        if (gc0 + j < N) {
+          atomicAdd(stats_sum + gc0 + j, s1[j]);
+          atomicAdd(stats_sumsq + gc0 + j, s2[j]);
+        }
+      }
+    }
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -454,6 +484,10 @@ struct GemmArgs {
   int splits;     // >1 => atomic split-K accumulate into fp32 C (zeroed by caller)
   int gather;     // 0 plain; 1 A = implicit im2col of NHWC input; 2 B likewise
   ConvGeom geom;
+  // optional fused column statistics (BatchNorm): per-column sum / sum-of-
+  // squares of C accumulated by the epilogue (atomics; caller zeroes)
+  float* stats_sum = nullptr;
+  float* stats_sumsq = nullptr;
 };
 
 template <int BM, int BN, int BK, int WM, int WN, bool TA, bool TB,
@@ -476,7 +510,7 @@ static void launch_cfg(const GemmArgs& g, hipStream_t s) {
       gemm_kernel<BM, BN, BK, WM, WN, TA, TB, EPI, OUT_F32, ATOMIC, GATHER>),
       grid, block, lds, s,
       (const bf16*)g.A, g.lda, (const bf16*)g.B, g.ldb, g.C, g.ldc, g.bias,
-      g.M, g.N, g.K, 0, kChunk, g.geom);
+      g.M, g.N, g.K, 0, kChunk, g.geom, g.stats_sum, g.stats_sumsq);
 }
 
 bool launch_gemm256(const void* A, long lda, const void* B, long ldb, void* C,

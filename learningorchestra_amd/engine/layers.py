@@ -113,27 +113,34 @@ class Conv2dNHWC(Layer):
         if getattr(self, "_wt_buf", None) is not None:
             self._wt_buf.copy_(self.arena.p(self.name + ".w").t())
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor,
+                stats: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """``stats``: optional [2, out_c] fp32 workspace — on GPU the GEMM
+        epilogue fills per-channel sum/sumsq of the output (the following
+        BatchNorm skips its stats pass)."""
         B, H, W, C = x.shape
         assert C == self.in_c
         bufs = self._alloc(B, H, W, x.device, x.dtype)
         self._x = x
         bias = self.arena.pf(self.name + ".b") if self.bias else None
+        st = stats if x.is_cuda else None
         if self._is_1x1:
             bufs["col"] = x.view(B * H * W, C)
             F.gemm(bufs["col"], self.arena.p(self.name + ".w"), tb=True,
-                   bias=bias, relu=self.relu, out=bufs["y"])
+                   bias=bias, relu=self.relu, out=bufs["y"], stats=st)
         elif x.is_cuda and self.implicit:
             # implicit conv: im2col gathered inside the GEMM staging
             F.conv2d_fwd_implicit(x, self.arena.p(self.name + ".w"), self.kh,
                                   self.kw, self.stride, self.stride, self.pad,
                                   self.pad, bias=bias, relu=self.relu,
                                   out=bufs["y"])
+            st = None
         else:
             F.im2col(x, self.kh, self.kw, self.stride, self.stride, self.pad,
                      self.pad, self.kpad, out=bufs["col"])
             F.gemm(bufs["col"], self.arena.p(self.name + ".w"), tb=True,
-                   bias=bias, relu=self.relu, out=bufs["y"])
+                   bias=bias, relu=self.relu, out=bufs["y"], stats=st)
+        self.stats_filled = st is not None
         return bufs["y"].view(B, bufs["OH"], bufs["OW"], self.out_c)
 
     def backward(self, dy: torch.Tensor) -> Optional[torch.Tensor]:
@@ -395,6 +402,13 @@ class BatchNormReLU(Layer):
         arena.add(self.name + ".g", (self.c,), torch.ones(self.c))
         arena.add(self.name + ".b", (self.c,), torch.zeros(self.c))
 
+    def scratch(self, dev) -> torch.Tensor:
+        """[2, C] sum/sumsq workspace — exposed so the producing conv's GEMM
+        epilogue can fill it (fused bn_stats)."""
+        if getattr(self, "_scratch", None) is None:
+            self._scratch = torch.zeros((2, self.c), device=dev)
+        return self._scratch
+
     def _alloc(self, shape, dev):
         if self._bufs.get("shape") != shape:
             M = 1
@@ -406,13 +420,13 @@ class BatchNormReLU(Layer):
                 "dx": torch.empty((M, self.c), device=dev, dtype=torch.bfloat16),
                 "mean": torch.zeros(self.c, device=dev),
                 "invstd": torch.ones(self.c, device=dev),
-                "scratch": torch.zeros((2, self.c), device=dev),
+                "scratch": self.scratch(dev),
                 "running_mean": torch.zeros(self.c, device=dev),
                 "running_var": torch.ones(self.c, device=dev),
             }
         return self._bufs
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, stats_ready: bool = False) -> torch.Tensor:
         bufs = self._alloc(tuple(x.shape), x.device)
         x2 = x.reshape(bufs["M"], self.c)
         self._x2 = x2
@@ -420,7 +434,8 @@ class BatchNormReLU(Layer):
         beta = self.arena.pf(self.name + ".b")
         if self.training:
             F.bn_fwd_train(x2, gamma, beta, self.eps, bufs["y"], bufs["mean"],
-                           bufs["invstd"], bufs["scratch"], self.relu)
+                           bufs["invstd"], bufs["scratch"], self.relu,
+                           stats_ready=stats_ready and x.is_cuda)
             m = self.momentum
             bufs["running_mean"].mul_(1 - m).add_(bufs["mean"], alpha=m)
             var = bufs["invstd"].square().reciprocal() - self.eps

@@ -55,14 +55,20 @@ class Bottleneck:
         for lay in self.layers():
             lay.build(arena)
 
+    @staticmethod
+    def _conv_bn(conv, bn, x):
+        # the conv GEMM epilogue fills bn's sum/sumsq scratch (fused bn_stats)
+        h = conv.forward(x, stats=bn.scratch(x.device) if x.is_cuda else None)
+        return bn.forward(h, stats_ready=conv.stats_filled)
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if self.downsample:
-            idt = self.downsample[1].forward(self.downsample[0].forward(x))
+            idt = self._conv_bn(self.downsample[0], self.downsample[1], x)
         else:
             idt = x
-        h = self.bn1.forward(self.conv1.forward(x))
-        h = self.bn2.forward(self.conv2.forward(h))
-        h = self.bn3.forward(self.conv3.forward(h))
+        h = self._conv_bn(self.conv1, self.bn1, x)
+        h = self._conv_bn(self.conv2, self.bn2, h)
+        h = self._conv_bn(self.conv3, self.bn3, h)
         if self._z is None or self._z.shape != h.shape:
             self._z = torch.empty_like(h)
             self._dsum = torch.empty_like(h)
@@ -133,7 +139,9 @@ class ResNet:
             bn.training = training
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        h = self.stem_bn.forward(self.stem_conv.forward(x))
+        h = self.stem_conv.forward(x, stats=self.stem_bn.scratch(x.device)
+                                   if x.is_cuda else None)
+        h = self.stem_bn.forward(h, stats_ready=self.stem_conv.stats_filled)
         h = self.stem_pool.forward(h)
         for blk in self.blocks:
             h = blk.forward(h)

@@ -27,7 +27,7 @@ def _is_gpu(t: torch.Tensor) -> bool:
 def gemm(A: torch.Tensor, B: torch.Tensor, *, ta: bool = False, tb: bool = False,
          bias: Optional[torch.Tensor] = None, relu: bool = False,
          out: Optional[torch.Tensor] = None, out_dtype: Optional[torch.dtype] = None,
-         splits: int = 1) -> torch.Tensor:
+         splits: int = 1, stats: Optional[torch.Tensor] = None) -> torch.Tensor:
     """C[M,N] = op(A) @ op(B) (+bias) (+relu). bf16 inputs, fp32 accumulate.
 
     op(A) = A.T if ta (A stored [K,M]); op(B) = B.T if tb (B stored [N,K]).
@@ -43,7 +43,10 @@ def gemm(A: torch.Tensor, B: torch.Tensor, *, ta: bool = False, tb: bool = False
         lo = require_ext()
         if splits > 1:
             out.zero_()
-        ok = lo.gemm(A, B, out, bias, ta, tb, 1 if relu else 0, splits)
+        if stats is not None:
+            stats.zero_()  # epilogue accumulates per-column sum/sumsq
+        ok = lo.gemm(A, B, out, bias, ta, tb, 1 if relu else 0, splits,
+                     stats.view(-1) if stats is not None else None)
         if ok:
             return out
         # cold-path shapes: plain library GEMM (rocBLAS via torch.matmul)
@@ -300,13 +303,16 @@ def embedding_bwd(ids: torch.Tensor, dy: torch.Tensor,
 def bn_fwd_train(x2d: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
                  eps: float, out: torch.Tensor, mean: torch.Tensor,
                  invstd: torch.Tensor, scratch: torch.Tensor,
-                 relu: bool = True) -> None:
+                 relu: bool = True, stats_ready: bool = False) -> None:
     """Training-mode BN over [M, C] (+fused ReLU). Writes out (bf16), mean,
-    invstd (fp32 [C]); ``scratch`` is a [2, C] fp32 workspace (sum/sumsq)."""
+    invstd (fp32 [C]); ``scratch`` is a [2, C] fp32 workspace (sum/sumsq).
+    ``stats_ready``: scratch was already filled by the producing GEMM's fused
+    epilogue — skip the bn_stats pass."""
     M, C = x2d.shape
     if _is_gpu(x2d):
         lo = require_ext()
-        lo.bn_stats(x2d, scratch[0], scratch[1])
+        if not stats_ready:
+            lo.bn_stats(x2d, scratch[0], scratch[1])
         mean.copy_(scratch[0] / M)
         invstd.copy_((scratch[1] / M - mean.square()).clamp_(min=0)
                      .add_(eps).rsqrt_())
