@@ -423,11 +423,17 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
   // BN): per-column partial sums + one global atomic per column per block —
   // saves the separate bn_stats pass over the activation.
   if (!OUT_F32 && stats_sum != nullptr) {
+    // LDS pre-reduction first: direct global atomics from every row-lane
+    // serialized on the few per-channel addresses (measured 10x whole-model
+    // slowdown); one global atomic per column per block only.
+    float* lacc = (float*)(ct + BM * CROWB + 16);   // [2][BN]
+    for (int i = tid; i < 2 * BN; i += T) lacc[i] = 0.f;
+    __syncthreads();
     const int tc = tid % (BN / 8);
     const int tr = tid / (BN / 8);
     const int rowsPerBlock = T / (BN / 8);
     const int gc0 = n0 + tc * 8;
-    if (gc0 < N) {
+    if (gc0 < N && tr < rowsPerBlock) {
       float s1[8] = {}, s2[8] = {};
       for (int rowL = tr; rowL < BM; rowL += rowsPerBlock) {
         if (m0 + rowL >= M) break;
@@ -441,11 +447,14 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
       }
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        if (gc0 + j < N) {
-          atomicAdd(stats_sum + gc0 + j, s1[j]);
-          atomicAdd(stats_sumsq + gc0 + j, s2[j]);
-        }
+        atomicAdd(lacc + tc * 8 + j, s1[j]);
+        atomicAdd(lacc + BN + tc * 8 + j, s2[j]);
       }
+    }
+    __syncthreads();
+    for (int i = tid; i < BN && n0 + i < N; i += T) {
+      atomicAdd(stats_sum + n0 + i, lacc[i]);
+      atomicAdd(stats_sumsq + n0 + i, lacc[BN + i]);
     }
   }
 }
@@ -504,7 +513,8 @@ static void launch_cfg(const GemmArgs& g, hipStream_t s) {
   const int zs = g.splits > 1 ? cdiv(g.K, kChunk) : 1;
   dim3 grid(mb * nb, 1, zs), block(WM * WN * 64);
   size_t lds_stage = (size_t)(BM + BN) * BK * 2 * 2;  // double-buffered
-  size_t lds_epi = (size_t)BM * (BN * (g.out_f32 ? 4 : 2) + 16);
+  size_t lds_epi = (size_t)BM * (BN * (g.out_f32 ? 4 : 2) + 16)
+                   + 2 * BN * 4 + 32;  // +fused-stats LDS reduction
   size_t lds = lds_stage > lds_epi ? lds_stage : lds_epi;
   hipLaunchKernelGGL(HIP_KERNEL_NAME(
       gemm_kernel<BM, BN, BK, WM, WN, TA, TB, EPI, OUT_F32, ATOMIC, GATHER>),
