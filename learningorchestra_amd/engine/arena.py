@@ -84,6 +84,17 @@ class ParamArena:
         """fp32 grad view (backward accumulates here)."""
         return self._view(self.grad, name)
 
+    def grad_slice(self, names) -> Optional[torch.Tensor]:
+        """Contiguous flat-grad slice covering ``names`` (params are laid
+        out in registration order, so a layer's params are contiguous) —
+        the unit of the DDP overlap all-reduce."""
+        offs = [self._offsets[n] for n in names if n in self._offsets]
+        if not offs:
+            return None
+        start = min(o for o, _ in offs)
+        end = max(o + _pad8(n) for o, n in offs)
+        return self.grad[start:end]
+
     def zero_grad(self) -> None:
         self.grad.zero_()
 

@@ -30,6 +30,9 @@ class Layer:
     def build(self, arena: ParamArena) -> None:  # register params
         pass
 
+    def param_names(self) -> List[str]:
+        return []
+
     def post_opt_step(self) -> None:
         """Refresh derived compute state after an optimizer step (e.g. the
         transposed-weight mirror that keeps dX GEMMs on the fast k-contiguous
@@ -82,6 +85,9 @@ class Conv2dNHWC(Layer):
         arena.add(self.name + ".w", (self.out_c, self.kpad), init_w)
         if self.bias:
             arena.add(self.name + ".b", (self.out_c,), torch.zeros(self.out_c))
+
+    def param_names(self) -> List[str]:
+        return [self.name + ".w"] + ([self.name + ".b"] if self.bias else [])
 
     def _alloc(self, B: int, H: int, W: int, dev, dtype):
         OH = (H + 2 * self.pad - self.kh) // self.stride + 1
@@ -233,6 +239,9 @@ class Linear(Layer):
                   torch.randn((self.out_f, self.in_f), generator=g) * std)
         arena.add(self.name + ".b", (self.out_f,), torch.zeros(self.out_f))
 
+    def param_names(self) -> List[str]:
+        return [self.name + ".w", self.name + ".b"]
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         M = x.shape[0]
         if self._bufs.get("M") != M:
@@ -306,9 +315,12 @@ class SequentialClassifier:
         return x
 
     def train_step(self, x: torch.Tensor, y: torch.Tensor,
-                   gscale: Optional[float] = None) -> Tuple[torch.Tensor, torch.Tensor]:
+                   gscale: Optional[float] = None,
+                   grad_hook=None) -> Tuple[torch.Tensor, torch.Tensor]:
         """One fused fwd+bwd. Returns (loss_sum, correct) device tensors;
-        optimizer step is separate (Trainer composes allreduce between)."""
+        optimizer step is separate (Trainer composes allreduce between).
+        ``grad_hook(param_names)`` fires after each layer's grads are final
+        (the DDP overlap point)."""
         if gscale is None:
             gscale = 1.0 / x.shape[0]
         logits = self.forward(x)
@@ -321,6 +333,8 @@ class SequentialClassifier:
         dy = self._dlogits
         for lay in reversed(self.layers):
             dy = lay.backward(dy)
+            if grad_hook and lay.param_names():
+                grad_hook(lay.param_names())
             if dy is None:
                 break
         return self.loss_sum, self.correct
@@ -362,6 +376,9 @@ class Embedding(Layer):
         arena.add(self.name + ".w", (self.vocab, self.dim),
                   torch.randn((self.vocab, self.dim), generator=g) * 0.05)
 
+    def param_names(self):
+        return [self.name + ".w"]
+
     def forward(self, ids: torch.Tensor) -> torch.Tensor:
         from ..ops import functional as F_
         self._ids = ids
@@ -401,6 +418,9 @@ class BatchNormReLU(Layer):
         self.arena = arena
         arena.add(self.name + ".g", (self.c,), torch.ones(self.c))
         arena.add(self.name + ".b", (self.c,), torch.zeros(self.c))
+
+    def param_names(self):
+        return [self.name + ".g", self.name + ".b"]
 
     def scratch(self, dev) -> torch.Tensor:
         """[2, C] sum/sumsq workspace — exposed so the producing conv's GEMM

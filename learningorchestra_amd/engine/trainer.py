@@ -15,6 +15,14 @@ from .arena import SGD, Adam
 from .layers import SequentialClassifier
 
 
+def _fn_params(fn):
+    import inspect
+    try:
+        return inspect.signature(fn).parameters
+    except (TypeError, ValueError):
+        return {}
+
+
 class Trainer:
     def __init__(self, model: SequentialClassifier, optimizer, device="cpu",
                  use_graph: bool = False):
@@ -30,8 +38,34 @@ class Trainer:
     def _step_body(self, x: torch.Tensor, y: torch.Tensor) -> None:
         world = get_world_size()
         gscale = 1.0 / (x.shape[0] * world)
-        self.model.train_step(x, y, gscale=gscale)
-        all_reduce_grads(self.model.arena.grad)
+        arena = self.model.arena
+        if is_distributed():
+            # per-layer async all-reduce overlapped with the backward walk;
+            # any params not covered by hooks reduce in one trailing op
+            works = []
+            covered = []
+
+            def hook(names):
+                sl = arena.grad_slice(names)
+                if sl is not None and sl.numel():
+                    covered.extend(n for n in names if n in arena._offsets)
+                    works.append(all_reduce_grads(sl, async_op=True))
+
+            accepts_hook = "grad_hook" in _fn_params(self.model.train_step)
+            if accepts_hook:
+                self.model.train_step(x, y, gscale=gscale, grad_hook=hook)
+            else:
+                self.model.train_step(x, y, gscale=gscale)
+            leftover = [n for n, _, _ in arena._specs if n not in covered]
+            if leftover:
+                sl = arena.grad_slice(leftover)
+                if sl is not None and sl.numel():
+                    works.append(all_reduce_grads(sl, async_op=True))
+            for w in works:
+                if w is not None:
+                    w.wait()
+        else:
+            self.model.train_step(x, y, gscale=gscale)
         self.opt.step()
         post = getattr(self.model, "post_opt_step", None)
         if post is not None:

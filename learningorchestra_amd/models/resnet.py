@@ -51,6 +51,12 @@ class Bottleneck:
             out += list(self.downsample)
         return out
 
+    def param_names(self):
+        names = []
+        for lay in self.layers():
+            names += lay.param_names()
+        return names
+
     def build(self, arena: ParamArena) -> None:
         for lay in self.layers():
             lay.build(arena)
@@ -149,7 +155,7 @@ class ResNet:
         return self.fc.forward(h)
 
     def train_step(self, x: torch.Tensor, y: torch.Tensor,
-                   gscale: Optional[float] = None):
+                   gscale: Optional[float] = None, grad_hook=None):
         if gscale is None:
             gscale = 1.0 / x.shape[0]
         logits = self.forward(x)
@@ -160,12 +166,18 @@ class ResNet:
         F.softmax_ce(logits, y, self._dlogits, self.loss_sum, self.correct,
                      cvalid=self.num_classes, gscale=gscale)
         dy = self.fc.backward(self._dlogits)
+        if grad_hook:
+            grad_hook(self.fc.param_names())
         dy = self.avgpool.backward(dy)
         for blk in reversed(self.blocks):
             dy = blk.backward(dy)
+            if grad_hook:
+                grad_hook(blk.param_names())
         dy = self.stem_pool.backward(dy)
         dy = self.stem_bn.backward(dy)
         self.stem_conv.backward(dy)
+        if grad_hook:
+            grad_hook(self.stem_bn.param_names() + self.stem_conv.param_names())
         return self.loss_sum, self.correct
 
     def post_opt_step(self) -> None:

@@ -72,7 +72,8 @@ class TextCNN:
         return self.fc.forward(bufs["cat"])
 
     def train_step(self, ids: torch.Tensor, y: torch.Tensor,
-                   gscale: Optional[float] = None) -> Tuple[torch.Tensor, torch.Tensor]:
+                   gscale: Optional[float] = None,
+                   grad_hook=None) -> Tuple[torch.Tensor, torch.Tensor]:
         B, S = ids.shape
         if gscale is None:
             gscale = 1.0 / B
@@ -83,6 +84,8 @@ class TextCNN:
         F.softmax_ce(logits, y, bufs["dlogits"], self.loss_sum, self.correct,
                      cvalid=self.num_classes, gscale=gscale)
         dcat = self.fc.backward(bufs["dlogits"])      # [B, 3F]
+        if grad_hook:
+            grad_hook(self.fc.param_names())
         Fn = self.filters
         dxe = bufs["dxe"]
         for i, (k, conv, pool) in enumerate(zip(self.kernel_sizes, self.convs,
@@ -90,11 +93,15 @@ class TextCNN:
             dpool = dcat[:, i * Fn:(i + 1) * Fn].contiguous().view(B, 1, 1, Fn)
             dconv = pool.backward(dpool)              # [B, S-k+1, 1, F]
             dbranch = conv.backward(dconv)            # [B, S, 1, emb]
+            if grad_hook:
+                grad_hook(conv.param_names())
             if i == 0:
                 dxe.copy_(dbranch)
             else:
                 dxe.add_(dbranch)
         self.emb.backward(dxe.view(B, S, self.emb_dim))
+        if grad_hook:
+            grad_hook(self.emb.param_names())
         return self.loss_sum, self.correct
 
     def post_opt_step(self) -> None:
