@@ -40,7 +40,6 @@ from ..data.csv_ingest import CsvIngest
 from ..executor.execution import Execution, ValidationError
 from ..executor.scheduler import JobScheduler
 from ..storage import ArtifactStore, Data, Metadata, connect
-from ..storage.metadata import METADATA_ROW_ID
 from .services import (BuilderService, DataTypeService, HistogramService,
                        ProjectionService)
 

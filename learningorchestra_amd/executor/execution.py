@@ -26,7 +26,7 @@ import sys
 import time
 import traceback
 from contextlib import redirect_stdout
-from typing import Any, Dict, Optional, Tuple
+from typing import Any, Dict, Optional
 
 from .parameters import Parameters
 from .scheduler import JobScheduler

@@ -14,7 +14,7 @@ binary_executor_image/binary_execution.py:18-89). Protocol, implemented once:
 """
 from __future__ import annotations
 
-from typing import Any, Dict, List
+from typing import Any, Dict
 
 DATASET_MARKER = "$"
 CODE_MARKER = "#"

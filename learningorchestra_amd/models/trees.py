@@ -22,7 +22,7 @@ from typing import List, Optional, Tuple
 
 import torch
 
-from ..ops._ext import has_ext, require_ext
+from ..ops._ext import require_ext
 
 MAX_BINS = 255
 

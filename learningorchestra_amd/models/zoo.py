@@ -9,7 +9,7 @@ runs on the MI355X engine (HIP kernels on GPU, torch fp32 reference on CPU).
 """
 from __future__ import annotations
 
-from typing import Optional, Sequence, Tuple
+from typing import Optional, Sequence
 
 import numpy as np
 import torch

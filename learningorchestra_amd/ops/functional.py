@@ -16,7 +16,7 @@ from typing import Optional, Tuple
 
 import torch
 
-from ._ext import has_ext, require_ext
+from ._ext import require_ext
 
 
 def _is_gpu(t: torch.Tensor) -> bool:
