@@ -56,18 +56,18 @@ def main() -> None:
         torch.cuda.set_device(local_rank)
 
     if args.model == "mnist-cnn":
-        batch = args.batch or 8192
+        batch = args.batch or 32768
         model = build_mnist_cnn(device, seed=0)
         model_desc = "MNIST-CNN (LeNet-style, conv32-conv64-fc256)"
         extra_cfg = {"image": "28x28x1"}
     elif args.model == "textcnn":
-        batch = args.batch or 2048
+        batch = args.batch or 4096
         model = build_textcnn(device, seed=0)
         model_desc = "TextCNN (IMDb sentiment, emb128, filters 128 x k3/4/5)"
         extra_cfg = {"seq_len": 256, "vocab": 20000}
     else:
         from learningorchestra_amd.models.resnet import build_resnet50
-        batch = args.batch or 128
+        batch = args.batch or 512
         model = build_resnet50(device, seed=0)
         model_desc = "ResNet-50 (bottleneck v1, 1000 classes)"
         extra_cfg = {"image": "224x224x3"}
