@@ -75,11 +75,14 @@ __global__ void bn_stats_kernel(const bf16* __restrict__ x, float* __restrict__ 
 // fixed-chunk thread layout: tc/tr computed once, per-channel params hoisted
 // out of the row loop (the flat-index form was VALU-bound on 64-bit div/mod
 // and re-loaded 4-5 scalar params per element — 18% of the ResNet step).
+// residual: optional second input added before the (optional) ReLU — the
+// ResNet bottleneck join z = relu(bn(x) + idt) fused into the BN pass.
 __global__ void bn_fwd_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
                               const float* __restrict__ mean,
                               const float* __restrict__ invstd,
                               const float* __restrict__ gamma,
                               const float* __restrict__ beta,
+                              const bf16* __restrict__ residual,
                               long M, int C, int relu) {
   const int nch = C / 8;
   for (int phase = 0; phase * (int)blockDim.x < nch; ++phase) {
@@ -98,10 +101,13 @@ __global__ void bn_fwd_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
     const long rStride = (long)gridDim.x * rowsPerBlock;
     for (long r = (long)blockIdx.x * rowsPerBlock + tr; r < M; r += rStride) {
       bf16x8 v = *(const bf16x8*)(x + r * C + c0);
+      bf16x8 res;
+      if (residual) res = *(const bf16x8*)(residual + r * C + c0);
       bf16x8 o;
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float f = (tofloat(v[j]) - mn[j]) * is[j] * gm[j] + bt[j];
+        if (residual) f += tofloat(res[j]);
         if (relu) f = fmaxf(f, 0.f);
         o[j] = tobf16(f);
       }
@@ -260,15 +266,15 @@ void launch_bn_stats(const void* x, void* sum, void* sumsq, long M, int C,
 }
 
 void launch_bn_fwd(const void* x, void* y, const void* mean, const void* invstd,
-                   const void* gamma, const void* beta, long M, int C, int relu,
-                   hipStream_t s) {
+                   const void* gamma, const void* beta, const void* residual,
+                   long M, int C, int relu, hipStream_t s) {
   const int block = 256;
   const int rowsPerBlock = max(1, block / (C / 8));
   const int grid = (int)min((M + rowsPerBlock - 1) / rowsPerBlock, (long)2048);
   hipLaunchKernelGGL(bn_fwd_kernel, dim3(grid), dim3(block),
                      0, s, (const bf16*)x, (bf16*)y, (const float*)mean,
                      (const float*)invstd, (const float*)gamma,
-                     (const float*)beta, M, C, relu);
+                     (const float*)beta, (const bf16*)residual, M, C, relu);
 }
 
 void launch_bn_bwd_reduce(const void* dy, const void* y, const void* x,

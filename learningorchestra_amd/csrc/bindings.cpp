@@ -44,8 +44,8 @@ void launch_maxpool_bwd(const void* dy, const void* idx, void* dx, int B, int H,
 void launch_bn_stats(const void* x, void* sum, void* sumsq, long M, int C,
                      hipStream_t s);
 void launch_bn_fwd(const void* x, void* y, const void* mean, const void* invstd,
-                   const void* gamma, const void* beta, long M, int C, int relu,
-                   hipStream_t s);
+                   const void* gamma, const void* beta, const void* residual,
+                   long M, int C, int relu, hipStream_t s);
 void launch_bn_bwd_reduce(const void* dy, const void* y, const void* x,
                           const void* mean, const void* invstd, void* dbeta,
                           void* dgamma, long M, int C, int relu, hipStream_t s);
@@ -257,13 +257,20 @@ void bn_stats(at::Tensor x, at::Tensor sum, at::Tensor sumsq) {
 }
 
 void bn_fwd(at::Tensor x, at::Tensor y, at::Tensor mean, at::Tensor invstd,
-            at::Tensor gamma, at::Tensor beta, bool relu) {
+            at::Tensor gamma, at::Tensor beta, bool relu,
+            c10::optional<at::Tensor> residual = c10::nullopt) {
   check_bf16(x, "x");
   check_bf16(y, "y");
   const int C = (int)x.size(-1);
+  const void* res = nullptr;
+  if (residual.has_value()) {
+    check_bf16(*residual, "residual");
+    TORCH_CHECK(residual->numel() == x.numel(), "residual shape");
+    res = residual->data_ptr();
+  }
   lo::launch_bn_fwd(x.data_ptr(), y.data_ptr(), mean.data_ptr(),
                     invstd.data_ptr(), gamma.data_ptr(), beta.data_ptr(),
-                    x.numel() / C, C, relu ? 1 : 0, stream());
+                    res, x.numel() / C, C, relu ? 1 : 0, stream());
 }
 
 void bn_bwd_reduce(at::Tensor dy, at::Tensor y, at::Tensor x, at::Tensor mean,
@@ -469,7 +476,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tree_hist", &tree_hist);
   m.def("embedding_fwd", &embedding_fwd);
   m.def("bn_stats", &bn_stats);
-  m.def("bn_fwd", &bn_fwd);
+  m.def("bn_fwd", &bn_fwd, py::arg("x"), py::arg("y"), py::arg("mean"),
+        py::arg("invstd"), py::arg("gamma"), py::arg("beta"),
+        py::arg("relu"), py::arg("residual") = py::none());
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
   m.def("bn_bwd_dx", &bn_bwd_dx);
   m.def("add_relu", &add_relu);

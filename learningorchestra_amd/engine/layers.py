@@ -408,6 +408,7 @@ class BatchNormReLU(Layer):
         self.name = name
         self.c = channels
         self.relu = relu
+        self.residual_relu = True   # fused-join z = relu(bn(x) + residual)
         self.eps = eps
         self.momentum = momentum
         self.training = True
@@ -446,23 +447,31 @@ class BatchNormReLU(Layer):
             }
         return self._bufs
 
-    def forward(self, x: torch.Tensor, stats_ready: bool = False) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, stats_ready: bool = False,
+                residual: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """``residual``: fused join z = relu(bn(x) + residual) — the ResNet
+        bottleneck add without a separate add_relu pass. The residual only
+        changes the forward output; bn backward w.r.t. x is unaffected
+        (callers mask dz by z themselves, as with add_relu)."""
         bufs = self._alloc(tuple(x.shape), x.device)
         x2 = x.reshape(bufs["M"], self.c)
         self._x2 = x2
+        relu = self.relu or (residual is not None and self.residual_relu)
         gamma = self.arena.pf(self.name + ".g")
         beta = self.arena.pf(self.name + ".b")
         if self.training:
             F.bn_fwd_train(x2, gamma, beta, self.eps, bufs["y"], bufs["mean"],
-                           bufs["invstd"], bufs["scratch"], self.relu,
-                           stats_ready=stats_ready and x.is_cuda)
+                           bufs["invstd"], bufs["scratch"], relu,
+                           stats_ready=stats_ready and x.is_cuda,
+                           residual=residual)
             m = self.momentum
             bufs["running_mean"].mul_(1 - m).add_(bufs["mean"], alpha=m)
             var = bufs["invstd"].square().reciprocal() - self.eps
             bufs["running_var"].mul_(1 - m).add_(var, alpha=m)
         else:
             F.bn_fwd_eval(x2, gamma, beta, bufs["running_mean"],
-                          bufs["running_var"], self.eps, bufs["y"], self.relu)
+                          bufs["running_var"], self.eps, bufs["y"], relu,
+                          residual=residual)
         return bufs["y"].view(x.shape)
 
     def backward(self, dy: torch.Tensor) -> torch.Tensor:

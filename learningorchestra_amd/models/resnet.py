@@ -44,6 +44,7 @@ class Bottleneck:
                 BatchNormReLU(f"{name}.dsbn", out_c, relu=False))
         self._z = None
         self._dx = None
+        self._dsum = None
 
     def layers(self):
         out = [self.conv1, self.bn1, self.conv2, self.bn2, self.conv3, self.bn3]
@@ -74,12 +75,15 @@ class Bottleneck:
             idt = x
         h = self._conv_bn(self.conv1, self.bn1, x)
         h = self._conv_bn(self.conv2, self.bn2, h)
-        h = self._conv_bn(self.conv3, self.bn3, h)
-        if self._z is None or self._z.shape != h.shape:
-            self._z = torch.empty_like(h)
-            self._dsum = torch.empty_like(h)
-        F.add_relu(h, idt, out=self._z, relu=True)
-        return self._z
+        # bottleneck join fused into bn3: z = relu(bn3(conv3) + idt)
+        c3 = self.conv3.forward(h, stats=self.bn3.scratch(h.device)
+                                if h.is_cuda else None)
+        z = self.bn3.forward(c3, stats_ready=self.conv3.stats_filled,
+                             residual=idt)
+        self._z = z
+        if self._dsum is None or self._dsum.shape != z.shape:
+            self._dsum = torch.empty_like(z)
+        return z
 
     def backward(self, dz: torch.Tensor) -> torch.Tensor:
         dsum = F.relu_bwd(dz, self._z, out=self._dsum)

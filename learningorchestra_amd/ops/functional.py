@@ -303,7 +303,8 @@ def embedding_bwd(ids: torch.Tensor, dy: torch.Tensor,
 def bn_fwd_train(x2d: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
                  eps: float, out: torch.Tensor, mean: torch.Tensor,
                  invstd: torch.Tensor, scratch: torch.Tensor,
-                 relu: bool = True, stats_ready: bool = False) -> None:
+                 relu: bool = True, stats_ready: bool = False,
+                 residual: Optional[torch.Tensor] = None) -> None:
     """Training-mode BN over [M, C] (+fused ReLU). Writes out (bf16), mean,
     invstd (fp32 [C]); ``scratch`` is a [2, C] fp32 workspace (sum/sumsq).
     ``stats_ready``: scratch was already filled by the producing GEMM's fused
@@ -316,12 +317,15 @@ def bn_fwd_train(x2d: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
         mean.copy_(scratch[0] / M)
         invstd.copy_((scratch[1] / M - mean.square()).clamp_(min=0)
                      .add_(eps).rsqrt_())
-        lo.bn_fwd(x2d, out, mean, invstd, gamma, beta, relu)
+        lo.bn_fwd(x2d, out, mean, invstd, gamma, beta, relu,
+                  residual.reshape(M, C) if residual is not None else None)
         return
     xf = x2d.float()
     mean.copy_(xf.mean(0))
     invstd.copy_((xf.var(0, unbiased=False) + eps).rsqrt())
     y = (xf - mean) * invstd * gamma + beta
+    if residual is not None:
+        y = y + residual.reshape(M, C).float()
     if relu:
         y = torch.relu(y)
     out.copy_(y.to(out.dtype))
@@ -329,14 +333,18 @@ def bn_fwd_train(x2d: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
 
 def bn_fwd_eval(x2d: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
                 running_mean: torch.Tensor, running_var: torch.Tensor,
-                eps: float, out: torch.Tensor, relu: bool = True) -> None:
+                eps: float, out: torch.Tensor, relu: bool = True,
+                residual: Optional[torch.Tensor] = None) -> None:
     invstd = (running_var + eps).rsqrt()
     if _is_gpu(x2d):
         lo = require_ext()
         lo.bn_fwd(x2d, out, running_mean.contiguous(), invstd.contiguous(),
-                  gamma, beta, relu)
+                  gamma, beta, relu,
+                  residual.reshape(x2d.shape) if residual is not None else None)
         return
     y = (x2d.float() - running_mean) * invstd * gamma + beta
+    if residual is not None:
+        y = y + residual.reshape(x2d.shape).float()
     if relu:
         y = torch.relu(y)
     out.copy_(y.to(out.dtype))
