@@ -14,6 +14,11 @@ Design (LightGBM/XGBoost-style, one GPU):
 GBT: logistic loss (grad = p - y, hess = p(1-p)).
 RF:  squared-loss trees on bootstrap + feature subsample, averaged.
 DT:  a single deeper tree.
+Data parallelism (the 8-GPU BASELINE config): each rank holds a row shard;
+bin edges broadcast from rank 0 and the per-level (grad,hess) histograms are
+all-reduced (they are tiny — nodes x F x 256 x 2 fp32 — so the xGMI cost is
+negligible next to the local tree_hist build), after which every rank grows
+an identical tree.
 """
 from __future__ import annotations
 
@@ -27,13 +32,23 @@ from ..ops._ext import require_ext
 MAX_BINS = 255
 
 
+def _distributed() -> bool:
+    import torch.distributed as dist
+    return dist.is_available() and dist.is_initialized()
+
+
 def quantize(X: torch.Tensor, n_bins: int = MAX_BINS,
              sample: int = 100_000) -> Tuple[torch.Tensor, torch.Tensor]:
-    """X [N,F] float -> (binned uint8 [N,F], bin_edges [F, n_bins-1])."""
+    """X [N,F] float -> (binned uint8 [N,F], bin_edges [F, n_bins-1]).
+    Distributed: rank 0's quantile grid is broadcast so shards bin
+    identically."""
     N, F = X.shape
     idx = torch.randperm(N, device=X.device)[: min(N, sample)]
     qs = torch.linspace(0, 1, n_bins + 1, device=X.device)[1:-1]
     edges = torch.quantile(X[idx].float(), qs, dim=0).T.contiguous()  # [F, n_bins-1]
+    if _distributed():
+        import torch.distributed as dist
+        dist.broadcast(edges, src=0)
     binned = torch.searchsorted(edges, X.T.contiguous().float()).T  # [N,F] in [0,n_bins-1]
     return binned.to(torch.uint8).contiguous(), edges
 
@@ -126,10 +141,16 @@ class TreeLearner:
             node_of = torch.where(sample_mask, node_of,
                                   torch.full_like(node_of, -1))
         level_start = 0
+        sync = _distributed()
         for depth in range(self.max_depth + 1):
             level_nodes = 2 ** depth
             rel_node = node_of - level_start
             hist = build_histograms(binned, rel_node, grad, hess, level_nodes)
+            if sync:
+                # data-parallel trees: shards contribute partial histograms;
+                # the reduced histogram makes every rank's splits identical
+                import torch.distributed as dist
+                dist.all_reduce(hist, op=dist.ReduceOp.SUM)
             # each sample lands once per feature; per-node totals from feature 0
             Gn = hist[:, 0, :, 0].sum(-1)
             Hn = hist[:, 0, :, 1].sum(-1)
@@ -201,7 +222,14 @@ class GBTClassifier(_TreeEnsembleBase):
         X = self._to_device_tensor(X)
         y = self._to_device_tensor(y).clamp(0, 1)
         binned, self.edges = quantize(X)
-        p0 = y.mean().clamp(1e-4, 1 - 1e-4)
+        if _distributed():
+            import torch.distributed as dist
+            t = torch.stack([y.sum(), torch.tensor(float(y.numel()),
+                                                   device=y.device)])
+            dist.all_reduce(t, op=dist.ReduceOp.SUM)
+            p0 = (t[0] / t[1]).clamp(1e-4, 1 - 1e-4)
+        else:
+            p0 = y.mean().clamp(1e-4, 1 - 1e-4)
         self.base_score = float(torch.log(p0 / (1 - p0)))
         raw = torch.full_like(y, self.base_score)
         learner = TreeLearner(self.max_depth, lr=self.lr)
