@@ -1,5 +1,10 @@
 """Race screen for the new 8-phase 256^2 GEMM (sync-structure change -> guide
 two-lane discipline: multi-run refcheck at several sizes)."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 from learningorchestra_amd.ops import functional as F
 

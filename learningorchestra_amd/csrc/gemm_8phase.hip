@@ -101,27 +101,30 @@ __global__ __launch_bounds__(512) void gemm256_kernel(
   // the NEXT tile's stages), so its lgkmcnt wait spans two barriers and the
   // staging section — the per-phase LDS latency is hidden (PMC showed
   // WAIT_INST_LDS ~= busy cycles with reads issued in their own phase).
-  for (int s = 0; s < ntiles; ++s) {
-    const int buf = s & 1;
-    bf16x8 afr[2][2][2];                    // [phase parity][mi-pair][kk]
-    auto read_quad = [&](int q, int pp) {
+  bf16x8 afr[2][2][2];                      // [phase parity][mi-pair][kk]
+  auto read_quad = [&](int buf_, int q, int pp) {
+    #pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int row = wr * 128 + (q * 2 + j) * 16 + fr;
       #pragma unroll
-      for (int j = 0; j < 2; ++j) {
-        const int row = wr * 128 + (q * 2 + j) * 16 + fr;
-        #pragma unroll
-        for (int kk = 0; kk < 2; ++kk)
-          afr[pp][j][kk] = lds_frag(buf, 0, row, kk);
-      }
-    };
-    // tile prologue (q0 section): B frags + quadrant 0
+      for (int kk = 0; kk < 2; ++kk)
+        afr[pp][j][kk] = lds_frag(buf_, 0, row, kk);
+    }
+  };
+  auto read_bfr = [&](int buf_) {
     #pragma unroll
     for (int ni = 0; ni < NFRAG; ++ni) {
       const int row = wc * 64 + ni * 16 + fr;
       #pragma unroll
       for (int kk = 0; kk < 2; ++kk)
-        bfr[ni][kk] = lds_frag(buf, BM * BKB, row, kk);
+        bfr[ni][kk] = lds_frag(buf_, BM * BKB, row, kk);
     }
-    read_quad(0, 0);
+  };
+  // tile-0 prologue reads (later tiles prefetch during the previous q3)
+  read_bfr(0);
+  read_quad(0, 0, 0);
+  for (int s = 0; s < ntiles; ++s) {
+    const int buf = s & 1;
     #pragma unroll
     for (int q = 0; q < 4; ++q) {
       // phase staging (schedule in the header comment)
@@ -143,7 +146,14 @@ __global__ __launch_bounds__(512) void gemm256_kernel(
             acc[q * 2 + j][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 afr[q & 1][j][kk], bfr[ni][kk], acc[q * 2 + j][ni], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
-      if (q < 3) read_quad(q + 1, (q + 1) & 1);
+      if (q < 3) {
+        read_quad(buf, q + 1, (q + 1) & 1);
+      } else if (s + 1 < ntiles) {
+        // next tile's frags: content visible after this phase's vmcnt +
+        // opening barrier; bfr/afr[0] are dead after q3's MFMA cluster
+        read_bfr(buf ^ 1);
+        read_quad(buf ^ 1, 0, 0);
+      }
       asm volatile("s_barrier" ::: "memory");
     }
   }
