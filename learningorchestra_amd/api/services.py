@@ -154,7 +154,7 @@ class BuilderService:
     MI355X-native tabular engines in models/tabular.py / models/trees.py.
     """
 
-    CLASSIFIERS = ("lr", "dt", "rf", "gb", "nb")
+    CLASSIFIERS = ("lr", "dt", "rf", "gb", "nb", "mlp")
 
     def __init__(self, database, artifacts, scheduler, allow_user_code=True,
                  device: Optional[str] = None):

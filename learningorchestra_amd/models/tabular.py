@@ -158,4 +158,83 @@ def make_classifier(kind: str, device: Optional[str] = None, **kw):
         return RandomForestClassifier(device=device, **kw)
     if kind == "gb":
         return GBTClassifier(device=device, **kw)
-    raise ValueError(f"unknown classifier '{kind}' (use lr/dt/rf/gb/nb)")
+    if kind == "mlp":
+        return MLPClassifier(device=device, **kw)
+    raise ValueError(f"unknown classifier '{kind}' (use lr/dt/rf/gb/nb/mlp)")
+
+
+class MLPClassifier:
+    """Configurable multi-layer perceptron on the engine (native analog of
+    sklearn.neural_network.MLPClassifier / a keras Dense stack): ReLU hidden
+    layers + padded softmax head, every op a gfx950 kernel on GPU."""
+
+    def __init__(self, hidden=(256, 128), epochs: int = 20,
+                 batch_size: int = 8192, lr: float = 0.05, momentum: float = 0.9,
+                 device: Optional[str] = None, seed: int = 0,
+                 standardize: bool = True):
+        self.hidden = tuple(hidden)
+        self.epochs = epochs
+        self.batch_size = batch_size
+        self.lr, self.momentum = lr, momentum
+        self.device = _pick_device(device)
+        self.seed = seed
+        self.standardize = standardize
+        self.model: Optional[SequentialClassifier] = None
+        self._mu = self._sigma = None
+        self.classes_: Optional[np.ndarray] = None
+
+    def _prep(self, X) -> torch.Tensor:
+        X = torch.as_tensor(np.asarray(X, dtype=np.float32))
+        X = torch.nan_to_num(X, nan=0.0)
+        if self.standardize and self._mu is not None:
+            X = (X - self._mu) / self._sigma
+        fpad = _pad8(X.shape[1])
+        if fpad != X.shape[1]:
+            X = torch.nn.functional.pad(X, (0, fpad - X.shape[1]))
+        return X.to(self.device, torch.bfloat16)
+
+    def fit(self, X, y, epochs: Optional[int] = None) -> "MLPClassifier":
+        Xt = torch.as_tensor(np.asarray(X, dtype=np.float32))
+        Xt = torch.nan_to_num(Xt, nan=0.0)
+        if self.standardize:
+            self._mu = Xt.mean(0, keepdim=True)
+            self._sigma = Xt.std(0, keepdim=True).clamp(min=1e-6)
+        yt = np.asarray(y)
+        self.classes_ = np.unique(yt)
+        ymap = {c: i for i, c in enumerate(self.classes_)}
+        yi = torch.as_tensor([ymap[v] for v in yt], dtype=torch.long,
+                             device=self.device)
+        Xb = self._prep(X)
+        dims = [Xb.shape[1], *[_pad8(h) for h in self.hidden]]
+        ncls = len(self.classes_)
+        cpad = max(8, _pad8(ncls))
+        layers = [Linear(f"mlp.l{i}", dims[i], dims[i + 1], relu=True)
+                  for i in range(len(dims) - 1)]
+        layers.append(Linear("mlp.head", dims[-1], cpad))
+        self.model = SequentialClassifier(layers, ncls, device=self.device,
+                                          seed=self.seed)
+        trainer = Trainer(self.model, make_sgd(self.model, lr=self.lr,
+                                               momentum=self.momentum),
+                          self.device)
+        n = Xb.shape[0]
+        bs = min(self.batch_size, n)
+        g = torch.Generator(device="cpu").manual_seed(self.seed)
+        for _ in range(int(epochs or self.epochs)):
+            perm = torch.randperm(n, generator=g).to(self.device)
+            for i in range(0, n - bs + 1, bs):
+                sel = perm[i:i + bs]
+                trainer.step_async(Xb[sel].contiguous(), yi[sel].contiguous())
+        return self
+
+    def predict(self, X):
+        with torch.no_grad():
+            logits = self.model.forward(self._prep(X)).float()
+        return self.classes_[logits[:, :len(self.classes_)].argmax(1).cpu().numpy()]
+
+    def predict_proba(self, X):
+        with torch.no_grad():
+            logits = self.model.forward(self._prep(X)).float()
+        return torch.softmax(logits[:, :len(self.classes_)], 1).cpu().numpy()
+
+    def score(self, X, y):
+        return float((self.predict(X) == np.asarray(y)).mean())

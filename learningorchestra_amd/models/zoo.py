@@ -16,7 +16,8 @@ import torch
 
 from ..engine.trainer import Trainer, make_sgd
 from .mnist_cnn import build_mnist_cnn
-from .tabular import (GaussianNBClassifier, LogisticRegressionClassifier)  # noqa: F401 - re-export
+from .tabular import (GaussianNBClassifier, LogisticRegressionClassifier,  # noqa: F401 - re-export
+                      MLPClassifier)
 from .trees import (DecisionTreeClassifier, GBTClassifier,  # noqa: F401 - re-export
                     RandomForestClassifier)
 

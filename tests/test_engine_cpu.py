@@ -177,3 +177,16 @@ def test_grid_search_tuning():
     assert len(gs.results_) == 2
     assert gs.best_score_ > 0.8
     assert gs.predict(X[:10]).shape == (10,)
+
+
+def test_mlp_classifier():
+    from learningorchestra_amd.models.tabular import MLPClassifier
+    import numpy as np
+    rng = np.random.RandomState(0)
+    X = rng.randn(600, 6).astype("float32")
+    y = ((X[:, 0] + X[:, 1] * X[:, 2]) > 0).astype("int64")
+    clf = MLPClassifier(hidden=(32, 16), epochs=30, batch_size=128,
+                        device="cpu", lr=0.1)
+    clf.fit(X[:500], y[:500])
+    assert clf.score(X[500:], y[500:]) > 0.7
+    assert clf.predict_proba(X[:5]).shape == (5, 2)
