@@ -11,6 +11,8 @@ fc 1024->256 (ReLU) -> fc 256->16 (10 valid classes, padded head).
 """
 from __future__ import annotations
 
+import os
+
 from ..engine.layers import (Conv2dNHWC, Flatten, Linear, MaxPool2dNHWC,
                              SequentialClassifier)
 
@@ -21,10 +23,13 @@ CPAD = 16  # classifier head padded for 16-B-aligned GEMM rows
 def build_mnist_cnn(device="cpu", seed: int = 0,
                     channels=(32, 64), fc_width: int = 256) -> SequentialClassifier:
     c1, c2 = channels
+    # LO_IMPLICIT_CONV=1: conv2 gathers im2col inside the GEMM staging
+    # (no col matrix) — A/B'd on GPU, see PERFORMANCE.md
+    imp = os.environ.get("LO_IMPLICIT_CONV", "0") == "1"
     layers = [
         Conv2dNHWC("conv1", 1, c1, 5, 5, relu=True, first=True),    # 28 -> 24
         MaxPool2dNHWC(2),                                           # 24 -> 12
-        Conv2dNHWC("conv2", c1, c2, 5, 5, relu=True),               # 12 -> 8
+        Conv2dNHWC("conv2", c1, c2, 5, 5, relu=True, implicit=imp),  # 12 -> 8
         MaxPool2dNHWC(2),                                           # 8 -> 4
         Flatten(),
         Linear("fc1", 4 * 4 * c2, fc_width, relu=True),

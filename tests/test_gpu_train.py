@@ -224,3 +224,17 @@ def test_resnet50_gpu_step():
     loss, _ = m.train_step(x, y)
     assert t.isfinite(loss).all() and loss.item() > 0
     assert t.isfinite(m.arena.grad).all()
+
+
+@pytest.mark.gpu
+def test_mlp_classifier_gpu():
+    """Native MLPClassifier end-to-end on the HIP engine."""
+    import numpy as np
+    from learningorchestra_amd.models.tabular import MLPClassifier
+    rng = np.random.RandomState(0)
+    X = rng.randn(4000, 16).astype("float32")
+    y = ((X[:, 0] + X[:, 1] * X[:, 2]) > 0).astype("int64")
+    clf = MLPClassifier(hidden=(64, 32), epochs=15, batch_size=512,
+                        device="cuda", lr=0.1)
+    clf.fit(X[:3200], y[:3200])
+    assert clf.score(X[3200:], y[3200:]) > 0.8
