@@ -23,11 +23,15 @@ CPAD = 16  # classifier head padded for 16-B-aligned GEMM rows
 def build_mnist_cnn(device="cpu", seed: int = 0,
                     channels=(32, 64), fc_width: int = 256) -> SequentialClassifier:
     c1, c2 = channels
-    # LO_IMPLICIT_CONV=1: conv2 gathers im2col inside the GEMM staging
-    # (no col matrix) — A/B'd on GPU, see PERFORMANCE.md
-    imp = os.environ.get("LO_IMPLICIT_CONV", "0") == "1"
+    # conv2 defaults to implicit-GEMM conv (im2col gathered inside the GEMM
+    # staging, no col matrix): measured +5% end-to-end vs materialized col at
+    # B=32768 (PERFORMANCE.md). LO_IMPLICIT_CONV=0 restores the col path;
+    # LO_IMPLICIT_CONV1=1 additionally tries conv1 (C=1, scalar gather).
+    imp = os.environ.get("LO_IMPLICIT_CONV", "1") == "1"
+    imp1 = os.environ.get("LO_IMPLICIT_CONV1", "0") == "1"
     layers = [
-        Conv2dNHWC("conv1", 1, c1, 5, 5, relu=True, first=True),    # 28 -> 24
+        Conv2dNHWC("conv1", 1, c1, 5, 5, relu=True, first=True,
+                   implicit=imp1),                                  # 28 -> 24
         MaxPool2dNHWC(2),                                           # 24 -> 12
         Conv2dNHWC("conv2", c1, c2, 5, 5, relu=True, implicit=imp),  # 12 -> 8
         MaxPool2dNHWC(2),                                           # 8 -> 4

@@ -10,6 +10,8 @@ branch concat/split and the branch-grad sum are device memcpy/add glue.
 """
 from __future__ import annotations
 
+import os
+
 from typing import Optional, Sequence, Tuple
 
 import torch
@@ -31,7 +33,9 @@ class TextCNN:
 
         self.arena = ParamArena(device)
         self.emb = Embedding("emb", vocab, emb_dim)
-        self.convs = [Conv2dNHWC(f"conv{k}", emb_dim, filters, k, 1, relu=True)
+        imp = os.environ.get("LO_IMPLICIT_CONV", "0") == "1"
+        self.convs = [Conv2dNHWC(f"conv{k}", emb_dim, filters, k, 1,
+                                 relu=True, implicit=imp)
                       for k in self.kernel_sizes]
         self.pools = [MaxPool2dNHWC(1) for _ in self.kernel_sizes]  # k set per fwd
         self.fc = Linear("fc", filters * len(self.kernel_sizes), self.cpad)
