@@ -32,6 +32,10 @@ void launch_mfma_probe(const void* A, const void* B, float* D, hipStream_t s);
 void launch_im2col(const void* in, void* col, int B, int H, int W, int C,
                    int KH, int KW, int SH, int SW, int PH, int PW,
                    int OH, int OW, int Kpad, hipStream_t s);
+bool launch_conv_dx(const void* dy2, long ldy, const void* wt, long ldw,
+                    void* dx, int B, int H, int W, int C, int KH, int KW,
+                    int SH, int SW, int PH, int PW, int OH, int OW, int outC,
+                    hipStream_t s);
 void launch_col2im(const void* dcol, void* dx, int B, int H, int W, int C,
                    int KH, int KW, int SH, int SW, int PH, int PW, int OH,
                    int OW, int Kpad, hipStream_t s);
@@ -215,6 +219,27 @@ at::Tensor col2im(at::Tensor dcol, int64_t B, int64_t H, int64_t W, int64_t C,
                     (int)KH, (int)KW, (int)SH, (int)SW, (int)PH, (int)PW, OH, OW,
                     (int)dcol.size(1), stream());
   return dx;
+}
+
+// fused conv dX: dy2 [B*OH*OW, outC] @ wt[kpad, outC]^T scattered into
+// dx [B,H,W,C] through an LDS fp32 accumulator (no dcol matrix).
+// Returns false when the shape is not eligible (caller falls back).
+bool conv_dx(at::Tensor dy2, at::Tensor wt, at::Tensor dx, int64_t KH,
+             int64_t KW, int64_t SH, int64_t SW, int64_t PH, int64_t PW) {
+  check_bf16(dy2, "dy2");
+  check_bf16(wt, "wt");
+  check_bf16(dx, "dx");
+  const int B = (int)dx.size(0), H = (int)dx.size(1), W = (int)dx.size(2),
+            C = (int)dx.size(3);
+  const int OH = (H + 2 * (int)PH - (int)KH) / (int)SH + 1;
+  const int OW = (W + 2 * (int)PW - (int)KW) / (int)SW + 1;
+  const int outC = (int)dy2.size(1);
+  TORCH_CHECK(dy2.size(0) == (long)B * OH * OW, "dy2 rows");
+  TORCH_CHECK(wt.size(0) >= (long)KH * KW * C && wt.size(1) == outC, "wt shape");
+  return lo::launch_conv_dx(dy2.data_ptr(), dy2.stride(0), wt.data_ptr(),
+                            wt.stride(0), dx.data_ptr(), B, H, W, C, (int)KH,
+                            (int)KW, (int)SH, (int)SW, (int)PH, (int)PW, OH,
+                            OW, outC, stream());
 }
 
 std::vector<at::Tensor> maxpool_fwd(at::Tensor in, int64_t KH, int64_t KW,
@@ -464,6 +489,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_conv_dw", &gemm_conv_dw);
   m.def("mfma_probe", &mfma_probe);
   m.def("im2col", &im2col);
+  m.def("conv_dx", &conv_dx, "fused conv dX (LDS-accumulated scatter)");
   m.def("col2im", &col2im);
   m.def("maxpool_fwd", &maxpool_fwd);
   m.def("maxpool_bwd", &maxpool_bwd);

@@ -105,9 +105,16 @@ class Conv2dNHWC(Layer):
                 self._bufs["col"] = torch.zeros((M, self.kpad), device=dev, dtype=dtype)
             if not self.first:
                 self._bufs["dx"] = torch.empty((B, H, W, self.in_c), device=dev, dtype=dtype)
-                if not self._is_1x1:
+                if not self._is_1x1 and not (dev.type == "cuda"
+                                             and self._dx_fused_ok(H, W)):
                     self._bufs["dcol"] = torch.empty((M, self.kpad), device=dev, dtype=dtype)
         return self._bufs
+
+    def _dx_fused_ok(self, H: int, W: int) -> bool:
+        # mirror of launch_conv_dx eligibility: whole-image dx fits LDS
+        hwc = H * W * self.in_c
+        return (hwc * 4 <= 48 * 1024 and self.in_c % 16 == 0 and
+                hwc % 8 == 0 and self.out_c <= 64 and self.out_c % 8 == 0)
 
     def _wt(self) -> torch.Tensor:
         # transposed mirror [kpad, out_c] (see Linear._wt)
@@ -173,6 +180,12 @@ class Conv2dNHWC(Layer):
             dx = bufs["dx"]
             F.gemm(dy2, self._wt(), tb=True, out=dx.view(M, self.in_c))
             return dx
+        if dy2.is_cuda and self._dx_fused_ok(bufs["H"], bufs["W"]) \
+                and F.conv2d_dx_fused(
+                dy2, self._wt(), bufs["B"], bufs["H"], bufs["W"], self.in_c,
+                self.kh, self.kw, self.stride, self.stride, self.pad,
+                self.pad, out=bufs["dx"]):
+            return bufs["dx"]
         F.gemm(dy2, self._wt(), tb=True, out=bufs["dcol"])
         F.col2im(bufs["dcol"], bufs["B"], bufs["H"], bufs["W"], self.in_c,
                  self.kh, self.kw, self.stride, self.stride, self.pad,

@@ -109,6 +109,20 @@ def col2im(dcol: torch.Tensor, B: int, H: int, W: int, C: int, kh: int, kw: int,
     return out
 
 
+def conv2d_dx_fused(dy2: torch.Tensor, wt: torch.Tensor, B: int, H: int,
+                    W: int, C: int, kh: int, kw: int, sh: int, sw: int,
+                    ph: int, pw: int, out: torch.Tensor) -> bool:
+    """GPU-only fused conv dX: dcol = dy2 @ wt^T computed per image with MFMA
+    and scattered straight into an LDS fp32 dx accumulator — no dcol matrix
+    (replaces gemm-NT + col2im).  wt is the [kpad, outC] transposed weight
+    mirror.  Returns False when the shape is not eligible (whole-image dx
+    must fit LDS: H*W*C*4 <= 48 KB, C%16==0, outC <= 64) — caller falls back."""
+    if not dy2.is_cuda:
+        return False
+    lo = require_ext()
+    return bool(lo.conv_dx(dy2, wt, out, kh, kw, sh, sw, ph, pw))
+
+
 # ------------------------------------------------------------------ maxpool
 def maxpool2d(x: torch.Tensor, kh: int, kw: int, sh: int, sw: int,
               ph: int = 0, pw: int = 0) -> Tuple[torch.Tensor, torch.Tensor]:

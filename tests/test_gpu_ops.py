@@ -254,3 +254,42 @@ def test_gemm_8phase_race_screen(lo):
             ref = A.float() @ B.float().t()
             rel = ((out.float() - ref).norm() / ref.norm()).item()
             assert rel < 2e-2, (trial, Msz, Nsz, Ksz, rel)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [
+    # (B, H, W, C, KH, KW, S, P, outC) — eligibility: H*W*C*4 <= 48KB
+    (4, 12, 12, 32, 5, 5, 1, 0, 64),     # MNIST conv2 geometry
+    (3, 12, 12, 32, 5, 5, 1, 2, 64),     # with padding
+    (2, 16, 16, 16, 3, 3, 1, 1, 48),     # C=16, outC not 64
+    (2, 15, 15, 16, 3, 3, 2, 1, 32),     # stride 2, odd spatial
+    (5, 9, 7, 16, 3, 3, 1, 0, 8),        # tiny outC, R not %64
+])
+def test_conv_dx_fused_matches_col2im(shape):
+    """Fused conv dX (MFMA + LDS scatter) vs the dcol+col2im reference."""
+    from learningorchestra_amd.ops import functional as F
+    B, H, W, C, KH, KW, S, P, outC = shape
+    torch.manual_seed(0)
+    OH = (H + 2 * P - KH) // S + 1
+    OW = (W + 2 * P - KW) // S + 1
+    kdim = KH * KW * C
+    kpad = (kdim + 7) // 8 * 8
+    dy2 = torch.randn(B * OH * OW, outC, device="cuda").to(torch.bfloat16)
+    wt = torch.zeros(kpad, outC, device="cuda", dtype=torch.bfloat16)
+    wt[:kdim] = torch.randn(kdim, outC, device="cuda").to(torch.bfloat16)
+    dx = torch.empty(B, H, W, C, device="cuda", dtype=torch.bfloat16)
+    ok = F.conv2d_dx_fused(dy2, wt, B, H, W, C, KH, KW, S, S, P, P, out=dx)
+    assert ok, "shape should be eligible"
+    dcol = F.gemm(dy2, wt, tb=True)
+    ref = F.col2im(dcol, B, H, W, C, KH, KW, S, S, P, P)
+    torch.testing.assert_close(dx.float(), ref.float(), atol=5e-2, rtol=5e-2)
+
+
+@pytest.mark.gpu
+def test_conv_dx_fused_rejects_oversize():
+    from learningorchestra_amd.ops import functional as F
+    dy2 = torch.randn(2 * 62 * 62, 64, device="cuda").to(torch.bfloat16)
+    wt = torch.randn(9 * 128, 64, device="cuda").to(torch.bfloat16)
+    dx = torch.empty(2, 64, 64, 128, device="cuda", dtype=torch.bfloat16)
+    assert not F.conv2d_dx_fused(dy2, wt, 2, 64, 64, 128, 3, 3, 1, 1, 0, 0,
+                                 out=dx)
