@@ -216,22 +216,43 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
     bf16* __restrict__ dx,                    // [B, H, W, C]
     int H, int W, int C, int KH, int KW, int SH, int SW, int PH, int PW,
     int OH, int OW, int outC) {
-  extern __shared__ float ldx[];              // H*W*C fp32 accumulator
+  // dynamic LDS: fp32 dx accumulator (HWC) + 2 x 4 KB wt tile buffers
+  // (32 rows x 64 k, XOR-swizzled like the GEMM's TB operand)
+  extern __shared__ float ldx[];
   const int img = blockIdx.x;
   const int R = OH * OW;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wave = tid >> 6;
   const int HWC = H * W * C;
+  char* smW = (char*)(ldx + HWC);
   for (int i = tid; i < HWC; i += 256) ldx[i] = 0.f;
-  __syncthreads();
 
   const bf16* dyi = dy2 + (long)img * OW * OH * ldy;
   const int CB = C > 32 ? 32 : C;             // n-tile: one (kh,kw) C-slice
+  const int NSTEP = KH * KW * (C / CB);       // (kh,kw,cb) tiles per m-chunk
+
+  // wt-tile staging: thread t fetches row t/8, 16-B k-chunk t%8 of the
+  // 32x64 tile; rows beyond CB / k beyond outC stage zeros.  The LDS image
+  // is row*128B with the ((row&7)<<4) XOR swizzle the fragment reads undo.
+  const int srow = tid >> 3, skc = tid & 7;
+  auto tile_n0 = [&](int step) {
+    const int cslices = C / CB;
+    const int khkw = step / cslices, cb = (step % cslices) * CB;
+    return (khkw) * C + cb;                   // first wt row of the tile
+  };
+  auto fetch_tile = [&](int step) -> bf16x8 {
+    bf16x8 v = {};
+    if (step < NSTEP && srow < CB && skc * 8 < outC)
+      v = *(const bf16x8*)(wt + (long)(tile_n0(step) + srow) * ldw + skc * 8);
+    return v;
+  };
+  auto write_tile = [&](int buf, bf16x8 v) {
+    *(bf16x8*)(smW + buf * 4096 + srow * 128 + ((skc * 16) ^ ((srow & 7) << 4))) = v;
+  };
+
   for (int m0 = 0; m0 < R; m0 += 64) {
-    // the wave's 16 A-rows and their (oh,ow); rows beyond R contribute 0
+    // the wave's 16 A-rows: row=lane%16, k=(lane/16)*8+j (outC <= 64)
     const int arow = m0 + wave * 16 + (lane & 15);
-    // A fragments for the whole m-chunk: row=lane%16, k=(lane/16)*8+j.
-    // outC <= 64 here (two k-chunks); guard rows/k-range.
     bf16x8 af[2] = {};
     #pragma unroll
     for (int kc = 0; kc < 2; ++kc) {
@@ -239,7 +260,7 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
       if (arow < R && k < outC)
         af[kc] = *(const bf16x8*)(dyi + (long)arow * ldy + k);
     }
-    // the 4 accumulator rows this lane will scatter: m = m0+wave*16+(lane>>4)*4+reg
+    // the 4 accumulator rows this lane scatters: m = m0+wave*16+(lane>>4)*4+r
     int oh4[4], ow4[4];
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -247,43 +268,48 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
       oh4[r] = m / OW;
       ow4[r] = m - oh4[r] * OW;
     }
-    const int nkc = (outC + 31) / 32;
-    for (int kh = 0; kh < KH; ++kh) {
-      for (int kw = 0; kw < KW; ++kw) {
-        for (int cb = 0; cb < C; cb += CB) {
-          const int n0 = (kh * KW + kw) * C + cb;     // first wt row (col idx)
-          f32x4 acc[2] = {};
-          #pragma unroll
-          for (int kc = 0; kc < 2; ++kc) {
-            if (kc >= nkc) break;
-            const int k = kc * 32 + (lane >> 4) * 8;
-            #pragma unroll
-            for (int ni = 0; ni < 2; ++ni) {
-              bf16x8 bfr = {};
-              const int wr = n0 + ni * 16 + (lane & 15);
-              if (k < outC && ni * 16 < CB)
-                bfr = *(const bf16x8*)(wt + (long)wr * ldw + k);
-              acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  af[kc], bfr, acc[ni], 0, 0, 0);
-            }
-          }
-          // scatter: D col=lane&15 -> channel, row=(lane>>4)*4+reg -> m
-          #pragma unroll
-          for (int ni = 0; ni < 2; ++ni) {
-            if (ni * 16 >= CB) break;
-            const int c = cb + ni * 16 + (lane & 15);
-            #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-              const int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
-              if (m >= R) break;
-              const int h = oh4[r] * SH - PH + kh;
-              const int w = ow4[r] * SW - PW + kw;
-              if (h >= 0 && h < H && w >= 0 && w < W)
-                atomicAdd(ldx + (h * W + w) * C + c, acc[ni][r]);
-            }
-          }
+    const int mok = (m0 + wave * 16 + (lane >> 4) * 4 + 3 < R) ? 4
+                    : max(0, R - (m0 + wave * 16 + (lane >> 4) * 4));
+
+    bf16x8 stg = fetch_tile(0);
+    write_tile(0, stg);
+    for (int step = 0; step < NSTEP; ++step) {
+      const int buf = step & 1;
+      stg = fetch_tile(step + 1);             // prefetch next tile
+      __syncthreads();                        // tile `buf` visible
+      const int cslices = C / CB;
+      const int khkw = step / cslices, cb = (step % cslices) * CB;
+      const int kh = khkw / KW, kw = khkw - kh * KW;
+      f32x4 acc[2] = {};
+      #pragma unroll
+      for (int kc = 0; kc < 2; ++kc) {
+        const int k = kc * 32 + (lane >> 4) * 8;
+        #pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          const int row = ni * 16 + (lane & 15);
+          const bf16x8 bfr = *(const bf16x8*)(
+              smW + buf * 4096 + row * 128 + ((k * 2) ^ ((row & 7) << 4)));
+          acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[kc], bfr, acc[ni], 0, 0, 0);
         }
       }
+      // scatter: D col=lane&15 -> channel, row=(lane>>4)*4+reg -> m
+      const int hb = -PH + kh, wb = -PW + kw;
+      #pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        if (ni * 16 >= CB) break;
+        const int c = cb + ni * 16 + (lane & 15);
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          if (r >= mok) break;
+          const int h = oh4[r] * SH + hb;
+          const int w = ow4[r] * SW + wb;
+          if (h >= 0 && h < H && w >= 0 && w < W)
+            atomicAdd(ldx + (h * W + w) * C + c, acc[ni][r]);
+        }
+      }
+      __syncthreads();                        // scatter done before overwrite
+      write_tile(buf ^ 1, stg);
     }
   }
   __syncthreads();
@@ -301,10 +327,10 @@ bool launch_conv_dx(const void* dy2, long ldy, const void* wt, long ldw,
                     int SH, int SW, int PH, int PW, int OH, int OW, int outC,
                     hipStream_t s) {
   const int HWC = H * W * C;
-  if ((long)HWC * 4 > 48 * 1024 || C % 16 != 0 || HWC % 8 != 0 || outC > 64 ||
-      outC % 8 != 0)
+  if ((long)HWC * 4 > 48 * 1024 || (C % 32 != 0 && C != 16) || HWC % 8 != 0 ||
+      outC > 64 || outC % 8 != 0)
     return false;
-  hipLaunchKernelGGL(conv_dx_kernel, dim3(B), dim3(256), HWC * 4, s,
+  hipLaunchKernelGGL(conv_dx_kernel, dim3(B), dim3(256), HWC * 4 + 8192, s,
                      (const bf16*)dy2, ldy, (const bf16*)wt, ldw, (bf16*)dx,
                      H, W, C, KH, KW, SH, SW, PH, PW, OH, OW, outC);
   return true;

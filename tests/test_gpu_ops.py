@@ -282,7 +282,14 @@ def test_conv_dx_fused_matches_col2im(shape):
     assert ok, "shape should be eligible"
     dcol = F.gemm(dy2, wt, tb=True)
     ref = F.col2im(dcol, B, H, W, C, KH, KW, S, S, P, P)
-    torch.testing.assert_close(dx.float(), ref.float(), atol=5e-2, rtol=5e-2)
+    # fp32 reference (no bf16 dcol rounding): fused accumulates in fp32 so it
+    # must be at least as close to it as the two-pass path is
+    dcol32 = (dy2.float() @ wt.float().t()).cpu()
+    ref32 = F.col2im(dcol32, B, H, W, C, KH, KW, S, S, P, P,
+                     out=torch.empty(B, H, W, C))
+    err_fused = (dx.float().cpu() - ref32).abs().max().item()
+    err_ref = (ref.float().cpu() - ref32).abs().max().item()
+    assert err_fused <= max(2 * err_ref, 1e-3), (err_fused, err_ref)
 
 
 @pytest.mark.gpu
