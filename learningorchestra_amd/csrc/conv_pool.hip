@@ -205,27 +205,38 @@ void launch_col2im(const void* dcol, void* dx, int B, int H, int W, int C,
 // ------------------------------------------------------- fused conv dX -----
 // dX without the dcol matrix: one block per image computes the dcol tiles
 // (dY_img[R, outC] @ Wt[kpad, outC]^T) with MFMA and scatters every element
-// straight into an LDS-resident fp32 dx accumulator (ds atomics — targets
-// within one fragment are unique, across (kh,kw) steps they collide), then
-// writes the image's dx once.  Replaces NT-GEMM-store(dcol) + col2im
-// re-read: for MNIST conv2 that is 2x3.4 GB of HBM traffic eliminated.
-// Eligible when the whole image's dx fits LDS (H*W*C*4 <= 48 KB, C%16==0).
+// straight into an LDS-resident fp32 dx accumulator, then writes the
+// image's dx once.  Replaces NT-GEMM-store(dcol) + col2im re-read: for
+// MNIST conv2 that is 2x3.4 GB of HBM traffic eliminated.
+//
+// The scatter is PLAIN read-add-write, not ds atomics: within one
+// (kh,kw,cb) step every element's target (h,w,c) is unique across the
+// whole block (distinct m -> distinct (h,w) at fixed (kh,kw); distinct c),
+// and steps are barrier-separated, so non-atomic RMW is race-free.
+// (Measured: LDS atomicAdd ran at ~0.3 lane-ops/cycle/CU and was 34x the
+// cost of everything else in the kernel combined - see PERFORMANCE.md.)
+// The accumulator rows are padded to C+1 floats so the 4 row-groups of a
+// fragment land in different banks (consecutive m -> +1 row -> +1 bank).
+// Eligible when the whole image's dx fits LDS (H*W*(C+1)*4 <= 56 KB).
 __global__ __launch_bounds__(256) void conv_dx_kernel(
     const bf16* __restrict__ dy2, long ldy,   // [B*OH*OW, outC]
     const bf16* __restrict__ wt, long ldw,    // [kpad, outC] row-major
     bf16* __restrict__ dx,                    // [B, H, W, C]
     int H, int W, int C, int KH, int KW, int SH, int SW, int PH, int PW,
     int OH, int OW, int outC) {
-  // dynamic LDS: fp32 dx accumulator (HWC) + 2 x 4 KB wt tile buffers
-  // (32 rows x 64 k, XOR-swizzled like the GEMM's TB operand)
+  // dynamic LDS: fp32 dx accumulator (H*W rows of C+1 floats - the +1
+  // staggers banks across rows) + 2 x 4 KB wt tile buffers (32 rows x 64 k,
+  // XOR-swizzled like the GEMM's TB operand)
   extern __shared__ float ldx[];
   const int img = blockIdx.x;
   const int R = OH * OW;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wave = tid >> 6;
   const int HWC = H * W * C;
-  char* smW = (char*)(ldx + HWC);
-  for (int i = tid; i < HWC; i += 256) ldx[i] = 0.f;
+  const int CP = C + 1;
+  const int HWCP = H * W * CP;
+  char* smW = (char*)(ldx + HWCP);
+  for (int i = tid; i < HWCP; i += 256) ldx[i] = 0.f;
 
   const bf16* dyi = dy2 + (long)img * OW * OH * ldy;
   const int CB = C > 32 ? 32 : C;             // n-tile: one (kh,kw) C-slice
@@ -293,21 +304,27 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
               af[kc], bfr, acc[ni], 0, 0, 0);
         }
       }
-      // scatter: D col=lane&15 -> channel, row=(lane>>4)*4+reg -> m
+      // scatter: D col=lane&15 -> channel, row=(lane>>4)*4+reg -> m.
+      // Targets are unique block-wide within this step -> plain RMW.
       const int hb = -PH + kh, wb = -PW + kw;
+      float* tgt[2][4];
       #pragma unroll
       for (int ni = 0; ni < 2; ++ni) {
-        if (ni * 16 >= CB) break;
         const int c = cb + ni * 16 + (lane & 15);
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          if (r >= mok) break;
           const int h = oh4[r] * SH + hb;
           const int w = ow4[r] * SW + wb;
-          if (h >= 0 && h < H && w >= 0 && w < W)
-            atomicAdd(ldx + (h * W + w) * C + c, acc[ni][r]);
+          const bool ok = (ni * 16 < CB) && (r < mok) &&
+                          h >= 0 && h < H && w >= 0 && w < W;
+          tgt[ni][r] = ok ? ldx + (h * W + w) * CP + c : nullptr;
         }
       }
+      #pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        #pragma unroll
+        for (int r = 0; r < 4; ++r)
+          if (tgt[ni][r]) *tgt[ni][r] += acc[ni][r];
       __syncthreads();                        // scatter done before overwrite
       write_tile(buf ^ 1, stg);
     }
@@ -315,9 +332,11 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
   __syncthreads();
   bf16* dxi = dx + (long)img * HWC;
   for (int i = tid * 8; i < HWC; i += 256 * 8) {
+    const int hw = i / C, c0 = i - hw * C;  // C%8==0: chunk stays in-row
+    const float* src = ldx + hw * CP + c0;
     bf16x8 v;
     #pragma unroll
-    for (int j = 0; j < 8; ++j) v[j] = tobf16(ldx[i + j]);
+    for (int j = 0; j < 8; ++j) v[j] = tobf16(src[j]);
     *(bf16x8*)(dxi + i) = v;
   }
 }
@@ -327,10 +346,11 @@ bool launch_conv_dx(const void* dy2, long ldy, const void* wt, long ldw,
                     int SH, int SW, int PH, int PW, int OH, int OW, int outC,
                     hipStream_t s) {
   const int HWC = H * W * C;
-  if ((long)HWC * 4 > 48 * 1024 || (C % 32 != 0 && C != 16) || HWC % 8 != 0 ||
-      outC > 64 || outC % 8 != 0)
+  const int HWCP = H * W * (C + 1);
+  if ((long)HWCP * 4 > 56 * 1024 || (C % 32 != 0 && C != 16) || HWC % 8 != 0 ||
+      outC > 64 || outC % 8 != 0 || C % 8 != 0)
     return false;
-  hipLaunchKernelGGL(conv_dx_kernel, dim3(B), dim3(256), HWC * 4 + 8192, s,
+  hipLaunchKernelGGL(conv_dx_kernel, dim3(B), dim3(256), HWCP * 4 + 8192, s,
                      (const bf16*)dy2, ldy, (const bf16*)wt, ldw, (bf16*)dx,
                      H, W, C, KH, KW, SH, SW, PH, PW, OH, OW, outC);
   return true;
