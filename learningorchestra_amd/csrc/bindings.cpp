@@ -32,6 +32,11 @@ void launch_mfma_probe(const void* A, const void* B, float* D, hipStream_t s);
 void launch_im2col(const void* in, void* col, int B, int H, int W, int C,
                    int KH, int KW, int SH, int SW, int PH, int PW,
                    int OH, int OW, int Kpad, hipStream_t s);
+bool launch_conv_fwd_small(const void* x, const void* w, long ldw,
+                           const void* bias, void* y, long ldy, int B, int H,
+                           int W, int C, int KH, int KW, int SH, int SW,
+                           int PH, int PW, int OH, int OW, int outC, int relu,
+                           hipStream_t s);
 bool launch_conv_dx(const void* dy2, long ldy, const void* wt, long ldw,
                     void* dx, int B, int H, int W, int C, int KH, int KW,
                     int SH, int SW, int PH, int PW, int OH, int OW, int outC,
@@ -219,6 +224,33 @@ at::Tensor col2im(at::Tensor dcol, int64_t B, int64_t H, int64_t W, int64_t C,
                     (int)KH, (int)KW, (int)SH, (int)SW, (int)PH, (int)PW, OH, OW,
                     (int)dcol.size(1), stream());
   return dx;
+}
+
+// small-image fused conv forward: x image staged in LDS, w tiles
+// double-buffered, y written once.  Returns false if shape not eligible.
+bool conv_fwd_small(at::Tensor x, at::Tensor w,
+                    c10::optional<at::Tensor> bias, at::Tensor y, int64_t KH,
+                    int64_t KW, int64_t SH, int64_t SW, int64_t PH,
+                    int64_t PW, bool relu) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  check_bf16(y, "y");
+  const int B = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2),
+            C = (int)x.size(3);
+  const int OH = (H + 2 * (int)PH - (int)KH) / (int)SH + 1;
+  const int OW = (W + 2 * (int)PW - (int)KW) / (int)SW + 1;
+  const int outC = (int)y.size(1);
+  TORCH_CHECK(y.size(0) == (long)B * OH * OW, "y rows");
+  TORCH_CHECK(w.size(0) == outC && w.size(1) >= (long)KH * KW * C, "w shape");
+  const void* bp = nullptr;
+  if (bias) {
+    TORCH_CHECK(bias->scalar_type() == at::kFloat && bias->numel() >= outC);
+    bp = bias->data_ptr();
+  }
+  return lo::launch_conv_fwd_small(
+      x.data_ptr(), w.data_ptr(), w.stride(0), bp, y.data_ptr(), y.stride(0),
+      B, H, W, C, (int)KH, (int)KW, (int)SH, (int)SW, (int)PH, (int)PW, OH,
+      OW, outC, relu ? 1 : 0, stream());
 }
 
 // fused conv dX: dy2 [B*OH*OW, outC] @ wt[kpad, outC]^T scattered into
@@ -489,6 +521,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_conv_dw", &gemm_conv_dw);
   m.def("mfma_probe", &mfma_probe);
   m.def("im2col", &im2col);
+  m.def("conv_fwd_small", &conv_fwd_small, "small-image fused conv fwd");
   m.def("conv_dx", &conv_dx, "fused conv dX (LDS-accumulated scatter)");
   m.def("col2im", &col2im);
   m.def("maxpool_fwd", &maxpool_fwd);

@@ -110,6 +110,14 @@ class Conv2dNHWC(Layer):
                     self._bufs["dcol"] = torch.empty((M, self.kpad), device=dev, dtype=dtype)
         return self._bufs
 
+    def _fwd_small_ok(self, H: int, W: int) -> bool:
+        # mirror of launch_conv_fwd_small eligibility (x image fits LDS)
+        return (H * W * (self.in_c * 2 + 16) + 8192 <= 56 * 1024 and
+                (self.in_c % 32 == 0 or self.in_c == 16) and
+                self.out_c <= 64 and self.out_c % 16 == 0 and
+                (self.kh * self.kw * self.in_c) % 32 == 0 and
+                (H * W * self.in_c) % 8 == 0)
+
     def _dx_fused_ok(self, H: int, W: int) -> bool:
         # mirror of launch_conv_dx eligibility: whole-image dx fits LDS
         hwc = H * W * self.in_c
@@ -143,11 +151,17 @@ class Conv2dNHWC(Layer):
             F.gemm(bufs["col"], self.arena.p(self.name + ".w"), tb=True,
                    bias=bias, relu=self.relu, out=bufs["y"], stats=st)
         elif x.is_cuda and self.implicit:
-            # implicit conv: im2col gathered inside the GEMM staging
-            F.conv2d_fwd_implicit(x, self.arena.p(self.name + ".w"), self.kh,
-                                  self.kw, self.stride, self.stride, self.pad,
-                                  self.pad, bias=bias, relu=self.relu,
-                                  out=bufs["y"])
+            # implicit conv, small-image fused kernel when the x image fits
+            # LDS; im2col gathered inside the GEMM staging otherwise
+            if not (self._fwd_small_ok(H, W) and F.conv2d_fwd_small(
+                    x, self.arena.p(self.name + ".w"), self.kh, self.kw,
+                    self.stride, self.stride, self.pad, self.pad, bias=bias,
+                    relu=self.relu, out=bufs["y"])):
+                F.conv2d_fwd_implicit(x, self.arena.p(self.name + ".w"),
+                                      self.kh, self.kw, self.stride,
+                                      self.stride, self.pad, self.pad,
+                                      bias=bias, relu=self.relu,
+                                      out=bufs["y"])
             st = None
         else:
             F.im2col(x, self.kh, self.kw, self.stride, self.stride, self.pad,

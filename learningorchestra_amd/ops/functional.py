@@ -109,6 +109,20 @@ def col2im(dcol: torch.Tensor, B: int, H: int, W: int, C: int, kh: int, kw: int,
     return out
 
 
+def conv2d_fwd_small(x: torch.Tensor, w: torch.Tensor, kh: int, kw: int,
+                     sh: int, sw: int, ph: int, pw: int,
+                     bias: Optional[torch.Tensor] = None, relu: bool = False,
+                     out: Optional[torch.Tensor] = None) -> bool:
+    """GPU-only small-image fused conv forward (x image LDS-resident, w
+    tiles double-buffered): pure read-x + write-y traffic.  Returns False
+    when not eligible (H*W*(2C+16)+8K > 56KB LDS, outC > 64, ...)."""
+    if not x.is_cuda:
+        return False
+    lo = require_ext()
+    return bool(lo.conv_fwd_small(x, w, bias, out, kh, kw, sh, sw, ph, pw,
+                                  relu))
+
+
 def conv2d_dx_fused(dy2: torch.Tensor, wt: torch.Tensor, B: int, H: int,
                     W: int, C: int, kh: int, kw: int, sh: int, sw: int,
                     ph: int, pw: int, out: torch.Tensor) -> bool:

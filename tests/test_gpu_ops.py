@@ -300,3 +300,33 @@ def test_conv_dx_fused_rejects_oversize():
     dx = torch.empty(2, 64, 64, 128, device="cuda", dtype=torch.bfloat16)
     assert not F.conv2d_dx_fused(dy2, wt, 2, 64, 64, 128, 3, 3, 1, 1, 0, 0,
                                  out=dx)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [
+    # (B, H, W, C, KH, KW, S, P, outC)
+    (4, 12, 12, 32, 5, 5, 1, 0, 64),     # MNIST conv2 geometry
+    (3, 12, 12, 32, 5, 5, 1, 2, 64),     # padding
+    (2, 16, 16, 16, 3, 3, 1, 1, 48),     # C=16
+    (2, 15, 15, 16, 3, 3, 2, 1, 32),     # stride 2, R not %64
+])
+def test_conv_fwd_small_matches_im2col_gemm(shape):
+    """Small-image fused conv fwd vs im2col + GEMM."""
+    from learningorchestra_amd.ops import functional as F
+    B, H, W, C, KH, KW, S, P, outC = shape
+    torch.manual_seed(1)
+    OH = (H + 2 * P - KH) // S + 1
+    OW = (W + 2 * P - KW) // S + 1
+    kdim = KH * KW * C
+    kpad = (kdim + 7) // 8 * 8
+    x = torch.randn(B, H, W, C, device="cuda").to(torch.bfloat16)
+    w = torch.zeros(outC, kpad, device="cuda", dtype=torch.bfloat16)
+    w[:, :kdim] = torch.randn(outC, kdim, device="cuda").to(torch.bfloat16) * 0.1
+    bias = torch.randn(outC, device="cuda", dtype=torch.float32)
+    y = torch.empty(B * OH * OW, outC, device="cuda", dtype=torch.bfloat16)
+    ok = F.conv2d_fwd_small(x, w, KH, KW, S, S, P, P, bias=bias, relu=True,
+                            out=y)
+    assert ok, "shape should be eligible"
+    col = F.im2col(x, KH, KW, S, S, P, P, kpad)
+    ref = F.gemm(col, w, tb=True, bias=bias, relu=True)
+    torch.testing.assert_close(y.float(), ref.float(), atol=2e-2, rtol=2e-2)
