@@ -390,7 +390,7 @@ __global__ __launch_bounds__(256) void conv_fwd_small_kernel(
   }
 
   const int NC = (KH * KW * C) / 32;          // 32-k chunks (C%32==0 or C==16)
-  const int srow = tid >> 3, skc = tid & 7;   // w-tile staging slot
+  const int srow = tid >> 2, skc = tid & 3;   // w-tile slot: 64 rows x 4x16B
   auto fetch_w = [&](int chunk) -> bf16x8 {
     bf16x8 v = {};
     if (chunk < NC && srow < outC)
