@@ -307,8 +307,9 @@ def test_conv_dx_fused_rejects_oversize():
     # (B, H, W, C, KH, KW, S, P, outC)
     (4, 12, 12, 32, 5, 5, 1, 0, 64),     # MNIST conv2 geometry
     (3, 12, 12, 32, 5, 5, 1, 2, 64),     # padding
-    (2, 16, 16, 16, 3, 3, 1, 1, 48),     # C=16
-    (2, 15, 15, 16, 3, 3, 2, 1, 32),     # stride 2, R not %64
+    (2, 16, 16, 16, 2, 2, 1, 0, 48),     # C=16 (kdim 64)
+    (2, 15, 15, 16, 2, 2, 2, 1, 32),     # stride 2, R not %64
+    (2, 18, 18, 64, 5, 5, 1, 2, 64),     # C=64, R=324 (multi-chunk + tail)
 ])
 def test_conv_fwd_small_matches_im2col_gemm(shape):
     """Small-image fused conv fwd vs im2col + GEMM."""
