@@ -47,7 +47,8 @@ void launch_col2im(const void* dcol, void* dx, int B, int H, int W, int C,
 void launch_maxpool_fwd(const void* in, void* out, void* idx, int B, int H, int W,
                         int C, int KH, int KW, int SH, int SW, int PH, int PW,
                         int OH, int OW, hipStream_t s);
-void launch_maxpool_bwd(const void* dy, const void* idx, void* dx, int B, int H,
+void launch_maxpool_bwd(const void* dy, const void* idx, void* dx,
+                        const void* relu_y, int B, int H,
                         int W, int C, int KH, int KW, int SH, int SW, int PH,
                         int PW, int OH, int OW, hipStream_t s);
 void launch_bn_stats(const void* x, void* sum, void* sumsq, long M, int C,
@@ -291,12 +292,19 @@ std::vector<at::Tensor> maxpool_fwd(at::Tensor in, int64_t KH, int64_t KW,
 
 at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor idx, int64_t H, int64_t W,
                        int64_t KH, int64_t KW, int64_t SH, int64_t SW,
-                       int64_t PH, int64_t PW, at::Tensor dx) {
+                       int64_t PH, int64_t PW, at::Tensor dx,
+                       c10::optional<at::Tensor> relu_y) {
   check_bf16(dy, "dy");
   check_bf16(dx, "dx");
   const int B = (int)dy.size(0), OH = (int)dy.size(1), OW = (int)dy.size(2),
             C = (int)dy.size(3);
-  lo::launch_maxpool_bwd(dy.data_ptr(), idx.data_ptr(), dx.data_ptr(), B, (int)H,
+  const void* yp = nullptr;
+  if (relu_y) {
+    check_bf16(*relu_y, "relu_y");
+    TORCH_CHECK(relu_y->numel() == dx.numel(), "relu_y shape");
+    yp = relu_y->data_ptr();
+  }
+  lo::launch_maxpool_bwd(dy.data_ptr(), idx.data_ptr(), dx.data_ptr(), yp, B, (int)H,
                          (int)W, C, (int)KH, (int)KW, (int)SH, (int)SW,
                          (int)PH, (int)PW, OH, OW, stream());
   return dx;

@@ -541,10 +541,14 @@ __global__ void maxpool_fwd_kernel(const bf16* __restrict__ in, bf16* __restrict
 
 // backward: per input element, sum dY of covering outputs whose argmax is it
 // (with SH==KH/SW==KW pooling each input has <=1 cover -> no atomics needed)
+// relu_y != null fuses the upstream conv's ReLU backward: dx is zeroed
+// where the conv output (the pool INPUT, same [b,h,w,c] position) was
+// clamped — saves the separate relu_bwd pass over the full conv activation.
 template <bool VEC8>
 __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
                                    const unsigned char* __restrict__ idx,
                                    bf16* __restrict__ dx,
+                                   const bf16* __restrict__ relu_y,
                                    int B, int H, int W, int C, int KH, int KW,
                                    int SH, int SW, int PH, int PW, int OH, int OW) {
   const int CV = VEC8 ? C / 8 : C;
@@ -583,11 +587,18 @@ __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
     }
     const long ibase = (((long)b * H + h) * W + w) * C + cu * NE;
     if (VEC8) {
+      if (relu_y) {
+        const bf16x8 yv = *(const bf16x8*)(relu_y + ibase);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (!(tofloat(yv[j]) > 0.f)) acc[j] = 0.f;
+      }
       bf16x8 o;
       #pragma unroll
       for (int j = 0; j < 8; ++j) o[j] = tobf16(acc[j]);
       *(bf16x8*)(dx + ibase) = o;
     } else {
+      if (relu_y && !(tofloat(relu_y[ibase]) > 0.f)) acc[0] = 0.f;
       dx[ibase] = tobf16(acc[0]);
     }
   }
@@ -681,7 +692,8 @@ void launch_maxpool_fwd(const void* in, void* out, void* idx, int B, int H, int 
                        B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
 }
 
-void launch_maxpool_bwd(const void* dy, const void* idx, void* dx, int B, int H,
+void launch_maxpool_bwd(const void* dy, const void* idx, void* dx,
+                        const void* relu_y, int B, int H,
                         int W, int C, int KH, int KW, int SH, int SW, int PH,
                         int PW, int OH, int OW, hipStream_t s) {
   const bool vec = (C % 8 == 0);
@@ -691,11 +703,11 @@ void launch_maxpool_bwd(const void* dy, const void* idx, void* dx, int B, int H,
   if (vec)
     hipLaunchKernelGGL(HIP_KERNEL_NAME(maxpool_bwd_kernel<true>), dim3(grid), dim3(block), 0, s,
                        (const bf16*)dy, (const unsigned char*)idx, (bf16*)dx,
-                       B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
+                       (const bf16*)relu_y, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
   else
     hipLaunchKernelGGL(HIP_KERNEL_NAME(maxpool_bwd_kernel<false>), dim3(grid), dim3(block), 0, s,
                        (const bf16*)dy, (const unsigned char*)idx, (bf16*)dx,
-                       B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
+                       (const bf16*)relu_y, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
 }
 
 }  // namespace lo

@@ -171,11 +171,14 @@ class Conv2dNHWC(Layer):
         self.stats_filled = st is not None
         return bufs["y"].view(B, bufs["OH"], bufs["OW"], self.out_c)
 
+    # True when the following pool's backward already applied our ReLU mask
+    relu_bwd_upstream = False
+
     def backward(self, dy: torch.Tensor) -> Optional[torch.Tensor]:
         bufs = self._bufs
         M = bufs["y"].shape[0]
         dy2 = dy.reshape(M, self.out_c)
-        if self.relu:
+        if self.relu and not self.relu_bwd_upstream:
             F.relu_bwd(dy2, bufs["y"], out=dy2)
         # weight grad: dW[outC, kpad] = dY^T @ col, split-K when M is deep
         gw = self.arena.g(self.name + ".w")
@@ -226,11 +229,18 @@ class MaxPool2dNHWC(Layer):
         self.sh = sh if sh is not None else kh
         self.sw = sw if sw is not None else kw
 
+    # set by SequentialClassifier when the preceding layer is a ReLU conv:
+    # the conv's ReLU backward mask folds into this kernel (relu_y = the
+    # pool input) and the conv skips its own relu_bwd pass.
+    fuse_relu = False
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         self._bufs["shape"] = x.shape
         out, idx = F.maxpool2d(x, self.kh, self.kw, self.sh, self.sw,
                                self.ph, self.pw)
         self._bufs["idx"] = idx
+        if self.fuse_relu:
+            self._x = x
         if "dx" not in self._bufs or self._bufs["dx"].shape != x.shape:
             self._bufs["dx"] = torch.empty_like(x)
         return out
@@ -239,7 +249,8 @@ class MaxPool2dNHWC(Layer):
         B, H, W, C = self._bufs["shape"]
         return F.maxpool2d_bwd(dy, self._bufs["idx"], H, W, self.kh, self.kw,
                                self.sh, self.sw, self.ph, self.pw,
-                               out=self._bufs["dx"])
+                               out=self._bufs["dx"],
+                               relu_y=self._x if self.fuse_relu else None)
 
 
 class Flatten(Layer):
@@ -323,6 +334,12 @@ class SequentialClassifier:
                  seed: int = 0):
         self.layers = layers
         self.num_classes = num_classes
+        # fuse conv-ReLU backward into the following maxpool's backward
+        for prev, nxt in zip(layers, layers[1:]):
+            if (isinstance(prev, Conv2dNHWC) and prev.relu
+                    and isinstance(nxt, MaxPool2dNHWC)):
+                nxt.fuse_relu = True
+                prev.relu_bwd_upstream = True
         self.arena = ParamArena(device)
         for lay in layers:
             lay.build(self.arena)

@@ -166,13 +166,17 @@ def maxpool2d(x: torch.Tensor, kh: int, kw: int, sh: int, sw: int,
 
 def maxpool2d_bwd(dy: torch.Tensor, idx: torch.Tensor, H: int, W: int,
                   kh: int, kw: int, sh: int, sw: int, ph: int = 0, pw: int = 0,
-                  out: Optional[torch.Tensor] = None) -> torch.Tensor:
+                  out: Optional[torch.Tensor] = None,
+                  relu_y: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """``relu_y`` (the pool input = upstream conv's ReLU output, same
+    [B,H,W,C]) fuses that conv's ReLU backward: dx is zeroed where
+    relu_y <= 0, replacing a separate relu_bwd pass."""
     B, OH, OW, C = dy.shape
     if out is None:
         out = torch.empty((B, H, W, C), device=dy.device, dtype=dy.dtype)
     if _is_gpu(dy):
         lo = require_ext()
-        lo.maxpool_bwd(dy, idx, H, W, kh, kw, sh, sw, ph, pw, out)
+        lo.maxpool_bwd(dy, idx, H, W, kh, kw, sh, sw, ph, pw, out, relu_y)
         return out
     dx = torch.zeros((B, H, W, C), dtype=torch.float32)
     rel = idx.long()
@@ -185,6 +189,8 @@ def maxpool2d_bwd(dy: torch.Tensor, idx: torch.Tensor, H: int, W: int,
     c = torch.arange(C).view(1, 1, 1, C).expand_as(rel)
     dx.index_put_((b.reshape(-1), hsrc.reshape(-1), wsrc.reshape(-1),
                    c.reshape(-1)), dy.float().reshape(-1), accumulate=True)
+    if relu_y is not None:
+        dx = dx * (relu_y.float() > 0)
     out.copy_(dx.to(out.dtype))
     return out
 
