@@ -365,6 +365,11 @@ bool launch_conv_dx(const void* dy2, long ldy, const void* wt, long ldw,
 // read-w(L2) + write-y.  Replaces im2col+GEMM / gather-GEMM for shapes
 // like MNIST conv2 (measured: the gather GEMM re-reads x through L2 per
 // (kh,kw) and ran at ~1.3 ms vs ~0.15 ms roofline for B=32768).
+// C1 = single-input-channel specialization (e.g. a 28x28x1 MNIST conv1):
+// x is staged unpadded (H*W bf16) and the A fragment is gathered with 8
+// scalar LDS reads per lane (the k-run spans (kh,kw) cells); kpad <= 32 so
+// the whole w fits one 32-k chunk.
+template <bool C1>
 __global__ __launch_bounds__(256) void conv_fwd_small_kernel(
     const bf16* __restrict__ x,               // [B, H, W, C]
     const bf16* __restrict__ w, long ldw,     // [outC, kpad] row-major
@@ -377,19 +382,20 @@ __global__ __launch_bounds__(256) void conv_fwd_small_kernel(
   const int R = OH * OW;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wave = tid >> 6;
-  const int XROW = C * 2 + 16;                // padded x row bytes
+  const int XROW = C1 ? 2 : C * 2 + 16;       // padded x row bytes
   char* smX = sm;
   char* smW = sm + H * W * XROW;
 
-  // stage x: thread t copies 16-B chunks (C*2 % 16 == 0 by eligibility)
+  // stage x: thread t copies 16-B chunks (C*2 % 16 == 0 by eligibility;
+  // C1: H*W % 8 == 0)
   const int HWC = H * W * C;
   for (int i = tid * 8; i < HWC; i += 256 * 8) {
-    const int hw = i / C, c0 = i - hw * C;
+    const int hw = C1 ? i : i / C, c0 = C1 ? 0 : i - (i / C) * C;
     *(bf16x8*)(smX + hw * XROW + c0 * 2) =
         *(const bf16x8*)(x + (long)img * HWC + i);
   }
 
-  const int NC = (KH * KW * C) / 32;          // 32-k chunks (C%32==0 or C==16)
+  const int NC = C1 ? 1 : (KH * KW * C) / 32; // 32-k chunks
   const int srow = tid >> 2, skc = tid & 3;   // w-tile slot: 64 rows x 4x16B
   auto fetch_w = [&](int chunk) -> bf16x8 {
     bf16x8 v = {};
@@ -418,7 +424,17 @@ __global__ __launch_bounds__(256) void conv_fwd_small_kernel(
       if (chunk) __syncthreads();
       // A fragment: k = chunk*32 + (lane>>4)*8 decoded to (kh,kw,c)
       bf16x8 af = {};
-      {
+      if (C1) {
+        // 8 scalar gathers: each k is its own (kh,kw) cell
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int k = (lane >> 4) * 8 + j;
+          const int kh = k / KW, kw = k - kh * KW;
+          const int h = aoh * SH - PH + kh, wx = aow * SW - PW + kw;
+          if (arow < R && k < KH * KW && h >= 0 && h < H && wx >= 0 && wx < W)
+            af[j] = *(const bf16*)(smX + (h * W + wx) * 2);
+        }
+      } else {
         const int k = chunk * 32 + (lane >> 4) * 8;
         const int kc = k / C, c = k - kc * C;
         const int kh = kc / KW, kw = kc - kh * KW;
@@ -473,11 +489,21 @@ bool launch_conv_fwd_small(const void* x, const void* w, long ldw,
                            int W, int C, int KH, int KW, int SH, int SW,
                            int PH, int PW, int OH, int OW, int outC, int relu,
                            hipStream_t s) {
+  if (C == 1 && KH * KW <= 32 && ldw >= 32 && outC <= 64 && outC % 16 == 0 &&
+      (H * W) % 8 == 0 && H * W * 2 + 8192 <= 56 * 1024) {
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(conv_fwd_small_kernel<true>),
+                       dim3(B), dim3(256), H * W * 2 + 8192, s,
+                       (const bf16*)x, (const bf16*)w, ldw, (const float*)bias,
+                       (bf16*)y, ldy, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW,
+                       outC, relu);
+    return true;
+  }
   const int lds = H * W * (C * 2 + 16) + 8192;
   if (lds > 56 * 1024 || (C % 32 != 0 && C != 16) || outC > 64 ||
       outC % 16 != 0 || (KH * KW * C) % 32 != 0 || (H * W * C) % 8 != 0)
     return false;
-  hipLaunchKernelGGL(conv_fwd_small_kernel, dim3(B), dim3(256), lds, s,
+  hipLaunchKernelGGL(HIP_KERNEL_NAME(conv_fwd_small_kernel<false>),
+                     dim3(B), dim3(256), lds, s,
                      (const bf16*)x, (const bf16*)w, ldw, (const float*)bias,
                      (bf16*)y, ldy, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW,
                      outC, relu);

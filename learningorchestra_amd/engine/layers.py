@@ -112,6 +112,10 @@ class Conv2dNHWC(Layer):
 
     def _fwd_small_ok(self, H: int, W: int) -> bool:
         # mirror of launch_conv_fwd_small eligibility (x image fits LDS)
+        if (self.in_c == 1 and self.kh * self.kw <= 32 and self.kpad >= 32
+                and self.out_c <= 64 and self.out_c % 16 == 0
+                and (H * W) % 8 == 0 and H * W * 2 + 8192 <= 56 * 1024):
+            return True
         return (H * W * (self.in_c * 2 + 16) + 8192 <= 56 * 1024 and
                 (self.in_c % 32 == 0 or self.in_c == 16) and
                 self.out_c <= 64 and self.out_c % 16 == 0 and

@@ -310,6 +310,8 @@ def test_conv_dx_fused_rejects_oversize():
     (2, 16, 16, 16, 2, 2, 1, 0, 48),     # C=16 (kdim 64)
     (2, 15, 15, 16, 2, 2, 2, 1, 32),     # stride 2, R not %64
     (2, 18, 18, 64, 5, 5, 1, 2, 64),     # C=64, R=324 (multi-chunk + tail)
+    (4, 28, 28, 1, 5, 5, 1, 0, 32),      # C=1 scalar-gather (MNIST conv1)
+    (3, 28, 28, 1, 5, 5, 1, 2, 64),      # C=1 with padding
 ])
 def test_conv_fwd_small_matches_im2col_gemm(shape):
     """Small-image fused conv fwd vs im2col + GEMM."""
