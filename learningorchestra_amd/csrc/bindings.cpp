@@ -37,6 +37,10 @@ bool launch_conv_fwd_small(const void* x, const void* w, long ldw,
                            int W, int C, int KH, int KW, int SH, int SW,
                            int PH, int PW, int OH, int OW, int outC, int relu,
                            hipStream_t s);
+bool launch_conv_dw_c1(const void* dy2, long ldy, const void* x, void* dw,
+                       long ldw, int B, int H, int W, int KH, int KW, int SH,
+                       int SW, int PH, int PW, int OH, int OW, int outC,
+                       hipStream_t s);
 bool launch_conv_dx(const void* dy2, long ldy, const void* wt, long ldw,
                     void* dx, int B, int H, int W, int C, int KH, int KW,
                     int SH, int SW, int PH, int PW, int OH, int OW, int outC,
@@ -252,6 +256,27 @@ bool conv_fwd_small(at::Tensor x, at::Tensor w,
       x.data_ptr(), w.data_ptr(), w.stride(0), bp, y.data_ptr(), y.stride(0),
       B, H, W, C, (int)KH, (int)KW, (int)SH, (int)SW, (int)PH, (int)PW, OH,
       OW, outC, relu ? 1 : 0, stream());
+}
+
+// C=1 conv dW: dY^T @ im2col(x) with x images LDS-resident; dw fp32
+// accumulated with one atomicAdd per element per block (dw zeroed here).
+bool conv_dw_c1(at::Tensor dy2, at::Tensor x, at::Tensor dw, int64_t KH,
+                int64_t KW, int64_t SH, int64_t SW, int64_t PH, int64_t PW) {
+  check_bf16(dy2, "dy2");
+  check_bf16(x, "x");
+  TORCH_CHECK(dw.scalar_type() == at::kFloat, "dw must be fp32");
+  const int B = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2);
+  TORCH_CHECK(x.size(3) == 1, "conv_dw_c1 needs C=1");
+  const int OH = (H + 2 * (int)PH - (int)KH) / (int)SH + 1;
+  const int OW = (W + 2 * (int)PW - (int)KW) / (int)SW + 1;
+  const int outC = (int)dy2.size(1);
+  TORCH_CHECK(dy2.size(0) == (long)B * OH * OW, "dy2 rows");
+  TORCH_CHECK(dw.size(0) == outC && dw.size(1) >= KH * KW, "dw shape");
+  dw.zero_();
+  return lo::launch_conv_dw_c1(dy2.data_ptr(), dy2.stride(0), x.data_ptr(),
+                               dw.data_ptr(), dw.stride(0), B, H, W, (int)KH,
+                               (int)KW, (int)SH, (int)SW, (int)PH, (int)PW,
+                               OH, OW, outC, stream());
 }
 
 // fused conv dX: dy2 [B*OH*OW, outC] @ wt[kpad, outC]^T scattered into
@@ -530,6 +555,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe);
   m.def("im2col", &im2col);
   m.def("conv_fwd_small", &conv_fwd_small, "small-image fused conv fwd");
+  m.def("conv_dw_c1", &conv_dw_c1, "C=1 conv dW (x LDS-resident)");
   m.def("conv_dx", &conv_dx, "fused conv dX (LDS-accumulated scatter)");
   m.def("col2im", &col2im);
   m.def("maxpool_fwd", &maxpool_fwd);

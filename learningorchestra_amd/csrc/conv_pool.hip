@@ -510,6 +510,131 @@ bool launch_conv_fwd_small(const void* x, const void* w, long ldw,
   return true;
 }
 
+// ----------------------------------------------------- C=1 conv dW ---------
+// dW[outC, kpad] = dY^T @ im2col(x) for single-channel convs (MNIST conv1):
+// the reduction dim is the 18.9M-row batch.  Each block owns a group of
+// images: x image (H*W bf16, ~1.6 KB) is LDS-resident; each wave stages its
+// own 64x32 dy chunk TRANSPOSED into a private LDS buffer (wave-coherent,
+// no barrier) and gathers the B operand (col) with scalar LDS reads from x
+// (for C=1 consecutive rows are consecutive x addresses, but row wraps make
+// a b128 read unsafe).  Per-block partial dW is reduced across waves in LDS
+// and added to global fp32 dW with one atomicAdd per element.
+__global__ __launch_bounds__(256) void conv_dw_c1_kernel(
+    const bf16* __restrict__ dy2, long ldy,   // [B*OH*OW, outC]
+    const bf16* __restrict__ x,               // [B, H, W, 1]
+    float* __restrict__ dw, long ldw,         // [outC, kpad] fp32, pre-zeroed
+    int B, int H, int W, int KH, int KW, int SH, int SW, int PH, int PW,
+    int OH, int OW, int outC, int imgs_per_block) {
+  extern __shared__ char sm[];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const int R = OH * OW;
+  const int HW = H * W;
+  char* smX = sm;                              // HW*2 bytes (16-B aligned)
+  char* smT = sm + ((HW * 2 + 127) & ~127) + wave * 4096;  // 32 x 64 dyT
+
+  const int i0 = blockIdx.x * imgs_per_block;
+  const int i1 = min(B, i0 + imgs_per_block);
+  f32x4 acc[2][2] = {};                        // [mfrag(outC 32)][nfrag(k 32)]
+  for (int img = i0; img < i1; ++img) {
+    __syncthreads();                           // previous x uses done
+    for (int i = tid * 8; i < HW; i += 256 * 8)
+      *(bf16x8*)(smX + i * 2) = *(const bf16x8*)(x + (long)img * HW + i);
+    __syncthreads();
+    const bf16* dyi = dy2 + (long)img * R * ldy;
+    for (int m0 = wave * 64; m0 < R; m0 += 4 * 64) {
+      // stage dyT: wave reads dy[m0+r][o-run] and writes smT[o][m] —
+      // 4 runs per lane ((64 rows x 32 cols)/8 = 256 runs)
+      #pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        const int run = t * 64 + lane;         // 0..255
+        const int m = run >> 2, o0 = (run & 3) * 8;
+        bf16x8 v = {};
+        if (m0 + m < R && o0 < outC)
+          v = *(const bf16x8*)(dyi + (long)(m0 + m) * ldy + o0);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+          *(bf16*)(smT + (o0 + j) * 128 + ((m * 2) ^ (((o0 + j) & 7) << 4))) = v[j];
+      }
+      // A[o][m-run]: o = mi*16 + lane%16, m = (lane/16)*8 (+16 per kc)
+      __asm__ volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      #pragma unroll
+      for (int kc = 0; kc < 2; ++kc) {
+        bf16x8 af[2];
+        #pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+          const int o = mi * 16 + (lane & 15);
+          const int mb = (kc * 32 + (lane >> 4) * 8) * 2;
+          af[mi] = *(const bf16x8*)(smT + o * 128 + (mb ^ ((o & 7) << 4)));
+        }
+        #pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          // B[kcol][m-run] gathered from the x image
+          bf16x8 bfr = {};
+          const int kcol = ni * 16 + (lane & 15);
+          const int kh = kcol / KW, kw = kcol - kh * KW;
+          if (kcol < KH * KW) {
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              const int m = m0 + kc * 32 + (lane >> 4) * 8 + j;
+              const int oh = m / OW, ow = m - oh * OW;
+              const int h = oh * SH - PH + kh, wx = ow * SW - PW + kw;
+              if (m < R && h >= 0 && h < H && wx >= 0 && wx < W)
+                bfr[j] = *(const bf16*)(smX + (h * W + wx) * 2);
+            }
+          }
+          #pragma unroll
+          for (int mi = 0; mi < 2; ++mi)
+            acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[mi], bfr, acc[mi][ni], 0, 0, 0);
+        }
+      }
+    }
+  }
+  // cross-wave reduce in LDS (reuse the dyT area) then one atomic per elem.
+  // D layout: row(o) = mi*16 + (lane>>4)*4 + reg, col(k) = ni*16 + (lane&15)
+  __syncthreads();
+  float* red = (float*)sm;                     // 32x32 fp32 = 4 KB
+  for (int wsel = 0; wsel < 4; ++wsel) {
+    if (wave == wsel) {
+      #pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+        #pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int o = mi * 16 + (lane >> 4) * 4 + r;
+            const int k = ni * 16 + (lane & 15);
+            if (wsel == 0) red[o * 32 + k] = acc[mi][ni][r];
+            else red[o * 32 + k] += acc[mi][ni][r];
+          }
+    }
+    __syncthreads();
+  }
+  for (int i = tid; i < 32 * 32; i += 256) {
+    const int o = i >> 5, k = i & 31;
+    if (o < outC && k < KH * KW && red[i] != 0.f)
+      atomicAdd(dw + (long)o * ldw + k, red[i]);
+  }
+}
+
+bool launch_conv_dw_c1(const void* dy2, long ldy, const void* x, void* dw,
+                       long ldw, int B, int H, int W, int KH, int KW, int SH,
+                       int SW, int PH, int PW, int OH, int OW, int outC,
+                       hipStream_t s) {
+  if (outC > 32 || outC % 8 != 0 || KH * KW > 32 || (H * W) % 8 != 0 ||
+      H * W * 2 > 16 * 1024)
+    return false;
+  // ~2048 blocks fill the chip; each handles a contiguous image group
+  const int ipb = max(1, (B + 2047) / 2048);
+  const int blocks = (B + ipb - 1) / ipb;
+  const int lds = ((H * W * 2 + 127) & ~127) + 4 * 4096;
+  hipLaunchKernelGGL(conv_dw_c1_kernel, dim3(blocks), dim3(256), lds, s,
+                     (const bf16*)dy2, ldy, (const bf16*)x, (float*)dw, ldw,
+                     B, H, W, KH, KW, SH, SW, PH, PW, OH, OW, outC, ipb);
+  return true;
+}
+
 // ------------------------------------------------------------- maxpool -----
 template <bool VEC8>
 __global__ void maxpool_fwd_kernel(const bf16* __restrict__ in, bf16* __restrict__ out,

@@ -188,9 +188,12 @@ class Conv2dNHWC(Layer):
         gw = self.arena.g(self.name + ".w")
         splits = _splitk_heuristic(self.out_c, self.kpad, M)
         if dy2.is_cuda and self.implicit and not self._is_1x1:
-            F.conv2d_dw_implicit(dy2, self._x, gw, self.kh, self.kw,
-                                 self.stride, self.stride, self.pad, self.pad,
-                                 splits)
+            if not (self.in_c == 1 and self.out_c <= 32 and F.conv2d_dw_c1(
+                    dy2, self._x, gw, self.kh, self.kw, self.stride,
+                    self.stride, self.pad, self.pad)):
+                F.conv2d_dw_implicit(dy2, self._x, gw, self.kh, self.kw,
+                                     self.stride, self.stride, self.pad,
+                                     self.pad, splits)
         else:
             F.gemm(dy2, bufs["col"], ta=True, out=gw, splits=splits)
         if self.bias:

@@ -123,6 +123,17 @@ def conv2d_fwd_small(x: torch.Tensor, w: torch.Tensor, kh: int, kw: int,
                                   relu))
 
 
+def conv2d_dw_c1(dy2: torch.Tensor, x: torch.Tensor, gw: torch.Tensor,
+                 kh: int, kw: int, sh: int, sw: int, ph: int, pw: int) -> bool:
+    """GPU-only C=1 conv weight grad (dY^T @ im2col(x)) with x images
+    LDS-resident — no col matrix, no split-K pass structure.  gw fp32,
+    zeroed inside.  Returns False when not eligible (outC > 32, ...)."""
+    if not dy2.is_cuda:
+        return False
+    lo = require_ext()
+    return bool(lo.conv_dw_c1(dy2, x, gw, kh, kw, sh, sw, ph, pw))
+
+
 def conv2d_dx_fused(dy2: torch.Tensor, wt: torch.Tensor, B: int, H: int,
                     W: int, C: int, kh: int, kw: int, sh: int, sw: int,
                     ph: int, pw: int, out: torch.Tensor) -> bool:

@@ -333,3 +333,30 @@ def test_conv_fwd_small_matches_im2col_gemm(shape):
     col = F.im2col(x, KH, KW, S, S, P, P, kpad)
     ref = F.gemm(col, w, tb=True, bias=bias, relu=True)
     torch.testing.assert_close(y.float(), ref.float(), atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [
+    # (B, H, W, KH, KW, S, P, outC)
+    (70, 28, 28, 5, 5, 1, 0, 32),        # MNIST conv1 (R=576, group tail)
+    (9, 24, 24, 3, 3, 1, 1, 16),         # padding, outC 16
+    (5, 16, 16, 5, 5, 2, 2, 24),         # stride 2
+])
+def test_conv_dw_c1_matches_splitk(shape):
+    """C=1 fused dW vs im2col + split-K GEMM reference."""
+    from learningorchestra_amd.ops import functional as F
+    B, H, W, KH, KW, S, P, outC = shape
+    torch.manual_seed(2)
+    OH = (H + 2 * P - KH) // S + 1
+    OW = (W + 2 * P - KW) // S + 1
+    kdim = KH * KW
+    kpad = (kdim + 7) // 8 * 8
+    x = torch.randn(B, H, W, 1, device="cuda").to(torch.bfloat16)
+    dy2 = torch.randn(B * OH * OW, outC, device="cuda").to(torch.bfloat16)
+    gw = torch.empty(outC, kpad, device="cuda", dtype=torch.float32)
+    ok = F.conv2d_dw_c1(dy2, x, gw, KH, KW, S, S, P, P)
+    assert ok, "shape should be eligible"
+    col = F.im2col(x, KH, KW, S, S, P, P, kpad)
+    ref = F.gemm(dy2, col, ta=True, splits=4)
+    torch.testing.assert_close(gw[:, :kdim], ref[:, :kdim], atol=2e-1,
+                               rtol=2e-2)
