@@ -23,12 +23,13 @@ CPAD = 16  # classifier head padded for 16-B-aligned GEMM rows
 def build_mnist_cnn(device="cpu", seed: int = 0,
                     channels=(32, 64), fc_width: int = 256) -> SequentialClassifier:
     c1, c2 = channels
-    # conv2 defaults to implicit-GEMM conv (im2col gathered inside the GEMM
-    # staging, no col matrix): measured +5% end-to-end vs materialized col at
-    # B=32768 (PERFORMANCE.md). LO_IMPLICIT_CONV=0 restores the col path;
-    # LO_IMPLICIT_CONV1=1 additionally tries conv1 (C=1, scalar gather).
+    # Both convs default to implicit (no col matrix): conv2 runs the
+    # small-image fused fwd + gather dW, conv1 the C=1 fused fwd + fused dW
+    # (conv_fwd_small / conv_dw_c1) — measured 3.31M -> 5.43M samples/s at
+    # B=32768 over the materialized-col baseline (PERFORMANCE.md).
+    # LO_IMPLICIT_CONV=0 / LO_IMPLICIT_CONV1=0 restore the col paths.
     imp = os.environ.get("LO_IMPLICIT_CONV", "1") == "1"
-    imp1 = os.environ.get("LO_IMPLICIT_CONV1", "0") == "1"
+    imp1 = os.environ.get("LO_IMPLICIT_CONV1", "1") == "1"
     layers = [
         Conv2dNHWC("conv1", 1, c1, 5, 5, relu=True, first=True,
                    implicit=imp1),                                  # 28 -> 24
