@@ -395,6 +395,66 @@ __global__ __launch_bounds__(256) void conv_fwd_small_kernel(
         *(const bf16x8*)(x + (long)img * HWC + i);
   }
 
+  if (C1) {
+    // single w-chunk (kpad <= 32): stage w once, then NO barriers — each
+    // wave computes and stores through its PRIVATE epilogue buffer.
+    {
+      const int srow = tid >> 2, skc = tid & 3;
+      bf16x8 v = {};
+      if (srow < outC)
+        v = *(const bf16x8*)(w + (long)srow * ldw + skc * 8);
+      *(bf16x8*)(smW + srow * 64 +
+                 ((skc * 8) ^ ((srow & 3) << 3) ^ (((srow >> 2) & 3) << 3)) * 2) = v;
+    }
+    __syncthreads();                          // x + w staged
+    char* se = smW + 4096 + wave * 2048;      // private 16x64 bf16
+    for (int m0 = 0; m0 < R; m0 += 64) {
+      const int arow = m0 + wave * 16 + (lane & 15);
+      const int aoh = arow / OW, aow = arow - aoh * OW;
+      bf16x8 af = {};
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int k = (lane >> 4) * 8 + j;
+        const int kh = k / KW, kw = k - kh * KW;
+        const int h = aoh * SH - PH + kh, wx = aow * SW - PW + kw;
+        if (arow < R && k < KH * KW && h >= 0 && h < H && wx >= 0 && wx < W)
+          af[j] = *(const bf16*)(smX + (h * W + wx) * 2);
+      }
+      f32x4 acc[4] = {};
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int row = ni * 16 + (lane & 15);
+        const int k2 = (lane >> 4) * 8;
+        const bf16x8 bfr = *(const bf16x8*)(
+            smW + row * 64 +
+            ((k2 ^ ((row & 3) << 3) ^ (((row >> 2) & 3) << 3)) * 2));
+        acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr, acc[ni],
+                                                          0, 0, 0);
+      }
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int c = ni * 16 + (lane & 15);
+        const float b = bias ? bias[c < outC ? c : 0] : 0.f;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = acc[ni][r] + b;
+          if (relu) v = v > 0.f ? v : 0.f;
+          ((bf16*)se)[((lane >> 4) * 4 + r) * 64 + c] = tobf16(v);
+        }
+      }
+      // wave-private: ds RAW ordering handled by the compiler's waitcnts
+      #pragma unroll
+      for (int p = 0; p < 2; ++p) {
+        const int row = p * 8 + (lane >> 3);
+        const int m = m0 + wave * 16 + row;
+        if (m < R && (lane & 7) * 8 < outC)
+          *(bf16x8*)(y + (long)img * R * ldy + (long)m * ldy + (lane & 7) * 8) =
+              *(const bf16x8*)(se + row * 128 + (lane & 7) * 16);
+      }
+    }
+    return;
+  }
+
   const int NC = C1 ? 1 : (KH * KW * C) / 32; // 32-k chunks
   const int srow = tid >> 2, skc = tid & 3;   // w-tile slot: 64 rows x 4x16B
   auto fetch_w = [&](int chunk) -> bf16x8 {
@@ -490,9 +550,9 @@ bool launch_conv_fwd_small(const void* x, const void* w, long ldw,
                            int PH, int PW, int OH, int OW, int outC, int relu,
                            hipStream_t s) {
   if (C == 1 && KH * KW <= 32 && ldw >= 32 && outC <= 64 && outC % 16 == 0 &&
-      (H * W) % 8 == 0 && H * W * 2 + 8192 <= 56 * 1024) {
+      (H * W) % 8 == 0 && H * W * 2 + 12288 <= 56 * 1024) {
     hipLaunchKernelGGL(HIP_KERNEL_NAME(conv_fwd_small_kernel<true>),
-                       dim3(B), dim3(256), H * W * 2 + 8192, s,
+                       dim3(B), dim3(256), H * W * 2 + 12288, s,
                        (const bf16*)x, (const bf16*)w, ldw, (const float*)bias,
                        (bf16*)y, ldy, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW,
                        outC, relu);
@@ -574,13 +634,15 @@ __global__ __launch_bounds__(256) void conv_dw_c1_kernel(
           const int kcol = ni * 16 + (lane & 15);
           const int kh = kcol / KW, kw = kcol - kh * KW;
           if (kcol < KH * KW) {
+            // one division for the run; (oh,ow) advance incrementally
+            const int mb = m0 + kc * 32 + (lane >> 4) * 8;
+            int oh = mb / OW, ow = mb - oh * OW;
             #pragma unroll
             for (int j = 0; j < 8; ++j) {
-              const int m = m0 + kc * 32 + (lane >> 4) * 8 + j;
-              const int oh = m / OW, ow = m - oh * OW;
               const int h = oh * SH - PH + kh, wx = ow * SW - PW + kw;
-              if (m < R && h >= 0 && h < H && wx >= 0 && wx < W)
+              if (mb + j < R && h >= 0 && h < H && wx >= 0 && wx < W)
                 bfr[j] = *(const bf16*)(smX + (h * W + wx) * 2);
+              if (++ow == OW) { ow = 0; ++oh; }
             }
           }
           #pragma unroll
