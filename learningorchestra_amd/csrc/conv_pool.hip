@@ -325,8 +325,10 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
         #pragma unroll
         for (int r = 0; r < 4; ++r)
           if (tgt[ni][r]) *tgt[ni][r] += acc[ni][r];
-      __syncthreads();                        // scatter done before overwrite
-      write_tile(buf ^ 1, stg);
+      // the write targets the OPPOSITE buffer of every in-flight read and
+      // the next top-barrier publishes it — no fence needed within an m0
+      if (step + 1 < NSTEP) write_tile(buf ^ 1, stg);
+      else if (m0 + 64 < R) __syncthreads();  // next m0 restages tile 0
     }
   }
   __syncthreads();
@@ -511,8 +513,7 @@ __global__ __launch_bounds__(256) void conv_fwd_small_kernel(
         acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr, acc[ni],
                                                           0, 0, 0);
       }
-      __syncthreads();
-      write_w(buf ^ 1, stg);
+      if (chunk + 1 < NC) write_w(buf ^ 1, stg);
     }
     // epilogue: stage the wave's 16x64 tile through its LDS quarter
     // (fp32 -> bias/relu -> bf16), then row-contiguous 16-B stores
