@@ -37,6 +37,13 @@ bool launch_conv_fwd_small(const void* x, const void* w, long ldw,
                            int W, int C, int KH, int KW, int SH, int SW,
                            int PH, int PW, int OH, int OW, int outC, int relu,
                            hipStream_t s);
+bool launch_conv1d_fwd(const void* x, const void* w, long ldw,
+                       const void* bias, void* y, long ldy, int B, int H,
+                       int C, int KH, int PH, int OH, int outC, int relu,
+                       hipStream_t s);
+bool launch_conv1d_dx(const void* dy2, long ldy, const void* wt, long ldw,
+                      void* dx, int B, int H, int C, int KH, int PH, int OH,
+                      int outC, hipStream_t s);
 bool launch_conv_dw_c1(const void* dy2, long ldy, const void* x, void* dw,
                        long ldw, int B, int H, int W, int KH, int KW, int SH,
                        int SW, int PH, int PW, int OH, int OW, int outC,
@@ -256,6 +263,41 @@ bool conv_fwd_small(at::Tensor x, at::Tensor w,
       x.data_ptr(), w.data_ptr(), w.stride(0), bp, y.data_ptr(), y.stride(0),
       B, H, W, C, (int)KH, (int)KW, (int)SH, (int)SW, (int)PH, (int)PW, OH,
       OW, outC, relu ? 1 : 0, stream());
+}
+
+// 1-D (sequence) conv forward: W==1, KW==1, stride 1 — h-tiled x window
+// in LDS, outC in 64-wide slices.
+bool conv1d_fwd(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias,
+                at::Tensor y, int64_t KH, int64_t PH, bool relu) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  check_bf16(y, "y");
+  const int B = (int)x.size(0), H = (int)x.size(1), C = (int)x.size(3);
+  TORCH_CHECK(x.size(2) == 1, "conv1d needs W==1");
+  const int OH = H + 2 * (int)PH - (int)KH + 1;
+  const int outC = (int)y.size(1);
+  TORCH_CHECK(y.size(0) == (long)B * OH, "y rows");
+  const void* bp = nullptr;
+  if (bias) bp = bias->data_ptr();
+  return lo::launch_conv1d_fwd(x.data_ptr(), w.data_ptr(), w.stride(0), bp,
+                               y.data_ptr(), y.stride(0), B, H, C, (int)KH,
+                               (int)PH, OH, outC, relu ? 1 : 0, stream());
+}
+
+// 1-D conv dX: h-tiled LDS fp32 accumulator, non-atomic RMW scatter.
+bool conv1d_dx(at::Tensor dy2, at::Tensor wt, at::Tensor dx, int64_t KH,
+               int64_t PH) {
+  check_bf16(dy2, "dy2");
+  check_bf16(wt, "wt");
+  check_bf16(dx, "dx");
+  const int B = (int)dx.size(0), H = (int)dx.size(1), C = (int)dx.size(3);
+  TORCH_CHECK(dx.size(2) == 1, "conv1d needs W==1");
+  const int OH = H + 2 * (int)PH - (int)KH + 1;
+  const int outC = (int)dy2.size(1);
+  TORCH_CHECK(dy2.size(0) == (long)B * OH, "dy2 rows");
+  return lo::launch_conv1d_dx(dy2.data_ptr(), dy2.stride(0), wt.data_ptr(),
+                              wt.stride(0), dx.data_ptr(), B, H, C, (int)KH,
+                              (int)PH, OH, outC, stream());
 }
 
 // C=1 conv dW: dY^T @ im2col(x) with x images LDS-resident; dw fp32
@@ -555,6 +597,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe);
   m.def("im2col", &im2col);
   m.def("conv_fwd_small", &conv_fwd_small, "small-image fused conv fwd");
+  m.def("conv1d_fwd", &conv1d_fwd, "1-D conv fwd (h-tiled LDS window)");
+  m.def("conv1d_dx", &conv1d_dx, "1-D conv dX (h-tiled LDS RMW)");
   m.def("conv_dw_c1", &conv_dw_c1, "C=1 conv dW (x LDS-resident)");
   m.def("conv_dx", &conv_dx, "fused conv dX (LDS-accumulated scatter)");
   m.def("col2im", &col2im);

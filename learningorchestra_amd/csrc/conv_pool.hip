@@ -571,6 +571,243 @@ bool launch_conv_fwd_small(const void* x, const void* w, long ldw,
   return true;
 }
 
+// --------------------------------------------------------- 1-D convs -------
+// TextCNN-style sequence convs: W==1, KW==1, stride 1, C%32==0.  kw=1 means
+// no halo in the w dimension, so the sequence tiles cleanly over h: each
+// block owns 64 output rows of one image and stages the x window
+// (64+KH-1 rows) — or the dx tile — in LDS.  outC up to 128 is handled by
+// 64-wide output slices (fwd) / a 4-deep K loop (dX).
+
+__global__ __launch_bounds__(256) void conv1d_fwd_kernel(
+    const bf16* __restrict__ x,               // [B, H, 1, C]
+    const bf16* __restrict__ w, long ldw,     // [outC, kpad]
+    const float* __restrict__ bias,
+    bf16* __restrict__ y, long ldy,           // [B*OH, outC]
+    int H, int C, int KH, int PH, int OH, int outC, int relu) {
+  extern __shared__ char sm[];
+  const int img = blockIdx.x;
+  const int m0 = blockIdx.y * 64;             // this block's y rows
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const int XROW = C * 2 + 16;
+  const int XR = 64 + KH - 1;                 // x window rows
+  const int x0 = m0 - PH;                     // first window row (may be <0)
+  char* smX = sm;
+  char* smW = sm + ((XR * XROW + 127) & ~127);
+
+  // stage x window (rows outside [0,H) stage zeros)
+  for (int i = tid * 8; i < XR * C; i += 256 * 8) {
+    const int r = i / C, c0 = i - r * C;
+    const int h = x0 + r;
+    bf16x8 v = {};
+    if (h >= 0 && h < H)
+      v = *(const bf16x8*)(x + ((long)img * H + h) * C + c0);
+    *(bf16x8*)(smX + r * XROW + c0 * 2) = v;
+  }
+
+  const int NC = (KH * C) / 32;
+  const int srow = tid >> 2, skc = tid & 3;
+  auto wswz = [](int row, int kel) {
+    return (kel ^ ((row & 3) << 3) ^ (((row >> 2) & 3) << 3));
+  };
+  const int arow = m0 + wave * 16 + (lane & 15);
+
+  for (int os = 0; os < outC; os += 64) {
+    auto fetch_w = [&](int chunk) -> bf16x8 {
+      bf16x8 v = {};
+      if (chunk < NC && os + srow < outC)
+        v = *(const bf16x8*)(w + (long)(os + srow) * ldw + chunk * 32 + skc * 8);
+      return v;
+    };
+    auto write_w = [&](int buf, bf16x8 v) {
+      *(bf16x8*)(smW + buf * 4096 + srow * 64 + wswz(srow, skc * 8) * 2) = v;
+    };
+    f32x4 acc[4] = {};
+    bf16x8 stg = fetch_w(0);
+    write_w(0, stg);
+    __syncthreads();                          // x (first os) / se reuse + w
+    for (int chunk = 0; chunk < NC; ++chunk) {
+      const int buf = chunk & 1;
+      stg = fetch_w(chunk + 1);
+      if (chunk) __syncthreads();
+      bf16x8 af = {};
+      {
+        const int k = chunk * 32 + (lane >> 4) * 8;
+        const int kh = k / C, c = k - kh * C;
+        const int r = (arow - m0) + kh;       // row in the window
+        if (arow < OH)
+          af = *(const bf16x8*)(smX + r * XROW + c * 2);
+      }
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int row = ni * 16 + (lane & 15);
+        const int k2 = (lane >> 4) * 8;
+        const bf16x8 bfr = *(const bf16x8*)(
+            smW + buf * 4096 + row * 64 + wswz(row, k2) * 2);
+        acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr, acc[ni],
+                                                          0, 0, 0);
+      }
+      if (chunk + 1 < NC) write_w(buf ^ 1, stg);
+    }
+    __syncthreads();                          // w buffers -> epilogue staging
+    char* se = smW + wave * 2048;
+    #pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int c = os + ni * 16 + (lane & 15);
+      const float b = bias ? bias[c < outC ? c : 0] : 0.f;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float v = acc[ni][r] + b;
+        if (relu) v = v > 0.f ? v : 0.f;
+        ((bf16*)se)[((lane >> 4) * 4 + r) * 64 + (ni * 16 + (lane & 15))] =
+            tobf16(v);
+      }
+    }
+    __syncthreads();
+    #pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      const int row = p * 8 + (lane >> 3);
+      const int m = m0 + wave * 16 + row;
+      if (m < OH)
+        *(bf16x8*)(y + ((long)img * OH + m) * ldy + os + (lane & 7) * 8) =
+            *(const bf16x8*)(se + row * 128 + (lane & 7) * 16);
+    }
+    __syncthreads();                          // before next os reuses smW
+  }
+}
+
+bool launch_conv1d_fwd(const void* x, const void* w, long ldw,
+                       const void* bias, void* y, long ldy, int B, int H,
+                       int C, int KH, int PH, int OH, int outC, int relu,
+                       hipStream_t s) {
+  const int XR = 64 + KH - 1;
+  const int lds = ((XR * (C * 2 + 16) + 127) & ~127) + 8192;
+  if (C % 32 != 0 || outC % 64 != 0 || outC > 128 || (KH * C) % 32 != 0 ||
+      lds > 56 * 1024)
+    return false;
+  const int T = (OH + 63) / 64;
+  hipLaunchKernelGGL(conv1d_fwd_kernel, dim3(B, T), dim3(256), lds, s,
+                     (const bf16*)x, (const bf16*)w, ldw, (const float*)bias,
+                     (bf16*)y, ldy, H, C, KH, PH, OH, outC, relu);
+  return true;
+}
+
+// dX for 1-D convs: block owns dx rows [h0, h0+64) of one image; the
+// contributing dy rows span [h0-KH+1+PH, h0+63+PH] — up to two 64-row
+// MFMA chunks (the waste is MFMA-only; traffic is read-dy ~2x + write-dx).
+// Same barrier-separated non-atomic LDS RMW scatter as conv_dx_kernel.
+__global__ __launch_bounds__(256) void conv1d_dx_kernel(
+    const bf16* __restrict__ dy2, long ldy,   // [B*OH, outC]
+    const bf16* __restrict__ wt, long ldw,    // [kpad, outC]
+    bf16* __restrict__ dx,                    // [B, H, 1, C]
+    int H, int C, int KH, int PH, int OH, int outC) {
+  extern __shared__ float ldx[];              // 64 x (C+1) fp32
+  const int img = blockIdx.x;
+  const int h0 = blockIdx.y * 64;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const int CP = C + 1;
+  char* smW = (char*)(ldx + 64 * CP);
+  for (int i = tid; i < 64 * CP; i += 256) ldx[i] = 0.f;
+
+  const bf16* dyi = dy2 + (long)img * OH * ldy;
+  const int KCH = outC / 32;                  // <= 4
+  const int NSTEP = KH * (C / 32);
+  const int TB = 32 * outC * 2;               // wt tile bytes (<= 8 KB)
+  auto wswz = [](int row, int kel) {
+    return (kel ^ ((row & 7) << 3));
+  };
+  auto stage_tile = [&](int buf, int step) {
+    // tile: 32 kpad-rows x outC, k-contiguous; row stride outC*2 bytes
+    if (step >= NSTEP) return;
+    const int kh = step / (C / 32), cb = (step % (C / 32)) * 32;
+    const int n0 = kh * C + cb;
+    for (int i = tid * 8; i < 32 * outC; i += 256 * 8) {
+      const int r = i / outC, k0 = i - r * outC;
+      *(bf16x8*)(smW + buf * TB + r * outC * 2 + wswz(r, k0) * 2) =
+          *(const bf16x8*)(wt + (long)(n0 + r) * ldw + k0);
+    }
+  };
+
+  const int m_lo0 = max(0, h0 - KH + 1 + PH) & ~63;
+  const int m_hi = min(OH - 1, h0 + 63 + PH);
+  for (int m0 = m_lo0; m0 <= m_hi; m0 += 64) {
+    const int arow = m0 + wave * 16 + (lane & 15);
+    bf16x8 af[4] = {};
+    #pragma unroll
+    for (int kc = 0; kc < 4; ++kc) {
+      if (kc >= KCH) break;
+      const int k = kc * 32 + (lane >> 4) * 8;
+      if (arow < OH)
+        af[kc] = *(const bf16x8*)(dyi + (long)arow * ldy + k);
+    }
+    const int mrow = m0 + wave * 16 + (lane >> 4) * 4;  // first of 4 regs
+    stage_tile(0, 0);
+    for (int step = 0; step < NSTEP; ++step) {
+      const int buf = step & 1;
+      __syncthreads();                        // tile buf visible / RMW safe
+      const int kh = step / (C / 32), cb = (step % (C / 32)) * 32;
+      f32x4 acc[2] = {};
+      #pragma unroll
+      for (int kc = 0; kc < 4; ++kc) {
+        if (kc >= KCH) break;
+        const int k = kc * 32 + (lane >> 4) * 8;
+        #pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          const int row = ni * 16 + (lane & 15);  // row within the 32-tile
+          const bf16x8 bfr = *(const bf16x8*)(
+              smW + buf * TB + row * outC * 2 + wswz(row, k) * 2);
+          acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[kc], bfr, acc[ni], 0, 0, 0);
+        }
+      }
+      stage_tile(buf ^ 1, step + 1);
+      // scatter: h = oh - PH + kh, target row h - h0 in [0, 64)
+      float* tgt[2][4];
+      #pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        const int c = cb + ni * 16 + (lane & 15);
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int m = mrow + r;
+          const int h = m - PH + kh;
+          const bool ok = m < OH && h >= h0 && h < h0 + 64 && h < H;
+          tgt[ni][r] = ok ? ldx + (h - h0) * CP + c : nullptr;
+        }
+      }
+      #pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+        #pragma unroll
+        for (int r = 0; r < 4; ++r)
+          if (tgt[ni][r]) *tgt[ni][r] += acc[ni][r];
+    }
+    __syncthreads();                          // next m0 restages tile 0
+  }
+  bf16* dxi = dx + ((long)img * H + h0) * C;
+  const int HR = min(64, H - h0);
+  for (int i = tid * 8; i < HR * C; i += 256 * 8) {
+    const int r = i / C, c0 = i - r * C;
+    const float* src = ldx + r * CP + c0;
+    bf16x8 v;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) v[j] = tobf16(src[j]);
+    *(bf16x8*)(dxi + (long)r * C + c0) = v;
+  }
+}
+
+bool launch_conv1d_dx(const void* dy2, long ldy, const void* wt, long ldw,
+                      void* dx, int B, int H, int C, int KH, int PH, int OH,
+                      int outC, hipStream_t s) {
+  const int lds = 64 * (C + 1) * 4 + 2 * 32 * outC * 2;
+  if (C % 32 != 0 || outC % 32 != 0 || outC > 128 || lds > 56 * 1024)
+    return false;
+  const int T = (H + 63) / 64;
+  hipLaunchKernelGGL(conv1d_dx_kernel, dim3(B, T), dim3(256), lds, s,
+                     (const bf16*)dy2, ldy, (const bf16*)wt, ldw, (bf16*)dx,
+                     H, C, KH, PH, OH, outC);
+  return true;
+}
+
 // ----------------------------------------------------- C=1 conv dW ---------
 // dW[outC, kpad] = dY^T @ im2col(x) for single-channel convs (MNIST conv1):
 // the reduction dim is the 18.9M-row batch.  Each block owns a group of

@@ -106,7 +106,8 @@ class Conv2dNHWC(Layer):
             if not self.first:
                 self._bufs["dx"] = torch.empty((B, H, W, self.in_c), device=dev, dtype=dtype)
                 if not self._is_1x1 and not (dev.type == "cuda"
-                                             and self._dx_fused_ok(H, W)):
+                                             and (self._dx_fused_ok(H, W)
+                                                  or self._conv1d_ok(W))):
                     self._bufs["dcol"] = torch.empty((M, self.kpad), device=dev, dtype=dtype)
         return self._bufs
 
@@ -121,6 +122,15 @@ class Conv2dNHWC(Layer):
                 self.out_c <= 64 and self.out_c % 16 == 0 and
                 (self.kh * self.kw * self.in_c) % 32 == 0 and
                 (H * W * self.in_c) % 8 == 0)
+
+    def _conv1d_ok(self, W: int) -> bool:
+        # mirror of launch_conv1d_* eligibility (W==1 sequence convs)
+        return (W == 1 and self.kw == 1 and self.stride == 1 and
+                self.in_c % 32 == 0 and self.out_c % 64 == 0 and
+                self.out_c <= 128 and (self.kh * self.in_c) % 32 == 0 and
+                (64 + self.kh - 1) * (self.in_c * 2 + 16) + 128 + 8192
+                <= 56 * 1024 and
+                64 * (self.in_c + 1) * 4 + 128 * self.out_c <= 56 * 1024)
 
     def _dx_fused_ok(self, H: int, W: int) -> bool:
         # mirror of launch_conv_dx eligibility: whole-image dx fits LDS
@@ -155,9 +165,14 @@ class Conv2dNHWC(Layer):
             F.gemm(bufs["col"], self.arena.p(self.name + ".w"), tb=True,
                    bias=bias, relu=self.relu, out=bufs["y"], stats=st)
         elif x.is_cuda and self.implicit:
-            # implicit conv, small-image fused kernel when the x image fits
-            # LDS; im2col gathered inside the GEMM staging otherwise
-            if not (self._fwd_small_ok(H, W) and F.conv2d_fwd_small(
+            # implicit conv: 1-D tiled kernel (W==1 sequences), small-image
+            # fused kernel (x image fits LDS), or im2col gathered inside the
+            # GEMM staging as the fallback
+            if self._conv1d_ok(W) and F.conv1d_fwd(
+                    x, self.arena.p(self.name + ".w"), self.kh, self.pad,
+                    bias=bias, relu=self.relu, out=bufs["y"]):
+                pass
+            elif not (self._fwd_small_ok(H, W) and F.conv2d_fwd_small(
                     x, self.arena.p(self.name + ".w"), self.kh, self.kw,
                     self.stride, self.stride, self.pad, self.pad, bias=bias,
                     relu=self.relu, out=bufs["y"])):
@@ -205,6 +220,9 @@ class Conv2dNHWC(Layer):
             dx = bufs["dx"]
             F.gemm(dy2, self._wt(), tb=True, out=dx.view(M, self.in_c))
             return dx
+        if dy2.is_cuda and self._conv1d_ok(bufs["W"]) and F.conv1d_dx(
+                dy2, self._wt(), self.kh, self.pad, out=bufs["dx"]):
+            return bufs["dx"]
         if dy2.is_cuda and self._dx_fused_ok(bufs["H"], bufs["W"]) \
                 and F.conv2d_dx_fused(
                 dy2, self._wt(), bufs["B"], bufs["H"], bufs["W"], self.in_c,

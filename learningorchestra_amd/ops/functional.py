@@ -123,6 +123,27 @@ def conv2d_fwd_small(x: torch.Tensor, w: torch.Tensor, kh: int, kw: int,
                                   relu))
 
 
+def conv1d_fwd(x: torch.Tensor, w: torch.Tensor, kh: int, ph: int,
+               bias: Optional[torch.Tensor] = None, relu: bool = False,
+               out: Optional[torch.Tensor] = None) -> bool:
+    """GPU-only 1-D (W==1, KW==1, stride 1) conv forward: h-tiled x window
+    in LDS, outC in 64-wide slices.  False when not eligible."""
+    if not x.is_cuda:
+        return False
+    lo = require_ext()
+    return bool(lo.conv1d_fwd(x, w, bias, out, kh, ph, relu))
+
+
+def conv1d_dx(dy2: torch.Tensor, wt: torch.Tensor, kh: int, ph: int,
+              out: torch.Tensor) -> bool:
+    """GPU-only 1-D conv dX: h-tiled LDS fp32 accumulator, non-atomic RMW
+    scatter (no dcol matrix).  False when not eligible."""
+    if not dy2.is_cuda:
+        return False
+    lo = require_ext()
+    return bool(lo.conv1d_dx(dy2, wt, out, kh, ph))
+
+
 def conv2d_dw_c1(dy2: torch.Tensor, x: torch.Tensor, gw: torch.Tensor,
                  kh: int, kw: int, sh: int, sw: int, ph: int, pw: int) -> bool:
     """GPU-only C=1 conv weight grad (dY^T @ im2col(x)) with x images

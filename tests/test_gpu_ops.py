@@ -360,3 +360,45 @@ def test_conv_dw_c1_matches_splitk(shape):
     ref = F.gemm(dy2, col, ta=True, splits=4)
     torch.testing.assert_close(gw[:, :kdim], ref[:, :kdim], atol=2e-1,
                                rtol=2e-2)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [
+    # (B, H, C, KH, P, outC)
+    (8, 256, 128, 5, 0, 128),    # TextCNN k5
+    (8, 256, 128, 3, 0, 128),    # TextCNN k3
+    (4, 100, 32, 4, 2, 64),      # padding, odd H (tile tail)
+    (2, 64, 64, 3, 1, 128),      # single tile
+])
+def test_conv1d_fwd_dx_match_reference(shape):
+    """1-D tiled conv fwd + dX vs im2col GEMM / dcol col2im references."""
+    from learningorchestra_amd.ops import functional as F
+    B, H, C, KH, P, outC = shape
+    torch.manual_seed(3)
+    OH = H + 2 * P - KH + 1
+    kdim = KH * C
+    kpad = (kdim + 7) // 8 * 8
+    x = torch.randn(B, H, 1, C, device="cuda").to(torch.bfloat16)
+    w = torch.zeros(outC, kpad, device="cuda", dtype=torch.bfloat16)
+    w[:, :kdim] = torch.randn(outC, kdim, device="cuda").to(torch.bfloat16) * 0.1
+    bias = torch.randn(outC, device="cuda", dtype=torch.float32)
+    y = torch.empty(B * OH, outC, device="cuda", dtype=torch.bfloat16)
+    ok = F.conv1d_fwd(x, w, KH, P, bias=bias, relu=True, out=y)
+    assert ok, "fwd shape should be eligible"
+    col = F.im2col(x, KH, 1, 1, 1, P, 0, kpad)
+    ref = F.gemm(col, w, tb=True, bias=bias, relu=True)
+    torch.testing.assert_close(y.float(), ref.float(), atol=3e-2, rtol=3e-2)
+
+    dy2 = torch.randn(B * OH, outC, device="cuda").to(torch.bfloat16)
+    wt = w.t().contiguous()
+    dx = torch.empty(B, H, 1, C, device="cuda", dtype=torch.bfloat16)
+    ok = F.conv1d_dx(dy2, wt, KH, P, out=dx)
+    assert ok, "dx shape should be eligible"
+    dcol = F.gemm(dy2, wt, tb=True)
+    refdx = F.col2im(dcol, B, H, 1, C, KH, 1, 1, 1, P, 0)
+    dcol32 = (dy2.float() @ wt.float().t()).cpu()
+    ref32 = F.col2im(dcol32, B, H, 1, C, KH, 1, 1, 1, P, 0,
+                     out=torch.empty(B, H, 1, C))
+    err_fused = (dx.float().cpu() - ref32).abs().max().item()
+    err_ref = (refdx.float().cpu() - ref32).abs().max().item()
+    assert err_fused <= max(2 * err_ref, 1e-3), (err_fused, err_ref)
