@@ -38,6 +38,10 @@ class TextCNN:
                                  relu=True, implicit=imp)
                       for k in self.kernel_sizes]
         self.pools = [MaxPool2dNHWC(1) for _ in self.kernel_sizes]  # k set per fwd
+        for conv, pool in zip(self.convs, self.pools):
+            # conv-ReLU backward folds into the pool's backward (relu_y mask)
+            pool.fuse_relu = True
+            conv.relu_bwd_upstream = True
         self.fc = Linear("fc", filters * len(self.kernel_sizes), self.cpad)
         for lay in [self.emb, *self.convs, self.fc]:
             lay.build(self.arena)
