@@ -107,7 +107,8 @@ class Conv2dNHWC(Layer):
                 self._bufs["dx"] = torch.empty((B, H, W, self.in_c), device=dev, dtype=dtype)
                 if not self._is_1x1 and not (dev.type == "cuda"
                                              and (self._dx_fused_ok(H, W)
-                                                  or self._conv1d_ok(W))):
+                                                  or (self.implicit
+                                                      and self._conv1d_ok(W)))):
                     self._bufs["dcol"] = torch.empty((M, self.kpad), device=dev, dtype=dtype)
         return self._bufs
 
@@ -220,8 +221,9 @@ class Conv2dNHWC(Layer):
             dx = bufs["dx"]
             F.gemm(dy2, self._wt(), tb=True, out=dx.view(M, self.in_c))
             return dx
-        if dy2.is_cuda and self._conv1d_ok(bufs["W"]) and F.conv1d_dx(
-                dy2, self._wt(), self.kh, self.pad, out=bufs["dx"]):
+        if (dy2.is_cuda and self.implicit and self._conv1d_ok(bufs["W"])
+                and F.conv1d_dx(dy2, self._wt(), self.kh, self.pad,
+                                out=bufs["dx"])):
             return bufs["dx"]
         if dy2.is_cuda and self._dx_fused_ok(bufs["H"], bufs["W"]) \
                 and F.conv2d_dx_fused(
