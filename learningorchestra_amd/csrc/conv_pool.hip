@@ -410,16 +410,25 @@ __global__ __launch_bounds__(256) void conv_fwd_small_kernel(
     }
     __syncthreads();                          // x + w staged
     char* se = smW + 4096 + wave * 2048;      // private 16x64 bf16
+    // this lane's 8 k-cells are fixed across m0 — decode once
+    int kh8[8], kw8[8];
+    bool kok8[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int k = (lane >> 4) * 8 + j;
+      kh8[j] = k / KW;
+      kw8[j] = k - kh8[j] * KW;
+      kok8[j] = k < KH * KW;
+    }
     for (int m0 = 0; m0 < R; m0 += 64) {
       const int arow = m0 + wave * 16 + (lane & 15);
       const int aoh = arow / OW, aow = arow - aoh * OW;
+      const int hb = aoh * SH - PH, wb = aow * SW - PW;
       bf16x8 af = {};
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        const int k = (lane >> 4) * 8 + j;
-        const int kh = k / KW, kw = k - kh * KW;
-        const int h = aoh * SH - PH + kh, wx = aow * SW - PW + kw;
-        if (arow < R && k < KH * KW && h >= 0 && h < H && wx >= 0 && wx < W)
+        const int h = hb + kh8[j], wx = wb + kw8[j];
+        if (arow < R && kok8[j] && h >= 0 && h < H && wx >= 0 && wx < W)
           af[j] = *(const bf16*)(smX + (h * W + wx) * 2);
       }
       f32x4 acc[4] = {};
