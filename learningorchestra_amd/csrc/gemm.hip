@@ -595,7 +595,12 @@ static bool dispatch_conv_fwd(const GemmArgs& g, hipStream_t s) {
 // implicit-conv dW: C[M=outC, N=kpad] = dY^T @ im2col(x), split-K atomics
 static bool dispatch_conv_dw(const GemmArgs& g, hipStream_t s) {
   if (!g.out_f32 || g.epi != 0 || !g.ta || g.tb) return false;
-  launch_cfg<32, 64, 64, 1, 4, true, false, 0, true, true, 2>(g, s);
+  // BM=64 when outC allows: halves the number of M-tiles and with it the
+  // total conv_gather8 work (the B gather dominates this kernel)
+  if (g.M >= 64)
+    launch_cfg<64, 64, 64, 2, 2, true, false, 0, true, true, 2>(g, s);
+  else
+    launch_cfg<32, 64, 64, 1, 4, true, false, 0, true, true, 2>(g, s);
   return true;
 }
 
