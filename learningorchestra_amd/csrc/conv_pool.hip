@@ -224,16 +224,16 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
     bf16* __restrict__ dx,                    // [B, H, W, C]
     int H, int W, int C, int KH, int KW, int SH, int SW, int PH, int PW,
     int OH, int OW, int outC) {
-  // dynamic LDS: fp32 dx accumulator (H*W rows of C+1 floats - the +1
-  // staggers banks across rows) + 2 x 4 KB wt tile buffers (32 rows x 64 k,
-  // XOR-swizzled like the GEMM's TB operand)
+  // dynamic LDS: fp32 dx accumulator (H*W rows of C+4 floats - the +4
+  // keeps 16-B alignment for the vectorized RMW and staggers banks across
+  // rows) + 2 x 4 KB wt tile buffers (32 rows x 64 k, XOR-swizzled)
   extern __shared__ float ldx[];
   const int img = blockIdx.x;
   const int R = OH * OW;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wave = tid >> 6;
   const int HWC = H * W * C;
-  const int CP = C + 1;
+  const int CP = C + 4;   // +4: keeps 16-B alignment AND staggers banks
   const int HWCP = H * W * CP;
   char* smW = (char*)(ldx + HWCP);
   for (int i = tid; i < HWCP; i += 256) ldx[i] = 0.f;
@@ -262,25 +262,20 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
   };
 
   for (int m0 = 0; m0 < R; m0 += 64) {
-    // the wave's 16 A-rows: row=lane%16, k=(lane/16)*8+j (outC <= 64)
-    const int arow = m0 + wave * 16 + (lane & 15);
-    bf16x8 af[2] = {};
+    // SWAPPED operand roles: A = the wt tile (kpad-slice x outC), B = dy
+    // (outC x m) -> D[kcol][m].  A lane's 4 acc regs are then 4 CONSECUTIVE
+    // channels of ONE m-row, so the LDS RMW is a single b128 per fragment
+    // (the row-per-reg layout needed 8 scalar RMWs per step).
+    const int arow = m0 + wave * 16 + (lane & 15);     // this lane's m
+    bf16x8 dyf[2] = {};
     #pragma unroll
     for (int kc = 0; kc < 2; ++kc) {
       const int k = kc * 32 + (lane >> 4) * 8;
       if (arow < R && k < outC)
-        af[kc] = *(const bf16x8*)(dyi + (long)arow * ldy + k);
+        dyf[kc] = *(const bf16x8*)(dyi + (long)arow * ldy + k);
     }
-    // the 4 accumulator rows this lane scatters: m = m0+wave*16+(lane>>4)*4+r
-    int oh4[4], ow4[4];
-    #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
-      oh4[r] = m / OW;
-      ow4[r] = m - oh4[r] * OW;
-    }
-    const int mok = (m0 + wave * 16 + (lane >> 4) * 4 + 3 < R) ? 4
-                    : max(0, R - (m0 + wave * 16 + (lane >> 4) * 4));
+    const int aoh = arow / OW, aow = arow - aoh * OW;
+    const bool mok = arow < R;
 
     bf16x8 stg = fetch_tile(0);
     write_tile(0, stg);
@@ -296,35 +291,29 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
       for (int kc = 0; kc < 2; ++kc) {
         const int k = kc * 32 + (lane >> 4) * 8;
         #pragma unroll
-        for (int ni = 0; ni < 2; ++ni) {
-          const int row = ni * 16 + (lane & 15);
-          const bf16x8 bfr = *(const bf16x8*)(
+        for (int mi = 0; mi < 2; ++mi) {
+          const int row = mi * 16 + (lane & 15);
+          const bf16x8 wf = *(const bf16x8*)(
               smW + buf * 4096 + row * 128 + ((k * 2) ^ ((row & 7) << 4)));
-          acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[kc], bfr, acc[ni], 0, 0, 0);
+          acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              wf, dyf[kc], acc[mi], 0, 0, 0);
         }
       }
-      // scatter: D col=lane&15 -> channel, row=(lane>>4)*4+reg -> m.
-      // Targets are unique block-wide within this step -> plain RMW.
-      const int hb = -PH + kh, wb = -PW + kw;
-      float* tgt[2][4];
+      // D col=lane&15 -> m (this lane's arow), row=(lane>>4)*4+reg -> kcol.
+      // 4 regs = channels c..c+3 -> one vectorized RMW per fragment.
+      const int h = aoh * SH - PH + kh;
+      const int w = aow * SW - PW + kw;
+      const bool hok = mok && h >= 0 && h < H && w >= 0 && w < W;
       #pragma unroll
-      for (int ni = 0; ni < 2; ++ni) {
-        const int c = cb + ni * 16 + (lane & 15);
-        #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int h = oh4[r] * SH + hb;
-          const int w = ow4[r] * SW + wb;
-          const bool ok = (ni * 16 < CB) && (r < mok) &&
-                          h >= 0 && h < H && w >= 0 && w < W;
-          tgt[ni][r] = ok ? ldx + (h * W + w) * CP + c : nullptr;
+      for (int mi = 0; mi < 2; ++mi) {
+        const int c = cb + mi * 16 + (lane >> 4) * 4;
+        if (hok && mi * 16 < CB) {
+          float* t = ldx + (h * W + w) * CP + c;
+          f32x4 v = *(f32x4*)t;
+          v += acc[mi];
+          *(f32x4*)t = v;
         }
       }
-      #pragma unroll
-      for (int ni = 0; ni < 2; ++ni)
-        #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          if (tgt[ni][r]) *tgt[ni][r] += acc[ni][r];
       // the write targets the OPPOSITE buffer of every in-flight read and
       // the next top-barrier publishes it — no fence needed within an m0
       if (step + 1 < NSTEP) write_tile(buf ^ 1, stg);
@@ -348,7 +337,7 @@ bool launch_conv_dx(const void* dy2, long ldy, const void* wt, long ldw,
                     int SH, int SW, int PH, int PW, int OH, int OW, int outC,
                     hipStream_t s) {
   const int HWC = H * W * C;
-  const int HWCP = H * W * (C + 1);
+  const int HWCP = H * W * (C + 4);
   if ((long)HWCP * 4 > 56 * 1024 || (C % 32 != 0 && C != 16) || HWC % 8 != 0 ||
       outC > 64 || outC % 8 != 0 || C % 8 != 0)
     return false;

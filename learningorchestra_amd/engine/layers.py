@@ -136,7 +136,7 @@ class Conv2dNHWC(Layer):
     def _dx_fused_ok(self, H: int, W: int) -> bool:
         # mirror of launch_conv_dx eligibility: whole-image dx fits LDS
         hwc = H * W * self.in_c
-        return (hwc * 4 <= 48 * 1024 and
+        return (H * W * (self.in_c + 4) * 4 <= 56 * 1024 and
                 (self.in_c % 32 == 0 or self.in_c == 16) and
                 hwc % 8 == 0 and self.out_c <= 64 and self.out_c % 8 == 0)
 
