@@ -699,12 +699,12 @@ __global__ __launch_bounds__(256) void conv1d_dx_kernel(
     const bf16* __restrict__ wt, long ldw,    // [kpad, outC]
     bf16* __restrict__ dx,                    // [B, H, 1, C]
     int H, int C, int KH, int PH, int OH, int outC) {
-  extern __shared__ float ldx[];              // 64 x (C+1) fp32
+  extern __shared__ float ldx[];              // 64 x (C+4) fp32
   const int img = blockIdx.x;
   const int h0 = blockIdx.y * 64;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wave = tid >> 6;
-  const int CP = C + 1;
+  const int CP = C + 4;   // 16-B-aligned rows, banks staggered
   char* smW = (char*)(ldx + 64 * CP);
   for (int i = tid; i < 64 * CP; i += 256) ldx[i] = 0.f;
 
@@ -730,16 +730,18 @@ __global__ __launch_bounds__(256) void conv1d_dx_kernel(
   const int m_lo0 = max(0, h0 - KH + 1 + PH) & ~63;
   const int m_hi = min(OH - 1, h0 + 63 + PH);
   for (int m0 = m_lo0; m0 <= m_hi; m0 += 64) {
-    const int arow = m0 + wave * 16 + (lane & 15);
-    bf16x8 af[4] = {};
+    // swapped operand roles (see conv_dx_kernel): A = wt tile, B = dy ->
+    // a lane's 4 acc regs are 4 consecutive channels of its ONE m-row and
+    // the scatter is a single b128 RMW per fragment
+    const int arow = m0 + wave * 16 + (lane & 15);     // this lane's m
+    bf16x8 dyf[4] = {};
     #pragma unroll
     for (int kc = 0; kc < 4; ++kc) {
       if (kc >= KCH) break;
       const int k = kc * 32 + (lane >> 4) * 8;
       if (arow < OH)
-        af[kc] = *(const bf16x8*)(dyi + (long)arow * ldy + k);
+        dyf[kc] = *(const bf16x8*)(dyi + (long)arow * ldy + k);
     }
-    const int mrow = m0 + wave * 16 + (lane >> 4) * 4;  // first of 4 regs
     stage_tile(0, 0);
     for (int step = 0; step < NSTEP; ++step) {
       const int buf = step & 1;
@@ -751,33 +753,28 @@ __global__ __launch_bounds__(256) void conv1d_dx_kernel(
         if (kc >= KCH) break;
         const int k = kc * 32 + (lane >> 4) * 8;
         #pragma unroll
-        for (int ni = 0; ni < 2; ++ni) {
-          const int row = ni * 16 + (lane & 15);  // row within the 32-tile
-          const bf16x8 bfr = *(const bf16x8*)(
+        for (int mi = 0; mi < 2; ++mi) {
+          const int row = mi * 16 + (lane & 15);  // kcol within the 32-tile
+          const bf16x8 wf = *(const bf16x8*)(
               smW + buf * TB + row * outC * 2 + wswz(row, k) * 2);
-          acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[kc], bfr, acc[ni], 0, 0, 0);
+          acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              wf, dyf[kc], acc[mi], 0, 0, 0);
         }
       }
       stage_tile(buf ^ 1, step + 1);
-      // scatter: h = oh - PH + kh, target row h - h0 in [0, 64)
-      float* tgt[2][4];
+      // D col=lane&15 -> m (arow), row=(lane>>4)*4+reg -> kcol -> channel
+      const int h = arow - PH + kh;
+      const bool hok = arow < OH && h >= h0 && h < h0 + 64 && h < H;
       #pragma unroll
-      for (int ni = 0; ni < 2; ++ni) {
-        const int c = cb + ni * 16 + (lane & 15);
-        #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int m = mrow + r;
-          const int h = m - PH + kh;
-          const bool ok = m < OH && h >= h0 && h < h0 + 64 && h < H;
-          tgt[ni][r] = ok ? ldx + (h - h0) * CP + c : nullptr;
+      for (int mi = 0; mi < 2; ++mi) {
+        const int c = cb + mi * 16 + (lane >> 4) * 4;
+        if (hok) {
+          float* t = ldx + (h - h0) * CP + c;
+          f32x4 v = *(f32x4*)t;
+          v += acc[mi];
+          *(f32x4*)t = v;
         }
       }
-      #pragma unroll
-      for (int ni = 0; ni < 2; ++ni)
-        #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          if (tgt[ni][r]) *tgt[ni][r] += acc[ni][r];
     }
     __syncthreads();                          // next m0 restages tile 0
   }
@@ -796,7 +793,7 @@ __global__ __launch_bounds__(256) void conv1d_dx_kernel(
 bool launch_conv1d_dx(const void* dy2, long ldy, const void* wt, long ldw,
                       void* dx, int B, int H, int C, int KH, int PH, int OH,
                       int outC, hipStream_t s) {
-  const int lds = 64 * (C + 1) * 4 + 2 * 32 * outC * 2;
+  const int lds = 64 * (C + 4) * 4 + 2 * 32 * outC * 2;
   if (C % 32 != 0 || outC % 32 != 0 || outC > 128 || lds > 56 * 1024)
     return false;
   const int T = (H + 63) / 64;

@@ -131,7 +131,7 @@ class Conv2dNHWC(Layer):
                 self.out_c <= 128 and (self.kh * self.in_c) % 32 == 0 and
                 (64 + self.kh - 1) * (self.in_c * 2 + 16) + 128 + 8192
                 <= 56 * 1024 and
-                64 * (self.in_c + 1) * 4 + 128 * self.out_c <= 56 * 1024)
+                64 * (self.in_c + 4) * 4 + 128 * self.out_c <= 56 * 1024)
 
     def _dx_fused_ok(self, H: int, W: int) -> bool:
         # mirror of launch_conv_dx eligibility: whole-image dx fits LDS
