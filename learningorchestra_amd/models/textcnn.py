@@ -33,7 +33,10 @@ class TextCNN:
 
         self.arena = ParamArena(device)
         self.emb = Embedding("emb", vocab, emb_dim)
-        imp = os.environ.get("LO_IMPLICIT_CONV", "0") == "1"
+        # implicit default: conv1d fused fwd + 64x64 gather dW + conv1d
+        # fused dX (vectorized-RMW scatter) measured 649K vs 572K samples/s
+        # for the materialized-col path; LO_IMPLICIT_CONV=0 restores it
+        imp = os.environ.get("LO_IMPLICIT_CONV", "1") == "1"
         self.convs = [Conv2dNHWC(f"conv{k}", emb_dim, filters, k, 1,
                                  relu=True, implicit=imp)
                       for k in self.kernel_sizes]
