@@ -43,7 +43,7 @@ bool launch_conv1d_fwd(const void* x, const void* w, long ldw,
                        hipStream_t s);
 bool launch_conv1d_dx(const void* dy2, long ldy, const void* wt, long ldw,
                       void* dx, int B, int H, int C, int KH, int PH, int OH,
-                      int outC, hipStream_t s);
+                      int outC, int accumulate, hipStream_t s);
 bool launch_conv_dw_c1(const void* dy2, long ldy, const void* x, void* dw,
                        long ldw, int B, int H, int W, int KH, int KW, int SH,
                        int SW, int PH, int PW, int OH, int OW, int outC,
@@ -286,7 +286,7 @@ bool conv1d_fwd(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias,
 
 // 1-D conv dX: h-tiled LDS fp32 accumulator, non-atomic RMW scatter.
 bool conv1d_dx(at::Tensor dy2, at::Tensor wt, at::Tensor dx, int64_t KH,
-               int64_t PH) {
+               int64_t PH, bool accumulate) {
   check_bf16(dy2, "dy2");
   check_bf16(wt, "wt");
   check_bf16(dx, "dx");
@@ -297,7 +297,8 @@ bool conv1d_dx(at::Tensor dy2, at::Tensor wt, at::Tensor dx, int64_t KH,
   TORCH_CHECK(dy2.size(0) == (long)B * OH, "dy2 rows");
   return lo::launch_conv1d_dx(dy2.data_ptr(), dy2.stride(0), wt.data_ptr(),
                               wt.stride(0), dx.data_ptr(), B, H, C, (int)KH,
-                              (int)PH, OH, outC, stream());
+                              (int)PH, OH, outC, accumulate ? 1 : 0,
+                              stream());
 }
 
 // C=1 conv dW: dY^T @ im2col(x) with x images LDS-resident; dw fp32

@@ -698,7 +698,7 @@ __global__ __launch_bounds__(256) void conv1d_dx_kernel(
     const bf16* __restrict__ dy2, long ldy,   // [B*OH, outC]
     const bf16* __restrict__ wt, long ldw,    // [kpad, outC]
     bf16* __restrict__ dx,                    // [B, H, 1, C]
-    int H, int C, int KH, int PH, int OH, int outC) {
+    int H, int C, int KH, int PH, int OH, int outC, int accumulate) {
   extern __shared__ float ldx[];              // 64 x (C+4) fp32
   const int img = blockIdx.x;
   const int h0 = blockIdx.y * 64;
@@ -784,22 +784,28 @@ __global__ __launch_bounds__(256) void conv1d_dx_kernel(
     const int r = i / C, c0 = i - r * C;
     const float* src = ldx + r * CP + c0;
     bf16x8 v;
-    #pragma unroll
-    for (int j = 0; j < 8; ++j) v[j] = tobf16(src[j]);
+    if (accumulate) {                          // dx += tile (branch-grad sum)
+      const bf16x8 prev = *(const bf16x8*)(dxi + (long)r * C + c0);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = tobf16(src[j] + tofloat(prev[j]));
+    } else {
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = tobf16(src[j]);
+    }
     *(bf16x8*)(dxi + (long)r * C + c0) = v;
   }
 }
 
 bool launch_conv1d_dx(const void* dy2, long ldy, const void* wt, long ldw,
                       void* dx, int B, int H, int C, int KH, int PH, int OH,
-                      int outC, hipStream_t s) {
+                      int outC, int accumulate, hipStream_t s) {
   const int lds = 64 * (C + 4) * 4 + 2 * 32 * outC * 2;
   if (C % 32 != 0 || outC % 32 != 0 || outC > 128 || lds > 56 * 1024)
     return false;
   const int T = (H + 63) / 64;
   hipLaunchKernelGGL(conv1d_dx_kernel, dim3(B, T), dim3(256), lds, s,
                      (const bf16*)dy2, ldy, (const bf16*)wt, ldw, (bf16*)dx,
-                     H, C, KH, PH, OH, outC);
+                     H, C, KH, PH, OH, outC, accumulate);
   return true;
 }
 

@@ -194,7 +194,13 @@ class Conv2dNHWC(Layer):
     # True when the following pool's backward already applied our ReLU mask
     relu_bwd_upstream = False
 
-    def backward(self, dy: torch.Tensor) -> Optional[torch.Tensor]:
+    def backward(self, dy: torch.Tensor,
+                 dx_out: Optional[torch.Tensor] = None,
+                 dx_accumulate: bool = False) -> Optional[torch.Tensor]:
+        """``dx_out``/``dx_accumulate``: on the fused 1-D dX path, write (or
+        add) the input grad straight into the caller's buffer — fuses the
+        multi-branch grad sum (TextCNN).  Callers must check the returned
+        tensor: on fallback paths it is the layer's own dx buffer."""
         bufs = self._bufs
         M = bufs["y"].shape[0]
         dy2 = dy.reshape(M, self.out_c)
@@ -221,10 +227,11 @@ class Conv2dNHWC(Layer):
             dx = bufs["dx"]
             F.gemm(dy2, self._wt(), tb=True, out=dx.view(M, self.in_c))
             return dx
-        if (dy2.is_cuda and self.implicit and self._conv1d_ok(bufs["W"])
-                and F.conv1d_dx(dy2, self._wt(), self.kh, self.pad,
-                                out=bufs["dx"])):
-            return bufs["dx"]
+        if dy2.is_cuda and self.implicit and self._conv1d_ok(bufs["W"]):
+            tgt = dx_out if dx_out is not None else bufs["dx"]
+            if F.conv1d_dx(dy2, self._wt(), self.kh, self.pad, out=tgt,
+                           accumulate=dx_accumulate):
+                return tgt
         if dy2.is_cuda and self._dx_fused_ok(bufs["H"], bufs["W"]) \
                 and F.conv2d_dx_fused(
                 dy2, self._wt(), bufs["B"], bufs["H"], bufs["W"], self.in_c,

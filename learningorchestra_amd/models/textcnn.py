@@ -103,13 +103,16 @@ class TextCNN:
                                                 self.pools)):
             dpool = dcat[:, i * Fn:(i + 1) * Fn].contiguous().view(B, 1, 1, Fn)
             dconv = pool.backward(dpool)              # [B, S-k+1, 1, F]
-            dbranch = conv.backward(dconv)            # [B, S, 1, emb]
+            # fused path writes/adds straight into dxe (branch-grad sum)
+            dbranch = conv.backward(dconv, dx_out=dxe,
+                                    dx_accumulate=(i > 0))  # [B, S, 1, emb]
             if grad_hook:
                 grad_hook(conv.param_names())
-            if i == 0:
-                dxe.copy_(dbranch)
-            else:
-                dxe.add_(dbranch)
+            if dbranch is not dxe:
+                if i == 0:
+                    dxe.copy_(dbranch)
+                else:
+                    dxe.add_(dbranch)
         self.emb.backward(dxe.view(B, S, self.emb_dim))
         if grad_hook:
             grad_hook(self.emb.param_names())

@@ -135,13 +135,15 @@ def conv1d_fwd(x: torch.Tensor, w: torch.Tensor, kh: int, ph: int,
 
 
 def conv1d_dx(dy2: torch.Tensor, wt: torch.Tensor, kh: int, ph: int,
-              out: torch.Tensor) -> bool:
+              out: torch.Tensor, accumulate: bool = False) -> bool:
     """GPU-only 1-D conv dX: h-tiled LDS fp32 accumulator, non-atomic RMW
-    scatter (no dcol matrix).  False when not eligible."""
+    scatter (no dcol matrix).  ``accumulate`` adds into ``out`` instead of
+    overwriting (fuses the TextCNN branch-grad sum).  False when not
+    eligible."""
     if not dy2.is_cuda:
         return False
     lo = require_ext()
-    return bool(lo.conv1d_dx(dy2, wt, out, kh, ph))
+    return bool(lo.conv1d_dx(dy2, wt, out, kh, ph, accumulate))
 
 
 def conv2d_dw_c1(dy2: torch.Tensor, x: torch.Tensor, gw: torch.Tensor,
