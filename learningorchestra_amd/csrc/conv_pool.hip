@@ -701,7 +701,9 @@ __global__ __launch_bounds__(256) void conv1d_dx_kernel(
     int H, int C, int KH, int PH, int OH, int outC, int accumulate) {
   extern __shared__ float ldx[];              // 64 x (C+4) fp32
   const int img = blockIdx.x;
-  const int h0 = blockIdx.y * 64;
+  const int TS = 64 - KH + 1;                 // dx rows per tile: the
+  // contributing dy rows then span EXACTLY 64 rows -> one m-chunk
+  const int h0 = blockIdx.y * TS;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wave = tid >> 6;
   const int CP = C + 4;   // 16-B-aligned rows, banks staggered
@@ -727,9 +729,8 @@ __global__ __launch_bounds__(256) void conv1d_dx_kernel(
     }
   };
 
-  const int m_lo0 = max(0, h0 - KH + 1 + PH) & ~63;
-  const int m_hi = min(OH - 1, h0 + 63 + PH);
-  for (int m0 = m_lo0; m0 <= m_hi; m0 += 64) {
+  const int m0 = h0 - KH + 1 + PH;            // may be < 0; guards below
+  {
     // swapped operand roles (see conv_dx_kernel): A = wt tile, B = dy ->
     // a lane's 4 acc regs are 4 consecutive channels of its ONE m-row and
     // the scatter is a single b128 RMW per fragment
@@ -739,7 +740,7 @@ __global__ __launch_bounds__(256) void conv1d_dx_kernel(
     for (int kc = 0; kc < 4; ++kc) {
       if (kc >= KCH) break;
       const int k = kc * 32 + (lane >> 4) * 8;
-      if (arow < OH)
+      if (arow >= 0 && arow < OH)
         dyf[kc] = *(const bf16x8*)(dyi + (long)arow * ldy + k);
     }
     stage_tile(0, 0);
@@ -764,7 +765,8 @@ __global__ __launch_bounds__(256) void conv1d_dx_kernel(
       stage_tile(buf ^ 1, step + 1);
       // D col=lane&15 -> m (arow), row=(lane>>4)*4+reg -> kcol -> channel
       const int h = arow - PH + kh;
-      const bool hok = arow < OH && h >= h0 && h < h0 + 64 && h < H;
+      const bool hok = arow >= 0 && arow < OH && h >= h0 && h < h0 + TS
+                       && h < H;
       #pragma unroll
       for (int mi = 0; mi < 2; ++mi) {
         const int c = cb + mi * 16 + (lane >> 4) * 4;
@@ -776,10 +778,10 @@ __global__ __launch_bounds__(256) void conv1d_dx_kernel(
         }
       }
     }
-    __syncthreads();                          // next m0 restages tile 0
+    __syncthreads();                          // scatter done before readout
   }
   bf16* dxi = dx + ((long)img * H + h0) * C;
-  const int HR = min(64, H - h0);
+  const int HR = min(TS, H - h0);
   for (int i = tid * 8; i < HR * C; i += 256 * 8) {
     const int r = i / C, c0 = i - r * C;
     const float* src = ldx + r * CP + c0;
@@ -802,7 +804,8 @@ bool launch_conv1d_dx(const void* dy2, long ldy, const void* wt, long ldw,
   const int lds = 64 * (C + 4) * 4 + 2 * 32 * outC * 2;
   if (C % 32 != 0 || outC % 32 != 0 || outC > 128 || lds > 56 * 1024)
     return false;
-  const int T = (H + 63) / 64;
+  const int TS = 64 - KH + 1;
+  const int T = (H + TS - 1) / TS;
   hipLaunchKernelGGL(conv1d_dx_kernel, dim3(B, T), dim3(256), lds, s,
                      (const bf16*)dy2, ldy, (const bf16*)wt, ldw, (bf16*)dx,
                      H, C, KH, PH, OH, outC, accumulate);
