@@ -26,6 +26,7 @@ struct GemmArgs {
   ConvGeom geom;
   float* stats_sum = nullptr;
   float* stats_sumsq = nullptr;
+  const void* addend = nullptr;
 };
 bool gemm_dispatch(const GemmArgs& g, hipStream_t s);
 void launch_mfma_probe(const void* A, const void* B, float* D, hipStream_t s);
@@ -123,7 +124,8 @@ void check_f32(const at::Tensor& t, const char* name) {
 // config covers the shape (python falls back to a library GEMM).
 bool gemm(at::Tensor A, at::Tensor B, at::Tensor C,
           c10::optional<at::Tensor> bias, bool ta, bool tb, int64_t epi,
-          int64_t splits, c10::optional<at::Tensor> stats = c10::nullopt) {
+          int64_t splits, c10::optional<at::Tensor> stats = c10::nullopt,
+          c10::optional<at::Tensor> addend = c10::nullopt) {
   check_bf16(A, "A");
   check_bf16(B, "B");
   TORCH_CHECK(C.is_cuda() && C.is_contiguous(), "C must be contiguous GPU");
@@ -143,6 +145,12 @@ bool gemm(at::Tensor A, at::Tensor B, at::Tensor C,
   lo::GemmArgs g{A.data_ptr(), B.data_ptr(), C.data_ptr(), bias_p,
                  A.size(1), B.size(1), C.size(1), M, N, K, ta, tb,
                  (int)epi, out_f32, (int)splits, 0, {}};
+  if (addend.has_value()) {
+    check_bf16(*addend, "addend");
+    TORCH_CHECK(!out_f32 && splits == 1, "addend needs bf16 non-split-K out");
+    TORCH_CHECK(addend->sizes() == C.sizes(), "addend/C shape mismatch");
+    g.addend = addend->data_ptr();
+  }
   if (stats.has_value()) {
     check_f32(*stats, "stats");
     TORCH_CHECK(stats->numel() == 2 * N, "stats must be [2, N]");
@@ -592,7 +600,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm", &gemm, "bf16 MFMA GEMM (gfx950)",
         py::arg("A"), py::arg("B"), py::arg("C"), py::arg("bias") = py::none(),
         py::arg("ta") = false, py::arg("tb") = false, py::arg("epi") = 0,
-        py::arg("splits") = 1, py::arg("stats") = py::none());
+        py::arg("splits") = 1, py::arg("stats") = py::none(),
+        py::arg("addend") = py::none());
   m.def("gemm_conv_fwd", &gemm_conv_fwd);
   m.def("gemm_conv_dw", &gemm_conv_dw);
   m.def("mfma_probe", &mfma_probe);

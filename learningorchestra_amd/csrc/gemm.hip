@@ -83,7 +83,8 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
     const bf16* __restrict__ A, long lda, const bf16* __restrict__ B, long ldb,
     void* __restrict__ Cv, long ldc, const float* __restrict__ bias,
     int M, int N, int K, int kStart, int kChunk, ConvGeom geom,
-    float* __restrict__ stats_sum, float* __restrict__ stats_sumsq) {
+    float* __restrict__ stats_sum, float* __restrict__ stats_sumsq,
+    const bf16* __restrict__ addend) {
   constexpr int T = WM * WN * 64;
   constexpr int WTM = BM / WM;          // wave tile rows
   constexpr int WTN = BN / WN;          // wave tile cols
@@ -411,11 +412,21 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
     const char* src = ct + rowL * CROWB + ch * 16;
     if (gc + VE <= N) {
       if (OUT_F32) *(f32x4*)(Cf + (long)gr * ldc + gc) = *(const f32x4*)src;
-      else *(bf16x8*)(Cb + (long)gr * ldc + gc) = *(const bf16x8*)src;
+      else if (addend) {
+        const bf16x8 d = *(const bf16x8*)(addend + (long)gr * ldc + gc);
+        bf16x8 v = *(const bf16x8*)src;
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) v[e] = tobf16(tofloat(v[e]) + tofloat(d[e]));
+        *(bf16x8*)(Cb + (long)gr * ldc + gc) = v;
+      } else *(bf16x8*)(Cb + (long)gr * ldc + gc) = *(const bf16x8*)src;
     } else if (gc < N) {
       for (int e = 0; e < VE && gc + e < N; ++e) {
         if (OUT_F32) Cf[(long)gr * ldc + gc + e] = ((const float*)src)[e];
-        else Cb[(long)gr * ldc + gc + e] = ((const bf16*)src)[e];
+        else {
+          bf16 v = ((const bf16*)src)[e];
+          if (addend) v = tobf16(tofloat(v) + tofloat(addend[(long)gr * ldc + gc + e]));
+          Cb[(long)gr * ldc + gc + e] = v;
+        }
       }
     }
   }
@@ -497,6 +508,10 @@ struct GemmArgs {
   // squares of C accumulated by the epilogue (atomics; caller zeroes)
   float* stats_sum = nullptr;
   float* stats_sumsq = nullptr;
+  // optional fused residual add: C = op(A)@op(B) + addend (bf16, same
+  // shape/ldc as C; bf16-out, non-split-K only) — the ResNet join without
+  // a separate add pass
+  const void* addend = nullptr;
 };
 
 template <int BM, int BN, int BK, int WM, int WN, bool TA, bool TB,
@@ -520,23 +535,25 @@ static void launch_cfg(const GemmArgs& g, hipStream_t s) {
       gemm_kernel<BM, BN, BK, WM, WN, TA, TB, EPI, OUT_F32, ATOMIC, GATHER>),
       grid, block, lds, s,
       (const bf16*)g.A, g.lda, (const bf16*)g.B, g.ldb, g.C, g.ldc, g.bias,
-      g.M, g.N, g.K, 0, kChunk, g.geom, g.stats_sum, g.stats_sumsq);
+      g.M, g.N, g.K, 0, kChunk, g.geom, g.stats_sum, g.stats_sumsq,
+      (g.out_f32 || g.splits > 1) ? nullptr : (const bf16*)g.addend);
 }
 
 bool launch_gemm256(const void* A, long lda, const void* B, long ldb, void* C,
                     long ldc, const float* bias, int M, int N, int K, int epi,
-                    hipStream_t s);  // gemm_8phase.hip
+                    const void* addend, hipStream_t s);  // gemm_8phase.hip
 
 // Per-(TA,TB) tile-size selection. Instantiates only the combos the engine
 // uses (fwd = N,T; dX = N,N; dW = T,N split-K) plus (T,T) for completeness.
 template <bool TA, bool TB>
 static bool dispatch_tiles(const GemmArgs& g, hipStream_t s) {
-  if (!TA && TB && !g.out_f32 && g.M % 256 == 0 && g.N % 256 == 0 &&
+  if (!TA && TB && !g.out_f32 && g.stats_sum == nullptr &&
+      g.M % 256 == 0 && g.N % 256 == 0 &&
       g.K % 64 == 0 && g.K >= 256 &&
       (g.M / 256) * (g.N / 256) >= 192) {  // needs a chip-filling grid
     // deep-pipelined 256^2 8-phase path (gemm_8phase.hip)
     if (launch_gemm256(g.A, g.lda, g.B, g.ldb, g.C, g.ldc, g.bias,
-                       g.M, g.N, g.K, g.epi, s))
+                       g.M, g.N, g.K, g.epi, g.addend, s))
       return true;
   }
   #define LO_EPI_CASES_BK(BM_, BN_, BK_, WM_, WN_)                             \

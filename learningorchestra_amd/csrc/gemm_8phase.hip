@@ -206,11 +206,11 @@ bool launch_gemm256(const void* A, long lda, const void* B, long ldb, void* C,
   if (epi == 1)
     hipLaunchKernelGGL(HIP_KERNEL_NAME(gemm256_kernel<1>), dim3(grid), dim3(512),
                        lds, s, (const bf16*)A, lda, (const bf16*)B, ldb,
-                       (bf16*)C, ldc, bias, M, N, K);
+                       (bf16*)C, ldc, bias, M, N, K, (const bf16*)addend);
   else
     hipLaunchKernelGGL(HIP_KERNEL_NAME(gemm256_kernel<0>), dim3(grid), dim3(512),
                        lds, s, (const bf16*)A, lda, (const bf16*)B, ldb,
-                       (bf16*)C, ldc, bias, M, N, K);
+                       (bf16*)C, ldc, bias, M, N, K, (const bf16*)addend);
   return true;
 }
 

@@ -196,7 +196,8 @@ class Conv2dNHWC(Layer):
 
     def backward(self, dy: torch.Tensor,
                  dx_out: Optional[torch.Tensor] = None,
-                 dx_accumulate: bool = False) -> Optional[torch.Tensor]:
+                 dx_accumulate: bool = False,
+                 dx_addend: Optional[torch.Tensor] = None) -> Optional[torch.Tensor]:
         """``dx_out``/``dx_accumulate``: on the fused 1-D dX path, write (or
         add) the input grad straight into the caller's buffer — fuses the
         multi-branch grad sum (TextCNN).  Callers must check the returned
@@ -223,10 +224,15 @@ class Conv2dNHWC(Layer):
         if self.first:
             return None
         if self._is_1x1:
-            # 1x1/s1 conv: col IS x, so dcol IS dx
+            # 1x1/s1 conv: col IS x, so dcol IS dx; dx_addend fuses the
+            # residual-join add into the GEMM epilogue (all gemm paths --
+            # native, library fallback, CPU reference -- apply it)
             dx = bufs["dx"]
-            F.gemm(dy2, self._wt(), tb=True, out=dx.view(M, self.in_c))
+            F.gemm(dy2, self._wt(), tb=True, out=dx.view(M, self.in_c),
+                   addend=dx_addend.view(M, self.in_c)
+                   if dx_addend is not None else None)
             return dx
+        assert dx_addend is None or self._is_1x1, "dx_addend: 1x1 only"
         if dy2.is_cuda and self.implicit and self._conv1d_ok(bufs["W"]):
             tgt = dx_out if dx_out is not None else bufs["dx"]
             if F.conv1d_dx(dy2, self._wt(), self.kh, self.pad, out=tgt,

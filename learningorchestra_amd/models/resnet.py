@@ -92,15 +92,12 @@ class Bottleneck:
         dh = self.bn2.backward(dh)
         dh = self.conv2.backward(dh)
         dh = self.bn1.backward(dh)
-        dx_main = self.conv1.backward(dh)
         if self.downsample:
             d_idt = self.downsample[0].backward(self.downsample[1].backward(dsum))
         else:
             d_idt = dsum
-        if self._dx is None or self._dx.shape != dx_main.shape:
-            self._dx = torch.empty_like(dx_main)
-        F.add_relu(dx_main, d_idt, out=self._dx, relu=False)
-        return self._dx
+        # branch-grad sum fused into conv1's dX GEMM epilogue (C = A@B + D)
+        return self.conv1.backward(dh, dx_addend=d_idt)
 
 
 class ResNet:

@@ -27,7 +27,8 @@ def _is_gpu(t: torch.Tensor) -> bool:
 def gemm(A: torch.Tensor, B: torch.Tensor, *, ta: bool = False, tb: bool = False,
          bias: Optional[torch.Tensor] = None, relu: bool = False,
          out: Optional[torch.Tensor] = None, out_dtype: Optional[torch.dtype] = None,
-         splits: int = 1, stats: Optional[torch.Tensor] = None) -> torch.Tensor:
+         splits: int = 1, stats: Optional[torch.Tensor] = None,
+         addend: Optional[torch.Tensor] = None) -> torch.Tensor:
     """C[M,N] = op(A) @ op(B) (+bias) (+relu). bf16 inputs, fp32 accumulate.
 
     op(A) = A.T if ta (A stored [K,M]); op(B) = B.T if tb (B stored [N,K]).
@@ -46,7 +47,7 @@ def gemm(A: torch.Tensor, B: torch.Tensor, *, ta: bool = False, tb: bool = False
         if stats is not None:
             stats.zero_()  # epilogue accumulates per-column sum/sumsq
         ok = lo.gemm(A, B, out, bias, ta, tb, 1 if relu else 0, splits,
-                     stats.view(-1) if stats is not None else None)
+                     stats.view(-1) if stats is not None else None, addend)
         if ok:
             return out
         # cold-path shapes: plain library GEMM (rocBLAS via torch.matmul)
@@ -57,6 +58,8 @@ def gemm(A: torch.Tensor, B: torch.Tensor, *, ta: bool = False, tb: bool = False
         c = c + bias.float()
     if relu:
         c = torch.relu(c)
+    if addend is not None:
+        c = c + addend.reshape(c.shape).float()
     out.copy_(c.to(out.dtype))
     return out
 

@@ -402,3 +402,36 @@ def test_conv1d_fwd_dx_match_reference(shape):
     err_fused = (dx.float().cpu() - ref32).abs().max().item()
     err_ref = (refdx.float().cpu() - ref32).abs().max().item()
     assert err_fused <= max(2 * err_ref, 1e-3), (err_fused, err_ref)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("mnk", [(512, 64, 128), (51200, 256, 512)])
+def test_gemm_addend_fused(mnk):
+    """C = A@B^T + D epilogue (the ResNet join add) on both the tile and the
+    256^2 8-phase kernels."""
+    from learningorchestra_amd.ops import functional as F
+    M, N, K = mnk
+    torch.manual_seed(4)
+    A = torch.randn(M, K, device="cuda").to(torch.bfloat16)
+    B = torch.randn(N, K, device="cuda").to(torch.bfloat16)
+    D = torch.randn(M, N, device="cuda").to(torch.bfloat16)
+    out = F.gemm(A, B, tb=True, addend=D)
+    ref = (A.float() @ B.float().t()) + D.float()
+    torch.testing.assert_close(out.float(), ref, atol=2.0, rtol=2e-2)
+
+
+@pytest.mark.gpu
+def test_gemm_stats_on_8phase_eligible_shape():
+    """Fused column stats must be produced even for shapes the stats-less
+    8-phase kernel would otherwise grab (regression: the gate now checks)."""
+    from learningorchestra_amd.ops import functional as F
+    M, N, K = 51200, 256, 512
+    torch.manual_seed(5)
+    A = torch.randn(M, K, device="cuda").to(torch.bfloat16) * 0.1
+    B = torch.randn(N, K, device="cuda").to(torch.bfloat16) * 0.1
+    st = torch.empty(2, N, device="cuda", dtype=torch.float32)
+    out = F.gemm(A, B, tb=True, stats=st)
+    ref_sum = out.float().sum(0)
+    ref_sq = out.float().square().sum(0)
+    torch.testing.assert_close(st[0], ref_sum, atol=ref_sum.abs().max().item() * 2e-2 + 1.0, rtol=2e-2)
+    torch.testing.assert_close(st[1], ref_sq, atol=ref_sq.abs().max().item() * 2e-2 + 1.0, rtol=2e-2)
