@@ -31,7 +31,7 @@ template <int EPI>
 __global__ __launch_bounds__(512) void gemm256_kernel(
     const bf16* __restrict__ A, long lda, const bf16* __restrict__ B, long ldb,
     bf16* __restrict__ C, long ldc, const float* __restrict__ bias,
-    int M, int N, int K) {
+    int M, int N, int K, const bf16* __restrict__ addend) {
   constexpr int BM = 256, BN = 256, BK = 64;
   constexpr int BKB = BK * 2;               // 128-B LDS rows
   constexpr int MFRAG = 8, NFRAG = 4;       // per-wave 128x64 output
@@ -189,7 +189,13 @@ __global__ __launch_bounds__(512) void gemm256_kernel(
       const int rowL = c / NCH, ch = c % NCH;
       const long gr = m0 + h * 128 + rowL;
       const int gc = n0 + ch * 8;
-      *(bf16x8*)(C + gr * ldc + gc) = *(const bf16x8*)(ct + rowL * CROWB + ch * 16);
+      bf16x8 v = *(const bf16x8*)(ct + rowL * CROWB + ch * 16);
+      if (addend) {
+        const bf16x8 d = *(const bf16x8*)(addend + gr * ldc + gc);
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) v[e] = tobf16(tofloat(v[e]) + tofloat(d[e]));
+      }
+      *(bf16x8*)(C + gr * ldc + gc) = v;
     }
     __syncthreads();
   }
@@ -197,7 +203,7 @@ __global__ __launch_bounds__(512) void gemm256_kernel(
 
 bool launch_gemm256(const void* A, long lda, const void* B, long ldb, void* C,
                     long ldc, const float* bias, int M, int N, int K, int epi,
-                    hipStream_t s) {
+                    const void* addend, hipStream_t s) {
   if (M % 256 != 0 || N % 256 != 0 || K % 64 != 0 || K < 256) return false;
   const int grid = (M / 256) * (N / 256);
   constexpr size_t LDS_MAIN = 2 * (256 + 256) * 128;        // 128 KiB
