@@ -541,19 +541,20 @@ static void launch_cfg(const GemmArgs& g, hipStream_t s) {
 
 bool launch_gemm256(const void* A, long lda, const void* B, long ldb, void* C,
                     long ldc, const float* bias, int M, int N, int K, int epi,
-                    const void* addend, hipStream_t s);  // gemm_8phase.hip
+                    const void* addend, float* stats_sum, float* stats_sumsq,
+                    hipStream_t s);  // gemm_8phase.hip
 
 // Per-(TA,TB) tile-size selection. Instantiates only the combos the engine
 // uses (fwd = N,T; dX = N,N; dW = T,N split-K) plus (T,T) for completeness.
 template <bool TA, bool TB>
 static bool dispatch_tiles(const GemmArgs& g, hipStream_t s) {
-  if (!TA && TB && !g.out_f32 && g.stats_sum == nullptr &&
-      g.M % 256 == 0 && g.N % 256 == 0 &&
+  if (!TA && TB && !g.out_f32 && g.M % 256 == 0 && g.N % 256 == 0 &&
       g.K % 64 == 0 && g.K >= 256 &&
       (g.M / 256) * (g.N / 256) >= 192) {  // needs a chip-filling grid
     // deep-pipelined 256^2 8-phase path (gemm_8phase.hip)
     if (launch_gemm256(g.A, g.lda, g.B, g.ldb, g.C, g.ldc, g.bias,
-                       g.M, g.N, g.K, g.epi, g.addend, s))
+                       g.M, g.N, g.K, g.epi, g.addend, g.stats_sum,
+                       g.stats_sumsq, s))
       return true;
   }
   #define LO_EPI_CASES_BK(BM_, BN_, BK_, WM_, WN_)                             \

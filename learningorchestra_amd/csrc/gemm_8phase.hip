@@ -31,7 +31,8 @@ template <int EPI>
 __global__ __launch_bounds__(512) void gemm256_kernel(
     const bf16* __restrict__ A, long lda, const bf16* __restrict__ B, long ldb,
     bf16* __restrict__ C, long ldc, const float* __restrict__ bias,
-    int M, int N, int K, const bf16* __restrict__ addend) {
+    int M, int N, int K, const bf16* __restrict__ addend,
+    float* __restrict__ stats_sum, float* __restrict__ stats_sumsq) {
   constexpr int BM = 256, BN = 256, BK = 64;
   constexpr int BKB = BK * 2;               // 128-B LDS rows
   constexpr int MFRAG = 8, NFRAG = 4;       // per-wave 128x64 output
@@ -197,26 +198,57 @@ __global__ __launch_bounds__(512) void gemm256_kernel(
       }
       *(bf16x8*)(C + gr * ldc + gc) = v;
     }
+    // fused BatchNorm statistics over the staged half (same LDS
+    // pre-reduction pattern as gemm.hip's tile epilogue)
+    if (stats_sum != nullptr) {
+      float* lacc = (float*)(ct + 128 * CROWB + 16);   // [2][BN]
+      for (int i = tid; i < 2 * BN; i += 512) lacc[i] = 0.f;
+      __syncthreads();
+      const int tc = tid & 31, tr = tid >> 5;          // 32 col-chunks x 16
+      float s1[8] = {}, s2[8] = {};
+      for (int rowL = tr; rowL < 128; rowL += 16) {
+        bf16x8 vv = *(const bf16x8*)(ct + rowL * CROWB + tc * 16);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float f = tofloat(vv[j]);
+          s1[j] += f;
+          s2[j] += f * f;
+        }
+      }
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        atomicAdd(lacc + tc * 8 + j, s1[j]);
+        atomicAdd(lacc + BN + tc * 8 + j, s2[j]);
+      }
+      __syncthreads();
+      for (int i = tid; i < BN; i += 512) {
+        atomicAdd(stats_sum + n0 + i, lacc[i]);
+        atomicAdd(stats_sumsq + n0 + i, lacc[BN + i]);
+      }
+    }
     __syncthreads();
   }
 }
 
 bool launch_gemm256(const void* A, long lda, const void* B, long ldb, void* C,
                     long ldc, const float* bias, int M, int N, int K, int epi,
-                    const void* addend, hipStream_t s) {
+                    const void* addend, float* stats_sum, float* stats_sumsq,
+                    hipStream_t s) {
   if (M % 256 != 0 || N % 256 != 0 || K % 64 != 0 || K < 256) return false;
   const int grid = (M / 256) * (N / 256);
   constexpr size_t LDS_MAIN = 2 * (256 + 256) * 128;        // 128 KiB
-  constexpr size_t LDS_EPI = 128 * (256 * 2 + 16);
+  constexpr size_t LDS_EPI = 128 * (256 * 2 + 16) + 16 + 2 * 256 * 4;
   const size_t lds = LDS_MAIN > LDS_EPI ? LDS_MAIN : LDS_EPI;
   if (epi == 1)
     hipLaunchKernelGGL(HIP_KERNEL_NAME(gemm256_kernel<1>), dim3(grid), dim3(512),
                        lds, s, (const bf16*)A, lda, (const bf16*)B, ldb,
-                       (bf16*)C, ldc, bias, M, N, K, (const bf16*)addend);
+                       (bf16*)C, ldc, bias, M, N, K, (const bf16*)addend,
+                       stats_sum, stats_sumsq);
   else
     hipLaunchKernelGGL(HIP_KERNEL_NAME(gemm256_kernel<0>), dim3(grid), dim3(512),
                        lds, s, (const bf16*)A, lda, (const bf16*)B, ldb,
-                       (bf16*)C, ldc, bias, M, N, K, (const bf16*)addend);
+                       (bf16*)C, ldc, bias, M, N, K, (const bf16*)addend,
+                       stats_sum, stats_sumsq);
   return true;
 }
 
