@@ -1032,9 +1032,11 @@ __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
         const unsigned char want = (unsigned char)(kh * KW + kw);
         if (VEC8) {
           bf16x8 g = *(const bf16x8*)(dy + obase);
+          // one 8-byte load instead of 8 scalar u8 loads
+          const unsigned long long iv = *(const unsigned long long*)(idx + obase);
           #pragma unroll
           for (int j = 0; j < 8; ++j)
-            if (idx[obase + j] == want) acc[j] += tofloat(g[j]);
+            if (((iv >> (j * 8)) & 0xffu) == want) acc[j] += tofloat(g[j]);
         } else {
           if (idx[obase] == want) acc[0] += tofloat(dy[obase]);
         }
