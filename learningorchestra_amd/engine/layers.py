@@ -126,7 +126,10 @@ class Conv2dNHWC(Layer):
 
     def _conv1d_ok(self, W: int) -> bool:
         # mirror of launch_conv1d_* eligibility (W==1 sequence convs)
+        # pad must be 0: Conv2dNHWC pads BOTH dims, and w-padding of a
+        # W==1 input (OW = 2p+1) has no 1-D representation
         return (W == 1 and self.kw == 1 and self.stride == 1 and
+                self.pad == 0 and
                 self.in_c % 32 == 0 and self.out_c % 64 == 0 and
                 self.out_c <= 128 and (self.kh * self.in_c) % 32 == 0 and
                 (64 + self.kh - 1) * (self.in_c * 2 + 16) + 128 + 8192

@@ -61,8 +61,9 @@ def main(iters=60, seed=0):
             torch.testing.assert_close(y.float(), ref.float(), atol=0.1,
                                        rtol=5e-2, msg=lambda m: f"fwd {tag}\n{m}")
 
-        # --- fwd 1-D ----------------------------------------------------
-        if W == 1 and KW == 1 and S == 1:
+        # --- fwd 1-D (P applies to h only in the 1-D kernels; the fuzz
+        # dy2/x use the 2-D geometry, so only P==0 is comparable) ---------
+        if W == 1 and KW == 1 and S == 1 and P == 0:
             y1 = torch.empty(B * OH, outC, device="cuda", dtype=torch.bfloat16)
             if F.conv1d_fwd(x, w, KH, P, bias=bias, relu=False, out=y1):
                 tried["fwd1d"] += 1
@@ -84,7 +85,7 @@ def main(iters=60, seed=0):
             assert err < 0.03 * scale + 0.05, f"dx {tag}: err {err} scale {scale}"
 
         # --- dX 1-D -----------------------------------------------------
-        if W == 1 and KW == 1 and S == 1:
+        if W == 1 and KW == 1 and S == 1 and P == 0:
             dx1 = torch.empty(B, H, 1, C, device="cuda", dtype=torch.bfloat16)
             if F.conv1d_dx(dy2, wt, KH, P, out=dx1):
                 tried["dx1d"] += 1
