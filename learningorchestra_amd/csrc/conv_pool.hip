@@ -714,8 +714,11 @@ __global__ __launch_bounds__(256) void conv1d_dx_kernel(
   const int KCH = outC / 32;                  // <= 4
   const int NSTEP = KH * (C / 32);
   const int TB = 32 * outC * 2;               // wt tile bytes (<= 8 KB)
-  auto wswz = [](int row, int kel) {
-    return (kel ^ ((row & 7) << 3));
+  // swizzle must stay inside the row: mask to the row's 16-B chunk count
+  // (outC/8) — fuzz-found out-of-row reads for outC < 64
+  const int cmask = (outC / 8 - 1) & 7;
+  auto wswz = [cmask](int row, int kel) {
+    return (kel ^ ((row & cmask) << 3));
   };
   auto stage_tile = [&](int buf, int step) {
     // tile: 32 kpad-rows x outC, k-contiguous; row stride outC*2 bytes

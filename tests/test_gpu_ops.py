@@ -369,6 +369,7 @@ def test_conv_dw_c1_matches_splitk(shape):
     (8, 256, 128, 3, 0, 128),    # TextCNN k3
     (4, 100, 32, 4, 2, 64),      # padding, odd H (tile tail)
     (2, 64, 64, 3, 1, 128),      # single tile
+    (2, 8, 128, 1, 0, 32),       # outC 32 (swizzle row-mask regression)
 ])
 def test_conv1d_fwd_dx_match_reference(shape):
     """1-D tiled conv fwd + dX vs im2col GEMM / dcol col2im references."""
