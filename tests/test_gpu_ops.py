@@ -385,10 +385,12 @@ def test_conv1d_fwd_dx_match_reference(shape):
     bias = torch.randn(outC, device="cuda", dtype=torch.float32)
     y = torch.empty(B * OH, outC, device="cuda", dtype=torch.bfloat16)
     ok = F.conv1d_fwd(x, w, KH, P, bias=bias, relu=True, out=y)
-    assert ok, "fwd shape should be eligible"
-    col = F.im2col(x, KH, 1, 1, 1, P, 0, kpad)
-    ref = F.gemm(col, w, tb=True, bias=bias, relu=True)
-    torch.testing.assert_close(y.float(), ref.float(), atol=3e-2, rtol=3e-2)
+    assert ok == (outC % 64 == 0), "fwd eligibility (outC in 64-slices)"
+    if ok:
+        col = F.im2col(x, KH, 1, 1, 1, P, 0, kpad)
+        ref = F.gemm(col, w, tb=True, bias=bias, relu=True)
+        torch.testing.assert_close(y.float(), ref.float(), atol=3e-2,
+                                   rtol=3e-2)
 
     dy2 = torch.randn(B * OH, outC, device="cuda").to(torch.bfloat16)
     wt = w.t().contiguous()
