@@ -438,3 +438,17 @@ def test_gemm_stats_on_8phase_eligible_shape():
     ref_sq = out.float().square().sum(0)
     torch.testing.assert_close(st[0], ref_sum, atol=ref_sum.abs().max().item() * 2e-2 + 1.0, rtol=2e-2)
     torch.testing.assert_close(st[1], ref_sq, atol=ref_sq.abs().max().item() * 2e-2 + 1.0, rtol=2e-2)
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(300)
+def test_conv_family_shape_fuzz():
+    """Randomized-shape sweep of the fused conv kernels vs references
+    (tools/conv_fuzz.py; caught an out-of-row swizzle for outC<64)."""
+    import importlib.util
+    from pathlib import Path
+    spec = importlib.util.spec_from_file_location(
+        "conv_fuzz", Path(__file__).parent.parent / "tools" / "conv_fuzz.py")
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    mod.main(iters=60, seed=3)
