@@ -36,24 +36,23 @@ def _worker(rank: int, world: int, port: int, q):
         q.put((rank, f"ERROR: {exc!r}", False))
 
 
-@pytest.mark.timeout(120)
-def test_ddp_gloo_world2_params_stay_identical():
-    world = 2
-    port = 29511
+@pytest.mark.timeout(240)
+@pytest.mark.parametrize("world,port", [(2, 29511), (4, 29515)])
+def test_ddp_gloo_params_stay_identical(world, port):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_worker, args=(r, world, port, q))
              for r in range(world)]
     for p in procs:
         p.start()
-    results = [q.get(timeout=110) for _ in range(world)]
+    results = [q.get(timeout=220) for _ in range(world)]
     for p in procs:
         p.join(timeout=30)
     for rank, checksum, same in results:
         assert not isinstance(checksum, str), checksum
         assert same, "params diverged across ranks"
-    # both ranks ended with bit-identical master weights
-    assert abs(results[0][1] - results[1][1]) == 0.0
+    # every rank ended with bit-identical master weights
+    assert len({c for _, c, _ in results}) == 1
 
 
 def test_single_process_allreduce_noop():
