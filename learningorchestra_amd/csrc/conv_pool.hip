@@ -1002,7 +1002,9 @@ __global__ void maxpool_fwd_kernel(const bf16* __restrict__ in, bf16* __restrict
 // relu_y != null fuses the upstream conv's ReLU backward: dx is zeroed
 // where the conv output (the pool INPUT, same [b,h,w,c] position) was
 // clamped — saves the separate relu_bwd pass over the full conv activation.
-template <bool VEC8>
+// K22: compile-time 2x2/stride-2/pad-0 specialization — the generic window
+// bounds cost ~1 runtime integer division per element (half roofline).
+template <bool VEC8, bool K22 = false>
 __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
                                    const unsigned char* __restrict__ idx,
                                    bf16* __restrict__ dx,
@@ -1022,15 +1024,17 @@ __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
     float acc[NE];
     #pragma unroll
     for (int j = 0; j < NE; ++j) acc[j] = 0.f;
-    const int hp = h + PH, wp = w + PW;
-    const int oh_lo = max(0, (hp - KH + SH) / SH), oh_hi = min(OH - 1, hp / SH);
-    const int ow_lo = max(0, (wp - KW + SW) / SW), ow_hi = min(OW - 1, wp / SW);
+    const int hp = h + (K22 ? 0 : PH), wp = w + (K22 ? 0 : PW);
+    const int oh_lo = K22 ? (h >> 1) : max(0, (hp - KH + SH) / SH);
+    const int oh_hi = K22 ? min(OH - 1, h >> 1) : min(OH - 1, hp / SH);
+    const int ow_lo = K22 ? (w >> 1) : max(0, (wp - KW + SW) / SW);
+    const int ow_hi = K22 ? min(OW - 1, w >> 1) : min(OW - 1, wp / SW);
     for (int oh = oh_lo; oh <= oh_hi; ++oh) {
-      const int kh = hp - oh * SH;
-      if (kh < 0 || kh >= KH) continue;
+      const int kh = K22 ? (h & 1) : (hp - oh * SH);
+      if (!K22 && (kh < 0 || kh >= KH)) continue;
       for (int ow = ow_lo; ow <= ow_hi; ++ow) {
-        const int kw = wp - ow * SW;
-        if (kw < 0 || kw >= KW) continue;
+        const int kw = K22 ? (w & 1) : (wp - ow * SW);
+        if (!K22 && (kw < 0 || kw >= KW)) continue;
         const long obase = (((long)b * OH + oh) * OW + ow) * C + cu * NE;
         const unsigned char want = (unsigned char)(kh * KW + kw);
         if (VEC8) {
@@ -1160,7 +1164,13 @@ void launch_maxpool_bwd(const void* dy, const void* idx, void* dx,
   const long total = (long)B * H * W * (vec ? C / 8 : C);
   const int block = 256;
   const int grid = (int)min((total + block - 1) / block, (long)2048);
-  if (vec)
+  const bool k22 = (KH == 2 && KW == 2 && SH == 2 && SW == 2 &&
+                    PH == 0 && PW == 0);
+  if (vec && k22)
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(maxpool_bwd_kernel<true, true>), dim3(grid), dim3(block), 0, s,
+                       (const bf16*)dy, (const unsigned char*)idx, (bf16*)dx,
+                       (const bf16*)relu_y, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
+  else if (vec)
     hipLaunchKernelGGL(HIP_KERNEL_NAME(maxpool_bwd_kernel<true>), dim3(grid), dim3(block), 0, s,
                        (const bf16*)dy, (const unsigned char*)idx, (bf16*)dx,
                        (const bf16*)relu_y, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
