@@ -22,7 +22,6 @@ from __future__ import annotations
 import importlib
 import inspect
 import io
-import sys
 import time
 import traceback
 from contextlib import redirect_stdout

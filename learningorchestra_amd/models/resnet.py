@@ -5,7 +5,8 @@ scaling curve").
 Standard bottleneck-v1 topology; every hot op is a gfx950 HIP kernel:
 * convs: 1x1/s1 as direct MFMA GEMM (no im2col), others im2col + MFMA GEMM,
   backward-data via stride-general gather col2im;
-* BN+ReLU fused fwd/bwd (batchnorm.hip), residual joins via add_relu;
+* BN+ReLU fused fwd/bwd (batchnorm.hip); forward joins fused into bn3,
+  backward join add fused into conv1's dX GEMM epilogue;
 * 3x3/s2 maxpool with padding, global average pool, padded-head linear,
   fused softmax-CE (wave path, 1000 classes).
 """
@@ -43,7 +44,6 @@ class Bottleneck:
                            relu=False, bias=False),
                 BatchNormReLU(f"{name}.dsbn", out_c, relu=False))
         self._z = None
-        self._dx = None
         self._dsum = None
 
     def layers(self):
