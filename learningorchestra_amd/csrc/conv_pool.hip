@@ -1,15 +1,25 @@
-// Conv2d support kernels for gfx950 — NHWC layout throughout.
+// Conv2d kernels for gfx950 — NHWC layout throughout.
 //
-// The CNN path (SURVEY.md §2.9 "Conv2d fwd/bwd -> implicit-GEMM conv") is
-// im2col + the MFMA GEMM of gemm.hip: NHWC makes every im2col row
-// k-contiguous (kh,kw,c) and the GEMM output [B*OH*OW, outC] directly the
-// next layer's NHWC activation — no transposes anywhere in the hot loop.
+// The CNN path (SURVEY.md §2.9 "Conv2d fwd/bwd -> implicit-GEMM conv") has
+// two tiers:
 //
-// * im2col:  in[B,H,W,C] -> col[B*OH*OW, Kpad] (Kpad >= KH*KW*C, %8==0, the
-//            pad tail is pre-zeroed by the allocator and never written)
-// * col2im:  gather-style dcol -> dX (no atomics; each input element sums its
-//            covering patches)
-// * maxpool2d fwd/bwd with u8 argmax indices (non-overlap needs no atomics)
+// 1. FUSED kernels (no col/dcol matrices at all) for shapes whose working
+//    set fits LDS — these carry the MNIST and TextCNN hot loops:
+//    * conv_fwd_small:  one block per image, x LDS-resident, w tiles
+//      double-buffered (C=1 scalar-gather specialization for 28x28x1)
+//    * conv_dx:         per-image MFMA dcol tiles scattered into an LDS
+//      fp32 dx accumulator; swapped operand roles (D = Wt . dY^T) make a
+//      lane's 4 acc regs consecutive CHANNELS -> one b128 RMW per fragment
+//    * conv_dw_c1:      C=1 dW with per-wave transposed-dy LDS staging
+//    * conv1d_fwd/dx:   h-tiled variants for W==1 sequence convs (TextCNN)
+//
+// 2. The general fallback: im2col + the MFMA GEMM of gemm.hip (NHWC makes
+//    every im2col row k-contiguous) and gather-style col2im for dX.
+//    * im2col:  in[B,H,W,C] -> col[B*OH*OW, Kpad] (Kpad >= KH*KW*C, %8==0)
+//    * col2im:  dcol -> dX gather (no atomics)
+//
+// * maxpool2d fwd/bwd with u8 argmax indices (non-overlap needs no atomics;
+//   bwd optionally applies the upstream conv's ReLU mask)
 
 #include "lo_common.h"
 

@@ -46,7 +46,10 @@ class Layer:
 
 
 class Conv2dNHWC(Layer):
-    """conv = im2col + MFMA GEMM (+fused bias/ReLU epilogue)."""
+    """NHWC conv on the fused kernel family (conv_fwd_small / conv_dx /
+    conv_dw_c1 / conv1d_* when the shape qualifies — no col/dcol matrices)
+    with im2col + MFMA GEMM (+fused bias/ReLU epilogue) as the general
+    fallback; 1x1/s1 convs skip im2col entirely."""
 
     def __init__(self, name: str, in_c: int, out_c: int, kh: int, kw: int,
                  stride: int = 1, pad: int = 0, relu: bool = True,
@@ -54,9 +57,10 @@ class Conv2dNHWC(Layer):
                  implicit: bool = False):
         self.name = name
         self.bias = bias
-        # implicit-GEMM conv (col gathered inside GEMM staging): numerically
-        # identical, saves the col buffer entirely, but measured slightly
-        # slower than glds-staged materialized col on MI355X — opt-in.
+        # implicit=True: no col matrix — the fused conv kernels (or the
+        # gather-GEMM) handle fwd/dW/dX directly. Measured per model:
+        # MNIST + TextCNN default on, ResNet stays on materialized col
+        # (3x3 convs are compute-dense; see PERFORMANCE.md A/B entries).
         self.implicit = implicit
         self.in_c, self.out_c = in_c, out_c
         self.kh, self.kw, self.stride, self.pad = kh, kw, stride, pad
