@@ -11,7 +11,23 @@ struct ConvGeom {
   int B, H, W, C;
   int KH, KW, SH, SW, PH, PW;
   int OH, OW;
+  unsigned mgOW = 0, mgOH = 0;
+  int sOW = 0, sOH = 0;
 };
+
+// magic-division constants (see gemm.hip fdiv): exact for n < 2^31
+static inline void fill_fastdiv(lo::ConvGeom& g) {
+  auto mk = [](unsigned d, unsigned& mg, int& s) {
+    if (d <= 1) { mg = 0; s = 0; return; }
+    s = 0;
+    while ((1u << s) < d) ++s;
+    if ((1u << s) == d) { mg = 1; return; }          // power of two
+    const unsigned long long L = 1ull << (32 + s);
+    mg = (unsigned)((L + d - 1) / d);
+  };
+  mk((unsigned)g.OW, g.mgOW, g.sOW);
+  mk((unsigned)g.OH, g.mgOH, g.sOH);
+}
 struct GemmArgs {
   const void *A, *B;
   void* C;
@@ -180,6 +196,7 @@ bool gemm_conv_fwd(at::Tensor x, at::Tensor Wt, at::Tensor C,
   if (bias.has_value()) { check_f32(*bias, "bias"); bias_p = bias->data_ptr<float>(); }
   lo::ConvGeom geom{B, H, W, Ci, (int)KH, (int)KW, (int)SH, (int)SW,
                     (int)PH, (int)PW, OH, OW};
+  fill_fastdiv(geom);
   lo::GemmArgs g{x.data_ptr(), Wt.data_ptr(), C.data_ptr(), bias_p,
                  K, Wt.size(1), C.size(1), M, N, K, false, true,
                  relu ? 1 : 0, false, 1, 1, geom};
@@ -202,6 +219,7 @@ bool gemm_conv_dw(at::Tensor dy2, at::Tensor x, at::Tensor gw, int64_t splits,
   TORCH_CHECK(K == (long)B * OH * OW && M == dy2.size(1), "conv dW shapes");
   lo::ConvGeom geom{B, H, W, Ci, (int)KH, (int)KW, (int)SH, (int)SW,
                     (int)PH, (int)PW, OH, OW};
+  fill_fastdiv(geom);
   lo::GemmArgs g{dy2.data_ptr(), x.data_ptr(), gw.data_ptr(), nullptr,
                  dy2.size(1), N, gw.size(1), M, N, (int)K, true, false,
                  0, true, (int)std::max<int64_t>(splits, 2), 2, geom};

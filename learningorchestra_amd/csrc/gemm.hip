@@ -35,7 +35,18 @@ struct ConvGeom {
   int B, H, W, C;        // NHWC input
   int KH, KW, SH, SW, PH, PW;
   int OH, OW;
+  // magic-division constants for OW/OH (filled by the host for gather
+  // paths): mg==0 -> divisor 1 (q=n), mg==1 -> power of two (q=n>>s),
+  // else q = mulhi(n, mg) >> s  (exact for n < 2^31)
+  unsigned mgOW = 0, mgOH = 0;
+  int sOW = 0, sOH = 0;
 };
+
+LO_DEVICE unsigned fdiv(unsigned n, unsigned mg, int s) {
+  if (mg == 0) return n;
+  if (mg == 1) return n >> s;
+  return (unsigned)(((unsigned long long)n * mg) >> 32) >> s;
+}
 
 // gather 8 consecutive col elements of row `m` starting at column `k`
 // (C % 8 == 0: the run lies in one (kh,kw) channel segment; C < 8: scalar)
@@ -135,6 +146,28 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
   constexpr int CHA_T = TA ? (BK * BM / 8 + T - 1) / T : 1;
   constexpr int CHB_T = (!TB) ? (BK * BN / 8 + T - 1) / T : 1;
   bf16x8 stgA[CHA_T], stgB[CHB_T];
+
+  // GATHER==2: this thread's B-tile columns are invariant across k-steps —
+  // decode (kh,kw,c) ONCE; the per-step row decode uses magic division
+  // (the full per-run conv_gather8 decode measured 28 VALU per MFMA).
+  constexpr int G2N = (GATHER == 2) ? CHB_T : 1;
+  int g2c[G2N], g2dh[G2N], g2dw[G2N];
+  bool g2ok[G2N];
+  if (GATHER == 2) {
+    #pragma unroll
+    for (int i = 0; i < G2N; ++i) {
+      const int c = tid + i * T;
+      const int nc = c % (BN / 8);
+      const int gn0 = n0 + nc * 8;
+      g2ok[i] = (c < BK * BN / 8) && gn0 < N &&
+                gn0 < geom.KH * geom.KW * geom.C;
+      const int cc = g2ok[i] ? gn0 % geom.C : 0;
+      const int p = g2ok[i] ? gn0 / geom.C : 0;
+      g2c[i] = cc;
+      g2dh[i] = p / geom.KW - geom.PH;
+      g2dw[i] = p % geom.KW - geom.PW;
+    }
+  }
 
   auto stage_next = [&](int buf, int k0) {
     char* smA = smem + buf * BUFB;
@@ -236,7 +269,17 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
           const int k = c / (BN / 8), nc = c % (BN / 8);
           const int gk = k0 + k, gn0 = n0 + nc * 8;
           if (GATHER == 2) {
-            if (gk < K && gn0 < N) v = conv_gather8(B, geom, (long)gk, gn0);
+            if (g2ok[i] && gk < K) {
+              const unsigned q1 = fdiv((unsigned)gk, geom.mgOW, geom.sOW);
+              const int ow = gk - (int)q1 * geom.OW;
+              const unsigned b = fdiv(q1, geom.mgOH, geom.sOH);
+              const int oh = (int)q1 - (int)b * geom.OH;
+              const int h = oh * geom.SH + g2dh[i];
+              const int w = ow * geom.SW + g2dw[i];
+              if (h >= 0 && h < geom.H && w >= 0 && w < geom.W)
+                v = *(const bf16x8*)(
+                    B + (((long)b * geom.H + h) * geom.W + w) * geom.C + g2c[i]);
+            }
           } else if (gk < K) {
             if (gn0 + 8 <= N) v = *(const bf16x8*)(B + (long)gk * ldb + gn0);
             else if (gn0 < N)
