@@ -21,9 +21,12 @@ static inline void fill_fastdiv(lo::ConvGeom& g) {
     if (d <= 1) { mg = 0; s = 0; return; }
     s = 0;
     while ((1u << s) < d) ++s;
-    if ((1u << s) == d) { mg = 1; return; }          // power of two
-    const unsigned long long L = 1ull << (32 + s);
+    if ((1u << s) == d) { mg = 1; return; }          // power of two: q = n>>s
+    // mg = ceil(2^(31+s)/d) fits u32 (2^s/d in (1,2)); device computes
+    // q = mulhi(n, mg) >> (s-1) == (n*mg) >> (31+s), exact for n < 2^31
+    const unsigned long long L = 1ull << (31 + s);
     mg = (unsigned)((L + d - 1) / d);
+    s = s - 1;
   };
   mk((unsigned)g.OW, g.mgOW, g.sOW);
   mk((unsigned)g.OH, g.mgOH, g.sOH);

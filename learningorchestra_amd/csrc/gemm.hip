@@ -159,7 +159,9 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
       const int c = tid + i * T;
       const int nc = c % (BN / 8);
       const int gn0 = n0 + nc * 8;
-      g2ok[i] = (c < BK * BN / 8) && gn0 < N &&
+      // fast path needs the 8-run inside ONE (kh,kw) segment (C % 8 == 0);
+      // C < 8 falls back to the per-element conv_gather8 decode below
+      g2ok[i] = (geom.C % 8 == 0) && (c < BK * BN / 8) && gn0 < N &&
                 gn0 < geom.KH * geom.KW * geom.C;
       const int cc = g2ok[i] ? gn0 % geom.C : 0;
       const int p = g2ok[i] ? gn0 / geom.C : 0;
@@ -269,7 +271,9 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
           const int k = c / (BN / 8), nc = c % (BN / 8);
           const int gk = k0 + k, gn0 = n0 + nc * 8;
           if (GATHER == 2) {
-            if (g2ok[i] && gk < K) {
+            if (geom.C % 8 != 0) {
+              if (gk < K && gn0 < N) v = conv_gather8(B, geom, (long)gk, gn0);
+            } else if (g2ok[i] && gk < K) {
               const unsigned q1 = fdiv((unsigned)gk, geom.mgOW, geom.sOW);
               const int ow = gk - (int)q1 * geom.OW;
               const unsigned b = fdiv(q1, geom.mgOH, geom.sOH);
@@ -280,7 +284,7 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
                 v = *(const bf16x8*)(
                     B + (((long)b * geom.H + h) * geom.W + w) * geom.C + g2c[i]);
             }
-          } else if (gk < K) {
+          } else if (gk < K) {  // GATHER == 0
             if (gn0 + 8 <= N) v = *(const bf16x8*)(B + (long)gk * ldb + gn0);
             else if (gn0 < N)
               for (int j = 0; j < 8 && gn0 + j < N; ++j) v[j] = B[(long)gk * ldb + gn0 + j];
