@@ -620,37 +620,24 @@ __global__ __launch_bounds__(256) void conv1d_fwd_kernel(
   };
   const int arow = m0 + wave * 16 + (lane & 15);
 
-  // per-WAVE private w-tile double buffers: each wave stages the full
-  // 64x32 tile itself (4x the L2 reads, trivially cached), so the chunk
-  // loop runs with NO barriers — only lgkm waits the compiler inserts for
-  // the wave's own ds write->read dependences.
-  char* smWv = smW + wave * 8192;
-  __syncthreads();                            // x window staged
   for (int os = 0; os < outC; os += 64) {
-    auto fetch_w = [&](int chunk, int q) -> bf16x8 {
-      // lane's quarter-tile slot q=0..3: row = q*16 + lane/4, chunk lane&3
+    auto fetch_w = [&](int chunk) -> bf16x8 {
       bf16x8 v = {};
-      const int r = q * 16 + (lane >> 2);
-      if (chunk < NC && os + r < outC)
-        v = *(const bf16x8*)(w + (long)(os + r) * ldw + chunk * 32 +
-                             (lane & 3) * 8);
+      if (chunk < NC && os + srow < outC)
+        v = *(const bf16x8*)(w + (long)(os + srow) * ldw + chunk * 32 + skc * 8);
       return v;
     };
-    auto write_w = [&](int buf, int q, bf16x8 v) {
-      const int r = q * 16 + (lane >> 2);
-      *(bf16x8*)(smWv + buf * 4096 + r * 64 + wswz(r, (lane & 3) * 8) * 2) = v;
+    auto write_w = [&](int buf, bf16x8 v) {
+      *(bf16x8*)(smW + buf * 4096 + srow * 64 + wswz(srow, skc * 8) * 2) = v;
     };
     f32x4 acc[4] = {};
-    bf16x8 stg[4];
-    #pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      stg[q] = fetch_w(0, q);
-      write_w(0, q, stg[q]);
-    }
+    bf16x8 stg = fetch_w(0);
+    write_w(0, stg);
+    __syncthreads();                          // x (first os) / se reuse + w
     for (int chunk = 0; chunk < NC; ++chunk) {
       const int buf = chunk & 1;
-      #pragma unroll
-      for (int q = 0; q < 4; ++q) stg[q] = fetch_w(chunk + 1, q);
+      stg = fetch_w(chunk + 1);
+      if (chunk) __syncthreads();
       bf16x8 af = {};
       {
         const int k = chunk * 32 + (lane >> 4) * 8;
@@ -664,16 +651,14 @@ __global__ __launch_bounds__(256) void conv1d_fwd_kernel(
         const int row = ni * 16 + (lane & 15);
         const int k2 = (lane >> 4) * 8;
         const bf16x8 bfr = *(const bf16x8*)(
-            smWv + buf * 4096 + row * 64 + wswz(row, k2) * 2);
+            smW + buf * 4096 + row * 64 + wswz(row, k2) * 2);
         acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr, acc[ni],
                                                           0, 0, 0);
       }
-      if (chunk + 1 < NC)
-        #pragma unroll
-        for (int q = 0; q < 4; ++q) write_w(buf ^ 1, q, stg[q]);
+      if (chunk + 1 < NC) write_w(buf ^ 1, stg);
     }
-    // epilogue staging reuses the wave's own buffer — no barrier needed
-    char* se = smWv;
+    __syncthreads();                          // w buffers -> epilogue staging
+    char* se = smW + wave * 2048;
     #pragma unroll
     for (int ni = 0; ni < 4; ++ni) {
       const int c = os + ni * 16 + (lane & 15);
@@ -686,6 +671,7 @@ __global__ __launch_bounds__(256) void conv1d_fwd_kernel(
             tobf16(v);
       }
     }
+    __syncthreads();
     #pragma unroll
     for (int p = 0; p < 2; ++p) {
       const int row = p * 8 + (lane >> 3);
@@ -694,6 +680,7 @@ __global__ __launch_bounds__(256) void conv1d_fwd_kernel(
         *(bf16x8*)(y + ((long)img * OH + m) * ldy + os + (lane & 7) * 8) =
             *(const bf16x8*)(se + row * 128 + (lane & 7) * 16);
     }
+    __syncthreads();                          // before next os reuses smW
   }
 }
 
@@ -702,7 +689,7 @@ bool launch_conv1d_fwd(const void* x, const void* w, long ldw,
                        int C, int KH, int PH, int OH, int outC, int relu,
                        hipStream_t s) {
   const int XR = 64 + KH - 1;
-  const int lds = ((XR * (C * 2 + 16) + 127) & ~127) + 4 * 8192;
+  const int lds = ((XR * (C * 2 + 16) + 127) & ~127) + 8192;
   if (C % 32 != 0 || outC % 64 != 0 || outC > 128 || (KH * C) % 32 != 0 ||
       lds > 56 * 1024)
     return false;
