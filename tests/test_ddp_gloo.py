@@ -107,3 +107,49 @@ def test_distributed_gbt_identical_trees():
         assert not isinstance(acc, str), acc
         assert same, "trees diverged across ranks"
         assert acc > 0.7, acc
+
+
+def _textcnn_worker(rank: int, world: int, port: int, q):
+    os.environ.update({"RANK": str(rank), "WORLD_SIZE": str(world),
+                       "LOCAL_RANK": str(rank),
+                       "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port)})
+    import torch.distributed as dist
+    from learningorchestra_amd.parallel import init_distributed
+    from learningorchestra_amd.models.textcnn import TextCNN
+    from learningorchestra_amd.engine.trainer import Trainer, make_sgd
+    from learningorchestra_amd.data.synthetic import imdb_batch
+    try:
+        init_distributed(backend="gloo")
+        model = TextCNN(vocab=200, emb_dim=16, filters=16,
+                        kernel_sizes=(2, 3), device="cpu", seed=7)
+        trainer = Trainer(model, make_sgd(model, lr=0.05), device="cpu")
+        ids, y = imdb_batch(8, seq_len=32, vocab=200, seed=50 + rank)
+        for _ in range(2):
+            trainer.step_async(ids, y)
+        sig = model.arena.master[:64].clone()
+        gathered = [torch.empty_like(sig) for _ in range(world)]
+        dist.all_gather(gathered, sig)
+        same = all(torch.equal(gathered[0], g) for g in gathered)
+        q.put((rank, float(model.arena.master.double().sum()), same))
+        dist.destroy_process_group()
+    except Exception as exc:  # pragma: no cover
+        q.put((rank, f"ERROR: {exc!r}", False))
+
+
+@pytest.mark.timeout(180)
+def test_ddp_gloo_textcnn_embedding_grads_sync():
+    """TextCNN DDP: embedding + multi-branch conv grads all-reduce through
+    the same flat-arena hooks -> identical params on every rank."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_textcnn_worker, args=(r, world, 29721, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=170) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, checksum, same in results:
+        assert not isinstance(checksum, str), checksum
+        assert same, "textcnn params diverged across ranks"
