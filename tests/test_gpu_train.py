@@ -238,3 +238,18 @@ def test_mlp_classifier_gpu():
                         device="cuda", lr=0.1)
     clf.fit(X[:3200], y[:3200])
     assert clf.score(X[3200:], y[3200:]) > 0.8
+
+
+@pytest.mark.gpu
+def test_resnet_eval_mode_gpu_matches_cpu():
+    """BN eval mode (running stats) on the HIP bn_fwd kernel vs CPU ref."""
+    from learningorchestra_amd.models.resnet import build_resnet18ish
+    torch.manual_seed(0)
+    x = torch.randn(4, 32, 32, 3).bfloat16()
+    mc = build_resnet18ish("cpu", seed=5, num_classes=4, width=8)
+    mg = build_resnet18ish("cuda", seed=5, num_classes=4, width=8)
+    mc.set_training(False)
+    mg.set_training(False)
+    pc = mc.forward(x).float()
+    pg = mg.forward(x.cuda()).float().cpu()
+    torch.testing.assert_close(pg, pc, atol=5e-2, rtol=5e-2)
