@@ -253,3 +253,22 @@ def test_resnet_eval_mode_gpu_matches_cpu():
     pc = mc.forward(x).float()
     pg = mg.forward(x.cuda()).float().cpu()
     torch.testing.assert_close(pg, pc, atol=5e-2, rtol=5e-2)
+
+
+@pytest.mark.gpu
+def test_graph_capture_adam_step():
+    """Adam under hipGraph capture: the device-side step counter must keep
+    bias correction advancing across graph replays."""
+    from learningorchestra_amd.models.mnist_cnn import build_mnist_cnn
+    from learningorchestra_amd.engine.trainer import Trainer, make_adam
+    from learningorchestra_amd.data.synthetic import mnist_batch
+    model = build_mnist_cnn("cuda", seed=4, channels=(8, 8), fc_width=32)
+    trainer = Trainer(model, make_adam(model, lr=1e-3), device="cuda",
+                      use_graph=True)
+    x, y = mnist_batch(512, device="cuda", dtype=torch.bfloat16, seed=2)
+    losses = []
+    for _ in range(10):
+        loss, _ = trainer.step(x, y)
+        losses.append(loss)
+    assert all(math.isfinite(v) for v in losses)
+    assert losses[-1] < losses[0]
