@@ -25,6 +25,8 @@
 
 namespace lo {
 
+typedef short s16x4 __attribute__((ext_vector_type(4)));
+
 // --------------------------------------------------------------- im2col ----
 // small-kpad fast path (e.g. C=1 5x5 convs, kpad<=64): one thread assembles
 // a FULL col row in registers (the input tile is tiny and L1/L2-hot) and
@@ -858,29 +860,40 @@ __global__ __launch_bounds__(256) void conv_dw_c1_kernel(
     __syncthreads();
     const bf16* dyi = dy2 + (long)img * R * ldy;
     for (int m0 = wave * 64; m0 < R; m0 += 4 * 64) {
-      // stage dyT: wave reads dy[m0+r][o-run] and writes smT[o][m] —
-      // 4 runs per lane ((64 rows x 32 cols)/8 = 256 runs)
+      // stage dy NATURALLY (vector loads) into the tr16 k-major image the
+      // hardware-transpose reads expect (same slot permutation as the
+      // GEMM's TA staging; k-dim = m chunk of 64, m-dim = outC 32) — the
+      // previous per-element LDS transpose was 32 scalar ds writes per run
       #pragma unroll
       for (int t = 0; t < 4; ++t) {
-        const int run = t * 64 + lane;         // 0..255
-        const int m = run >> 2, o0 = (run & 3) * 8;
+        const int c = t * 64 + lane;           // 0..255 16-B chunks
+        const int k = c >> 2, mc = c & 3;      // m-row, o-octet
         bf16x8 v = {};
-        if (m0 + m < R && o0 < outC)
-          v = *(const bf16x8*)(dyi + (long)(m0 + m) * ldy + o0);
-        #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          *(bf16*)(smT + (o0 + j) * 128 + ((m * 2) ^ (((o0 + j) & 7) << 4))) = v[j];
+        if (m0 + k < R && mc * 8 < outC)
+          v = *(const bf16x8*)(dyi + (long)(m0 + k) * ldy + mc * 8);
+        const int msub = mc >> 1, mrem = (mc & 1) * 8;
+        const int step = k >> 5, kb = (k & 31) >> 2;
+        const int slot = step * 8 + (kb >> 1) + (kb & 1) * 4;
+        *(bf16x8*)(smT + (msub * 64 * 16 + slot * 64 + (k & 3) * 16
+                          + mrem) * 2) = v;
       }
-      // A[o][m-run]: o = mi*16 + lane%16, m = (lane/16)*8 (+16 per kc)
       __asm__ volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
         bf16x8 af[2];
         #pragma unroll
         for (int mi = 0; mi < 2; ++mi) {
-          const int o = mi * 16 + (lane & 15);
-          const int mb = (kc * 32 + (lane >> 4) * 8) * 2;
-          af[mi] = *(const bf16x8*)(smT + o * 128 + (mb ^ ((o & 7) << 4)));
+          const char* tb = smT + (mi * 64 * 16) * 2
+                           + (kc * 8 + (lane >> 4)) * 128 + (lane & 15) * 8;
+          s16x4 lo4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (__attribute__((address_space(3))) s16x4*)tb);
+          s16x4 hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (__attribute__((address_space(3))) s16x4*)(tb + 512));
+          #pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            ((short*)&af[mi])[j] = lo4[j];
+            ((short*)&af[mi])[j + 4] = hi4[j];
+          }
         }
         #pragma unroll
         for (int ni = 0; ni < 2; ++ni) {
