@@ -51,7 +51,7 @@ def test_ingest_roundtrip(headers, rows):
     for i, r in enumerate(rows):
         doc = docs[i + 1]
         for h, v in zip(san, r):
-            if not h:       # fully-sanitized-away header: column dropped or empty key
+            if not h or san.count(h) > 1:   # sanitized collision: last wins
                 continue
             want = _parse_value(v)
             got = doc.get(h)
