@@ -56,6 +56,9 @@ class CsvIngest:
         for row in reader:
             if headers is None:
                 headers = [_sanitize_header(h) for h in row]
+                # a user column literally named "_id" would collide with the
+                # row-document key (fuzz-found: DuplicateKeyError mid-batch)
+                headers = [h if h != "_id" else "_id_" for h in headers]
                 self._metadata.update_file_headers(name, headers)
                 continue
             rowcount += 1
