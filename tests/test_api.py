@@ -238,3 +238,48 @@ def test_explore_plot_png(client, tmp_path):
     assert resp.status_code == 200
     assert resp.headers["content-type"] == "image/png"
     assert resp.content[:8] == b"\x89PNG\r\n\x1a\n"
+
+
+def test_concurrent_creates_unique_names(client):
+    """The create race the reference had (TOCTOU): N threads
+    POSTing the SAME name must yield exactly one 201 and the rest 409."""
+    import threading
+    csv_text = titanic_csv(40)
+    codes = []
+    lock = threading.Lock()
+
+    def post():
+        r = client.post(f"{PREFIX}/dataset/generic",
+                        json={"datasetName": "race", "datasetURI":
+                              "data:text/plain," + csv_text})
+        with lock:
+            codes.append(r.status_code)
+
+    threads = [threading.Thread(target=post) for _ in range(8)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert sorted(codes).count(201) == 1, codes
+    assert all(c in (201, 409) for c in codes), codes
+
+
+def test_concurrent_distinct_creates_all_succeed(client):
+    import threading
+    csv_text = titanic_csv(30)
+    codes = {}
+
+    def post(i):
+        r = client.post(f"{PREFIX}/dataset/generic",
+                        json={"datasetName": f"cc{i}", "datasetURI":
+                              "data:text/plain," + csv_text})
+        codes[i] = r.status_code
+
+    threads = [threading.Thread(target=post, args=(i,)) for i in range(8)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert all(v == 201 for v in codes.values()), codes
+    for i in range(8):
+        wait_finished(client, f"cc{i}")
