@@ -283,3 +283,40 @@ def test_concurrent_distinct_creates_all_succeed(client):
     assert all(v == 201 for v in codes.values()), codes
     for i in range(8):
         wait_finished(client, f"cc{i}")
+
+
+def test_negative_paths_status_codes(client):
+    """Reference status-code parity: 406 invalid tool/fields, 404 unknown."""
+    # unknown tool = unmapped gateway path in the reference -> 404
+    r = client.post(f"{PREFIX}/dataset/nosuchtool",
+                    json={"datasetName": "x", "datasetURI": "file:///nope"})
+    assert r.status_code == 404
+    # missing required field
+    r = client.post(f"{PREFIX}/dataset/csv", json={"datasetName": "x"})
+    assert r.status_code == 406
+    # model verb with an invalid executor tool (unmapped path)
+    r = client.post(f"{PREFIX}/model/spark",
+                    json={"modelName": "m", "modulePath": "m", "class": "C"})
+    assert r.status_code == 404
+    # binary verb on an unknown parent
+    r = client.post(f"{PREFIX}/train/torch",
+                    json={"name": "t1", "parentName": "ghost",
+                          "method": "fit"})
+    assert r.status_code in (404, 406)
+    # observe unknown name
+    assert client.get(f"{PREFIX}/observe/ghost").status_code == 404
+    # metadata of unknown name
+    assert client.get(f"{PREFIX}/train/torch/ghost/metadata").status_code == 404
+    # builder with an invalid classifier
+    r = client.post(f"{PREFIX}/builder/sparkml",
+                    json={"trainDatasetName": "ghost", "testDatasetName":
+                          "ghost", "modelingCode": "", "classifiersList":
+                          ["nope"]})
+    assert r.status_code == 406
+    # invalid class parameters on a real module (constructor check)
+    r = client.post(f"{PREFIX}/model/scikitlearn",
+                    json={"modelName": "badparam",
+                          "modulePath": "sklearn.linear_model",
+                          "class": "LogisticRegression",
+                          "classParameters": {"definitely_not_an_arg": 1}})
+    assert r.status_code == 406
