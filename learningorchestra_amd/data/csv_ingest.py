@@ -98,7 +98,8 @@ class CsvIngest:
             if re.match(r"^https?://", source):
                 self.ingest_url(name, source)
             else:
-                self.ingest_path(name, source)
+                # accept file:// URIs and plain paths alike
+                self.ingest_path(name, re.sub(r"^file://", "", source))
             self._metadata.update_finished_flag(name, True)
 
         scheduler.submit(name, pipeline)

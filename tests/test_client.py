@@ -89,3 +89,16 @@ def test_client_error_mapping(ctx):
     with pytest.raises(LearningOrchestraError) as e:
         fn.wait("bad", timeout=30)
     assert "ZeroDivisionError" in str(e.value)
+
+
+def test_titanic_example_runs(tmp_path):
+    """examples/titanic_pipeline.py is the documented onboarding flow — it
+    must stay runnable end-to-end."""
+    import subprocess
+    import sys
+    from pathlib import Path
+    script = Path(__file__).parent.parent / "examples" / "titanic_pipeline.py"
+    out = subprocess.run([sys.executable, str(script)], capture_output=True,
+                         text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "accuracy=" in out.stdout
