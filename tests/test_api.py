@@ -320,3 +320,23 @@ def test_negative_paths_status_codes(client):
                           "class": "LogisticRegression",
                           "classParameters": {"definitely_not_an_arg": 1}})
     assert r.status_code == 406
+
+
+def test_observe_wait_timeout_flag(client, tmp_path):
+    """observe/{name}/wait returns timedOut=true when the flag stays false."""
+    client.rt.metadata.create_file("slowjob", "train/torch")
+    r = client.get(f"{PREFIX}/observe/slowjob/wait",
+                   params={"timeoutSeconds": 0.2})
+    assert r.status_code == 200
+    body = r.json()
+    assert body.get("timedOut") is True
+    assert body["result"]["finished"] is False
+
+
+def test_metrics_endpoint(client, tmp_path):
+    ingest_titanic(client, tmp_path)
+    r = client.get(f"{PREFIX}/metrics")
+    assert r.status_code == 200
+    m = r.json()["result"]
+    assert "artifactsByType" in m and "device" in m and "gpu" in m
+    assert m["collections"] >= 1
