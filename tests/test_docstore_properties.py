@@ -172,3 +172,27 @@ def test_update_set_inc_roundtrip(docs, delta):
 def test_match_function_agrees_with_model(docs, flt):
     for d in docs:
         assert match(d, flt) == model_match(d, flt), (d, flt)
+
+
+@settings(max_examples=150, deadline=None)
+@given(st.lists(st.tuples(st.sampled_from(["a", "b", "c"]),
+                          st.integers(-20, 20)), max_size=20))
+def test_group_aggregation_matches_model(pairs):
+    """$group {_id: "$k", count: {$sum: 1}, total: {$sum: "$v"}} — the
+    histogram verb's aggregation — vs a dict model."""
+    store = DocumentStore()
+    col = store["col"]
+    for k, v in pairs:
+        col.insert_one({"k": k, "v": v})
+    got = {d["_id"]: d for d in col.aggregate([
+        {"$group": {"_id": "$k", "count": {"$sum": 1},
+                    "total": {"$sum": "$v"}}}])}
+    want = {}
+    for k, v in pairs:
+        e = want.setdefault(k, {"count": 0, "total": 0})
+        e["count"] += 1
+        e["total"] += v
+    assert set(got) == set(want)
+    for k, e in want.items():
+        assert got[k]["count"] == e["count"]
+        assert got[k]["total"] == e["total"]
