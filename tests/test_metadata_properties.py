@@ -54,3 +54,21 @@ def test_walk_to_model_finds_root_module():
     md.create_file("p", "predict/torch", parentName="t")
     root = md.walk_to_model("p")
     assert root is not None and root.get("modulePath") == "mod.path"
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.dictionaries(names, st.one_of(st.integers(), st.text(max_size=6),
+                                        st.lists(st.floats(-5, 5),
+                                                 max_size=4)),
+                       max_size=5))
+def test_artifact_store_roundtrip(payload):
+    """ArtifactStore must round-trip arbitrary picklable objects under the
+    reference's /binaries/{type}/{name} layout."""
+    import tempfile
+    from learningorchestra_amd.storage.artifacts import ArtifactStore
+    store = ArtifactStore(tempfile.mkdtemp(prefix="lo_art_"))
+    store.save(payload, "obj", "train/torch")
+    assert store.load("obj", "train/torch") == payload
+    assert store.exists("obj", "train/torch")
+    store.delete("obj", "train/torch")
+    assert not store.exists("obj", "train/torch")
