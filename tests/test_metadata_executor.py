@@ -160,3 +160,54 @@ def test_validation_errors(db, artifacts):
         ex.create_model("x", "model/scikitlearn", "sklearn.linear_model", "Nope", {})
     with pytest.raises(ValidationError):
         ex.create_binary_execution("y", "train/scikitlearn", "missing_parent", "fit", {})
+
+
+def test_scheduler_device_serialization_and_concurrency():
+    """GPU-tagged jobs serialize per device; CPU jobs run concurrently;
+    failures are data (finished flag + exception), not crashes."""
+    import threading
+    import time as _t
+    from learningorchestra_amd.executor.scheduler import JobScheduler
+    from learningorchestra_amd.storage.docstore import DocumentStore
+    from learningorchestra_amd.storage.metadata import Metadata
+
+    md = Metadata(DocumentStore())
+    sched = JobScheduler(md, max_workers=8)
+
+    active = {"n": 0, "max": 0}
+    lock = threading.Lock()
+
+    def dev_job():
+        with lock:
+            active["n"] += 1
+            active["max"] = max(active["max"], active["n"])
+        _t.sleep(0.02)
+        with lock:
+            active["n"] -= 1
+
+    jobs = [sched.submit(f"d{i}", dev_job, device="cuda:0") for i in range(6)]
+    for j in jobs:
+        j.wait(10)
+    assert active["max"] == 1, "device jobs must serialize"
+
+    # concurrent CPU jobs overlap
+    active["max"] = 0
+    jobs = [sched.submit(f"c{i}", dev_job) for i in range(6)]
+    for j in jobs:
+        j.wait(10)
+    assert active["max"] >= 2, "cpu jobs should overlap"
+
+    # a failing job records its exception into metadata
+    md.create_file("boom", "train/torch")
+
+    def bad():
+        raise RuntimeError("kaput")
+
+    j = sched.submit("boom", bad)
+    try:
+        j.wait(10)
+    except RuntimeError:
+        pass
+    doc = md.get_metadata("boom")
+    assert doc.get("finished") is True
+    assert "kaput" in (doc.get("exception") or "")
