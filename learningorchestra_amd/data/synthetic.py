@@ -54,6 +54,18 @@ def tabular(rows: int, features: int = 28, seed: int = 0,
     return x.to(device), y.to(device)
 
 
+def tabular_multiclass(rows: int, features: int = 64, n_classes: int = 10,
+                       seed: int = 0, device="cpu"
+                       ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Multiclass blobs (MNIST-like 10-class tabular shape): class-dependent
+    feature means + unit noise — learnable by gini trees but not trivial."""
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    y = torch.randint(0, n_classes, (rows,), generator=g)
+    centers = torch.randn(n_classes, features, generator=g) * 1.6
+    x = centers[y] + torch.randn(rows, features, generator=g)
+    return x.to(device), y.to(device)
+
+
 def titanic_csv(rows: int = 891, seed: int = 0) -> str:
     """A Titanic-shaped CSV (same columns as the canonical dataset) as text,
     for the Dataset->Transform->LogisticRegression plumbing config."""

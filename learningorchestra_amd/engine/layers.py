@@ -511,6 +511,11 @@ class BatchNormReLU(Layer):
         self.arena = arena
         arena.add(self.name + ".g", (self.c,), torch.ones(self.c))
         arena.add(self.name + ".b", (self.c,), torch.zeros(self.c))
+        # running stats are keyed only on C and live for the model's lifetime
+        # (NOT in the shape-keyed _bufs dict: a batch-size change must not
+        # reset them, and load_extra_state must work before any forward)
+        self.running_mean = torch.zeros(self.c, device=arena.device)
+        self.running_var = torch.ones(self.c, device=arena.device)
 
     def param_names(self):
         return [self.name + ".g", self.name + ".b"]
@@ -534,8 +539,6 @@ class BatchNormReLU(Layer):
                 "mean": torch.zeros(self.c, device=dev),
                 "invstd": torch.ones(self.c, device=dev),
                 "scratch": self.scratch(dev),
-                "running_mean": torch.zeros(self.c, device=dev),
-                "running_var": torch.ones(self.c, device=dev),
             }
         return self._bufs
 
@@ -557,12 +560,12 @@ class BatchNormReLU(Layer):
                            stats_ready=stats_ready and x.is_cuda,
                            residual=residual)
             m = self.momentum
-            bufs["running_mean"].mul_(1 - m).add_(bufs["mean"], alpha=m)
+            self.running_mean.mul_(1 - m).add_(bufs["mean"], alpha=m)
             var = bufs["invstd"].square().reciprocal() - self.eps
-            bufs["running_var"].mul_(1 - m).add_(var, alpha=m)
+            self.running_var.mul_(1 - m).add_(var, alpha=m)
         else:
-            F.bn_fwd_eval(x2, gamma, beta, bufs["running_mean"],
-                          bufs["running_var"], self.eps, bufs["y"], relu,
+            F.bn_fwd_eval(x2, gamma, beta, self.running_mean,
+                          self.running_var, self.eps, bufs["y"], relu,
                           residual=residual)
         return bufs["y"].view(x.shape)
 
@@ -577,15 +580,15 @@ class BatchNormReLU(Layer):
 
     # checkpointable non-arena state
     def extra_state(self):
-        if not self._bufs:
-            return {}
-        return {self.name + ".running_mean": self._bufs["running_mean"].cpu(),
-                self.name + ".running_var": self._bufs["running_var"].cpu()}
+        return {self.name + ".running_mean": self.running_mean.cpu(),
+                self.name + ".running_var": self.running_var.cpu()}
 
     def load_extra_state(self, sd):
-        if self._bufs and self.name + ".running_mean" in sd:
-            self._bufs["running_mean"].copy_(sd[self.name + ".running_mean"])
-            self._bufs["running_var"].copy_(sd[self.name + ".running_var"])
+        if self.name + ".running_mean" in sd:
+            self.running_mean.copy_(sd[self.name + ".running_mean"].to(
+                self.running_mean.device))
+            self.running_var.copy_(sd[self.name + ".running_var"].to(
+                self.running_var.device))
 
 
 class AvgPoolGlobal(Layer):

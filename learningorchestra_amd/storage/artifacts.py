@@ -13,9 +13,23 @@ from __future__ import annotations
 
 import json
 import os
+import re
 from typing import Any, Optional
 
 import dill
+
+_NAME_RE = re.compile(r"^[A-Za-z0-9][A-Za-z0-9_.\- ]*$")
+
+
+def check_name(name: str) -> str:
+    """Reject artifact/dataset names that could escape the data root
+    (path separators, '..' traversal, hidden/empty names). Raised at the
+    API boundary too; this is defense in depth."""
+    if (not isinstance(name, str) or not _NAME_RE.match(name)
+            or ".." in name or len(name) > 200):
+        raise ValueError(f"invalid name {name!r}: names must match "
+                         "[A-Za-z0-9][A-Za-z0-9_.- ]* with no '..'")
+    return name
 
 
 class ArtifactStore:
@@ -32,7 +46,15 @@ class ArtifactStore:
         return d
 
     def path(self, name: str, service_type: str) -> str:
-        return os.path.join(self._dir(service_type), name)
+        check_name(name)
+        d = self._dir(service_type)
+        p = os.path.join(d, name)
+        # belt-and-braces: the final path must stay under the store root
+        if os.path.commonpath([os.path.realpath(os.path.dirname(p)),
+                               os.path.realpath(self._root)]) \
+                != os.path.realpath(self._root):
+            raise ValueError(f"name {name!r} escapes the artifact root")
+        return p
 
     # -- save ---------------------------------------------------------------
     def save(self, instance: Any, name: str, service_type: str) -> str:
