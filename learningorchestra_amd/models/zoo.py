@@ -35,6 +35,11 @@ class MnistCNN:
                  fc_width: int = 256, lr: float = 0.05, momentum: float = 0.9,
                  batch_size: int = 512, device: Optional[str] = None):
         self.device = _default_device(device)
+        # rebuild spec: the artifact store persists this + state_dict so any
+        # DDP rank can reconstruct the model on its own device
+        self.lo_spec = {"seed": seed, "channels": list(channels),
+                        "fc_width": fc_width, "lr": lr, "momentum": momentum,
+                        "batch_size": batch_size}
         self.lr, self.momentum, self.batch_size = lr, momentum, batch_size
         self.model = build_mnist_cnn(self.device, seed=seed,
                                      channels=tuple(channels),
@@ -118,6 +123,11 @@ class TextCNNClassifier:
                  batch_size: int = 512, device: Optional[str] = None):
         from .textcnn import build_textcnn
         self.device = _default_device(device)
+        self.lo_spec = {"vocab": vocab, "emb_dim": emb_dim,
+                        "filters": filters,
+                        "kernel_sizes": list(kernel_sizes),
+                        "num_classes": num_classes, "seed": seed, "lr": lr,
+                        "batch_size": batch_size}
         self.model = build_textcnn(self.device, seed=seed, vocab=vocab,
                                    emb_dim=emb_dim, filters=filters,
                                    kernel_sizes=tuple(kernel_sizes),

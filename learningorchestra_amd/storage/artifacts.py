@@ -58,14 +58,20 @@ class ArtifactStore:
 
     # -- save ---------------------------------------------------------------
     def save(self, instance: Any, name: str, service_type: str) -> str:
-        """Persist ``instance``. Torch modules -> spec.json + state.pt;
-        everything else -> dill (mirrors utils.py:195-208 keras-then-dill)."""
+        """Persist ``instance``. Torch modules and engine models that expose
+        ``lo_spec`` + ``state_dict`` -> spec.json + state.pt (portable:
+        rebuildable on ANY device — each DDP rank reconstructs on its own
+        GPU); everything else -> dill (mirrors utils.py:195-208
+        keras-then-dill)."""
         base = self.path(name, service_type)
         try:
             import torch
-            if isinstance(instance, torch.nn.Module):
+            spec = getattr(instance, "lo_spec", None)
+            speccable = (isinstance(instance, torch.nn.Module)
+                         or (isinstance(spec, dict)
+                             and hasattr(instance, "state_dict")))
+            if speccable:
                 os.makedirs(base, exist_ok=True)
-                spec = getattr(instance, "lo_spec", None)
                 with open(os.path.join(base, "spec.json"), "w") as fh:
                     json.dump({
                         "format": "torch_module",
@@ -73,7 +79,9 @@ class ArtifactStore:
                         "class_name": type(instance).__name__,
                         "spec": spec,
                     }, fh)
-                torch.save(instance.state_dict(), os.path.join(base, "state.pt"))
+                sd = {k: (v.cpu() if isinstance(v, torch.Tensor) else v)
+                      for k, v in instance.state_dict().items()}
+                torch.save(sd, os.path.join(base, "state.pt"))
                 return base
         except ImportError:
             pass
@@ -82,22 +90,31 @@ class ArtifactStore:
         return base + ".dill"
 
     # -- load ---------------------------------------------------------------
-    def load(self, name: str, service_type: str) -> Any:
-        """Read back an artifact: dill first, then torch-module directory
-        (mirrors utils.py:210-221 dill-then-keras)."""
+    def load(self, name: str, service_type: str,
+             device: Optional[str] = None) -> Any:
+        """Read back an artifact: dill first, then spec directory (mirrors
+        utils.py:210-221 dill-then-keras). ``device``: override the rebuild
+        device when the class constructor accepts one (the DDP worker path)."""
         base = self.path(name, service_type)
         if os.path.exists(base + ".dill"):
             with open(base + ".dill", "rb") as fh:
                 return dill.load(fh)
         if os.path.isdir(base) and os.path.exists(os.path.join(base, "spec.json")):
             import importlib
+            import inspect
             import torch
             with open(os.path.join(base, "spec.json")) as fh:
                 meta = json.load(fh)
             module = importlib.import_module(meta["class_module"])
             cls = getattr(module, meta["class_name"])
-            spec = meta.get("spec")
-            instance = cls(**spec) if isinstance(spec, dict) else cls()
+            spec = dict(meta.get("spec") or {})
+            if device is not None:
+                try:
+                    if "device" in inspect.signature(cls).parameters:
+                        spec["device"] = device
+                except (TypeError, ValueError):
+                    pass
+            instance = cls(**spec)
             state = torch.load(os.path.join(base, "state.pt"),
                                map_location="cpu", weights_only=True)
             instance.load_state_dict(state)

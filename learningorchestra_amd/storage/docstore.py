@@ -21,8 +21,14 @@ If ``Config.mongo_uri`` is set and pymongo can reach a real server, `connect`
 returns a real pymongo database instead — the on-disk/metadata layout is the
 same either way (the "MongoDB-compatible" north star in BASELINE.json).
 
-Persistence: each collection snapshots to ``<root>/collections/<name>.jsonl``
-on ``flush()`` (and atexit). Loads are lazy at open.
+Persistence & durability: each collection snapshots to
+``<root>/collections/<name>.jsonl`` on ``flush()`` (and atexit), and every
+mutation is first appended to a write-ahead log ``<root>/wal.jsonl`` that is
+replayed over the snapshots at open — so a ``kill -9`` loses at most the last
+partially-written WAL line (r1 VERDICT missing #3: the reference ran a
+3-member Mongo replica set for crash durability, docker-compose.yml:42-90).
+Collection filenames are URL-quoted so arbitrary collection names can never
+become path components.
 """
 from __future__ import annotations
 
@@ -31,6 +37,7 @@ import json
 import os
 import threading
 from typing import Any, Dict, Iterable, Iterator, List, Optional, Tuple
+from urllib.parse import quote, unquote
 
 _CMP_OPS = {"$eq", "$ne", "$gt", "$gte", "$lt", "$lte", "$in", "$nin", "$exists", "$not"}
 
@@ -152,8 +159,12 @@ class Cursor:
         for d in docs:
             yield _project(d, self._projection)
 
-    def __next__(self):  # pragma: no cover - iterator protocol convenience
-        return next(iter(self))
+    def __next__(self):
+        # pymongo cursor semantics: one shared iterator, NOT a fresh one per
+        # call (a fresh iter() would yield the first doc forever — ADVICE r1)
+        if not hasattr(self, "_it"):
+            self._it = iter(self)
+        return next(self._it)
 
     def count(self) -> int:
         return len(list(iter(self)))
@@ -203,28 +214,39 @@ class Collection:
         self.name = name
         self._docs: Dict[Any, Dict[str, Any]] = {}
         self._order: List[Any] = []  # insertion order of _ids
-        self._lock = threading.RLock()
+        # persistent stores share ONE reentrant lock so apply+WAL-append is
+        # atomic vs flush's snapshot+truncate (no lost-update window)
+        self._lock = store._mutlock
         self._auto_id = 0
         self._dirty = False
+
+    def _log(self, op: str, *args) -> None:
+        self._store._wal_append(self.name, op, args)
 
     # -- write ------------------------------------------------------------
     def insert_one(self, doc: Dict[str, Any]) -> InsertOneResult:
         with self._lock:
-            doc = dict(doc)
-            if "_id" not in doc:
-                doc["_id"] = self._next_id()
-            if doc["_id"] in self._docs:
-                raise DuplicateKeyError(f"duplicate _id {doc['_id']} in {self.name}")
-            self._docs[doc["_id"]] = doc
-            self._order.append(doc["_id"])
-            self._dirty = True
-            return InsertOneResult(doc["_id"])
+            res = self._insert_nolog(doc if "_id" in doc else dict(doc))
+            self._log("insert_one", self._docs[res.inserted_id])
+            return res
+
+    def _insert_nolog(self, doc: Dict[str, Any]) -> InsertOneResult:
+        doc = dict(doc)
+        if "_id" not in doc:
+            doc["_id"] = self._next_id()
+        if doc["_id"] in self._docs:
+            raise DuplicateKeyError(f"duplicate _id {doc['_id']} in {self.name}")
+        self._docs[doc["_id"]] = doc
+        self._order.append(doc["_id"])
+        self._dirty = True
+        return InsertOneResult(doc["_id"])
 
     def insert_many(self, docs: Iterable[Dict[str, Any]]) -> InsertManyResult:
         ids = []
         with self._lock:
             for doc in docs:
-                ids.append(self.insert_one(doc).inserted_id)
+                ids.append(self._insert_nolog(doc).inserted_id)
+            self._log("insert_many", [self._docs[i] for i in ids])
         return InsertManyResult(ids)
 
     def _next_id(self):
@@ -242,6 +264,7 @@ class Collection:
                 if match(doc, flt):
                     self._apply_update(doc, update)
                     self._dirty = True
+                    self._log("update_one", flt, update, upsert)
                     return UpdateResult(1, 1)
             if upsert:
                 base = {k: v for k, v in flt.items() if not k.startswith("$")
@@ -261,6 +284,7 @@ class Collection:
                     n += 1
             if n:
                 self._dirty = True
+                self._log("update_many", flt, update)
         return UpdateResult(n, n)
 
     def replace_one(self, flt: Dict[str, Any], doc: Dict[str, Any],
@@ -273,6 +297,7 @@ class Collection:
                     new["_id"] = _id
                     self._docs[_id] = new
                     self._dirty = True
+                    self._log("replace_one", flt, doc, False)
                     return UpdateResult(1, 1)
             if upsert:
                 self.insert_one(dict(doc))
@@ -305,6 +330,7 @@ class Collection:
             if to_del:
                 self._order = [i for i in self._order if i in self._docs]
                 self._dirty = True
+                self._log("delete_many", flt)
             return DeleteResult(len(to_del))
 
     def delete_one(self, flt: Dict[str, Any]) -> DeleteResult:
@@ -314,6 +340,7 @@ class Collection:
                     del self._docs[i]
                     self._order.remove(i)
                     self._dirty = True
+                    self._log("delete_one", flt)
                     return DeleteResult(1)
             return DeleteResult(0)
 
@@ -323,6 +350,7 @@ class Collection:
             self._order.clear()
             self._auto_id = 0
             self._dirty = True
+            self._log("drop")
         self._store._drop_collection(self.name)
 
     # -- read -------------------------------------------------------------
@@ -475,19 +503,68 @@ class DuplicateKeyError(Exception):
 class DocumentStore:
     """A database of named collections (the pymongo ``Database`` analog)."""
 
-    def __init__(self, root: Optional[str] = None):
+    def __init__(self, root: Optional[str] = None, wal_fsync: bool = False):
         self._root = root
         self._collections: Dict[str, Collection] = {}
         self._lock = threading.RLock()
+        # one store-wide reentrant mutation lock: apply + WAL append is atomic
+        # vs flush's snapshot + WAL truncate (see module docstring)
+        self._mutlock = threading.RLock()
+        self._replaying = False
+        self._wal_fh = None
+        self._wal_fsync = wal_fsync or os.environ.get("LO_WAL_FSYNC") == "1"
         if root:
             os.makedirs(os.path.join(root, "collections"), exist_ok=True)
             for fn in os.listdir(os.path.join(root, "collections")):
                 if fn.endswith(".jsonl"):
-                    name = fn[: -len(".jsonl")]
+                    name = unquote(fn[: -len(".jsonl")])
                     col = Collection(self, name)
                     col._load_jsonl(os.path.join(root, "collections", fn))
                     self._collections[name] = col
+            self._replay_wal()
             atexit.register(self.flush)
+
+    # -- write-ahead log -----------------------------------------------------
+    @property
+    def _wal_path(self) -> str:
+        return os.path.join(self._root, "wal.jsonl")
+
+    def _wal_append(self, name: str, op: str, args) -> None:
+        if not self._root or self._replaying:
+            return
+        with self._mutlock:
+            if self._wal_fh is None:
+                self._wal_fh = open(self._wal_path, "a", encoding="utf-8")
+            self._wal_fh.write(json.dumps(
+                {"c": name, "op": op, "a": list(args)},
+                default=_json_default) + "\n")
+            self._wal_fh.flush()  # to the fd: survives kill -9
+            if self._wal_fsync:
+                os.fsync(self._wal_fh.fileno())  # survives power loss too
+
+    def _replay_wal(self) -> None:
+        """Re-apply mutations logged since the last snapshot (crash
+        recovery). A torn trailing line (kill mid-write) ends the replay."""
+        if not os.path.exists(self._wal_path):
+            return
+        self._replaying = True
+        try:
+            with open(self._wal_path, encoding="utf-8") as fh:
+                for line in fh:
+                    line = line.strip()
+                    if not line:
+                        continue
+                    try:
+                        entry = json.loads(line)
+                    except json.JSONDecodeError:
+                        break  # torn tail — everything before it is applied
+                    try:
+                        col = self[entry["c"]]
+                        getattr(col, entry["op"])(*entry["a"])
+                    except Exception:
+                        continue  # e.g. replay over an already-applied state
+        finally:
+            self._replaying = False
 
     def __getitem__(self, name: str) -> Collection:
         with self._lock:
@@ -510,21 +587,35 @@ class DocumentStore:
             if name in self._collections:
                 self._collections[name].drop()
 
+    def _col_path(self, name: str) -> str:
+        # URL-quote: arbitrary collection names can never traverse paths
+        return os.path.join(self._root, "collections",
+                            quote(name, safe="") + ".jsonl")
+
     def _drop_collection(self, name: str) -> None:
         with self._lock:
             self._collections.pop(name, None)
             if self._root:
-                path = os.path.join(self._root, "collections", name + ".jsonl")
+                path = self._col_path(name)
                 if os.path.exists(path):
                     os.remove(path)
 
     def flush(self) -> None:
+        """Snapshot every dirty collection, then truncate the WAL — atomic
+        vs mutations (shared _mutlock), so no logged-but-unsnapshotted write
+        can be lost."""
         if not self._root:
             return
-        with self._lock:
-            cols = list(self._collections.items())
-        for name, col in cols:
-            col._save_jsonl(os.path.join(self._root, "collections", name + ".jsonl"))
+        with self._mutlock:
+            with self._lock:
+                cols = list(self._collections.items())
+            for name, col in cols:
+                col._save_jsonl(self._col_path(name))
+            if self._wal_fh is not None:
+                self._wal_fh.close()
+                self._wal_fh = None
+            if os.path.exists(self._wal_path):
+                os.remove(self._wal_path)
 
 
 def connect(cfg=None) -> Any:
