@@ -120,12 +120,17 @@ class Trainer:
                 all_reduce_grads(self.model.arena.grad)
                 self._opt_body()
             torch.cuda.synchronize()
+            # thread_local capture: the RCCL/NCCL watchdog thread polls
+            # events concurrently; global capture mode would invalidate the
+            # capture when it does
             self._graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(self._graph):
+            with torch.cuda.graph(self._graph,
+                                  capture_error_mode="thread_local"):
                 self.model.train_step(self._static_x, self._static_y,
                                       gscale=gscale)
             self._graph_opt = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(self._graph_opt):
+            with torch.cuda.graph(self._graph_opt,
+                                  capture_error_mode="thread_local"):
                 self._opt_body()
             torch.cuda.synchronize()
             return

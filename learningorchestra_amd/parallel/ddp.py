@@ -34,14 +34,21 @@ def get_world_size() -> int:
 def init_distributed(backend: Optional[str] = None, timeout_s: int = 300) -> int:
     """Initialize from torchrun env (RANK/WORLD_SIZE/LOCAL_RANK). Returns
     local rank. No-op (rank 0) when not launched distributed."""
-    if "RANK" not in os.environ or int(os.environ.get("WORLD_SIZE", "1")) == 1:
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if "RANK" not in os.environ or world == 1:
         return 0
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # RCCL (like NCCL) refuses two ranks on one device ("Duplicate GPU
+        # detected", verified on hardware) — oversubscribed runs (e.g. a
+        # 2-rank test on a 1-GPU lease) fall back to gloo, which supports
+        # CUDA tensors by host staging. One-rank-per-GPU runs use RCCL.
+        n_dev = torch.cuda.device_count() if torch.cuda.is_available() else 0
+        backend = "nccl" if 0 < world <= n_dev else "gloo"
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if torch.cuda.is_available():
+        local_rank %= max(torch.cuda.device_count(), 1)
+        torch.cuda.set_device(local_rank)
     if not dist.is_initialized():
-        if backend == "nccl":
-            torch.cuda.set_device(local_rank)
         dist.init_process_group(backend=backend,
                                 timeout=timedelta(seconds=timeout_s))
     return local_rank
@@ -49,8 +56,14 @@ def init_distributed(backend: Optional[str] = None, timeout_s: int = 300) -> int
 
 def all_reduce_grads(grad_flat: torch.Tensor, async_op: bool = False):
     """SUM all-reduce of the flat grad arena (grads carry 1/(B*world) so sum
-    = global-batch mean)."""
+    = global-batch mean). The gloo+CUDA combination (oversubscribed test
+    leases) stages through host memory explicitly."""
     if not is_distributed():
+        return None
+    if grad_flat.is_cuda and dist.get_backend() == "gloo":
+        host = grad_flat.detach().to("cpu")
+        dist.all_reduce(host, op=dist.ReduceOp.SUM)
+        grad_flat.copy_(host)
         return None
     return dist.all_reduce(grad_flat, op=dist.ReduceOp.SUM, async_op=async_op)
 

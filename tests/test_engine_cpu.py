@@ -190,3 +190,30 @@ def test_mlp_classifier():
     clf.fit(X[:500], y[:500])
     assert clf.score(X[500:], y[500:]) > 0.7
     assert clf.predict_proba(X[:5]).shape == (5, 2)
+
+
+def test_bn_running_stats_survive_batch_change():
+    """BN running stats must survive a batch-size change
+    and load into a fresh model BEFORE any forward."""
+    import torch
+
+    from learningorchestra_amd.models.resnet import build_resnet18ish
+    m = build_resnet18ish("cpu", seed=0, num_classes=8, width=8)
+    x = torch.randn(4, 32, 32, 3)
+    y = torch.randint(0, 8, (4,))
+    m.train_step(x, y)
+    rm = m.stem_bn.running_mean.clone()
+    assert rm.abs().sum() > 0
+    # different batch size: stats keep accumulating, not reset
+    m.train_step(torch.randn(2, 32, 32, 3), torch.randint(0, 8, (2,)))
+    assert not torch.equal(m.stem_bn.running_mean, rm)
+    assert m.stem_bn.running_mean.abs().sum() > 0
+    # checkpoint into a FRESH model with no forward yet, then eval
+    sd = m.state_dict()
+    m2 = build_resnet18ish("cpu", seed=1, num_classes=8, width=8)
+    m2.load_state_dict(sd)   # must apply running stats immediately
+    assert torch.equal(m2.stem_bn.running_mean, m.stem_bn.running_mean)
+    m2.set_training(False)
+    m.set_training(False)
+    xe = torch.randn(3, 32, 32, 3)
+    assert torch.equal(m.predict(xe), m2.predict(xe))

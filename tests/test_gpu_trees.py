@@ -75,3 +75,33 @@ def test_builder_native_classifiers_gpu(tmp_path):
         doc = b.wait(f"titanic{c}", timeout=240)
         assert doc.get("exception") in (None, ""), doc
         assert doc["accuracy"] > 0.55, (c, doc["accuracy"])
+
+
+@pytest.mark.gpu
+def test_class_histograms_gpu_vs_cpu():
+    """Multiclass per-class count histograms: GPU tree_hist packing (2
+    one-hot classes per kernel call) vs the CPU index_add reference."""
+    import torch
+
+    from learningorchestra_amd.models.trees import build_class_histograms
+    torch.manual_seed(0)
+    N, F, K = 20000, 6, 5
+    binned = torch.randint(0, 256, (N, F), dtype=torch.uint8)
+    node_of = torch.randint(-1, 4, (N,), dtype=torch.int32)
+    onehot = torch.nn.functional.one_hot(
+        torch.randint(0, K, (N,)), K).float()
+    ref = build_class_histograms(binned, node_of, onehot, 4)
+    got = build_class_histograms(binned.cuda(), node_of.cuda(),
+                                 onehot.cuda(), 4).cpu()
+    assert torch.allclose(ref, got, atol=1e-3), (ref - got).abs().max()
+
+
+@pytest.mark.gpu
+def test_multiclass_rf_gpu():
+    from learningorchestra_amd.data.synthetic import tabular_multiclass
+    from learningorchestra_amd.models.trees import RandomForestClassifier
+    X, y = tabular_multiclass(8000, 16, n_classes=10, seed=3)
+    clf = RandomForestClassifier(n_trees=8, max_depth=7,
+                                 device="cuda").fit(X.numpy(), y.numpy())
+    acc = (clf.predict(X.numpy()).astype(int) == y.numpy()).mean()
+    assert clf.n_classes == 10 and acc > 0.8, acc
