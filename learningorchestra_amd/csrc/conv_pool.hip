@@ -593,26 +593,30 @@ __global__ __launch_bounds__(256) void conv1d_fwd_kernel(
     const bf16* __restrict__ w, long ldw,     // [outC, kpad]
     const float* __restrict__ bias,
     bf16* __restrict__ y, long ldy,           // [B*OH, outC]
-    int H, int C, int KH, int PH, int OH, int outC, int relu) {
+    int B, int H, int C, int KH, int PH, int OH, int outC, int G, int relu) {
   extern __shared__ char sm[];
-  const int img = blockIdx.x;
+  const int img0 = blockIdx.x * G;            // G images share each w staging
   const int m0 = blockIdx.y * 64;             // this block's y rows
   const int tid = threadIdx.x;
   const int lane = tid & 63, wave = tid >> 6;
   const int XROW = C * 2 + 16;
   const int XR = 64 + KH - 1;                 // x window rows
   const int x0 = m0 - PH;                     // first window row (may be <0)
+  const int XB = (XR * XROW + 127) & ~127;    // per-image window bytes
   char* smX = sm;
-  char* smW = sm + ((XR * XROW + 127) & ~127);
+  char* smW = sm + XB * G;
 
-  // stage x window (rows outside [0,H) stage zeros)
-  for (int i = tid * 8; i < XR * C; i += 256 * 8) {
-    const int r = i / C, c0 = i - r * C;
-    const int h = x0 + r;
-    bf16x8 v = {};
-    if (h >= 0 && h < H)
-      v = *(const bf16x8*)(x + ((long)img * H + h) * C + c0);
-    *(bf16x8*)(smX + r * XROW + c0 * 2) = v;
+  // stage the G x windows (rows outside [0,H) / images >= B stage zeros)
+  for (int g = 0; g < G; ++g) {
+    const int img = img0 + g;
+    for (int i = tid * 8; i < XR * C; i += 256 * 8) {
+      const int r = i / C, c0 = i - r * C;
+      const int h = x0 + r;
+      bf16x8 v = {};
+      if (img < B && h >= 0 && h < H)
+        v = *(const bf16x8*)(x + ((long)img * H + h) * C + c0);
+      *(bf16x8*)(smX + g * XB + r * XROW + c0 * 2) = v;
+    }
   }
 
   const int NC = (KH * C) / 32;
@@ -632,7 +636,7 @@ __global__ __launch_bounds__(256) void conv1d_fwd_kernel(
     auto write_w = [&](int buf, bf16x8 v) {
       *(bf16x8*)(smW + buf * 4096 + srow * 64 + wswz(srow, skc * 8) * 2) = v;
     };
-    f32x4 acc[4] = {};
+    f32x4 acc0[4] = {}, acc1[4] = {};
     bf16x8 stg = fetch_w(0);
     write_w(0, stg);
     __syncthreads();                          // x (first os) / se reuse + w
@@ -640,13 +644,16 @@ __global__ __launch_bounds__(256) void conv1d_fwd_kernel(
       const int buf = chunk & 1;
       stg = fetch_w(chunk + 1);
       if (chunk) __syncthreads();
-      bf16x8 af = {};
+      bf16x8 af0 = {}, af1 = {};
       {
         const int k = chunk * 32 + (lane >> 4) * 8;
         const int kh = k / C, c = k - kh * C;
         const int r = (arow - m0) + kh;       // row in the window
-        if (arow < OH)
-          af = *(const bf16x8*)(smX + r * XROW + c * 2);
+        if (arow < OH) {
+          af0 = *(const bf16x8*)(smX + r * XROW + c * 2);
+          if (G > 1)
+            af1 = *(const bf16x8*)(smX + XB + r * XROW + c * 2);
+        }
       }
       #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
@@ -654,33 +661,42 @@ __global__ __launch_bounds__(256) void conv1d_fwd_kernel(
         const int k2 = (lane >> 4) * 8;
         const bf16x8 bfr = *(const bf16x8*)(
             smW + buf * 4096 + row * 64 + wswz(row, k2) * 2);
-        acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr, acc[ni],
-                                                          0, 0, 0);
+        acc0[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af0, bfr, acc0[ni],
+                                                           0, 0, 0);
+        if (G > 1)
+          acc1[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af1, bfr,
+                                                             acc1[ni], 0, 0, 0);
       }
       if (chunk + 1 < NC) write_w(buf ^ 1, stg);
     }
-    __syncthreads();                          // w buffers -> epilogue staging
-    char* se = smW + wave * 2048;
-    #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
-      const int c = os + ni * 16 + (lane & 15);
-      const float b = bias ? bias[c < outC ? c : 0] : 0.f;
+    // per-image epilogue: acc -> LDS staging -> coalesced y writes
+    for (int g = 0; g < G; ++g) {
+      if (img0 + g >= B) break;
+      __syncthreads();                        // w buffers / prev g -> staging
+      char* se = smW + wave * 2048;
       #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float v = acc[ni][r] + b;
-        if (relu) v = v > 0.f ? v : 0.f;
-        ((bf16*)se)[((lane >> 4) * 4 + r) * 64 + (ni * 16 + (lane & 15))] =
-            tobf16(v);
+      for (int ni = 0; ni < 4; ++ni) {
+        const int c = os + ni * 16 + (lane & 15);
+        const float b = bias ? bias[c < outC ? c : 0] : 0.f;
+        const f32x4 a = g ? acc1[ni] : acc0[ni];
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = a[r] + b;
+          if (relu) v = v > 0.f ? v : 0.f;
+          ((bf16*)se)[((lane >> 4) * 4 + r) * 64 + (ni * 16 + (lane & 15))] =
+              tobf16(v);
+        }
       }
-    }
-    __syncthreads();
-    #pragma unroll
-    for (int p = 0; p < 2; ++p) {
-      const int row = p * 8 + (lane >> 3);
-      const int m = m0 + wave * 16 + row;
-      if (m < OH)
-        *(bf16x8*)(y + ((long)img * OH + m) * ldy + os + (lane & 7) * 8) =
-            *(const bf16x8*)(se + row * 128 + (lane & 7) * 16);
+      __syncthreads();
+      #pragma unroll
+      for (int p = 0; p < 2; ++p) {
+        const int row = p * 8 + (lane >> 3);
+        const int m = m0 + wave * 16 + row;
+        if (m < OH)
+          *(bf16x8*)(y + ((long)(img0 + g) * OH + m) * ldy + os +
+                     (lane & 7) * 8) =
+              *(const bf16x8*)(se + row * 128 + (lane & 7) * 16);
+      }
     }
     __syncthreads();                          // before next os reuses smW
   }
@@ -691,14 +707,17 @@ bool launch_conv1d_fwd(const void* x, const void* w, long ldw,
                        int C, int KH, int PH, int OH, int outC, int relu,
                        hipStream_t s) {
   const int XR = 64 + KH - 1;
-  const int lds = ((XR * (C * 2 + 16) + 127) & ~127) + 8192;
-  if (C % 32 != 0 || outC % 64 != 0 || outC > 128 || (KH * C) % 32 != 0 ||
-      lds > 56 * 1024)
+  const int XB = (XR * (C * 2 + 16) + 127) & ~127;
+  if (C % 32 != 0 || outC % 64 != 0 || outC > 128 || (KH * C) % 32 != 0)
     return false;
+  int G = 2;                                  // halves the w re-staging
+  if (XB * G + 8192 > 56 * 1024) G = 1;
+  if (XB * G + 8192 > 56 * 1024) return false;
   const int T = (OH + 63) / 64;
-  hipLaunchKernelGGL(conv1d_fwd_kernel, dim3(B, T), dim3(256), lds, s,
+  hipLaunchKernelGGL(conv1d_fwd_kernel, dim3((B + G - 1) / G, T), dim3(256),
+                     XB * G + 8192, s,
                      (const bf16*)x, (const bf16*)w, ldw, (const float*)bias,
-                     (bf16*)y, ldy, H, C, KH, PH, OH, outC, relu);
+                     (bf16*)y, ldy, B, H, C, KH, PH, OH, outC, G, relu);
   return true;
 }
 
@@ -813,11 +832,202 @@ __global__ __launch_bounds__(256) void conv1d_dx_kernel(
   }
 }
 
+// conv1d_dx v2 — identity row ownership + register accumulation.
+//
+// The v1 kernel (above) scatters MFMA fragments into a 33 KB LDS fp32 dx
+// tile with barrier-separated RMW, and pays NSTEP x 8 KB of wt-tile staging
+// per 62 dx rows — measured 633 us/conv on TextCNN (31% of the step) vs a
+// ~130 us traffic floor (r1 profile; the wt L2->LDS re-streaming was the
+// named structural problem, ROUND2_PLAN §2).
+//
+// v2 restructures so each lane OWNS a fixed dx row hd (the MFMA m index is
+// hd itself, not the dy row):
+//   dx[hd, c] = sum_kh sum_oc wt[kh*C+c, oc] * dy[hd + PH - kh, oc]
+// The kh shift moves into the *B-fragment read* from an LDS-resident dy
+// window (64+KH-1 rows, staged once per block), so the accumulator lives in
+// REGISTERS across the whole kh loop and the LDS dx tile, its zeroing, the
+// RMW scatters and the readout pass all disappear. With the freed LDS the
+// block processes G=2 images per wt staging pass, halving the wt-tile
+// traffic per image. Stores are direct bf16x4 (8 B from consecutive lanes
+// coalesce into the row's contiguous 64 B segment).
+__global__ __launch_bounds__(256) void conv1d_dx2_kernel(
+    const bf16* __restrict__ dy2, long ldy,   // [B*OH, outC]
+    const bf16* __restrict__ wt, long ldw,    // [kpad, outC]
+    bf16* __restrict__ dx,                    // [B, H, 1, C]
+    int B, int H, int C, int KH, int PH, int OH, int outC, int G,
+    int accumulate) {
+  extern __shared__ char sm[];
+  const int img0 = blockIdx.x * G;
+  const int h0 = blockIdx.y * 64;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const int WR = 64 + KH - 1;                 // dy window rows
+  const int WROW = outC * 2 + 16;             // bytes, bank-staggered
+  const int TB = 32 * outC * 2;               // wt tile bytes (<= 8 KB)
+  char* smW = sm;                             // 2 x TB wt tiles
+  char* smY = sm + 2 * TB;                    // G dy windows
+  const int wstart = h0 + PH - (KH - 1);      // first dy row in the window
+
+  // stage the G dy windows (rows outside [0, OH) or images >= B are zeros)
+  for (int g = 0; g < G; ++g) {
+    const int img = img0 + g;
+    const bf16* dyi = dy2 + (long)img * OH * ldy;
+    for (int i = tid * 8; i < WR * outC; i += 256 * 8) {
+      const int r = i / outC, c0 = i - r * outC;
+      const int dyr = wstart + r;
+      bf16x8 v = {};
+      if (img < B && dyr >= 0 && dyr < OH)
+        v = *(const bf16x8*)(dyi + (long)dyr * ldy + c0);
+      *(bf16x8*)(smY + g * WR * WROW + r * WROW + c0 * 2) = v;
+    }
+  }
+
+  const int KCH = outC / 32;                  // oc chunks, <= 4
+  const int NCB = C / 32;                     // channel blocks
+  const int cmask = (outC / 8 - 1) & 7;
+  auto wswz = [cmask](int row, int kel) {
+    return (kel ^ ((row & cmask) << 3));
+  };
+  // async-stage split (r1 T14 trick): tile s+2 is FETCHED into registers
+  // before iteration s's compute and tile s+1 is WRITTEN to LDS after it, so
+  // every wt L2 read gets a full iteration (compute + barrier) of latency
+  // hiding instead of racing the very next barrier.
+  // Per-thread tile portion: 32*outC/(256*8) <= 2 bf16x8.
+  const int NP = (32 * outC + 256 * 8 - 1) / (256 * 8);  // 1 or 2
+  auto fetch_tile = [&](int kh, int cb, bf16x8 (&regs)[2]) {
+    // tile: 32 kpad-rows (channels cb*32..+32 of tap kh) x outC
+    const int n0 = kh * C + cb * 32;
+    #pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      if (p >= NP) break;
+      const int i = tid * 8 + p * 256 * 8;
+      if (i < 32 * outC) {
+        const int r = i / outC, k0 = i - r * outC;
+        regs[p] = *(const bf16x8*)(wt + (long)(n0 + r) * ldw + k0);
+      }
+    }
+  };
+  auto write_tile = [&](int buf, const bf16x8 (&regs)[2]) {
+    #pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      if (p >= NP) break;
+      const int i = tid * 8 + p * 256 * 8;
+      if (i < 32 * outC) {
+        const int r = i / outC, k0 = i - r * outC;
+        *(bf16x8*)(smW + buf * TB + r * outC * 2 + wswz(r, k0) * 2) = regs[p];
+      }
+    }
+  };
+
+  const int NT = NCB * KH;
+  // two NAMED in-flight tile register sets (a runtime-indexed array here
+  // sends them to scratch — measured 5x slower); parity-selected branches
+  // on the uniform sidx keep them in VGPRs
+  bf16x8 tA[2], tB[2];
+  {
+    bf16x8 t0[2];
+    fetch_tile(0, 0, t0);
+    write_tile(0, t0);
+  }
+  if (NT > 1) fetch_tile(1 % KH, 1 / KH, tB);
+  const int hrow = wave * 16 + (lane & 15);   // this lane's dx row - h0
+  const int kcol = (lane >> 4) * 8;           // oc sub-offset in fragments
+  int sidx = 0;
+  int f_kh = 2 % KH, f_cb = 2 / KH;           // next tile to fetch (s+2)
+  for (int cb = 0; cb < NCB; ++cb) {
+    f32x4 acc[2][2] = {};                     // [g][mi]; mi = channel half
+    for (int kh = 0; kh < KH; ++kh, ++sidx) {
+      const int buf = sidx & 1;
+      __syncthreads();                        // tile(buf) + window visible
+      if (sidx + 2 < NT) {
+        if (buf) fetch_tile(f_kh, f_cb, tB);
+        else     fetch_tile(f_kh, f_cb, tA);
+        if (++f_kh == KH) { f_kh = 0; ++f_cb; }
+      }
+      const int lrow = hrow + (KH - 1) - kh;  // dy row in the window
+      const char* win0 = smY + lrow * WROW;
+      #pragma unroll
+      for (int kc = 0; kc < 4; ++kc) {
+        if (kc >= KCH) break;
+        // wf fragments are g-invariant: read once, use for both images
+        // (16 LDS reads for 16 MFMAs; the per-g re-read was 24)
+        const bf16x8 dyf0 = *(const bf16x8*)(win0 + (kc * 32 + kcol) * 2);
+        bf16x8 dyf1 = {};
+        if (G > 1)
+          dyf1 = *(const bf16x8*)(win0 + WR * WROW + (kc * 32 + kcol) * 2);
+        #pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+          const int row = mi * 16 + (lane & 15);
+          const bf16x8 wf = *(const bf16x8*)(
+              smW + buf * TB + row * outC * 2 +
+              wswz(row, kc * 32 + kcol) * 2);
+          acc[0][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              wf, dyf0, acc[0][mi], 0, 0, 0);
+          if (G > 1)
+            acc[1][mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                wf, dyf1, acc[1][mi], 0, 0, 0);
+        }
+      }
+      // LDS-write tile s+1 (fetched last iteration) after the compute so
+      // the write waits on its global loads as late as possible
+      if (sidx + 1 < NT) {
+        if (buf) write_tile(buf ^ 1, tA);
+        else     write_tile(buf ^ 1, tB);
+      }
+    }
+    // store this cb's channels: lane owns dx row hd, channels c..c+3
+    const int hd = h0 + hrow;
+    if (hd < H) {
+      #pragma unroll
+      for (int g = 0; g < 2; ++g) {
+        if (g >= G || img0 + g >= B) break;
+        bf16* dxp = dx + ((long)(img0 + g) * H + hd) * C + cb * 32;
+        #pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+          const int c = mi * 16 + (lane >> 4) * 4;
+          bf16x4 v;
+          if (accumulate) {
+            const bf16x4 prev = *(const bf16x4*)(dxp + c);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j)
+              v[j] = tobf16(acc[g][mi][j] + tofloat(prev[j]));
+          } else {
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) v[j] = tobf16(acc[g][mi][j]);
+          }
+          *(bf16x4*)(dxp + c) = v;
+        }
+      }
+    }
+  }
+}
+
 bool launch_conv1d_dx(const void* dy2, long ldy, const void* wt, long ldw,
                       void* dx, int B, int H, int C, int KH, int PH, int OH,
                       int outC, int accumulate, hipStream_t s) {
+  if (C % 32 != 0 || outC % 32 != 0 || outC > 128)
+    return false;
+  static const bool use_v1 = [] {
+    const char* e = getenv("LO_CONV1D_DX_V1");
+    return e && e[0] == '1';
+  }();
+  if (!use_v1) {
+    const int WR = 64 + KH - 1;
+    const int WROW = outC * 2 + 16;
+    const int TB = 32 * outC * 2;
+    int G = 2;
+    if (2 * TB + G * WR * WROW > 56 * 1024) G = 1;
+    if (2 * TB + G * WR * WROW <= 56 * 1024) {
+      const int T = (H + 63) / 64;
+      hipLaunchKernelGGL(conv1d_dx2_kernel, dim3((B + G - 1) / G, T),
+                         dim3(256), 2 * TB + G * WR * WROW, s,
+                         (const bf16*)dy2, ldy, (const bf16*)wt, ldw,
+                         (bf16*)dx, B, H, C, KH, PH, OH, outC, G, accumulate);
+      return true;
+    }
+  }
   const int lds = 64 * (C + 4) * 4 + 2 * 32 * outC * 2;
-  if (C % 32 != 0 || outC % 32 != 0 || outC > 128 || lds > 56 * 1024)
+  if (lds > 56 * 1024)
     return false;
   const int TS = 64 - KH + 1;
   const int T = (H + TS - 1) / TS;

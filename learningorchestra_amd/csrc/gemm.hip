@@ -660,9 +660,17 @@ static bool dispatch_conv_fwd(const GemmArgs& g, hipStream_t s) {
 // implicit-conv dW: C[M=outC, N=kpad] = dY^T @ im2col(x), split-K atomics
 static bool dispatch_conv_dw(const GemmArgs& g, hipStream_t s) {
   if (!g.out_f32 || g.epi != 0 || !g.ta || g.tb) return false;
-  // BM=64 when outC allows: halves the number of M-tiles and with it the
-  // total conv_gather8 work (the B gather dominates this kernel)
-  if (g.M >= 64)
+  // Tile-width choice is a TRAFFIC decision for these tall-skinny deep-K
+  // shapes: the A operand (dY) is re-streamed once per N(kpad)-tile and the
+  // gathered B (x windows) once per M(outC)-tile, so wider tiles cut the
+  // dominant re-reads (TextCNN dW measured 4x over its traffic floor with
+  // 64x64 tiles, r2). BM=64 when outC allows also halves conv_gather8 work.
+  // (outC=64 shapes measured FASTER on 64x64 than 64x128 — MNIST conv2 dW
+  // regressed 6.78 -> 6.35 M samples/s on the wide tile — so wide tiles
+  // require outC >= 128)
+  if (g.M >= 128 && g.N >= 128 && g.K >= (1 << 18))
+    launch_cfg<128, 128, 64, 2, 2, true, false, 0, true, true, 2>(g, s);
+  else if (g.M >= 64)
     launch_cfg<64, 64, 64, 2, 2, true, false, 0, true, true, 2>(g, s);
   else
     launch_cfg<32, 64, 64, 1, 4, true, false, 0, true, true, 2>(g, s);
