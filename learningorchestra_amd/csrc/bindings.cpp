@@ -116,7 +116,7 @@ void launch_softmax_ce(const void* logits, const void* labels, void* dlogits,
                        float gscale, hipStream_t s);
 void launch_tree_hist(const void* binned, const void* node_of, const void* grad,
                       const void* hess, void* hist, long N, int F, int n_nodes,
-                      int B, hipStream_t s);
+                      int B, double gbound, double hbound, hipStream_t s);
 void launch_embedding_fwd(const void* ids, const void* table, void* out,
                           long n, int dim, hipStream_t s);
 void launch_embedding_bwd(const void* ids, const void* dy, void* gtable,
@@ -577,7 +577,8 @@ void softmax_ce(at::Tensor logits, at::Tensor labels, at::Tensor dlogits,
 }
 
 void tree_hist(at::Tensor binned, at::Tensor node_of, at::Tensor grad,
-               at::Tensor hess, at::Tensor hist, int64_t n_nodes, int64_t n_bins) {
+               at::Tensor hess, at::Tensor hist, int64_t n_nodes, int64_t n_bins,
+               double gbound = 0.0, double hbound = 0.0) {
   TORCH_CHECK(binned.is_cuda() && binned.scalar_type() == at::kByte &&
               binned.is_contiguous(), "binned must be contiguous u8 GPU");
   TORCH_CHECK(node_of.scalar_type() == at::kInt && node_of.is_contiguous());
@@ -590,7 +591,7 @@ void tree_hist(at::Tensor binned, at::Tensor node_of, at::Tensor grad,
   TORCH_CHECK(hist.numel() == n_nodes * F * n_bins * 2, "hist size");
   lo::launch_tree_hist(binned.data_ptr(), node_of.data_ptr(), grad.data_ptr(),
                        hess.data_ptr(), hist.data_ptr(), N, F, (int)n_nodes,
-                       (int)n_bins, stream());
+                       (int)n_bins, gbound, hbound, stream());
 }
 
 at::Tensor embedding_fwd(at::Tensor ids, at::Tensor table, at::Tensor out) {
@@ -641,7 +642,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum", &colsum);
   m.def("argmax_rows", &argmax_rows);
   m.def("accuracy_count", &accuracy_count);
-  m.def("tree_hist", &tree_hist);
+  m.def("tree_hist", &tree_hist, py::arg("binned"), py::arg("node_of"),
+        py::arg("grad"), py::arg("hess"), py::arg("hist"),
+        py::arg("n_nodes"), py::arg("n_bins"),
+        py::arg("gbound") = 0.0, py::arg("hbound") = 0.0);
   m.def("embedding_fwd", &embedding_fwd);
   m.def("bn_stats", &bn_stats);
   m.def("bn_fwd", &bn_fwd, py::arg("x"), py::arg("y"), py::arg("mean"),

@@ -55,15 +55,26 @@ def quantize(X: torch.Tensor, n_bins: int = MAX_BINS,
 
 def build_histograms(binned: torch.Tensor, node_of: torch.Tensor,
                      grad: torch.Tensor, hess: torch.Tensor,
-                     n_nodes: int, n_bins: int = MAX_BINS + 1) -> torch.Tensor:
+                     n_nodes: int, n_bins: int = MAX_BINS + 1,
+                     bounds: Optional[Tuple[float, float]] = None
+                     ) -> torch.Tensor:
     """-> hist [n_nodes, F, n_bins, 2] (sum grad, sum hess) for samples with
-    node_of >= 0. GPU: tree_hist HIP kernel; CPU: torch index_add reference."""
+    node_of >= 0. GPU: tree_hist HIP kernel; CPU: torch index_add reference.
+
+    ``bounds`` = (sum|g|, sum h) upper bounds over active rows: enables the
+    packed-u64 fixed-point LDS-atomic path (one 8-byte atomic per (row,
+    feature) instead of two fp32 — measured +22% on the 10M-row config; the
+    DS pipe still spends 2 bank cycles on the 8-byte op, so the issue-count
+    halving does not double throughput). Callers compute them once per tree
+    (loss-specific bounds) so no per-level device sync is needed."""
     N, F = binned.shape
     if binned.is_cuda:
         lo = require_ext()
         hist = torch.zeros(n_nodes * F * n_bins * 2, device=binned.device,
                            dtype=torch.float32)
-        lo.tree_hist(binned, node_of, grad, hess, hist, n_nodes, n_bins)
+        gb, hb = bounds if bounds is not None else (0.0, 0.0)
+        lo.tree_hist(binned, node_of, grad, hess, hist, n_nodes, n_bins,
+                     gbound=float(gb), hbound=float(hb))
         return hist.view(n_nodes, F, n_bins, 2)
     hist = torch.zeros(n_nodes * F * n_bins, 2, dtype=torch.float32)
     active = node_of >= 0
@@ -108,7 +119,10 @@ def build_class_histograms(binned: torch.Tensor, node_of: torch.Tensor,
         g = onehot[:, k0].contiguous()
         h = (onehot[:, k0 + 1].contiguous() if k0 + 1 < K
              else torch.zeros_like(g))
-        parts.append(build_histograms(binned, node_of, g, h, n_nodes, n_bins))
+        # one-hot columns: both sums are bounded by N (counts), sync-free
+        parts.append(build_histograms(binned, node_of, g, h, n_nodes, n_bins,
+                                      bounds=(float(onehot.shape[0]),
+                                              float(onehot.shape[0]))))
     hist = torch.cat(parts, dim=-1)
     return hist[..., :K].contiguous() if hist.shape[-1] != K else hist
 
@@ -176,6 +190,11 @@ class TreeLearner:
         device = binned.device
         total_nodes = 2 ** (self.max_depth + 1) - 1
         tree = Tree(total_nodes, device)
+        # per-tree quantization bounds (one sync per tree, valid for every
+        # level since the active set only shrinks)
+        bounds = None
+        if binned.is_cuda:
+            bounds = (float(grad.abs().sum()), float(hess.sum()))
         node_of = torch.zeros(N, dtype=torch.int32, device=device)
         if sample_mask is not None:
             node_of = torch.where(sample_mask, node_of,
@@ -185,7 +204,8 @@ class TreeLearner:
         for depth in range(self.max_depth + 1):
             level_nodes = 2 ** depth
             rel_node = node_of - level_start
-            hist = build_histograms(binned, rel_node, grad, hess, level_nodes)
+            hist = build_histograms(binned, rel_node, grad, hess, level_nodes,
+                                    bounds=bounds)
             if sync:
                 # data-parallel trees: shards contribute partial histograms;
                 # the reduced histogram makes every rank's splits identical
