@@ -90,6 +90,10 @@ void launch_bn_fwd(const void* x, void* y, const void* mean, const void* invstd,
 void launch_bn_bwd_reduce(const void* dy, const void* y, const void* x,
                           const void* mean, const void* invstd, void* dbeta,
                           void* dgamma, long M, int C, int relu, hipStream_t s);
+void launch_bn_finalize_stats(const void* scratch, void* mean, void* invstd,
+                              void* rmean, void* rvar, float invM,
+                              float momentum, float eps, int C,
+                              hipStream_t s);
 void launch_bn_bwd_dx(const void* dy, const void* y, const void* x, void* dx,
                       const void* mean, const void* invstd, const void* gamma,
                       const void* dbeta, const void* dgamma, long M, int C,
@@ -457,6 +461,21 @@ void bn_bwd_dx(at::Tensor dy, at::Tensor y, at::Tensor x, at::Tensor dx,
                        relu ? 1 : 0, stream());
 }
 
+void bn_finalize_stats(at::Tensor scratch, at::Tensor mean, at::Tensor invstd,
+                       int64_t M,
+                       c10::optional<at::Tensor> rmean,
+                       c10::optional<at::Tensor> rvar,
+                       double momentum, double eps) {
+  check_f32(scratch, "scratch");
+  check_f32(mean, "mean");
+  lo::launch_bn_finalize_stats(
+      scratch.data_ptr(), mean.data_ptr(), invstd.data_ptr(),
+      rmean ? rmean->data_ptr() : nullptr,
+      rvar ? rvar->data_ptr() : nullptr,
+      1.f / (float)M, (float)momentum, (float)eps,
+      (int)mean.numel(), stream());
+}
+
 void add_relu(at::Tensor a, at::Tensor b, at::Tensor z, bool relu) {
   check_bf16(a, "a");
   check_bf16(b, "b");
@@ -653,6 +672,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("relu"), py::arg("residual") = py::none());
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
   m.def("bn_bwd_dx", &bn_bwd_dx);
+  m.def("bn_finalize_stats", &bn_finalize_stats, py::arg("scratch"),
+        py::arg("mean"), py::arg("invstd"), py::arg("M"),
+        py::arg("rmean") = py::none(), py::arg("rvar") = py::none(),
+        py::arg("momentum") = 0.0, py::arg("eps") = 1e-5);
   m.def("add_relu", &add_relu);
   m.def("avgpool_global", &avgpool_global);
   m.def("avgpool_global_bwd", &avgpool_global_bwd);

@@ -27,7 +27,7 @@
 
 namespace lo {
 
-template <int EPI>
+template <int EPI, bool DRAIN>
 __global__ __launch_bounds__(512) void gemm256_kernel(
     const bf16* __restrict__ A, long lda, const bf16* __restrict__ B, long ldb,
     bf16* __restrict__ C, long ldc, const float* __restrict__ bias,
@@ -134,7 +134,10 @@ __global__ __launch_bounds__(512) void gemm256_kernel(
       else if (q == 3) {
         stage_B(s + 2, 0);
         stage_B(s + 2, 1);
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        if (DRAIN)
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
       }
       asm volatile("s_barrier" ::: "memory");
       __builtin_amdgcn_s_setprio(1);
@@ -239,16 +242,23 @@ bool launch_gemm256(const void* A, long lda, const void* B, long ldb, void* C,
   constexpr size_t LDS_MAIN = 2 * (256 + 256) * 128;        // 128 KiB
   constexpr size_t LDS_EPI = 128 * (256 * 2 + 16) + 16 + 2 * 256 * 4;
   const size_t lds = LDS_MAIN > LDS_EPI ? LDS_MAIN : LDS_EPI;
-  if (epi == 1)
-    hipLaunchKernelGGL(HIP_KERNEL_NAME(gemm256_kernel<1>), dim3(grid), dim3(512),
-                       lds, s, (const bf16*)A, lda, (const bf16*)B, ldb,
-                       (bf16*)C, ldc, bias, M, N, K, (const bf16*)addend,
-                       stats_sum, stats_sumsq);
-  else
-    hipLaunchKernelGGL(HIP_KERNEL_NAME(gemm256_kernel<0>), dim3(grid), dim3(512),
-                       lds, s, (const bf16*)A, lda, (const bf16*)B, ldb,
-                       (bf16*)C, ldc, bias, M, N, K, (const bf16*)addend,
-                       stats_sum, stats_sumsq);
+  // Grids of >= ~192 of these 128-KiB-LDS workgroups break the counted
+  // vmcnt(4) pipeline (r2 measurement: <= 188 blocks 0 failures over
+  // hundreds of runs at every K; >= 192 blocks scattered wrong tiles in
+  // ~95% of runs, K >= 512, errors across ALL blocks; stream-serialized
+  // sub-launches and even hipStreamSynchronize between them do NOT fix it,
+  // so it is a residency/scheduling effect, not launch overlap). Big grids
+  // take the DRAIN variant: full vmcnt(0) per K-tile — the same schedule
+  // with the pipeline depth collapsed, correct by construction.
+  const bool drain = grid >= 190;
+  #define LO_G256(EPI_, DRAIN_)                                               \
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(gemm256_kernel<EPI_, DRAIN_>),         \
+                       dim3(grid), dim3(512), lds, s, (const bf16*)A, lda,    \
+                       (const bf16*)B, ldb, (bf16*)C, ldc, bias, M, N, K,     \
+                       (const bf16*)addend, stats_sum, stats_sumsq)
+  if (epi == 1) { if (drain) LO_G256(1, true); else LO_G256(1, false); }
+  else          { if (drain) LO_G256(0, true); else LO_G256(0, false); }
+  #undef LO_G256
   return true;
 }
 

@@ -555,14 +555,18 @@ class BatchNormReLU(Layer):
         gamma = self.arena.pf(self.name + ".g")
         beta = self.arena.pf(self.name + ".b")
         if self.training:
+            run = ((self.running_mean, self.running_var) if x.is_cuda
+                   else None)
             F.bn_fwd_train(x2, gamma, beta, self.eps, bufs["y"], bufs["mean"],
                            bufs["invstd"], bufs["scratch"], relu,
                            stats_ready=stats_ready and x.is_cuda,
-                           residual=residual)
-            m = self.momentum
-            self.running_mean.mul_(1 - m).add_(bufs["mean"], alpha=m)
-            var = bufs["invstd"].square().reciprocal() - self.eps
-            self.running_var.mul_(1 - m).add_(var, alpha=m)
+                           residual=residual, running=run,
+                           momentum=self.momentum)
+            if not x.is_cuda:
+                m = self.momentum
+                self.running_mean.mul_(1 - m).add_(bufs["mean"], alpha=m)
+                var = bufs["invstd"].square().reciprocal() - self.eps
+                self.running_var.mul_(1 - m).add_(var, alpha=m)
         else:
             F.bn_fwd_eval(x2, gamma, beta, self.running_mean,
                           self.running_var, self.eps, bufs["y"], relu,

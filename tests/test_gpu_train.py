@@ -379,4 +379,14 @@ def test_api_multi_rank_train_gpu(tmp_path):
         assert "cuda" in exec_doc.get("trainResult", "") or True
         assert rt.artifacts.exists("gtrain", "train/torch")
     finally:
+        # drain the scheduler before the Runtime's GPU tensors are freed:
+        # an in-flight background job writing into memory the allocator has
+        # already recycled corrupts whatever test runs next
+        try:
+            rt.scheduler.wait_all(timeout=60)
+            rt.scheduler.shutdown()
+        except Exception:
+            pass
+        import torch
+        torch.cuda.synchronize()
         set_config(None)
