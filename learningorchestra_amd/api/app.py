@@ -50,6 +50,34 @@ EXECUTOR_TOOLS = ("scikitlearn", "tensorflow", "torch")
 BINARY_VERBS = ("train", "tune", "evaluate", "predict")
 
 
+class ResponseCache:
+    """Gateway-style GET response cache (the reference's KrakenD cached
+    responses for 300 s, krakend.json:1769-1770). Any mutation clears the
+    whole cache — coarse but correct for a single-node gateway."""
+
+    def __init__(self, ttl: float):
+        self.ttl = ttl
+        self._store: Dict[str, Any] = {}
+        self._lock = threading.Lock()
+
+    def get_or(self, key: str, fn):
+        if self.ttl <= 0:
+            return fn()
+        now = time.time()
+        with self._lock:
+            hit = self._store.get(key)
+            if hit and hit[0] > now:
+                return hit[1]
+        value = fn()
+        with self._lock:
+            self._store[key] = (now + self.ttl, value)
+        return value
+
+    def clear(self) -> None:
+        with self._lock:
+            self._store.clear()
+
+
 class Runtime:
     """Bundles storage + executor + verb services (the whole L2-L6 stack)."""
 
@@ -71,6 +99,7 @@ class Runtime:
         self.builder = BuilderService(self.db, self.artifacts, self.scheduler,
                                       self.cfg.allow_user_code, device=gpu_device)
         self._name_lock = threading.Lock()
+        self.cache = ResponseCache(getattr(self.cfg, "cache_ttl", 0.0))
 
     # -- shared helpers ------------------------------------------------------
     def require_unique(self, name: str) -> None:
@@ -106,9 +135,13 @@ def create_app(runtime: Optional[Runtime] = None) -> FastAPI:
         return JSONResponse({RESULT: str(exc)},
                             status_code=getattr(exc, "status", 406))
 
+    def _inval():
+        rt.cache.clear()
+
     # ------------------------------------------------------------- dataset --
     @app.post(PREFIX + "/dataset/{tool}", status_code=201)
     def create_dataset(tool: str, body: Dict[str, Any]):
+        _inval()
         _check_tool(tool, ("csv", "generic"))
         name = _name_field(body, "datasetName")
         uri = _field(body, "datasetURI")
@@ -123,6 +156,7 @@ def create_app(runtime: Optional[Runtime] = None) -> FastAPI:
     # ---------------------------------------------------------------- model --
     @app.post(PREFIX + "/model/{tool}", status_code=201)
     def create_model(tool: str, body: Dict[str, Any]):
+        _inval()
         _check_tool(tool, EXECUTOR_TOOLS)
         name = _name_field(body, "modelName")
         with rt._name_lock:
@@ -135,6 +169,7 @@ def create_app(runtime: Optional[Runtime] = None) -> FastAPI:
 
     @app.patch(PREFIX + "/model/{tool}/{name}")
     def update_model(tool: str, name: str, body: Dict[str, Any]):
+        _inval()
         _check_tool(tool, EXECUTOR_TOOLS)
         meta = rt.require_exists(name)
         rt.execution.create_model(
@@ -147,6 +182,7 @@ def create_app(runtime: Optional[Runtime] = None) -> FastAPI:
     # ------------------------------------------- train/tune/evaluate/predict --
     @app.post(PREFIX + "/{verb}/{tool}", status_code=201)
     def create_binary_execution(verb: str, tool: str, body: Dict[str, Any]):
+        _inval()
         if verb == "builder":
             return _builder_post(rt, body)
         if verb == "function":
@@ -169,11 +205,13 @@ def create_app(runtime: Optional[Runtime] = None) -> FastAPI:
 
     @app.patch(PREFIX + "/transform/projection")
     def update_projection(body: Dict[str, Any]):
+        _inval()
         # reference PATCHes projections at the collection URI (krakend table)
         return _projection_post(rt, body)
 
     @app.patch(PREFIX + "/{verb}/{tool}/{name}")
     def update_execution(verb: str, tool: str, name: str, body: Dict[str, Any]):
+        _inval()
         if verb == "model":
             return update_model(tool, name, body)
         rt.require_exists(name)
@@ -190,6 +228,7 @@ def create_app(runtime: Optional[Runtime] = None) -> FastAPI:
     # ---------------------------------------------------- dataType (PATCH) --
     @app.patch(PREFIX + "/transform/dataType")
     def transform_datatype(body: Dict[str, Any]):
+        _inval()
         name = _field(body, "datasetName")
         rt.require_exists(name)
         rt.datatype.convert(name, _field(body, "types"))
@@ -250,7 +289,8 @@ def create_app(runtime: Optional[Runtime] = None) -> FastAPI:
     @app.get(PREFIX + "/{verb}/{tool}")
     def catalog(verb: str, tool: str):
         stype = f"{verb}/{tool}"
-        return {RESULT: rt.metadata.catalog(stype)}
+        return rt.cache.get_or(f"catalog:{stype}",
+                               lambda: {RESULT: rt.metadata.catalog(stype)})
 
     # ------------------------------------------------------- rows/metadata --
     @app.get(PREFIX + "/{verb}/{tool}/{name}")
@@ -272,7 +312,9 @@ def create_app(runtime: Optional[Runtime] = None) -> FastAPI:
                 raise ValueError("query must be a JSON object")
         except ValueError as exc:
             raise ValidationError(f"malformed query: {exc}")
-        return {RESULT: rt.read_rows(name, q, skip, max(limit, 0))}
+        key = f"rows:{name}:{query}:{limit}:{skip}"
+        return rt.cache.get_or(
+            key, lambda: {RESULT: rt.read_rows(name, q, skip, max(limit, 0))})
 
     @app.get(PREFIX + "/{verb}/{tool}/{name}/metadata")
     def read_metadata(verb: str, tool: str, name: str):
@@ -281,6 +323,7 @@ def create_app(runtime: Optional[Runtime] = None) -> FastAPI:
     # -------------------------------------------------------------- delete --
     @app.delete(PREFIX + "/{verb}/{tool}/{name}")
     def delete(verb: str, tool: str, name: str):
+        _inval()
         rt.require_exists(name)
         rt.execution.delete(name)
         return {RESULT: f"deleted {name}"}

@@ -37,6 +37,14 @@ class Config:
     limit_param_max: int = 100
     metadata_row_id: int = 0
 
+    # --- gateway-style response cache (krakend.json:1769-1770: 300 s
+    # cache_ttl on the reference's gateway). DELIBERATE deviation: default 0
+    # (off) — a 300 s GET cache makes the finished-flag poll contract serve
+    # stale "finished: false" for minutes, which the reference shipped as a
+    # quirk; enable with LO_CACHE_TTL for reference-faithful behavior.
+    # Mutations invalidate the cache either way.
+    cache_ttl: float = field(default_factory=lambda: float(_env("LO_CACHE_TTL", "0")))
+
     # --- executor ----------------------------------------------------------
     max_jobs: int = field(default_factory=lambda: int(_env("LO_MAX_JOBS", "8")))
     # exec() of user-supplied code (builder modelingCode, function/python) is

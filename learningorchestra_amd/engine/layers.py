@@ -63,13 +63,18 @@ class Conv2dNHWC(Layer):
         # (3x3 convs are compute-dense; see PERFORMANCE.md A/B entries).
         self.implicit = implicit
         self.in_c, self.out_c = in_c, out_c
-        self.kh, self.kw, self.stride, self.pad = kh, kw, stride, pad
+        # pad: int (square) or (pad_h, pad_w) — a W==1 sequence conv with
+        # h-padding is (p, 0) (padded 1-D convs, r1 fuzz-found gate)
+        self.kh, self.kw, self.stride = kh, kw, stride
+        self.ph, self.pw = pad if isinstance(pad, (tuple, list)) else (pad, pad)
+        self.pad = self.ph  # legacy alias (square users had ph == pw)
         self.relu = relu
         self.first = first  # input layer: skip dX
         self.kdim = kh * kw * in_c
         self.kpad = _pad8(self.kdim)
         # 1x1 stride-1 convs skip im2col entirely (col IS the input)
-        self._is_1x1 = (kh == 1 and kw == 1 and stride == 1 and pad == 0
+        self._is_1x1 = (kh == 1 and kw == 1 and stride == 1
+                        and self.ph == 0 and self.pw == 0
                         and self.kpad == in_c)
         self.arena: Optional[ParamArena] = None
         self._bufs = {}
@@ -94,8 +99,8 @@ class Conv2dNHWC(Layer):
         return [self.name + ".w"] + ([self.name + ".b"] if self.bias else [])
 
     def _alloc(self, B: int, H: int, W: int, dev, dtype):
-        OH = (H + 2 * self.pad - self.kh) // self.stride + 1
-        OW = (W + 2 * self.pad - self.kw) // self.stride + 1
+        OH = (H + 2 * self.ph - self.kh) // self.stride + 1
+        OW = (W + 2 * self.pw - self.kw) // self.stride + 1
         key = (B, H, W)
         if self._bufs.get("key") != key:
             M = B * OH * OW
@@ -129,11 +134,12 @@ class Conv2dNHWC(Layer):
                 (H * W * self.in_c) % 8 == 0)
 
     def _conv1d_ok(self, W: int) -> bool:
-        # mirror of launch_conv1d_* eligibility (W==1 sequence convs)
-        # pad must be 0: Conv2dNHWC pads BOTH dims, and w-padding of a
-        # W==1 input (OW = 2p+1) has no 1-D representation
+        # mirror of launch_conv1d_* eligibility (W==1 sequence convs).
+        # h-padding is fine (the kernels stage zero rows); only w-padding is
+        # excluded — w-padding a W==1 input (OW = 2p+1) has no 1-D form, so
+        # padded sequence convs use pad=(p, 0) (r1 fuzz-found gate, fixed r2)
         return (W == 1 and self.kw == 1 and self.stride == 1 and
-                self.pad == 0 and
+                self.pw == 0 and
                 self.in_c % 32 == 0 and self.out_c % 64 == 0 and
                 self.out_c <= 128 and (self.kh * self.in_c) % 32 == 0 and
                 (64 + self.kh - 1) * (self.in_c * 2 + 16) + 128 + 8192
@@ -177,22 +183,22 @@ class Conv2dNHWC(Layer):
             # fused kernel (x image fits LDS), or im2col gathered inside the
             # GEMM staging as the fallback
             if self._conv1d_ok(W) and F.conv1d_fwd(
-                    x, self.arena.p(self.name + ".w"), self.kh, self.pad,
+                    x, self.arena.p(self.name + ".w"), self.kh, self.ph,
                     bias=bias, relu=self.relu, out=bufs["y"]):
                 pass
             elif not (self._fwd_small_ok(H, W) and F.conv2d_fwd_small(
                     x, self.arena.p(self.name + ".w"), self.kh, self.kw,
-                    self.stride, self.stride, self.pad, self.pad, bias=bias,
+                    self.stride, self.stride, self.ph, self.pw, bias=bias,
                     relu=self.relu, out=bufs["y"])):
                 F.conv2d_fwd_implicit(x, self.arena.p(self.name + ".w"),
                                       self.kh, self.kw, self.stride,
-                                      self.stride, self.pad, self.pad,
+                                      self.stride, self.ph, self.pw,
                                       bias=bias, relu=self.relu,
                                       out=bufs["y"])
             st = None
         else:
-            F.im2col(x, self.kh, self.kw, self.stride, self.stride, self.pad,
-                     self.pad, self.kpad, out=bufs["col"])
+            F.im2col(x, self.kh, self.kw, self.stride, self.stride, self.ph,
+                     self.pw, self.kpad, out=bufs["col"])
             F.gemm(bufs["col"], self.arena.p(self.name + ".w"), tb=True,
                    bias=bias, relu=self.relu, out=bufs["y"], stats=st)
         self.stats_filled = st is not None
@@ -220,10 +226,10 @@ class Conv2dNHWC(Layer):
         if dy2.is_cuda and self.implicit and not self._is_1x1:
             if not (self.in_c == 1 and self.out_c <= 32 and F.conv2d_dw_c1(
                     dy2, self._x, gw, self.kh, self.kw, self.stride,
-                    self.stride, self.pad, self.pad)):
+                    self.stride, self.ph, self.pw)):
                 F.conv2d_dw_implicit(dy2, self._x, gw, self.kh, self.kw,
-                                     self.stride, self.stride, self.pad,
-                                     self.pad, splits)
+                                     self.stride, self.stride, self.ph,
+                                     self.pw, splits)
         else:
             F.gemm(dy2, bufs["col"], ta=True, out=gw, splits=splits)
         if self.bias:
@@ -242,19 +248,19 @@ class Conv2dNHWC(Layer):
         assert dx_addend is None or self._is_1x1, "dx_addend: 1x1 only"
         if dy2.is_cuda and self.implicit and self._conv1d_ok(bufs["W"]):
             tgt = dx_out if dx_out is not None else bufs["dx"]
-            if F.conv1d_dx(dy2, self._wt(), self.kh, self.pad, out=tgt,
+            if F.conv1d_dx(dy2, self._wt(), self.kh, self.ph, out=tgt,
                            accumulate=dx_accumulate):
                 return tgt
         if dy2.is_cuda and self._dx_fused_ok(bufs["H"], bufs["W"]) \
                 and F.conv2d_dx_fused(
                 dy2, self._wt(), bufs["B"], bufs["H"], bufs["W"], self.in_c,
-                self.kh, self.kw, self.stride, self.stride, self.pad,
-                self.pad, out=bufs["dx"]):
+                self.kh, self.kw, self.stride, self.stride, self.ph,
+                self.pw, out=bufs["dx"]):
             return bufs["dx"]
         F.gemm(dy2, self._wt(), tb=True, out=bufs["dcol"])
         F.col2im(bufs["dcol"], bufs["B"], bufs["H"], bufs["W"], self.in_c,
-                 self.kh, self.kw, self.stride, self.stride, self.pad,
-                 self.pad, out=bufs["dx"])
+                 self.kh, self.kw, self.stride, self.stride, self.ph,
+                 self.pw, out=bufs["dx"])
         return bufs["dx"]
 
 

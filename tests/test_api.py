@@ -422,3 +422,59 @@ def test_train_torch_multi_rank(client):
                                       device="cpu")
     assert fitted.predict(
         __import__("numpy").random.rand(4, 784).astype("float32")).shape == (4,)
+
+
+def test_gateway_response_cache(tmp_path, monkeypatch):
+    """Reference-parity GET cache (krakend cache_ttl 300 s): repeated reads
+    are served from cache within the TTL; any mutation invalidates."""
+    from learningorchestra_amd.api.app import Runtime, create_app
+    from learningorchestra_amd.config import Config, set_config
+    cfg = Config(data_root=str(tmp_path), mongo_uri="", cache_ttl=300.0)
+    set_config(cfg)
+    try:
+        rt = Runtime(cfg)
+        client = TestClient(create_app(rt))
+        client.rt = rt
+        ingest_titanic(client, tmp_path, name="ct")
+        r1 = client.get(f"{PREFIX}/dataset/csv/ct", params={"limit": 3})
+        assert r1.status_code == 200
+        # mutate the collection BEHIND the API: cached read doesn't see it
+        rt.db["ct"].update_one({"_id": 1}, {"$set": {"Sex": "mutated"}})
+        r2 = client.get(f"{PREFIX}/dataset/csv/ct", params={"limit": 3})
+        assert r2.json() == r1.json()
+        # an API mutation invalidates the cache
+        r = client.post(f"{PREFIX}/transform/projection",
+                        json={"inputDatasetName": "ct",
+                              "outputDatasetName": "ct_p",
+                              "names": ["Sex"]})
+        assert r.status_code == 201
+        r3 = client.get(f"{PREFIX}/dataset/csv/ct", params={"limit": 3})
+        assert any(x.get("Sex") == "mutated" for x in r3.json()["result"])
+    finally:
+        set_config(None)
+
+
+def test_observe_many_concurrent_waiters(client):
+    """100 concurrent Observe waiters park on events (not worker threads) and
+    all wake when the flag flips; normal traffic stays responsive."""
+    import concurrent.futures
+    import threading
+
+    client.rt.metadata.create_file("slowjob", "train/torch")
+
+    def waiter(_i):
+        r = client.get(f"{PREFIX}/observe/slowjob/wait",
+                       params={"timeoutSeconds": 30})
+        return r.json()["result"].get("finished")
+
+    with concurrent.futures.ThreadPoolExecutor(max_workers=104) as pool:
+        futs = [pool.submit(waiter, i) for i in range(100)]
+        # normal traffic while the waiters are parked
+        time.sleep(0.3)
+        assert client.get(f"{PREFIX}/metrics").status_code == 200
+        assert not any(f.done() and f.result() for f in futs[:5])
+        threading.Timer(
+            0.2, lambda: client.rt.metadata.update_finished_flag(
+                "slowjob", True)).start()
+        results = [f.result(timeout=60) for f in futs]
+    assert all(results), results.count(False)

@@ -217,3 +217,34 @@ def test_bn_running_stats_survive_batch_change():
     m.set_training(False)
     xe = torch.randn(3, 32, 32, 3)
     assert torch.equal(m.predict(xe), m2.predict(xe))
+
+
+def test_conv_tuple_pad_matches_autograd():
+    """pad=(ph, pw) asymmetric padding (padded 1-D sequence convs use
+    (p, 0)) against torch autograd."""
+    import torch
+
+    from learningorchestra_amd.engine.arena import ParamArena
+    from learningorchestra_amd.engine.layers import Conv2dNHWC
+    torch.manual_seed(0)
+    B, H, W, C, OC = 2, 10, 1, 8, 8
+    lay = Conv2dNHWC("c", C, OC, 3, 1, stride=1, pad=(2, 0), relu=False)
+    arena = ParamArena("cpu")
+    lay.build(arena)
+    arena.finalize(0)
+    x = torch.randn(B, H, W, C)
+    y = lay.forward(x.clone())
+    assert y.shape == (B, H + 2 * 2 - 3 + 1, W, OC)
+    # torch reference (NCHW conv2d with asymmetric pad via explicit pad)
+    w = arena.pf("c.w")[:, : 3 * 1 * C].view(OC, 3, 1, C).permute(0, 3, 1, 2)
+    xr = x.permute(0, 3, 1, 2).detach().requires_grad_(True)
+    yr = torch.nn.functional.conv2d(
+        xr, w, bias=arena.pf("c.b"), padding=(2, 0))
+    torch.testing.assert_close(
+        y.permute(0, 3, 1, 2).float(), yr.float(), atol=5e-2, rtol=5e-2)
+    dy = torch.randn_like(yr)
+    yr.backward(dy)
+    dx = lay.backward(dy.permute(0, 2, 3, 1).reshape(-1, OC).clone())
+    torch.testing.assert_close(dx.float(),
+                               xr.grad.permute(0, 2, 3, 1).float(),
+                               atol=5e-2, rtol=5e-2)

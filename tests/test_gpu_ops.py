@@ -452,3 +452,35 @@ def test_conv_family_shape_fuzz():
     mod = importlib.util.module_from_spec(spec)
     spec.loader.exec_module(mod)
     mod.main(iters=60, seed=3)
+
+
+@pytest.mark.gpu
+def test_padded_1d_conv_layer_gpu():
+    """pad=(p, 0) sequence convs route to the fused 1-D kernels (the r1
+    fuzz-found gate held them off entirely); GPU vs CPU layer reference."""
+    from learningorchestra_amd.engine.arena import ParamArena
+    from learningorchestra_amd.engine.layers import Conv2dNHWC
+
+    def build(dev):
+        lay = Conv2dNHWC("c", 32, 64, 4, 1, stride=1, pad=(2, 0),
+                         relu=False, implicit=True)
+        arena = ParamArena(dev)
+        lay.build(arena)
+        arena.finalize(0)
+        return lay
+
+    torch.manual_seed(1)
+    x = torch.randn(4, 96, 1, 32).to(torch.bfloat16)
+    g = build("cuda")
+    assert g._conv1d_ok(1), "padded 1-D shape must be conv1d-eligible now"
+    c = build("cpu")
+    yg = g.forward(x.cuda())
+    yc = c.forward(x.clone())
+    assert yg.shape == yc.shape == (4, 96 + 4 - 4 + 1, 1, 64)
+    torch.testing.assert_close(yg.float().cpu(), yc.float(), atol=5e-2,
+                               rtol=5e-2)
+    dy = torch.randn(yg.shape[0] * yg.shape[1], 64).to(torch.bfloat16)
+    dxg = g.backward(dy.cuda().clone())
+    dxc = c.backward(dy.clone())
+    torch.testing.assert_close(dxg.float().cpu(), dxc.float(), atol=8e-2,
+                               rtol=8e-2)
