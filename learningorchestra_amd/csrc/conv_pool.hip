@@ -1365,6 +1365,43 @@ __global__ void maxpool_global_kernel(const bf16* __restrict__ in,
   }
 }
 
+// backward of the global max-over-time pool: dx is 1/HW-sparse, so the
+// generic kernel's full relu_y stream + per-element window walk is ~3x the
+// write floor (measured 160 us vs ~45 on the TextCNN pool). Specialized:
+// one pass over dx; dy/idx rows are tiny [B, C] and L1-resident per image;
+// relu_y is read ONLY at argmax hits (~1/HW of elements).
+__global__ void maxpool_global_bwd_kernel(const bf16* __restrict__ dy,
+                                          const unsigned char* __restrict__ idx,
+                                          const bf16* __restrict__ relu_y,
+                                          bf16* __restrict__ dx,
+                                          int B, int HW, int C8) {
+  const long total = (long)B * HW * C8;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    unsigned int r32 = (unsigned int)i;
+    const int cu = r32 % C8; r32 /= C8;
+    const int h = r32 % HW;
+    const int b = r32 / HW;
+    const int C = C8 * 8;
+    const unsigned long long iv =
+        *(const unsigned long long*)(idx + (long)b * C + cu * 8);
+    const bf16x8 g = *(const bf16x8*)(dy + (long)b * C + cu * 8);
+    bf16x8 o = {};
+    const unsigned char hh = (unsigned char)h;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      if (((iv >> (j * 8)) & 0xffu) == hh) {
+        float v = tofloat(g[j]);
+        if (relu_y &&
+            !(tofloat(relu_y[((long)b * HW + h) * C + cu * 8 + j]) > 0.f))
+          v = 0.f;
+        o[j] = tobf16(v);
+      }
+    }
+    *(bf16x8*)(dx + ((long)b * HW + h) * C + cu * 8) = o;
+  }
+}
+
 void launch_maxpool_fwd(const void* in, void* out, void* idx, int B, int H, int W,
                         int C, int KH, int KW, int SH, int SW, int PH, int PW,
                         int OH, int OW, hipStream_t s) {
@@ -1397,6 +1434,13 @@ void launch_maxpool_bwd(const void* dy, const void* idx, void* dx,
   const long total = (long)B * H * W * (vec ? C / 8 : C);
   const int block = 256;
   const int grid = (int)min((total + block - 1) / block, (long)2048);
+  if (OH == 1 && OW == 1 && PH == 0 && PW == 0 && KH == H && KW == W &&
+      vec && (long)H * W <= 255) {
+    hipLaunchKernelGGL(maxpool_global_bwd_kernel, dim3(grid), dim3(block),
+                       0, s, (const bf16*)dy, (const unsigned char*)idx,
+                       (const bf16*)relu_y, (bf16*)dx, B, H * W, C / 8);
+    return;
+  }
   const bool k22 = (KH == 2 && KW == 2 && SH == 2 && SW == 2 &&
                     PH == 0 && PW == 0);
   if (vec && k22)

@@ -692,7 +692,13 @@ bool gemm_dispatch(const GemmArgs& g, hipStream_t s) {
       if (g.M >= 64) launch_cfg<64, 64, 64, 2, 2, false, true, 0, true, true>(g, s);
       else launch_cfg<32, 64, 64, 1, 4, false, true, 0, true, true>(g, s);
     } else if (g.ta && !g.tb) {
-      if (g.M >= 64) launch_cfg<64, 64, 64, 2, 2, true, false, 0, true, true>(g, s);
+      // wide tiles for big deep-K dW shapes: A(dy) is re-read once per
+      // N(kpad)-tile, B(col) once per M(outC)-tile (same traffic argument
+      // as dispatch_conv_dw; ResNet's dW GEMMs are 18% of its step)
+      if (g.M >= 128 && g.N >= 128 && g.K >= (1 << 17))
+        launch_cfg<128, 128, 64, 2, 2, true, false, 0, true, true>(g, s);
+      else if (g.M >= 64)
+        launch_cfg<64, 64, 64, 2, 2, true, false, 0, true, true>(g, s);
       else launch_cfg<32, 64, 64, 1, 4, true, false, 0, true, true>(g, s);
     } else {
       return false;

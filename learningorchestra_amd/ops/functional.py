@@ -526,5 +526,10 @@ def conv2d_dw_implicit(dy2: torch.Tensor, x: torch.Tensor, gw: torch.Tensor,
     gw.zero_()
     ok = lo.gemm_conv_dw(dy2, x, gw, max(splits, 2), kh, kw, sh, sw, ph, pw)
     if not ok:
-        raise RuntimeError("gemm_conv_dw: no config for this shape")
+        # shapes the gather path can't take (e.g. B*OH*OW % 8 != 0):
+        # materialize col for this call — allocates, so only odd shapes
+        # (never the steady-state bench models) pay it
+        kpad = gw.shape[1]
+        col = im2col(x, kh, kw, sh, sw, ph, pw, kpad)
+        gemm(dy2, col, ta=True, out=gw, splits=max(splits, 2))
     return gw
