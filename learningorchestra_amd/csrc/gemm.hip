@@ -595,9 +595,21 @@ bool launch_gemm256(const void* A, long lda, const void* B, long ldb, void* C,
 // uses (fwd = N,T; dX = N,N; dW = T,N split-K) plus (T,T) for completeness.
 template <bool TA, bool TB>
 static bool dispatch_tiles(const GemmArgs& g, hipStream_t s) {
-  if (!TA && TB && !g.out_f32 && g.M % 256 == 0 && g.N % 256 == 0 &&
+  static const bool use256 = [] {
+    const char* e = getenv("LO_GEMM256");
+    return e && e[0] == '1';
+  }();
+  // The 8-phase kernel is RETIRED from dispatch (r2): its counted-vmcnt
+  // pipeline silently corrupts tiles in a residency/state-dependent way —
+  // reliably at >= ~192-block grids, and once observed even at 160 blocks
+  // on a warmed box (tools/gemm256_raceprobe.py). The provably-safe DRAIN
+  // variant measured BELOW this 128^2 tile kernel on every big-grid shape
+  // (ResNet 6294 vs 6389 samples/s, 8192^3 706 vs 793 TF), so the tile
+  // path is both the correct and the faster choice. LO_GEMM256=1 re-enables
+  // the fast variant for experiments only.
+  if (use256 && !TA && TB && !g.out_f32 && g.M % 256 == 0 && g.N % 256 == 0 &&
       g.K % 64 == 0 && g.K >= 256 &&
-      (g.M / 256) * (g.N / 256) >= 192) {  // needs a chip-filling grid
+      (g.M / 256) * (g.N / 256) >= 128) {
     // deep-pipelined 256^2 8-phase path (gemm_8phase.hip)
     if (launch_gemm256(g.A, g.lda, g.B, g.ldb, g.C, g.ldc, g.bias,
                        g.M, g.N, g.K, g.epi, g.addend, g.stats_sum,

@@ -49,3 +49,36 @@ def verify_fix():
     for M in (49152, 51200, 65536, 102400):
         probe(M, K=512, iters=30)
         probe(M, K=1024, iters=10)
+
+
+def stale_lds_probe():
+    """Distinguish a stale-LDS dependence from a timing race. Prediction if
+    some LDS region is consumed before being written this launch: after a
+    DIFFERENT-shape gemm256 poisons the LDS, the first iterations fail and
+    later ones self-heal (stale bytes == this shape's bytes again); with a
+    poison run before EVERY iteration, every iteration fails. A timing race
+    would give randomly scattered bad iterations instead."""
+    import torch
+
+    def run_case(M, poison_each, iters=16, K=512, N=256):
+        torch.manual_seed(4)
+        A = torch.randn(M, K, device="cuda").to(torch.bfloat16)
+        B = torch.randn(N, K, device="cuda").to(torch.bfloat16)
+        ref = (A.float() @ B.float().t())
+        Ap = torch.randn(33024, K, device="cuda").to(torch.bfloat16)
+        refp = None
+        F.gemm(Ap, B, tb=True)                  # poison once (129 blocks)
+        bad = []
+        for i in range(iters):
+            if poison_each:
+                F.gemm(Ap, B, tb=True)
+            out = F.gemm(A, B, tb=True)
+            err = (out.float() - ref).abs().max().item()
+            if err > 2.0:
+                bad.append(i)
+        print(f"M={M} poison_each={poison_each}: bad iters {bad}", flush=True)
+
+    for rep in range(2):
+        run_case(40960, False)
+    for rep in range(2):
+        run_case(40960, True)
