@@ -702,6 +702,156 @@ __global__ __launch_bounds__(256) void conv1d_fwd_kernel(
   }
 }
 
+// conv1d_fwd v2 — full-outC w tiles, single chunk loop.
+// The v1 kernel loops outC in 64-wide slices, re-reading the x-window
+// fragments and re-staging/re-barriering per slice (2x for outC=128).
+// v2 stages one [outC x 32k] tile per chunk (both slices share it), reads
+// each af once, and uses the dx2-style named-register fetch/write split —
+// half the barriers, half the af reads, half the staging passes.
+__global__ __launch_bounds__(256) void conv1d_fwd2_kernel(
+    const bf16* __restrict__ x,               // [B, H, 1, C]
+    const bf16* __restrict__ w, long ldw,     // [outC, kpad]
+    const float* __restrict__ bias,
+    bf16* __restrict__ y, long ldy,           // [B*OH, outC]
+    int B, int H, int C, int KH, int PH, int OH, int outC, int G, int relu) {
+  extern __shared__ char sm[];
+  const int img0 = blockIdx.x * G;
+  const int m0 = blockIdx.y * 64;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const int XROW = C * 2 + 16;
+  const int XR = 64 + KH - 1;
+  const int x0 = m0 - PH;
+  const int XB = (XR * XROW + 127) & ~127;
+  const int TB = outC * 64;                   // [outC rows x 32 k] bytes
+  char* smX = sm;
+  char* smW = sm + XB * G;
+
+  for (int g = 0; g < G; ++g) {
+    const int img = img0 + g;
+    for (int i = tid * 8; i < XR * C; i += 256 * 8) {
+      const int r = i / C, c0 = i - r * C;
+      const int h = x0 + r;
+      bf16x8 v = {};
+      if (img < B && h >= 0 && h < H)
+        v = *(const bf16x8*)(x + ((long)img * H + h) * C + c0);
+      *(bf16x8*)(smX + g * XB + r * XROW + c0 * 2) = v;
+    }
+  }
+
+  const int NC = (KH * C) / 32;
+  const int NOS = outC / 64;                  // 1 or 2
+  auto wswz = [](int row, int kel) {
+    return (kel ^ ((row & 3) << 3) ^ (((row >> 2) & 3) << 3));
+  };
+  // per-thread tile portion: outC*32/(256*8) <= 2 bf16x8
+  const int NP = (outC * 32 + 256 * 8 - 1) / (256 * 8);
+  auto fetch_tile = [&](int chunk, bf16x8 (&regs)[2]) {
+    #pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      if (p >= NP) break;
+      const int i = tid * 8 + p * 256 * 8;
+      if (i < outC * 32) {
+        const int r = i / 32, k0 = i - r * 32;
+        regs[p] = *(const bf16x8*)(w + (long)r * ldw + chunk * 32 + k0);
+      }
+    }
+  };
+  auto write_tile = [&](int buf, const bf16x8 (&regs)[2]) {
+    #pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      if (p >= NP) break;
+      const int i = tid * 8 + p * 256 * 8;
+      if (i < outC * 32) {
+        const int r = i / 32, k0 = i - r * 32;
+        *(bf16x8*)(smW + buf * TB + r * 64 + wswz(r, k0) * 2) = regs[p];
+      }
+    }
+  };
+
+  bf16x8 tA[2], tB[2];
+  {
+    bf16x8 t0[2];
+    fetch_tile(0, t0);
+    write_tile(0, t0);
+  }
+  if (NC > 1) fetch_tile(1, tB);
+  const int arow = m0 + wave * 16 + (lane & 15);
+  const int kcol = (lane >> 4) * 8;
+  f32x4 acc[2][2][4] = {};                    // [g][os][ni]
+  for (int chunk = 0; chunk < NC; ++chunk) {
+    const int buf = chunk & 1;
+    __syncthreads();                          // tile(buf) + window visible
+    if (chunk + 2 < NC) {
+      if (buf) fetch_tile(chunk + 2, tB);
+      else     fetch_tile(chunk + 2, tA);
+    }
+    bf16x8 af0 = {}, af1 = {};
+    {
+      const int k = chunk * 32 + kcol;
+      const int kh = k / C, c = k - kh * C;
+      const int r = (arow - m0) + kh;
+      if (arow < OH) {
+        af0 = *(const bf16x8*)(smX + r * XROW + c * 2);
+        if (G > 1)
+          af1 = *(const bf16x8*)(smX + XB + r * XROW + c * 2);
+      }
+    }
+    #pragma unroll
+    for (int os = 0; os < 2; ++os) {
+      if (os >= NOS) break;
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int row = os * 64 + ni * 16 + (lane & 15);
+        const bf16x8 bfr = *(const bf16x8*)(
+            smW + buf * TB + row * 64 + wswz(row, kcol) * 2);
+        acc[0][os][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af0, bfr, acc[0][os][ni], 0, 0, 0);
+        if (G > 1)
+          acc[1][os][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af1, bfr, acc[1][os][ni], 0, 0, 0);
+      }
+    }
+    if (chunk + 1 < NC) {
+      if (buf) write_tile(buf ^ 1, tA);
+      else     write_tile(buf ^ 1, tB);
+    }
+  }
+  // epilogue: per (g, os): acc -> LDS staging (reuse smW) -> coalesced y
+  for (int g = 0; g < G; ++g) {
+    if (img0 + g >= B) break;
+    #pragma unroll
+    for (int os = 0; os < 2; ++os) {          // unrolled: runtime-indexed
+      if (os >= NOS) break;                   // acc would spill to scratch
+      __syncthreads();
+      char* se = smW + wave * 2048;
+      #pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int c = os * 64 + ni * 16 + (lane & 15);
+        const float b = bias ? bias[c < outC ? c : 0] : 0.f;
+        const f32x4 a = g ? acc[1][os][ni] : acc[0][os][ni];
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float v = a[r] + b;
+          if (relu) v = v > 0.f ? v : 0.f;
+          ((bf16*)se)[((lane >> 4) * 4 + r) * 64 + (ni * 16 + (lane & 15))] =
+              tobf16(v);
+        }
+      }
+      __syncthreads();
+      #pragma unroll
+      for (int p = 0; p < 2; ++p) {
+        const int row = p * 8 + (lane >> 3);
+        const int m = m0 + wave * 16 + row;
+        if (m < OH)
+          *(bf16x8*)(y + ((long)(img0 + g) * OH + m) * ldy + os * 64 +
+                     (lane & 7) * 8) =
+              *(const bf16x8*)(se + row * 128 + (lane & 7) * 16);
+      }
+    }
+  }
+}
+
 bool launch_conv1d_fwd(const void* x, const void* w, long ldw,
                        const void* bias, void* y, long ldy, int B, int H,
                        int C, int KH, int PH, int OH, int outC, int relu,
@@ -710,10 +860,23 @@ bool launch_conv1d_fwd(const void* x, const void* w, long ldw,
   const int XB = (XR * (C * 2 + 16) + 127) & ~127;
   if (C % 32 != 0 || outC % 64 != 0 || outC > 128 || (KH * C) % 32 != 0)
     return false;
+  static const bool use_v1 = [] {
+    const char* e = getenv("LO_CONV1D_FWD_V1");
+    return e && e[0] == '1';
+  }();
   int G = 2;                                  // halves the w re-staging
   if (XB * G + 8192 > 56 * 1024) G = 1;
   if (XB * G + 8192 > 56 * 1024) return false;
   const int T = (OH + 63) / 64;
+  const int lds = XB * G + 2 * outC * 64;     // v2: 2 full-outC tiles
+  if (!use_v1 && lds <= 56 * 1024) {
+    hipLaunchKernelGGL(conv1d_fwd2_kernel, dim3((B + G - 1) / G, T),
+                       dim3(256), lds, s,
+                       (const bf16*)x, (const bf16*)w, ldw,
+                       (const float*)bias, (bf16*)y, ldy, B, H, C, KH, PH,
+                       OH, outC, G, relu);
+    return true;
+  }
   hipLaunchKernelGGL(conv1d_fwd_kernel, dim3((B + G - 1) / G, T), dim3(256),
                      XB * G + 8192, s,
                      (const bf16*)x, (const bf16*)w, ldw, (const float*)bias,
