@@ -47,9 +47,14 @@ class Config:
 
     # --- executor ----------------------------------------------------------
     max_jobs: int = field(default_factory=lambda: int(_env("LO_MAX_JOBS", "8")))
-    # exec() of user-supplied code (builder modelingCode, function/python) is
-    # part of the reference API (builder.py:99, code_execution.py:185). It is
-    # gated behind this flag; the trusted-cluster assumption is documented.
+    # exec() of user-supplied code (builder modelingCode, function/python,
+    # '#' parameters) is part of the reference API (builder.py:99,
+    # code_execution.py:185) and is gated behind this flag. IMPORTANT SCOPE
+    # NOTE: LO_ALLOW_USER_CODE=0 does NOT make the API safe for
+    # untrusted callers — the model/explore/transform/binary verbs remain a
+    # reflective importlib surface by reference design (any module path +
+    # callable with caller kwargs). The deployment assumption is a trusted
+    # cluster, exactly as the reference's (which shipped 4 ungated exec()s).
     allow_user_code: bool = field(default_factory=lambda: _env("LO_ALLOW_USER_CODE", "1") == "1")
 
     # --- compute -----------------------------------------------------------

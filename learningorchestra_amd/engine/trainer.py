@@ -84,7 +84,20 @@ class Trainer:
     def step_async(self, x: torch.Tensor, y: torch.Tensor) -> None:
         if self.use_graph:
             if self._graph is None:
-                self._capture(x, y)
+                try:
+                    self._capture(x, y)
+                except RuntimeError as exc:
+                    # capture can fail in environments we cannot pre-test
+                    # (e.g. a driver/runtime combination on the 8-GPU node);
+                    # a slower eager run beats a crashed scaling bench
+                    import sys
+                    print(f"[trainer] graph capture failed ({exc}); "
+                          "falling back to eager", file=sys.stderr)
+                    self.use_graph = False
+                    self._graph = None
+                    torch.cuda.synchronize()
+                    self._step_body(x, y)
+                return
             else:
                 self._static_x.copy_(x, non_blocking=True)
                 self._static_y.copy_(y, non_blocking=True)
