@@ -708,12 +708,17 @@ __global__ __launch_bounds__(256) void conv1d_fwd_kernel(
 // v2 stages one [outC x 32k] tile per chunk (both slices share it), reads
 // each af once, and uses the dx2-style named-register fetch/write split —
 // half the barriers, half the af reads, half the staging passes.
+template <int TC = 0, int TOC = 0, int TKH = 0>  // 0 = runtime
 __global__ __launch_bounds__(256) void conv1d_fwd2_kernel(
     const bf16* __restrict__ x,               // [B, H, 1, C]
     const bf16* __restrict__ w, long ldw,     // [outC, kpad]
     const float* __restrict__ bias,
     bf16* __restrict__ y, long ldy,           // [B*OH, outC]
-    int B, int H, int C, int KH, int PH, int OH, int outC, int G, int relu) {
+    int B, int H, int C_, int KH_, int PH, int OH, int outC_, int G,
+    int relu) {
+  const int C = TC ? TC : C_;
+  const int KH = TKH ? TKH : KH_;
+  const int outC = TOC ? TOC : outC_;
   extern __shared__ char sm[];
   const int img0 = blockIdx.x * G;
   const int m0 = blockIdx.y * 64;
@@ -870,11 +875,17 @@ bool launch_conv1d_fwd(const void* x, const void* w, long ldw,
   const int T = (OH + 63) / 64;
   const int lds = XB * G + 2 * outC * 64;     // v2: 2 full-outC tiles
   if (!use_v1 && lds <= 56 * 1024) {
-    hipLaunchKernelGGL(conv1d_fwd2_kernel, dim3((B + G - 1) / G, T),
-                       dim3(256), lds, s,
-                       (const bf16*)x, (const bf16*)w, ldw,
-                       (const float*)bias, (bf16*)y, ldy, B, H, C, KH, PH,
-                       OH, outC, G, relu);
+    #define LO_C1FW(TC_, TOC_, TKH_)                                          \
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(conv1d_fwd2_kernel<TC_, TOC_, TKH_>),\
+                         dim3((B + G - 1) / G, T), dim3(256), lds, s,         \
+                         (const bf16*)x, (const bf16*)w, ldw,                 \
+                         (const float*)bias, (bf16*)y, ldy, B, H, C, KH, PH, \
+                         OH, outC, G, relu)
+    if (C == 128 && outC == 128 && KH == 3) LO_C1FW(128, 128, 3);
+    else if (C == 128 && outC == 128 && KH == 4) LO_C1FW(128, 128, 4);
+    else if (C == 128 && outC == 128 && KH == 5) LO_C1FW(128, 128, 5);
+    else LO_C1FW(0, 0, 0);
+    #undef LO_C1FW
     return true;
   }
   hipLaunchKernelGGL(conv1d_fwd_kernel, dim3((B + G - 1) / G, T), dim3(256),
@@ -1013,12 +1024,18 @@ __global__ __launch_bounds__(256) void conv1d_dx_kernel(
 // block processes G=2 images per wt staging pass, halving the wt-tile
 // traffic per image. Stores are direct bf16x4 (8 B from consecutive lanes
 // coalesce into the row's contiguous 64 B segment).
+template <int TC = 0, int TOC = 0, int TKH = 0>  // 0 = runtime (compile-
+// time C/outC/KH turn the index divisions and loop bounds into constants —
+// PMC measured 14 VALU per MFMA on the generic version, address math)
 __global__ __launch_bounds__(256) void conv1d_dx2_kernel(
     const bf16* __restrict__ dy2, long ldy,   // [B*OH, outC]
     const bf16* __restrict__ wt, long ldw,    // [kpad, outC]
     bf16* __restrict__ dx,                    // [B, H, 1, C]
-    int B, int H, int C, int KH, int PH, int OH, int outC, int G,
+    int B, int H, int C_, int KH_, int PH, int OH, int outC_, int G,
     int accumulate) {
+  const int C = TC ? TC : C_;
+  const int KH = TKH ? TKH : KH_;
+  const int outC = TOC ? TOC : outC_;
   extern __shared__ char sm[];
   const int img0 = blockIdx.x * G;
   const int h0 = blockIdx.y * 64;
@@ -1182,10 +1199,17 @@ bool launch_conv1d_dx(const void* dy2, long ldy, const void* wt, long ldw,
     if (2 * TB + G * WR * WROW > 56 * 1024) G = 1;
     if (2 * TB + G * WR * WROW <= 56 * 1024) {
       const int T = (H + 63) / 64;
-      hipLaunchKernelGGL(conv1d_dx2_kernel, dim3((B + G - 1) / G, T),
-                         dim3(256), 2 * TB + G * WR * WROW, s,
-                         (const bf16*)dy2, ldy, (const bf16*)wt, ldw,
-                         (bf16*)dx, B, H, C, KH, PH, OH, outC, G, accumulate);
+      #define LO_C1DX(TC_, TOC_, TKH_)                                        \
+        hipLaunchKernelGGL(                                                   \
+            HIP_KERNEL_NAME(conv1d_dx2_kernel<TC_, TOC_, TKH_>),              \
+            dim3((B + G - 1) / G, T), dim3(256), 2 * TB + G * WR * WROW, s,   \
+            (const bf16*)dy2, ldy, (const bf16*)wt, ldw, (bf16*)dx,           \
+            B, H, C, KH, PH, OH, outC, G, accumulate)
+      if (C == 128 && outC == 128 && KH == 3) LO_C1DX(128, 128, 3);
+      else if (C == 128 && outC == 128 && KH == 4) LO_C1DX(128, 128, 4);
+      else if (C == 128 && outC == 128 && KH == 5) LO_C1DX(128, 128, 5);
+      else LO_C1DX(0, 0, 0);
+      #undef LO_C1DX
       return true;
     }
   }
