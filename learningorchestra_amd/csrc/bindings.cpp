@@ -77,7 +77,7 @@ void launch_col2im(const void* dcol, void* dx, int B, int H, int W, int C,
                    int OW, int Kpad, hipStream_t s);
 void launch_maxpool_fwd(const void* in, void* out, void* idx, int B, int H, int W,
                         int C, int KH, int KW, int SH, int SW, int PH, int PW,
-                        int OH, int OW, hipStream_t s);
+                        int OH, int OW, int relu_sentinel, hipStream_t s);
 void launch_maxpool_bwd(const void* dy, const void* idx, void* dx,
                         const void* relu_y, int B, int H,
                         int W, int C, int KH, int KW, int SH, int SW, int PH,
@@ -377,7 +377,8 @@ bool conv_dx(at::Tensor dy2, at::Tensor wt, at::Tensor dx, int64_t KH,
 }
 
 std::vector<at::Tensor> maxpool_fwd(at::Tensor in, int64_t KH, int64_t KW,
-                                    int64_t SH, int64_t SW, int64_t PH, int64_t PW) {
+                                    int64_t SH, int64_t SW, int64_t PH, int64_t PW,
+                                    bool relu_sentinel = false) {
   check_bf16(in, "in");
   const int B = (int)in.size(0), H = (int)in.size(1), W = (int)in.size(2),
             C = (int)in.size(3);
@@ -387,7 +388,7 @@ std::vector<at::Tensor> maxpool_fwd(at::Tensor in, int64_t KH, int64_t KW,
   auto idx = at::empty({B, OH, OW, C}, in.options().dtype(at::kByte));
   lo::launch_maxpool_fwd(in.data_ptr(), out.data_ptr(), idx.data_ptr(), B, H, W, C,
                          (int)KH, (int)KW, (int)SH, (int)SW, (int)PH, (int)PW,
-                         OH, OW, stream());
+                         OH, OW, relu_sentinel ? 1 : 0, stream());
   return {out, idx};
 }
 
@@ -653,7 +654,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_dw_c1", &conv_dw_c1, "C=1 conv dW (x LDS-resident)");
   m.def("conv_dx", &conv_dx, "fused conv dX (LDS-accumulated scatter)");
   m.def("col2im", &col2im);
-  m.def("maxpool_fwd", &maxpool_fwd);
+  m.def("maxpool_fwd", &maxpool_fwd, py::arg("in"), py::arg("KH"),
+        py::arg("KW"), py::arg("SH"), py::arg("SW"), py::arg("PH"),
+        py::arg("PW"), py::arg("relu_sentinel") = false);
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("relu_bwd", &relu_bwd);
   m.def("sgd_step", &sgd_step);

@@ -1367,7 +1367,8 @@ template <bool VEC8>
 __global__ void maxpool_fwd_kernel(const bf16* __restrict__ in, bf16* __restrict__ out,
                                    unsigned char* __restrict__ idx,
                                    int B, int H, int W, int C, int KH, int KW,
-                                   int SH, int SW, int PH, int PW, int OH, int OW) {
+                                   int SH, int SW, int PH, int PW, int OH, int OW,
+                                   int relu_sentinel) {
   const int CV = VEC8 ? C / 8 : C;
   const long total = (long)B * OH * OW * CV;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
@@ -1401,6 +1402,15 @@ __global__ void maxpool_fwd_kernel(const bf16* __restrict__ in, bf16* __restrict
           if (f > best[0]) { best[0] = f; bidx[0] = (unsigned char)(kh * KW + kw); }
         }
       }
+    }
+    // relu_sentinel: the upstream conv's ReLU bwd folds into the pool's
+    // idx — a non-positive max (ReLU-clamped everywhere) gets sentinel 255
+    // (never a valid kh*KW+kw), so the BACKWARD needs no relu_y stream at
+    // all (that stream was ~40% of the pool-bwd traffic on MNIST)
+    if (relu_sentinel) {
+      #pragma unroll
+      for (int j = 0; j < NE; ++j)
+        if (!(best[j] > 0.f)) bidx[j] = 255;
     }
     const long obase = (((long)b * OH + oh) * OW + ow) * C + cu * NE;
     if (VEC8) {
@@ -1494,7 +1504,7 @@ __global__ void maxpool_bwd_kernel(const bf16* __restrict__ dy,
 __global__ void maxpool_global_kernel(const bf16* __restrict__ in,
                                       bf16* __restrict__ out,
                                       unsigned char* __restrict__ idx,
-                                      int HW, int C) {
+                                      int HW, int C, int relu_sentinel) {
   extern __shared__ __attribute__((aligned(16))) float lmx[];  // [C][2] max+idx
   const int b = blockIdx.x;
   const int nch = C / 8;
@@ -1547,8 +1557,11 @@ __global__ void maxpool_global_kernel(const bf16* __restrict__ in,
   for (int i = threadIdx.x; i < C; i += blockDim.x) {
     const unsigned u = lk[i];
     const unsigned raw = (u & 0x80000000u) ? (u & 0x7fffffffu) : ~u;
-    out[(long)b * C + i] = tobf16(__uint_as_float(raw));
-    idx[(long)b * C + i] = (unsigned char)min(lk[C + i], 255u);
+    const float mx = __uint_as_float(raw);
+    out[(long)b * C + i] = tobf16(mx);
+    unsigned char bi = (unsigned char)min(lk[C + i], 254u);
+    if (relu_sentinel && !(mx > 0.f)) bi = 255;  // fused ReLU bwd sentinel
+    idx[(long)b * C + i] = bi;
   }
 }
 
@@ -1591,12 +1604,12 @@ __global__ void maxpool_global_bwd_kernel(const bf16* __restrict__ dy,
 
 void launch_maxpool_fwd(const void* in, void* out, void* idx, int B, int H, int W,
                         int C, int KH, int KW, int SH, int SW, int PH, int PW,
-                        int OH, int OW, hipStream_t s) {
+                        int OH, int OW, int relu_sentinel, hipStream_t s) {
   if (OH == 1 && OW == 1 && PH == 0 && PW == 0 && KH == H && KW == W &&
       C % 8 == 0 && C <= 1024 && (long)H * W <= 255) {
     hipLaunchKernelGGL(maxpool_global_kernel, dim3(B), dim3(256),
                        C * 2 * sizeof(float), s, (const bf16*)in, (bf16*)out,
-                       (unsigned char*)idx, H * W, C);
+                       (unsigned char*)idx, H * W, C, relu_sentinel);
     return;
   }
   const bool vec = (C % 8 == 0);
@@ -1606,11 +1619,13 @@ void launch_maxpool_fwd(const void* in, void* out, void* idx, int B, int H, int 
   if (vec)
     hipLaunchKernelGGL(HIP_KERNEL_NAME(maxpool_fwd_kernel<true>), dim3(grid), dim3(block), 0, s,
                        (const bf16*)in, (bf16*)out, (unsigned char*)idx,
-                       B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
+                       B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW,
+                       relu_sentinel);
   else
     hipLaunchKernelGGL(HIP_KERNEL_NAME(maxpool_fwd_kernel<false>), dim3(grid), dim3(block), 0, s,
                        (const bf16*)in, (bf16*)out, (unsigned char*)idx,
-                       B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW);
+                       B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW,
+                       relu_sentinel);
 }
 
 void launch_maxpool_bwd(const void* dy, const void* idx, void* dx,

@@ -289,11 +289,11 @@ class MaxPool2dNHWC(Layer):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         self._bufs["shape"] = x.shape
+        # fuse_relu: the upstream conv's ReLU bwd folds into the argmax idx
+        # (sentinel 255 where max <= 0), so backward needs no relu_y stream
         out, idx = F.maxpool2d(x, self.kh, self.kw, self.sh, self.sw,
-                               self.ph, self.pw)
+                               self.ph, self.pw, relu_mask=self.fuse_relu)
         self._bufs["idx"] = idx
-        if self.fuse_relu:
-            self._x = x
         if "dx" not in self._bufs or self._bufs["dx"].shape != x.shape:
             self._bufs["dx"] = torch.empty_like(x)
         return out
@@ -302,8 +302,7 @@ class MaxPool2dNHWC(Layer):
         B, H, W, C = self._bufs["shape"]
         return F.maxpool2d_bwd(dy, self._bufs["idx"], H, W, self.kh, self.kw,
                                self.sh, self.sw, self.ph, self.pw,
-                               out=self._bufs["dx"],
-                               relu_y=self._x if self.fuse_relu else None)
+                               out=self._bufs["dx"], relu_y=None)
 
 
 class Flatten(Layer):

@@ -176,12 +176,17 @@ def conv2d_dx_fused(dy2: torch.Tensor, wt: torch.Tensor, B: int, H: int,
 
 # ------------------------------------------------------------------ maxpool
 def maxpool2d(x: torch.Tensor, kh: int, kw: int, sh: int, sw: int,
-              ph: int = 0, pw: int = 0) -> Tuple[torch.Tensor, torch.Tensor]:
+              ph: int = 0, pw: int = 0,
+              relu_mask: bool = False) -> Tuple[torch.Tensor, torch.Tensor]:
     """NHWC maxpool (zero-pad treated as -inf); returns (out, idx u8 of
-    kh*KW+kw argmax)."""
+    kh*KW+kw argmax). ``relu_mask``: positions whose max is <= 0 get
+    sentinel idx 255, folding the upstream conv's ReLU backward into the
+    index — the pool BACKWARD then needs no relu_y stream (that stream was
+    ~40% of the pool-bwd traffic on MNIST)."""
     if _is_gpu(x):
         lo = require_ext()
-        out, idx = lo.maxpool_fwd(x, kh, kw, sh, sw, ph, pw)
+        out, idx = lo.maxpool_fwd(x, kh, kw, sh, sw, ph, pw,
+                                  relu_sentinel=relu_mask)
         return out, idx
     B, H, W, C = x.shape
     OH = (H + 2 * ph - kh) // sh + 1
@@ -198,6 +203,8 @@ def maxpool2d(x: torch.Tensor, kh: int, kw: int, sh: int, sw: int,
     ow = torch.arange(OW, device=x.device).view(1, 1, 1, OW)
     rel = (hh - (oh * sh - ph)) * kw + (ww - (ow * sw - pw))
     idx = rel.permute(0, 2, 3, 1).to(torch.uint8).contiguous()
+    if relu_mask:
+        idx[out <= 0] = 255
     return out, idx
 
 
@@ -217,6 +224,8 @@ def maxpool2d_bwd(dy: torch.Tensor, idx: torch.Tensor, H: int, W: int,
         return out
     dx = torch.zeros((B, H, W, C), dtype=torch.float32)
     rel = idx.long()
+    valid = rel != 255                 # 255 = fused-ReLU sentinel (no grad)
+    rel = torch.where(valid, rel, torch.zeros_like(rel))
     khh, kww = rel // kw, rel % kw
     oh = torch.arange(OH).view(1, OH, 1, 1)
     ow = torch.arange(OW).view(1, 1, OW, 1)
@@ -224,8 +233,9 @@ def maxpool2d_bwd(dy: torch.Tensor, idx: torch.Tensor, H: int, W: int,
     wsrc = ow * sw + kww - pw
     b = torch.arange(B).view(B, 1, 1, 1).expand_as(rel)
     c = torch.arange(C).view(1, 1, 1, C).expand_as(rel)
+    src = torch.where(valid, dy.float(), torch.zeros_like(dy, dtype=torch.float32))
     dx.index_put_((b.reshape(-1), hsrc.reshape(-1), wsrc.reshape(-1),
-                   c.reshape(-1)), dy.float().reshape(-1), accumulate=True)
+                   c.reshape(-1)), src.reshape(-1), accumulate=True)
     if relu_y is not None:
         dx = dx * (relu_y.float() > 0)
     out.copy_(dx.to(out.dtype))
