@@ -230,12 +230,19 @@ void launch_col2im(const void* dcol, void* dx, int B, int H, int W, int C,
 // The accumulator rows are padded to C+1 floats so the 4 row-groups of a
 // fragment land in different banks (consecutive m -> +1 row -> +1 bank).
 // Eligible when the whole image's dx fits LDS (H*W*(C+1)*4 <= 56 KB).
+template <int TH = 0, int TW = 0, int TC = 0, int TK = 0, int TOC = 0>
 __global__ __launch_bounds__(256) void conv_dx_kernel(
     const bf16* __restrict__ dy2, long ldy,   // [B*OH*OW, outC]
     const bf16* __restrict__ wt, long ldw,    // [kpad, outC] row-major
     bf16* __restrict__ dx,                    // [B, H, W, C]
-    int H, int W, int C, int KH, int KW, int SH, int SW, int PH, int PW,
-    int OH, int OW, int outC) {
+    int H_, int W_, int C_, int KH_, int KW_, int SH_, int SW_, int PH_,
+    int PW_, int OH_, int OW_, int outC_) {
+  const int H = TH ? TH : H_, W = TW ? TW : W_, C = TC ? TC : C_;
+  const int KH = TK ? TK : KH_, KW = TK ? TK : KW_;
+  const int SH = TH ? 1 : SH_, SW = TH ? 1 : SW_;
+  const int PH = TH ? 0 : PH_, PW = TH ? 0 : PW_;
+  const int OH = TH ? (TH - TK + 1) : OH_, OW = TW ? (TW - TK + 1) : OW_;
+  const int outC = TOC ? TOC : outC_;
   // dynamic LDS: fp32 dx accumulator (H*W rows of C+4 floats - the +4
   // keeps 16-B alignment for the vectorized RMW and staggers banks across
   // rows) + 2 x 4 KB wt tile buffers (32 rows x 64 k, XOR-swizzled)
@@ -353,9 +360,18 @@ bool launch_conv_dx(const void* dy2, long ldy, const void* wt, long ldw,
   if ((long)HWCP * 4 > 56 * 1024 || (C % 32 != 0 && C != 16) || HWC % 8 != 0 ||
       outC > 64 || outC % 8 != 0 || C % 8 != 0)
     return false;
-  hipLaunchKernelGGL(conv_dx_kernel, dim3(B), dim3(256), HWCP * 4 + 8192, s,
-                     (const bf16*)dy2, ldy, (const bf16*)wt, ldw, (bf16*)dx,
-                     H, W, C, KH, KW, SH, SW, PH, PW, OH, OW, outC);
+  #define LO_CDX(...)                                                         \
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(conv_dx_kernel<__VA_ARGS__>),          \
+                       dim3(B), dim3(256), HWCP * 4 + 8192, s,                \
+                       (const bf16*)dy2, ldy, (const bf16*)wt, ldw,           \
+                       (bf16*)dx, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW,    \
+                       outC)
+  if (H == 12 && W == 12 && C == 32 && KH == 5 && KW == 5 && SH == 1 &&
+      SW == 1 && PH == 0 && PW == 0 && outC == 64)
+    LO_CDX(12, 12, 32, 5, 64);          // MNIST conv2 dX
+  else
+    LO_CDX(0);
+  #undef LO_CDX
   return true;
 }
 
@@ -372,14 +388,21 @@ bool launch_conv_dx(const void* dy2, long ldy, const void* wt, long ldw,
 // x is staged unpadded (H*W bf16) and the A fragment is gathered with 8
 // scalar LDS reads per lane (the k-run spans (kh,kw) cells); kpad <= 32 so
 // the whole w fits one 32-k chunk.
-template <bool C1>
+template <bool C1, int TH = 0, int TW = 0, int TC = 0, int TK = 0,
+          int TOC = 0>   // TH!=0 -> compile-time shape, stride 1, pad 0
 __global__ __launch_bounds__(256) void conv_fwd_small_kernel(
     const bf16* __restrict__ x,               // [B, H, W, C]
     const bf16* __restrict__ w, long ldw,     // [outC, kpad] row-major
     const float* __restrict__ bias,           // [outC] or null
     bf16* __restrict__ y, long ldy,           // [B*OH*OW, outC]
-    int H, int W, int C, int KH, int KW, int SH, int SW, int PH, int PW,
-    int OH, int OW, int outC, int relu) {
+    int H_, int W_, int C_, int KH_, int KW_, int SH_, int SW_, int PH_,
+    int PW_, int OH_, int OW_, int outC_, int relu) {
+  const int H = TH ? TH : H_, W = TW ? TW : W_, C = TC ? TC : C_;
+  const int KH = TK ? TK : KH_, KW = TK ? TK : KW_;
+  const int SH = TH ? 1 : SH_, SW = TH ? 1 : SW_;
+  const int PH = TH ? 0 : PH_, PW = TH ? 0 : PW_;
+  const int OH = TH ? (TH - TK + 1) : OH_, OW = TW ? (TW - TK + 1) : OW_;
+  const int outC = TOC ? TOC : outC_;
   extern __shared__ char sm[];                // x image + 2 x 4 KB w tiles
   const int img = blockIdx.x;
   const int R = OH * OW;
@@ -562,22 +585,38 @@ bool launch_conv_fwd_small(const void* x, const void* w, long ldw,
                            hipStream_t s) {
   if (C == 1 && KH * KW <= 32 && ldw >= 32 && outC <= 64 && outC % 16 == 0 &&
       (H * W) % 8 == 0 && H * W * 2 + 12288 <= 56 * 1024) {
-    hipLaunchKernelGGL(HIP_KERNEL_NAME(conv_fwd_small_kernel<true>),
-                       dim3(B), dim3(256), H * W * 2 + 12288, s,
-                       (const bf16*)x, (const bf16*)w, ldw, (const float*)bias,
-                       (bf16*)y, ldy, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW,
-                       outC, relu);
+    #define LO_CFS1(...)                                                      \
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(conv_fwd_small_kernel<__VA_ARGS__>), \
+                         dim3(B), dim3(256), H * W * 2 + 12288, s,            \
+                         (const bf16*)x, (const bf16*)w, ldw,                 \
+                         (const float*)bias, (bf16*)y, ldy, H, W, C, KH, KW,  \
+                         SH, SW, PH, PW, OH, OW, outC, relu)
+    // MNIST conv1 shape gets the compile-time variant (PMC: the generic
+    // C=1 scalar-gather path ran 79 VALU per MFMA — shape index math)
+    if (H == 28 && W == 28 && C == 1 && KH == 5 && KW == 5 && SH == 1 &&
+        SW == 1 && PH == 0 && PW == 0 && outC == 32)
+      LO_CFS1(true, 28, 28, 1, 5, 32);
+    else
+      LO_CFS1(true);
+    #undef LO_CFS1
     return true;
   }
   const int lds = H * W * (C * 2 + 16) + 8192;
   if (lds > 56 * 1024 || (C % 32 != 0 && C != 16) || outC > 64 ||
       outC % 16 != 0 || (KH * KW * C) % 32 != 0 || (H * W * C) % 8 != 0)
     return false;
-  hipLaunchKernelGGL(HIP_KERNEL_NAME(conv_fwd_small_kernel<false>),
-                     dim3(B), dim3(256), lds, s,
-                     (const bf16*)x, (const bf16*)w, ldw, (const float*)bias,
-                     (bf16*)y, ldy, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW,
-                     outC, relu);
+  #define LO_CFS(...)                                                         \
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(conv_fwd_small_kernel<__VA_ARGS__>),   \
+                       dim3(B), dim3(256), lds, s,                            \
+                       (const bf16*)x, (const bf16*)w, ldw,                   \
+                       (const float*)bias, (bf16*)y, ldy, H, W, C, KH, KW,    \
+                       SH, SW, PH, PW, OH, OW, outC, relu)
+  if (H == 12 && W == 12 && C == 32 && KH == 5 && KW == 5 && SH == 1 &&
+      SW == 1 && PH == 0 && PW == 0 && outC == 64)
+    LO_CFS(false, 12, 12, 32, 5, 64);   // MNIST conv2
+  else
+    LO_CFS(false);
+  #undef LO_CFS
   return true;
 }
 
@@ -1233,12 +1272,19 @@ bool launch_conv1d_dx(const void* dy2, long ldy, const void* wt, long ldw,
 // (for C=1 consecutive rows are consecutive x addresses, but row wraps make
 // a b128 read unsafe).  Per-block partial dW is reduced across waves in LDS
 // and added to global fp32 dW with one atomicAdd per element.
+template <int TH = 0, int TW = 0, int TK = 0, int TOC = 0>
 __global__ __launch_bounds__(256) void conv_dw_c1_kernel(
     const bf16* __restrict__ dy2, long ldy,   // [B*OH*OW, outC]
     const bf16* __restrict__ x,               // [B, H, W, 1]
     float* __restrict__ dw, long ldw,         // [outC, kpad] fp32, pre-zeroed
-    int B, int H, int W, int KH, int KW, int SH, int SW, int PH, int PW,
-    int OH, int OW, int outC, int imgs_per_block) {
+    int B, int H_, int W_, int KH_, int KW_, int SH_, int SW_, int PH_,
+    int PW_, int OH_, int OW_, int outC_, int imgs_per_block) {
+  const int H = TH ? TH : H_, W = TW ? TW : W_;
+  const int KH = TK ? TK : KH_, KW = TK ? TK : KW_;
+  const int SH = TH ? 1 : SH_, SW = TH ? 1 : SW_;
+  const int PH = TH ? 0 : PH_, PW = TH ? 0 : PW_;
+  const int OH = TH ? (TH - TK + 1) : OH_, OW = TW ? (TW - TK + 1) : OW_;
+  const int outC = TOC ? TOC : outC_;
   extern __shared__ char sm[];
   const int tid = threadIdx.x;
   const int lane = tid & 63, wave = tid >> 6;
@@ -1356,9 +1402,18 @@ bool launch_conv_dw_c1(const void* dy2, long ldy, const void* x, void* dw,
   const int ipb = max(1, (B + 2047) / 2048);
   const int blocks = (B + ipb - 1) / ipb;
   const int lds = ((H * W * 2 + 127) & ~127) + 4 * 4096;
-  hipLaunchKernelGGL(conv_dw_c1_kernel, dim3(blocks), dim3(256), lds, s,
-                     (const bf16*)dy2, ldy, (const bf16*)x, (float*)dw, ldw,
-                     B, H, W, KH, KW, SH, SW, PH, PW, OH, OW, outC, ipb);
+  #define LO_CDW(...)                                                         \
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(conv_dw_c1_kernel<__VA_ARGS__>),       \
+                       dim3(blocks), dim3(256), lds, s,                       \
+                       (const bf16*)dy2, ldy, (const bf16*)x, (float*)dw,     \
+                       ldw, B, H, W, KH, KW, SH, SW, PH, PW, OH, OW, outC,    \
+                       ipb)
+  if (H == 28 && W == 28 && KH == 5 && KW == 5 && SH == 1 && SW == 1 &&
+      PH == 0 && PW == 0 && outC == 32)
+    LO_CDW(28, 28, 5, 32);              // MNIST conv1 dW
+  else
+    LO_CDW(0);
+  #undef LO_CDW
   return true;
 }
 
