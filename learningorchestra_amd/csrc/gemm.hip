@@ -680,6 +680,10 @@ static bool dispatch_conv_dw(const GemmArgs& g, hipStream_t s) {
   // (outC=64 shapes measured FASTER on 64x64 than 64x128 — MNIST conv2 dW
   // regressed 6.78 -> 6.35 M samples/s on the wide tile — so wide tiles
   // require outC >= 128)
+  // (re-measured r2 after the shape specializations: <64,128> for outC=64
+  // STILL regresses MNIST — 7.14 vs 7.79 M samples/s — despite halving the
+  // dY re-reads; the wide-BN gather's per-thread column state doubles and
+  // eats the win. Wide tiles stay gated to outC >= 128.)
   if (g.M >= 128 && g.N >= 128 && g.K >= (1 << 18))
     launch_cfg<128, 128, 64, 2, 2, true, false, 0, true, true, 2>(g, s);
   else if (g.M >= 64)
