@@ -71,17 +71,26 @@ template <bool VEC8>
 __global__ void im2col_kernel(const bf16* __restrict__ in, bf16* __restrict__ col,
                               int B, int H, int W, int C, int KH, int KW,
                               int SH, int SW, int PH, int PW,
-                              int OH, int OW, int Kpad) {
+                              int OH, int OW, int Kpad,
+                              FDiv fCV, FDiv fKW, FDiv fKH, FDiv fOW,
+                              FDiv fOH) {
   const int CV = VEC8 ? C / 8 : C;            // channel units per position
   const long total = (long)B * OH * OW * KH * KW * CV;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
     unsigned int r32 = (unsigned int)i;   // totals < 2^31 by launch contract
-    const int cu = r32 % CV; r32 /= CV;
-    const int kw = r32 % KW; r32 /= KW;
-    const int kh = r32 % KH; r32 /= KH;
-    const int ow = r32 % OW; r32 /= OW;
-    const int oh = r32 % OH; r32 /= OH;
+    // host-computed magic division: the 5-deep runtime div/mod chain was
+    // the kernel's VALU bound (same disease PMC showed on the conv family)
+    unsigned q = fdiv2(r32, fCV);
+    const int cu = (int)(r32 - q * CV); r32 = q;
+    q = fdiv2(r32, fKW);
+    const int kw = (int)(r32 - q * KW); r32 = q;
+    q = fdiv2(r32, fKH);
+    const int kh = (int)(r32 - q * KH); r32 = q;
+    q = fdiv2(r32, fOW);
+    const int ow = (int)(r32 - q * OW); r32 = q;
+    q = fdiv2(r32, fOH);
+    const int oh = (int)(r32 - q * OH); r32 = q;
     const int b = r32;
     const int h = oh * SH - PH + kh, w = ow * SW - PW + kw;
     const long row = ((long)b * OH + oh) * OW + ow;
@@ -138,14 +147,20 @@ void launch_im2col(const void* in, void* col, int B, int H, int W, int C,
   const long total = (long)B * OH * OW * KH * KW * (vec ? C / 8 : C);
   const int block = 256;
   const int grid = (int)min((total + block - 1) / block, (long)2048);
+  FDiv fCV, fKW, fKH, fOW, fOH;
+  mkmagic((unsigned)(vec ? C / 8 : C), fCV);
+  mkmagic((unsigned)KW, fKW);
+  mkmagic((unsigned)KH, fKH);
+  mkmagic((unsigned)OW, fOW);
+  mkmagic((unsigned)OH, fOH);
   if (vec)
     hipLaunchKernelGGL(HIP_KERNEL_NAME(im2col_kernel<true>), dim3(grid), dim3(block), 0, s,
                        (const bf16*)in, (bf16*)col, B, H, W, C, KH, KW, SH, SW,
-                       PH, PW, OH, OW, Kpad);
+                       PH, PW, OH, OW, Kpad, fCV, fKW, fKH, fOW, fOH);
   else
     hipLaunchKernelGGL(HIP_KERNEL_NAME(im2col_kernel<false>), dim3(grid), dim3(block), 0, s,
                        (const bf16*)in, (bf16*)col, B, H, W, C, KH, KW, SH, SW,
-                       PH, PW, OH, OW, Kpad);
+                       PH, PW, OH, OW, Kpad, fCV, fKW, fKH, fOW, fOH);
 }
 
 // --------------------------------------------------------------- col2im ----
@@ -156,15 +171,18 @@ template <bool VEC8>
 __global__ void col2im_kernel(const bf16* __restrict__ dcol, bf16* __restrict__ dx,
                               int B, int H, int W, int C, int KH, int KW,
                               int SH, int SW, int PH, int PW, int OH, int OW,
-                              int Kpad) {
+                              int Kpad, FDiv fCV, FDiv fW, FDiv fH) {
   const int CV = VEC8 ? C / 8 : C;
   const long total = (long)B * H * W * CV;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
     unsigned int r32 = (unsigned int)i;
-    const int cu = r32 % CV; r32 /= CV;
-    const int w = r32 % W; r32 /= W;
-    const int h = r32 % H; r32 /= H;
+    unsigned q = fdiv2(r32, fCV);
+    const int cu = (int)(r32 - q * CV); r32 = q;
+    q = fdiv2(r32, fW);
+    const int w = (int)(r32 - q * W); r32 = q;
+    q = fdiv2(r32, fH);
+    const int h = (int)(r32 - q * H); r32 = q;
     const int b = r32;
     float acc[VEC8 ? 8 : 1] = {};
     for (int kh = 0; kh < KH; ++kh) {
@@ -206,12 +224,16 @@ void launch_col2im(const void* dcol, void* dx, int B, int H, int W, int C,
   const long total = (long)B * H * W * (vec ? C / 8 : C);
   const int block = 256;
   const int grid = (int)min((total + block - 1) / block, (long)2048);
+  FDiv fCV, fW, fH;
+  mkmagic((unsigned)(vec ? C / 8 : C), fCV);
+  mkmagic((unsigned)W, fW);
+  mkmagic((unsigned)H, fH);
   if (vec)
     hipLaunchKernelGGL(HIP_KERNEL_NAME(col2im_kernel<true>), dim3(grid), dim3(block), 0, s,
-                       (const bf16*)dcol, (bf16*)dx, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW, Kpad);
+                       (const bf16*)dcol, (bf16*)dx, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW, Kpad, fCV, fW, fH);
   else
     hipLaunchKernelGGL(HIP_KERNEL_NAME(col2im_kernel<false>), dim3(grid), dim3(block), 0, s,
-                       (const bf16*)dcol, (bf16*)dx, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW, Kpad);
+                       (const bf16*)dcol, (bf16*)dx, B, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW, Kpad, fCV, fW, fH);
 }
 
 // ------------------------------------------------------- fused conv dX -----

@@ -119,11 +119,19 @@ __global__ __launch_bounds__(WM * WN * 64) void gemm_kernel(
 
   const int nb = (N + BN - 1) / BN;
   const int mb = (M + BM - 1) / BM;
+  // (rejected r2 experiment, kept as a note: remapping (tile, z) so each
+  // XCD sweeps one k-slice's tiles before moving on — to keep the A(dy)
+  // slice L2-resident across its N-tile re-reads — measured WORSE on every
+  // model (TextCNN -12%, MNIST/ResNet -2%): the i%8 -> XCD round-robin
+  // assumption evidently does not describe the MI355X workgroup dispatcher,
+  // and per-slice tile counts are too small to amortize the lost
+  // cross-slice parallelism.)
   int bid = xcd_swizzle(blockIdx.x, mb * nb);
   const int bm = bid % mb, bn = bid / mb;   // consecutive blocks share B panel
   const int m0 = bm * BM, n0 = bn * BN;
 
   const int kBegin = kStart + blockIdx.z * kChunk;
+  if (kBegin >= K) return;                  // padded z-slice (block-uniform)
   const int kEnd = min(K, kBegin + kChunk);
 
   f32x4 acc[MFRAG][NFRAG] = {};

@@ -49,4 +49,26 @@ LO_DEVICE int xcd_swizzle(int bid, int nblocks) {
   return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
 }
 
+// ---- fast integer division (magic numbers) --------------------------------
+// q = n / d exact for n < 2^31: mg==0 -> d==1; mg==1 -> d==2^s (q = n>>s);
+// else q = (n*mg) >> (32+s) with mg = ceil(2^(31+s0)/d), s = s0-1 (the
+// 2^(32+s) formulation overflows u32 — r1 hard-won invariant).
+struct FDiv { unsigned mg = 0; int s = 0; };
+
+LO_DEVICE unsigned fdiv2(unsigned n, FDiv f) {
+  if (f.mg == 0) return n;
+  if (f.mg == 1) return n >> f.s;
+  return (unsigned)(((unsigned long long)n * f.mg) >> 32) >> f.s;
+}
+
+inline void mkmagic(unsigned d, FDiv& f) {
+  if (d <= 1) { f.mg = 0; f.s = 0; return; }
+  int s = 0;
+  while ((1u << s) < d) ++s;
+  if ((1u << s) == d) { f.mg = 1; f.s = s; return; }
+  const unsigned long long L = 1ull << (31 + s);
+  f.mg = (unsigned)((L + d - 1) / d);
+  f.s = s - 1;
+}
+
 }  // namespace lo
