@@ -466,6 +466,9 @@ __global__ __launch_bounds__(256) void conv_fwd_small_kernel(
       kw8[j] = k - kh8[j] * KW;
       kok8[j] = k < KH * KW;
     }
+    // ceil(outC/16) MFMA fragments only — the fixed ni<4 loop computed 64
+    // output columns even for outC=32 (HALF the conv1 MFMAs were garbage)
+    const int NI = (outC + 15) >> 4;
     for (int m0 = 0; m0 < R; m0 += 64) {
       const int arow = m0 + wave * 16 + (lane & 15);
       const int aoh = arow / OW, aow = arow - aoh * OW;
@@ -474,12 +477,18 @@ __global__ __launch_bounds__(256) void conv_fwd_small_kernel(
       #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int h = hb + kh8[j], wx = wb + kw8[j];
-        if (arow < R && kok8[j] && h >= 0 && h < H && wx >= 0 && wx < W)
+        // TH!=0 implies stride 1 / pad 0: h = aoh + kh <= (OH-1)+(KH-1)
+        // < H and likewise wx — the bounds are provably true
+        const bool ok = TH ? (arow < R && kok8[j])
+                           : (arow < R && kok8[j] && h >= 0 && h < H &&
+                              wx >= 0 && wx < W);
+        if (ok)
           af[j] = *(const bf16*)(smX + (h * W + wx) * 2);
       }
       f32x4 acc[4] = {};
       #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
+        if (ni >= NI) break;
         const int row = ni * 16 + (lane & 15);
         const int k2 = (lane >> 4) * 8;
         const bf16x8 bfr = *(const bf16x8*)(
@@ -490,6 +499,7 @@ __global__ __launch_bounds__(256) void conv_fwd_small_kernel(
       }
       #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
+        if (ni >= NI) break;
         const int c = ni * 16 + (lane & 15);
         const float b = bias ? bias[c < outC ? c : 0] : 0.f;
         #pragma unroll
