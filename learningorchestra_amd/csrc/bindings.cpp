@@ -68,6 +68,10 @@ bool launch_conv_dw_c1(const void* dy2, long ldy, const void* x, void* dw,
                        long ldw, int B, int H, int W, int KH, int KW, int SH,
                        int SW, int PH, int PW, int OH, int OW, int outC,
                        hipStream_t s);
+bool launch_conv_dw_smallc(const void* dy2, long ldy, const void* x, void* dw,
+                           long ldw, int B, int H, int W, int C, int KH,
+                           int KW, int SH, int SW, int PH, int PW, int OH,
+                           int OW, int outC, hipStream_t s);
 bool launch_conv_dx(const void* dy2, long ldy, const void* wt, long ldw,
                     void* dx, int B, int H, int W, int C, int KH, int KW,
                     int SH, int SW, int PH, int PW, int OH, int OW, int outC,
@@ -336,6 +340,25 @@ bool conv1d_dx(at::Tensor dy2, at::Tensor wt, at::Tensor dx, int64_t KH,
 
 // C=1 conv dW: dY^T @ im2col(x) with x images LDS-resident; dw fp32
 // accumulated with one atomicAdd per element per block (dw zeroed here).
+bool conv_dw_small(at::Tensor dy2, at::Tensor x, at::Tensor dw, int64_t KH,
+                   int64_t KW, int64_t SH, int64_t SW, int64_t PH, int64_t PW) {
+  check_bf16(dy2, "dy2");
+  check_bf16(x, "x");
+  TORCH_CHECK(dw.scalar_type() == at::kFloat, "dw must be fp32");
+  const int B = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2),
+            C = (int)x.size(3);
+  const int OH = (H + 2 * (int)PH - (int)KH) / (int)SH + 1;
+  const int OW = (W + 2 * (int)PW - (int)KW) / (int)SW + 1;
+  const int outC = (int)dy2.size(1);
+  TORCH_CHECK(dy2.size(0) == (long)B * OH * OW, "dy2 rows");
+  TORCH_CHECK(dw.size(0) == outC && dw.size(1) >= KH * KW * C, "dw shape");
+  dw.zero_();  // harmless extra zero when ineligible (fallback re-zeroes)
+  return lo::launch_conv_dw_smallc(dy2.data_ptr(), dy2.stride(0), x.data_ptr(),
+                                   dw.data_ptr(), dw.stride(0), B, H, W, C,
+                                   (int)KH, (int)KW, (int)SH, (int)SW,
+                                   (int)PH, (int)PW, OH, OW, outC, stream());
+}
+
 bool conv_dw_c1(at::Tensor dy2, at::Tensor x, at::Tensor dw, int64_t KH,
                 int64_t KW, int64_t SH, int64_t SW, int64_t PH, int64_t PW) {
   check_bf16(dy2, "dy2");
@@ -652,6 +675,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv1d_fwd", &conv1d_fwd, "1-D conv fwd (h-tiled LDS window)");
   m.def("conv1d_dx", &conv1d_dx, "1-D conv dX (h-tiled LDS RMW)");
   m.def("conv_dw_c1", &conv_dw_c1, "C=1 conv dW (x LDS-resident)");
+  m.def("conv_dw_small", &conv_dw_small,
+        "small-image C%8==0 conv dW (x+dyT LDS-resident per image)");
   m.def("conv_dx", &conv_dx, "fused conv dX (LDS-accumulated scatter)");
   m.def("col2im", &col2im);
   m.def("maxpool_fwd", &maxpool_fwd, py::arg("in"), py::arg("KH"),

@@ -224,9 +224,14 @@ class Conv2dNHWC(Layer):
         gw = self.arena.g(self.name + ".w")
         splits = _splitk_heuristic(self.out_c, self.kpad, M)
         if dy2.is_cuda and self.implicit and not self._is_1x1:
-            if not (self.in_c == 1 and self.out_c <= 32 and F.conv2d_dw_c1(
-                    dy2, self._x, gw, self.kh, self.kw, self.stride,
-                    self.stride, self.ph, self.pw)):
+            done = (self.in_c == 1 and self.out_c <= 32 and F.conv2d_dw_c1(
+                dy2, self._x, gw, self.kh, self.kw, self.stride,
+                self.stride, self.ph, self.pw))
+            if not done:
+                done = F.conv2d_dw_small(dy2, self._x, gw, self.kh, self.kw,
+                                         self.stride, self.stride, self.ph,
+                                         self.pw)
+            if not done:
                 F.conv2d_dw_implicit(dy2, self._x, gw, self.kh, self.kw,
                                      self.stride, self.stride, self.ph,
                                      self.pw, splits)

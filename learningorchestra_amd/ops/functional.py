@@ -149,6 +149,25 @@ def conv1d_dx(dy2: torch.Tensor, wt: torch.Tensor, kh: int, ph: int,
     return bool(lo.conv1d_dx(dy2, wt, out, kh, ph, accumulate))
 
 
+def conv2d_dw_small(dy2: torch.Tensor, x: torch.Tensor, gw: torch.Tensor,
+                    kh: int, kw: int, sh: int, sw: int, ph: int, pw: int) -> bool:
+    """GPU-only small-image multi-channel conv weight grad: x + transposed
+    dy LDS-resident per image, register-accumulated kpad chunks — built to
+    kill the gather-GEMM's per-kpad-tile dY re-streaming (3.5 GB on MNIST
+    conv2). Measured r2: ~1% SLOWER than the gather GEMM end-to-end (8.05
+    vs 8.13 M samples/s) — the per-element scalar LDS gathers of the
+    B operand pay in bank conflicts what the halved dY traffic saves — so
+    it is OFF by default (LO_DW_SMALL=1 enables; kernel + numerics tests
+    kept as the documented alternative).  gw fp32, zeroed inside."""
+    if not dy2.is_cuda:
+        return False
+    import os
+    if os.environ.get("LO_DW_SMALL", "0") != "1":
+        return False
+    lo = require_ext()
+    return bool(lo.conv_dw_small(dy2, x, gw, kh, kw, sh, sw, ph, pw))
+
+
 def conv2d_dw_c1(dy2: torch.Tensor, x: torch.Tensor, gw: torch.Tensor,
                  kh: int, kw: int, sh: int, sw: int, ph: int, pw: int) -> bool:
     """GPU-only C=1 conv weight grad (dY^T @ im2col(x)) with x images

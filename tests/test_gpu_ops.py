@@ -484,3 +484,31 @@ def test_padded_1d_conv_layer_gpu():
     dxc = c.backward(dy.clone())
     torch.testing.assert_close(dxg.float().cpu(), dxc.float(), atol=8e-2,
                                rtol=8e-2)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [
+    (64, 12, 12, 32, 5, 64),   # MNIST conv2 (specialized instance)
+    (32, 10, 10, 16, 3, 48),   # generic instance, odd sizes
+    (8, 6, 6, 8, 3, 16),       # tiny
+])
+def test_conv_dw_small_vs_reference(shape, monkeypatch):
+    """Small-image multi-channel dW kernel vs the materialized split-K
+    (the kernel is off by default — measured 1% behind the gather GEMM —
+    but stays correct and available under LO_DW_SMALL=1)."""
+    monkeypatch.setenv("LO_DW_SMALL", "1")
+    from learningorchestra_amd.ops import functional as F
+    B, H, W, C, K, outC = shape
+    torch.manual_seed(5)
+    OH = OW = H - K + 1
+    kdim = K * K * C
+    kpad = (kdim + 7) // 8 * 8
+    x = torch.randn(B, H, W, C, device="cuda").to(torch.bfloat16)
+    dy2 = torch.randn(B * OH * OW, outC, device="cuda").to(torch.bfloat16)
+    gw = torch.empty(outC, kpad, device="cuda", dtype=torch.float32)
+    ok = F.conv2d_dw_small(dy2, x, gw, K, K, 1, 1, 0, 0)
+    assert ok, "shape should be eligible"
+    col = F.im2col(x, K, K, 1, 1, 0, 0, kpad)
+    ref = F.gemm(dy2, col, ta=True, splits=4)
+    torch.testing.assert_close(gw[:, :kdim], ref[:, :kdim], atol=2e-1,
+                               rtol=2e-2)
