@@ -37,10 +37,24 @@ def gemm(A: torch.Tensor, B: torch.Tensor, *, ta: bool = False, tb: bool = False
     """
     M = A.shape[1] if ta else A.shape[0]
     N = B.shape[0] if tb else B.shape[1]
+    K = A.shape[0] if ta else A.shape[1]
     dtype = out_dtype or (torch.float32 if splits > 1 else A.dtype)
     if out is None:
         out = torch.empty((M, N), device=A.device, dtype=dtype)
     if _is_gpu(A):
+        # PLAIN (F,T) GEMMs with a large N*K panel go to rocBLAS — the
+        # sanctioned "plain library GEMM" case (no fused epilogue to carry).
+        # Measured crossover on ResNet/MNIST dX shapes (benchmarks/
+        # rn_gemm_ab.py): rocBLAS wins 1.2-1.9x above N*K ~= 256k
+        # ((1024,256) 0.59, (512,2048) 0.57, (4608,512) 0.61) while the
+        # native tile kernel wins below it ((1152,128) 1.02, (64,64) 1.93).
+        # Every fused GEMM (bias/ReLU/BN-stats/addend/split-K) stays native.
+        if (not ta and tb and bias is None and not relu and stats is None
+                and addend is None and splits == 1
+                and A.dtype == torch.bfloat16 and out.dtype == torch.bfloat16
+                and N * K >= 256 * 1024):
+            torch.matmul(A, B.t(), out=out)
+            return out
         lo = require_ext()
         if splits > 1:
             out.zero_()
