@@ -60,6 +60,8 @@ class ResponseCache:
         self._store: Dict[str, Any] = {}
         self._lock = threading.Lock()
 
+    MAX_ENTRIES = 4096
+
     def get_or(self, key: str, fn):
         if self.ttl <= 0:
             return fn()
@@ -70,6 +72,10 @@ class ResponseCache:
                 return hit[1]
         value = fn()
         with self._lock:
+            if len(self._store) >= self.MAX_ENTRIES:
+                # drop expired entries first; fall back to clearing
+                live = {k: v for k, v in self._store.items() if v[0] > now}
+                self._store = live if len(live) < self.MAX_ENTRIES else {}
             self._store[key] = (now + self.ttl, value)
         return value
 

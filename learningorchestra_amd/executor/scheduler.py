@@ -218,6 +218,12 @@ class JobScheduler:
 
         with self._lock:
             self._jobs[name] = job
+            # long-running servers: prune the oldest finished jobs so the
+            # registry cannot grow without bound
+            if len(self._jobs) > 4096:
+                done = [n for n, j in self._jobs.items() if j.done()]
+                for n in done[: len(done) // 2]:
+                    self._jobs.pop(n, None)
         job.future = self._pool.submit(run)
         if job.timeout is not None:
             self._ensure_watchdog()

@@ -50,6 +50,11 @@ def launch_distributed_train(scheduler, metadata, cfg, *, name: str,
     metadata doc with finished=false; this fills the execution document and
     flips the flag when the process tree exits."""
     import time as _time
+
+    # a wedged multi-rank job must not hold its device slots forever: give
+    # every distributed train a ceiling unless the caller set one
+    if timeout is None:
+        timeout = 24 * 3600.0
     workdir = tempfile.mkdtemp(prefix="lo_ddp_")
     result_path = os.path.join(workdir, "result.json")
     spec = {
@@ -91,6 +96,8 @@ def launch_distributed_train(scheduler, metadata, cfg, *, name: str,
         if os.path.exists(result_path):
             with open(result_path) as fh:
                 result = json.load(fh)
+        import shutil
+        shutil.rmtree(workdir, ignore_errors=True)
         metadata.create_execution_document(
             name, description or f"{method} on {parent_name} (dp{gpus})",
             {"methodParameters": {k: repr(v)[:200] for k, v in

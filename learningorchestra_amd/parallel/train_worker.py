@@ -101,7 +101,7 @@ def main() -> None:
         if spec.get("result_path"):
             tmp = spec["result_path"] + ".tmp"
             with open(tmp, "w") as fh:
-                json.dump(out, fh)
+                json.dump(out, fh, default=str)  # numpy scalars etc.
             os.replace(tmp, spec["result_path"])
         # metadata is written by the API process when the job completes
         _ = Metadata  # imported for parity; reserved for future direct writes
