@@ -153,3 +153,29 @@ def test_ddp_gloo_textcnn_embedding_grads_sync():
     for rank, checksum, same in results:
         assert not isinstance(checksum, str), checksum
         assert same, "textcnn params diverged across ranks"
+
+
+def test_bench_torchrun_world4_cpu(tmp_path):
+    """The EXACT command shape the round driver uses for the N-GPU scaling
+    ladder, at world 4 on CPU/gloo: one JSON line, whole-job aggregate,
+    dp4 parallelism reported."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    from learningorchestra_amd.parallel.launch import free_port
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", str(free_port()), "bench.py", "--gpus", "4",
+         "--steps", "2", "--warmup", "1", "--batch", "64"],
+        capture_output=True, text=True, timeout=420, cwd=repo)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    line = [ln for ln in r.stdout.splitlines() if ln.startswith("{")][-1]
+    doc = json.loads(line)
+    assert doc["n_gpus"] == 4
+    assert doc["config"]["parallelism"] == "dp4"
+    assert doc["config"]["global_batch"] == 4 * 64
+    assert doc["value"] > 0
