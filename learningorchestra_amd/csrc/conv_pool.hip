@@ -257,6 +257,7 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
     const bf16* __restrict__ dy2, long ldy,   // [B*OH*OW, outC]
     const bf16* __restrict__ wt, long ldw,    // [kpad, outC] row-major
     bf16* __restrict__ dx,                    // [B, H, W, C]
+    int B, int G,                             // G images share wt stagings
     int H_, int W_, int C_, int KH_, int KW_, int SH_, int SW_, int PH_,
     int PW_, int OH_, int OW_, int outC_) {
   const int H = TH ? TH : H_, W = TW ? TW : W_, C = TC ? TC : C_;
@@ -269,17 +270,19 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
   // keeps 16-B alignment for the vectorized RMW and staggers banks across
   // rows) + 2 x 4 KB wt tile buffers (32 rows x 64 k, XOR-swizzled)
   extern __shared__ float ldx[];
-  const int img = blockIdx.x;
+  const int img0 = blockIdx.x * G;            // wt stagings amortized over G
   const int R = OH * OW;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wave = tid >> 6;
   const int HWC = H * W * C;
   const int CP = C + 4;   // +4: keeps 16-B alignment AND staggers banks
   const int HWCP = H * W * CP;
-  char* smW = (char*)(ldx + HWCP);
-  for (int i = tid; i < HWCP; i += 256) ldx[i] = 0.f;
+  char* smW = (char*)(ldx + G * HWCP);        // (G <= 2)
+  for (int i = tid; i < G * HWCP; i += 256) ldx[i] = 0.f;
 
-  const bf16* dyi = dy2 + (long)img * OW * OH * ldy;
+  const bf16* dyi0 = dy2 + (long)img0 * OW * OH * ldy;
+  const bf16* dyi1 = dy2 + (long)(img0 + 1) * OW * OH * ldy;
+  const bool g1ok = G > 1 && img0 + 1 < B;
   const int CB = C > 32 ? 32 : C;             // n-tile: one (kh,kw) C-slice
   const int NSTEP = KH * KW * (C / CB);       // (kh,kw,cb) tiles per m-chunk
 
@@ -308,12 +311,15 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
     // channels of ONE m-row, so the LDS RMW is a single b128 per fragment
     // (the row-per-reg layout needed 8 scalar RMWs per step).
     const int arow = m0 + wave * 16 + (lane & 15);     // this lane's m
-    bf16x8 dyf[2] = {};
+    bf16x8 dyf[2] = {}, dyf1[2] = {};
     #pragma unroll
     for (int kc = 0; kc < 2; ++kc) {
       const int k = kc * 32 + (lane >> 4) * 8;
-      if (arow < R && k < outC)
-        dyf[kc] = *(const bf16x8*)(dyi + (long)arow * ldy + k);
+      if (arow < R && k < outC) {
+        dyf[kc] = *(const bf16x8*)(dyi0 + (long)arow * ldy + k);
+        if (g1ok)
+          dyf1[kc] = *(const bf16x8*)(dyi1 + (long)arow * ldy + k);
+      }
     }
     const int aoh = arow / OW, aow = arow - aoh * OW;
     const bool mok = arow < R;
@@ -327,7 +333,7 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
       const int cslices = C / CB;
       const int khkw = step / cslices, cb = (step % cslices) * CB;
       const int kh = khkw / KW, kw = khkw - kh * KW;
-      f32x4 acc[2] = {};
+      f32x4 acc[2] = {}, acc1[2] = {};
       #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
         const int k = kc * 32 + (lane >> 4) * 8;
@@ -338,6 +344,9 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
               smW + buf * 4096 + row * 128 + ((k * 2) ^ ((row & 7) << 4)));
           acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               wf, dyf[kc], acc[mi], 0, 0, 0);
+          if (g1ok)
+            acc1[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                wf, dyf1[kc], acc1[mi], 0, 0, 0);
         }
       }
       // D col=lane&15 -> m (this lane's arow), row=(lane>>4)*4+reg -> kcol.
@@ -353,6 +362,12 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
           f32x4 v = *(f32x4*)t;
           v += acc[mi];
           *(f32x4*)t = v;
+          if (g1ok) {
+            float* t1 = ldx + HWCP + (h * W + w) * CP + c;
+            f32x4 v1 = *(f32x4*)t1;
+            v1 += acc1[mi];
+            *(f32x4*)t1 = v1;
+          }
         }
       }
       // the write targets the OPPOSITE buffer of every in-flight read and
@@ -362,14 +377,18 @@ __global__ __launch_bounds__(256) void conv_dx_kernel(
     }
   }
   __syncthreads();
-  bf16* dxi = dx + (long)img * HWC;
-  for (int i = tid * 8; i < HWC; i += 256 * 8) {
-    const int hw = i / C, c0 = i - hw * C;  // C%8==0: chunk stays in-row
-    const float* src = ldx + hw * CP + c0;
-    bf16x8 v;
-    #pragma unroll
-    for (int j = 0; j < 8; ++j) v[j] = tobf16(src[j]);
-    *(bf16x8*)(dxi + i) = v;
+  for (int g = 0; g < G; ++g) {
+    if (img0 + g >= B) break;
+    bf16* dxi = dx + (long)(img0 + g) * HWC;
+    const float* base = ldx + g * HWCP;
+    for (int i = tid * 8; i < HWC; i += 256 * 8) {
+      const int hw = i / C, c0 = i - hw * C;  // C%8==0: chunk stays in-row
+      const float* src = base + hw * CP + c0;
+      bf16x8 v;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = tobf16(src[j]);
+      *(bf16x8*)(dxi + i) = v;
+    }
   }
 }
 
@@ -382,12 +401,21 @@ bool launch_conv_dx(const void* dy2, long ldy, const void* wt, long ldw,
   if ((long)HWCP * 4 > 56 * 1024 || (C % 32 != 0 && C != 16) || HWC % 8 != 0 ||
       outC > 64 || outC % 8 != 0 || C % 8 != 0)
     return false;
+  // G=2 images per block share every wt-tile staging (the conv1d_dx2
+  // lesson: per-image restaging was NSTEP x 4 KB x B of L2 traffic)
+  static const int forceG = [] {
+    const char* e = getenv("LO_CONVDX_G");
+    return e ? atoi(e) : 0;
+  }();
+  int G = forceG ? forceG : 2;
+  if (G * HWCP * 4 + 8192 > 56 * 1024) G = 1;
   #define LO_CDX(...)                                                         \
     hipLaunchKernelGGL(HIP_KERNEL_NAME(conv_dx_kernel<__VA_ARGS__>),          \
-                       dim3(B), dim3(256), HWCP * 4 + 8192, s,                \
+                       dim3((B + G - 1) / G), dim3(256),                      \
+                       G * HWCP * 4 + 8192, s,                                \
                        (const bf16*)dy2, ldy, (const bf16*)wt, ldw,           \
-                       (bf16*)dx, H, W, C, KH, KW, SH, SW, PH, PW, OH, OW,    \
-                       outC)
+                       (bf16*)dx, B, G, H, W, C, KH, KW, SH, SW, PH, PW,      \
+                       OH, OW, outC)
   if (H == 12 && W == 12 && C == 32 && KH == 5 && KW == 5 && SH == 1 &&
       SW == 1 && PH == 0 && PW == 0 && outC == 64)
     LO_CDX(12, 12, 32, 5, 64);          // MNIST conv2 dX
