@@ -210,11 +210,19 @@ class Conv2dNHWC(Layer):
     def backward(self, dy: torch.Tensor,
                  dx_out: Optional[torch.Tensor] = None,
                  dx_accumulate: bool = False,
-                 dx_addend: Optional[torch.Tensor] = None) -> Optional[torch.Tensor]:
+                 dx_addend: Optional[torch.Tensor] = None,
+                 bias_grad_src: Optional[torch.Tensor] = None
+                 ) -> Optional[torch.Tensor]:
         """``dx_out``/``dx_accumulate``: on the fused 1-D dX path, write (or
         add) the input grad straight into the caller's buffer — fuses the
         multi-branch grad sum (TextCNN).  Callers must check the returned
-        tensor: on fallback paths it is the layer's own dx buffer."""
+        tensor: on fallback paths it is the layer's own dx buffer.
+
+        ``bias_grad_src``: a small [rows, out_c] matrix whose column sums
+        EQUAL colsum(dy) — e.g. the relu-masked pool-level grad when a
+        global max pool follows this conv (the pool scatters each value at
+        most once, so the sums agree); the bias grad then reads KBs instead
+        of re-streaming the full dy (266 MB per TextCNN conv)."""
         bufs = self._bufs
         M = bufs["y"].shape[0]
         dy2 = dy.reshape(M, self.out_c)
@@ -238,7 +246,8 @@ class Conv2dNHWC(Layer):
         else:
             F.gemm(dy2, bufs["col"], ta=True, out=gw, splits=splits)
         if self.bias:
-            F.colsum(dy2, out=self.arena.g(self.name + ".b"))
+            F.colsum(bias_grad_src if bias_grad_src is not None else dy2,
+                     out=self.arena.g(self.name + ".b"))
         if self.first:
             return None
         if self._is_1x1:

@@ -103,9 +103,18 @@ class TextCNN:
                                                 self.pools)):
             dpool = dcat[:, i * Fn:(i + 1) * Fn].contiguous().view(B, 1, 1, Fn)
             dconv = pool.backward(dpool)              # [B, S-k+1, 1, F]
+            # bias grad from the TINY pool-level grad: the global max pool
+            # scatters each dpool value at most once (relu-sentinel masked),
+            # so colsum(dconv) == colsum(dpool masked by idx != 255) — this
+            # replaces a full 266 MB dconv re-read per conv with a [B, F] one
+            idx = pool._bufs["idx"].view(B, Fn)
+            bias_src = torch.where(idx != 255, dpool.view(B, Fn),
+                                   torch.zeros((), dtype=dpool.dtype,
+                                               device=dpool.device))
             # fused path writes/adds straight into dxe (branch-grad sum)
             dbranch = conv.backward(dconv, dx_out=dxe,
-                                    dx_accumulate=(i > 0))  # [B, S, 1, emb]
+                                    dx_accumulate=(i > 0),
+                                    bias_grad_src=bias_src)  # [B, S, 1, emb]
             if grad_hook:
                 grad_hook(conv.param_names())
             if dbranch is not dxe:
