@@ -115,6 +115,8 @@ void launch_adam(void* master, const void* grad, void* m1, void* m2, void* mirro
                  const void* step_dev, long n, float lr, float b1, float b2,
                  float eps, float wd, float gscale, hipStream_t s);
 void launch_colsum(const void* dy, void* out, long M, int N, long ldy, hipStream_t s);
+void launch_colsum_masked(const void* dy, const void* mask, void* out, long M,
+                          int N, long ldy, hipStream_t s);
 void launch_argmax_rows(const void* x, void* out, long M, int C, int Cvalid,
                         long ldx, hipStream_t s);
 void launch_accuracy_count(const void* pred, const void* label, void* out, long M,
@@ -572,10 +574,20 @@ void adam_step(at::Tensor master, at::Tensor grad, at::Tensor m1, at::Tensor m2,
                   (float)gscale, stream());
 }
 
-at::Tensor colsum(at::Tensor dy, at::Tensor out) {
+at::Tensor colsum(at::Tensor dy, at::Tensor out,
+                  c10::optional<at::Tensor> mask = c10::nullopt) {
   check_bf16(dy, "dy");
   check_f32(out, "out");
   TORCH_CHECK(dy.dim() == 2 && out.numel() == dy.size(1));
+  if (mask) {
+    TORCH_CHECK(mask->scalar_type() == at::kByte && mask->is_contiguous() &&
+                mask->numel() == dy.numel(), "mask: u8, same shape");
+    TORCH_CHECK(dy.size(1) <= 256, "masked colsum: N <= 256");
+    lo::launch_colsum_masked(dy.data_ptr(), mask->data_ptr(), out.data_ptr(),
+                             dy.size(0), (int)dy.size(1), dy.size(1),
+                             stream());
+    return out;
+  }
   lo::launch_colsum(dy.data_ptr(), out.data_ptr(), dy.size(0), (int)dy.size(1),
                     dy.size(1), stream());
   return out;
@@ -686,7 +698,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_bwd", &relu_bwd);
   m.def("sgd_step", &sgd_step);
   m.def("adam_step", &adam_step);
-  m.def("colsum", &colsum);
+  m.def("colsum", &colsum, py::arg("dy"), py::arg("out"),
+        py::arg("mask") = py::none());
   m.def("argmax_rows", &argmax_rows);
   m.def("accuracy_count", &accuracy_count);
   m.def("tree_hist", &tree_hist, py::arg("binned"), py::arg("node_of"),

@@ -130,7 +130,12 @@ void launch_adam(void* master, const void* grad, void* m1, void* m2, void* mirro
 //    per block (guide G12). The naive column-per-thread form left 224/256
 //    lanes idle at N=32 and was 30% of the MNIST step.
 //  * N > 256: column-per-thread over grid.y row chunks (coalesced across N).
+// mask: optional u8 array, same shape as dy; value 255 excludes the
+// element (the maxpool ReLU-sentinel) — colsum(pool-level grad, idx mask)
+// equals colsum of the scattered full-resolution grad, at a fraction of
+// the read traffic (each pool grad value scatters at most once).
 __global__ void colsum_small_kernel(const bf16* __restrict__ dy,
+                                    const unsigned char* __restrict__ mask,
                                     float* __restrict__ out,
                                     long M, int N, long ldy) {
   extern __shared__ float lacc[];  // N floats
@@ -145,8 +150,16 @@ __global__ void colsum_small_kernel(const bf16* __restrict__ dy,
     const long rStride = (long)gridDim.x * rowsPerBlock;
     for (long r = (long)blockIdx.x * rowsPerBlock + tr; r < M; r += rStride) {
       bf16x8 v = *(const bf16x8*)(dy + r * ldy + tc * 8);
-      #pragma unroll
-      for (int j = 0; j < 8; ++j) acc[j] += tofloat(v[j]);
+      if (mask) {
+        const unsigned long long iv =
+            *(const unsigned long long*)(mask + r * ldy + tc * 8);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (((iv >> (j * 8)) & 0xffu) != 0xffu) acc[j] += tofloat(v[j]);
+      } else {
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) acc[j] += tofloat(v[j]);
+      }
     }
   }
   #pragma unroll
@@ -167,14 +180,23 @@ __global__ void colsum_wide_kernel(const bf16* __restrict__ dy, float* __restric
   else atomicAdd(out + col, acc);
 }
 
+void launch_colsum_masked(const void* dy, const void* mask, void* out, long M,
+                          int N, long ldy, hipStream_t s);
+
 void launch_colsum(const void* dy, void* out, long M, int N, long ldy, hipStream_t s) {
+  launch_colsum_masked(dy, nullptr, out, M, N, ldy, s);
+}
+
+void launch_colsum_masked(const void* dy, const void* mask, void* out, long M,
+                          int N, long ldy, hipStream_t s) {
   const int block = 256;
   if (N <= 256 && N % 8 == 0) {
     const int rowsPerBlock = block / (N / 8);
     const int grid = (int)min((M + rowsPerBlock - 1) / rowsPerBlock, (long)1024);
     hipMemsetAsync(out, 0, sizeof(float) * N, s);
     hipLaunchKernelGGL(colsum_small_kernel, dim3(grid), dim3(block),
-                       N * sizeof(float), s, (const bf16*)dy, (float*)out, M, N, ldy);
+                       N * sizeof(float), s, (const bf16*)dy,
+                       (const unsigned char*)mask, (float*)out, M, N, ldy);
     return;
   }
   const int gx = (N + block - 1) / block;

@@ -218,11 +218,12 @@ class Conv2dNHWC(Layer):
         multi-branch grad sum (TextCNN).  Callers must check the returned
         tensor: on fallback paths it is the layer's own dx buffer.
 
-        ``bias_grad_src``: a small [rows, out_c] matrix whose column sums
-        EQUAL colsum(dy) — e.g. the relu-masked pool-level grad when a
-        global max pool follows this conv (the pool scatters each value at
-        most once, so the sums agree); the bias grad then reads KBs instead
-        of re-streaming the full dy (266 MB per TextCNN conv)."""
+        ``bias_grad_src``: either a small [rows, out_c] matrix or a
+        (matrix, u8-mask) pair whose (masked) column sums EQUAL colsum(dy)
+        — e.g. the pool-level grad with the ReLU-sentinel idx mask when a
+        max pool follows this conv (the pool scatters each value at most
+        once, so the sums agree); the bias grad then reads the pool-level
+        tensor instead of re-streaming the full dy."""
         bufs = self._bufs
         M = bufs["y"].shape[0]
         dy2 = dy.reshape(M, self.out_c)
@@ -246,8 +247,13 @@ class Conv2dNHWC(Layer):
         else:
             F.gemm(dy2, bufs["col"], ta=True, out=gw, splits=splits)
         if self.bias:
-            F.colsum(bias_grad_src if bias_grad_src is not None else dy2,
-                     out=self.arena.g(self.name + ".b"))
+            if isinstance(bias_grad_src, tuple):
+                F.colsum(bias_grad_src[0], mask=bias_grad_src[1],
+                         out=self.arena.g(self.name + ".b"))
+            elif bias_grad_src is not None:
+                F.colsum(bias_grad_src, out=self.arena.g(self.name + ".b"))
+            else:
+                F.colsum(dy2, out=self.arena.g(self.name + ".b"))
         if self.first:
             return None
         if self._is_1x1:
@@ -314,6 +320,7 @@ class MaxPool2dNHWC(Layer):
 
     def backward(self, dy: torch.Tensor) -> torch.Tensor:
         B, H, W, C = self._bufs["shape"]
+        self._last_dy = dy                    # pool-level grad (bias source)
         return F.maxpool2d_bwd(dy, self._bufs["idx"], H, W, self.kh, self.kw,
                                self.sh, self.sw, self.ph, self.pw,
                                out=self._bufs["dx"], relu_y=None)
@@ -406,6 +413,7 @@ class SequentialClassifier:
                     and isinstance(nxt, MaxPool2dNHWC)):
                 nxt.fuse_relu = True
                 prev.relu_bwd_upstream = True
+                prev._fused_pool = nxt  # bias grad from the pool-level grad
         self.arena = ParamArena(device)
         for lay in layers:
             lay.build(self.arena)
@@ -443,7 +451,19 @@ class SequentialClassifier:
                      cvalid=self.num_classes, gscale=gscale)
         dy = self._dlogits
         for lay in reversed(self.layers):
-            dy = lay.backward(dy)
+            if (isinstance(lay, Conv2dNHWC)
+                    and getattr(lay, "_fused_pool", None) is not None
+                    and lay.bias and dy is not None):
+                # colsum(dconv) == colsum(pool-level dy masked by the
+                # ReLU-sentinel idx): sum the SMALL tensor (the non-overlap
+                # pool scatters each value at most once)
+                pool = lay._fused_pool
+                pdy, pidx = pool._last_dy, pool._bufs["idx"]
+                dy = lay.backward(dy, bias_grad_src=(
+                    pdy.reshape(-1, lay.out_c),
+                    pidx.reshape(-1, lay.out_c)))
+            else:
+                dy = lay.backward(dy)
             if grad_hook and lay.param_names():
                 grad_hook(lay.param_names())
             if dy is None:

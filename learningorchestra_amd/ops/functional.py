@@ -328,14 +328,22 @@ def adam_step(master: torch.Tensor, grad: torch.Tensor, m1: torch.Tensor,
 
 
 # --------------------------------------------------------------- reductions
-def colsum(dy: torch.Tensor, out: Optional[torch.Tensor] = None) -> torch.Tensor:
+def colsum(dy: torch.Tensor, out: Optional[torch.Tensor] = None,
+           mask: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """``mask``: optional u8 array (same shape); elements with value 255 are
+    EXCLUDED (the maxpool ReLU-sentinel) — lets bias grads sum the small
+    pool-level grad instead of the scattered full-resolution one."""
     if out is None:
         out = torch.empty(dy.shape[1], device=dy.device, dtype=torch.float32)
     if _is_gpu(dy):
         lo = require_ext()
-        lo.colsum(dy, out)
+        lo.colsum(dy, out, mask)
         return out
-    out.copy_(dy.float().sum(0))
+    d = dy.float()
+    if mask is not None:
+        d = torch.where(torch.as_tensor(mask).view(dy.shape) != 255, d,
+                        torch.zeros((), dtype=d.dtype))
+    out.copy_(d.sum(0))
     return out
 
 

@@ -115,7 +115,10 @@ def test_training_reduces_loss():
     a few SGD steps must cut the loss substantially."""
     torch.manual_seed(0)
     model = build_mnist_cnn("cpu", seed=1, channels=(8, 8), fc_width=32)
-    opt = make_sgd(model, lr=0.1, momentum=0.9)
+    # lr 0.05 + a 5-batch trailing average: the old lr=0.1 single-batch
+    # check sat on a chaotic oscillation and flipped on last-bit summation-
+    # order changes (e.g. the masked bias-colsum reorders fp adds)
+    opt = make_sgd(model, lr=0.05, momentum=0.9)
     trainer = Trainer(model, opt, device="cpu")
 
     def batch():
@@ -126,9 +129,11 @@ def test_training_reduces_loss():
 
     x0, y0 = batch()
     first, _ = trainer.step(x0, y0)
-    for _ in range(80):
+    for _ in range(150):
         trainer.step_async(*batch())
-    last, acc = trainer.step(*batch())
+    tail = [trainer.step(*batch()) for _ in range(5)]
+    last = sum(l for l, _ in tail) / 5
+    acc = sum(a for _, a in tail) / 5
     assert last < first * 0.75, (first, last)
     assert acc > 0.2
 
