@@ -10,7 +10,9 @@ from typing import Callable, Optional, Tuple
 
 import torch
 
-from ..parallel import all_reduce_grads, get_world_size, is_distributed
+from ..parallel import (all_reduce_grads, get_world_size, is_distributed,
+                        oversubscribed)
+from ..parallel.ddp import device_step_lock
 from .arena import SGD, Adam
 from .layers import SequentialClassifier
 
@@ -29,18 +31,55 @@ class Trainer:
         self.model = model
         self.opt = optimizer
         self.device = torch.device(device)
-        self.use_graph = use_graph and self.device.type == "cuda"
+        # graphs are disabled when ranks time-share one GPU: graph-dispatched
+        # kernels corrupt under mid-kernel preemption by the peer process
+        # (measured on MI355X, tools/nanverify.py — eager 0/30 bad iterations
+        # vs graph 28/30 under identical concurrency); production one-rank-
+        # per-GPU runs keep split-graph mode
+        self.use_graph = (use_graph and self.device.type == "cuda"
+                          and not oversubscribed())
         self._graph = None          # world==1: whole step; world>1: fwd+bwd
         self._graph_opt = None      # world>1: optimizer + mirror refresh
         self._split = False         # world>1 split-graph mode
         self._static_x = None
         self._static_y = None
+        # oversubscribed runs (more ranks than GPUs) serialize device work
+        # across ranks: mid-kernel preemption while the peer's kernels/copy
+        # tails are in flight corrupts long-running MFMA/LDS wavefront state
+        # (measured; see parallel.ddp.device_step_lock). No-op otherwise.
+        self._devlock = None
+
+    def _lock(self):
+        if self._devlock is None and self.device.type == "cuda" \
+                and oversubscribed():
+            self._devlock = device_step_lock()
+        if self._devlock is None:
+            class _Null:
+                def __enter__(self): return self
+                def __exit__(self, *a): return False
+            return _Null()
+        return self._devlock
 
     # full step on given tensors (eager)
     def _step_body(self, x: torch.Tensor, y: torch.Tensor) -> None:
         world = get_world_size()
         gscale = 1.0 / (x.shape[0] * world)
         arena = self.model.arena
+        if is_distributed() and oversubscribed():
+            # ranks time-share one GPU: serialize compute across ranks and
+            # reduce the whole arena once between the locked sections (the
+            # lock must not span the collective — the peer needs the device
+            # to reach its own all-reduce). Per-layer overlap buys nothing
+            # when staging through the host anyway.
+            with self._lock():
+                self.model.train_step(x, y, gscale=gscale)
+            all_reduce_grads(arena.grad)
+            with self._lock():
+                self.opt.step()
+                post = getattr(self.model, "post_opt_step", None)
+                if post is not None:
+                    post()
+            return
         if is_distributed():
             # per-layer async all-reduce overlapped with the backward walk;
             # any params not covered by hooks reduce in one trailing op
@@ -99,16 +138,23 @@ class Trainer:
                     self._step_body(x, y)
                 return
             else:
-                self._static_x.copy_(x, non_blocking=True)
-                self._static_y.copy_(y, non_blocking=True)
-                self._graph.replay()
                 if self._split:
                     # collectives live OUTSIDE the graphs: one flat SUM
                     # all-reduce of the whole grad arena between the fwd+bwd
                     # graph and the optimizer graph (r1 VERDICT weak #4 — the
-                    # world>1 step must not be launch-bound)
+                    # world>1 step must not be launch-bound). Oversubscribed
+                    # ranks serialize the replays (lock is a no-op otherwise)
+                    with self._lock():
+                        self._static_x.copy_(x, non_blocking=True)
+                        self._static_y.copy_(y, non_blocking=True)
+                        self._graph.replay()
                     all_reduce_grads(self.model.arena.grad)
-                    self._graph_opt.replay()
+                    with self._lock():
+                        self._graph_opt.replay()
+                else:
+                    self._static_x.copy_(x, non_blocking=True)
+                    self._static_y.copy_(y, non_blocking=True)
+                    self._graph.replay()
             return
         self._step_body(x, y)
 
@@ -128,10 +174,12 @@ class Trainer:
         gscale = 1.0 / (x.shape[0] * world)
         if self._split:
             for _ in range(3):
-                self.model.train_step(self._static_x, self._static_y,
-                                      gscale=gscale)
+                with self._lock():
+                    self.model.train_step(self._static_x, self._static_y,
+                                          gscale=gscale)
                 all_reduce_grads(self.model.arena.grad)
-                self._opt_body()
+                with self._lock():
+                    self._opt_body()
             torch.cuda.synchronize()
             # thread_local capture: the RCCL/NCCL watchdog thread polls
             # events concurrently; global capture mode would invalidate the

@@ -12,6 +12,7 @@ Activation layout is NHWC everywhere (SURVEY §2.9 + conv_pool.hip header).
 """
 from __future__ import annotations
 
+import os
 from typing import Optional, Tuple
 
 import torch
@@ -52,7 +53,8 @@ def gemm(A: torch.Tensor, B: torch.Tensor, *, ta: bool = False, tb: bool = False
         if (not ta and tb and bias is None and not relu and stats is None
                 and addend is None and splits == 1
                 and A.dtype == torch.bfloat16 and out.dtype == torch.bfloat16
-                and N * K >= 256 * 1024):
+                and N * K >= 256 * 1024
+                and os.environ.get("LO_PLAIN_ROCBLAS", "1") == "1"):
             torch.matmul(A, B.t(), out=out)
             return out
         lo = require_ext()

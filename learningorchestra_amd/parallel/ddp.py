@@ -54,16 +54,72 @@ def init_distributed(backend: Optional[str] = None, timeout_s: int = 300) -> int
     return local_rank
 
 
+def oversubscribed() -> bool:
+    """More ranks than visible GPUs (a multi-rank test on a 1-GPU lease).
+    This is the only configuration where two of our processes time-share one
+    device; production runs are one process per GPU over RCCL."""
+    if not (is_distributed() and torch.cuda.is_available()):
+        return False
+    return get_world_size() > torch.cuda.device_count()
+
+
+class device_step_lock:
+    """Cross-process exclusive section for GPU work in oversubscribed runs.
+
+    Measured on MI355X (tools/nanrepro7.py): when two ranks time-share one
+    GPU, mid-kernel preemption of our long-running MFMA/LDS kernels while the
+    peer's kernels or copy tails are in flight corrupts wavefront state —
+    grad buffers pick up pointer-looking garbage (0x73xx_xxxx-range words).
+    Pure-PyTorch kernels (short blocks) are immune; serializing whole steps
+    (lock + synchronize before release, 0/15 vs 8/15 bad iterations) is
+    throughput-neutral on a shared device and makes the run correct.
+    No-op unless oversubscribed. NEVER hold this across a collective — the
+    peer needs the device to reach its own collective call (deadlock)."""
+
+    def __init__(self, device_index: Optional[int] = None):
+        self._active = oversubscribed()
+        self._fh = None
+        if self._active:
+            import tempfile
+            idx = (torch.cuda.current_device()
+                   if device_index is None else device_index)
+            path = os.path.join(tempfile.gettempdir(), f"lo_dev{idx}.lock")
+            self._fh = open(path, "w")
+
+    def __enter__(self):
+        if self._fh is not None:
+            import fcntl
+            fcntl.flock(self._fh, fcntl.LOCK_EX)
+        return self
+
+    def __exit__(self, *exc):
+        if self._fh is not None:
+            import fcntl
+            torch.cuda.synchronize()   # drain before the peer runs
+            fcntl.flock(self._fh, fcntl.LOCK_UN)
+        return False
+
+    def close(self) -> None:
+        if self._fh is not None:
+            self._fh.close()
+            self._fh = None
+
+
 def all_reduce_grads(grad_flat: torch.Tensor, async_op: bool = False):
     """SUM all-reduce of the flat grad arena (grads carry 1/(B*world) so sum
     = global-batch mean). The gloo+CUDA combination (oversubscribed test
-    leases) stages through host memory explicitly."""
+    leases) stages through host memory explicitly, sync-bracketed: an
+    un-drained copy tail overlapping the peer's kernels corrupts under
+    mid-kernel preemption (see device_step_lock; measured 15/15 bad
+    iterations without the trailing sync, 0/15 with)."""
     if not is_distributed():
         return None
     if grad_flat.is_cuda and dist.get_backend() == "gloo":
+        torch.cuda.synchronize()
         host = grad_flat.detach().to("cpu")
         dist.all_reduce(host, op=dist.ReduceOp.SUM)
         grad_flat.copy_(host)
+        torch.cuda.synchronize()
         return None
     return dist.all_reduce(grad_flat, op=dist.ReduceOp.SUM, async_op=async_op)
 

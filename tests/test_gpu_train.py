@@ -310,8 +310,19 @@ def test_ddp_two_ranks_gpu(tmp_path):
         for _ in range(8):
             tr.step_async(x, y)
         torch.cuda.synchronize()
-        assert tr._split and tr._graph is not None \
-            and tr._graph_opt is not None
+        from learningorchestra_amd.parallel import oversubscribed
+        if oversubscribed():
+            # ranks share one GPU: graphs are disabled (graph-dispatched
+            # kernels corrupt under mid-kernel preemption by the peer;
+            # measured, see trainer.py) and steps serialize via the device
+            # lock — the eager distributed path must have run
+            assert not tr.use_graph and tr._graph is None
+        else:
+            assert tr._split and tr._graph is not None \
+                and tr._graph_opt is not None
+        # either way grads must be finite and sane after 8 steps
+        assert bool(torch.isfinite(m.arena.grad).all())
+        assert float(m.arena.grad.abs().max()) < 1e3
         # ranks hold identical params after all-reduced training
         p = m.arena.master.detach().cpu().clone()
         ref = p.clone()
