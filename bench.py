@@ -90,6 +90,9 @@ def main() -> None:
     use_graph = use_gpu and not args.no_graph
     trainer = Trainer(model, make_sgd(model, lr=args.lr), device=device,
                       use_graph=use_graph)
+    # the trainer disables graphs when ranks time-share one GPU (preemption
+    # corruption, see engine/trainer.py) — report the EFFECTIVE state
+    use_graph = trainer.use_graph
 
     if args.model == "mnist-cnn":
         x, y = mnist_batch(batch, device=device, dtype=torch.bfloat16,
