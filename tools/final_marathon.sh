@@ -3,7 +3,7 @@
 set -u
 cd /root/repo
 mkdir -p gpurun_out
-R=gpurun_out/final
+R=gpurun_out/final3
 mkdir -p $R
 echo "=== freshness ===" | tee $R/summary.log
 SO=$(ls learningorchestra_amd/_build/*.so 2>/dev/null | head -1)
@@ -40,7 +40,7 @@ timeout 240 python bench.py --gpus 1 --steps 40 --warmup 10 --model textcnn > $R
 tail -1 $R/bench_textcnn.json | tee -a $R/summary.log
 
 echo "=== bench: resnet ===" | tee -a $R/summary.log
-timeout 300 python bench.py --gpus 1 --steps 20 --warmup 5 --model resnet > $R/bench_resnet.json 2>$R/bench_resnet.err
+timeout 300 python bench.py --gpus 1 --steps 20 --warmup 5 --model resnet50 > $R/bench_resnet.json 2>$R/bench_resnet.err
 tail -1 $R/bench_resnet.json | tee -a $R/summary.log
 
 echo "=== bench: gbt ===" | tee -a $R/summary.log
