@@ -73,9 +73,11 @@ class _DeviceSlot:
 
 class Job:
     def __init__(self, name: str, device: Optional[str] = None,
-                 timeout: Optional[float] = None):
+                 timeout: Optional[float] = None,
+                 devices: Optional[List[str]] = None):
         self.name = name
         self.device = device
+        self.devices = devices  # multi-GPU jobs own every device they span
         self.timeout = timeout
         self.future: Optional[Future] = None
         self.cancel_event = threading.Event()
@@ -178,17 +180,25 @@ class JobScheduler:
 
         def run():
             job.started_at = time.time()
-            slot = self._device_slot(device) if device is not None else None
-            acquired = False
+            # a job may claim several devices (an N-rank torchrun train owns
+            # every GPU it spans): acquire in sorted order so two multi-
+            # device jobs can never deadlock. Cross-process GPU sharing is
+            # not allowed — two processes time-slicing one device corrupts
+            # long-running kernels under mid-kernel preemption (measured,
+            # parallel/ddp.py device_step_lock).
+            devices = sorted(getattr(job, "devices", None)
+                             or ([device] if device is not None else []))
+            slots = [self._device_slot(d) for d in devices]
+            held: List[_DeviceSlot] = []
             try:
                 if job.cancel_event.is_set():
                     job.outcome = job.outcome or "cancelled"
                     raise JobCancelled(name)
-                if slot is not None:
-                    acquired = slot.acquire(name, job.cancel_event)
-                    if not acquired:
+                for slot in slots:
+                    if not slot.acquire(name, job.cancel_event):
                         job.outcome = job.outcome or "cancelled"
                         raise JobCancelled(name)
+                    held.append(slot)
                 result = fn(*args, **kwargs)
                 job.outcome = job.outcome or "ok"
                 return result
@@ -213,7 +223,7 @@ class JobScheduler:
                 raise
             finally:
                 job.finished_at = time.time()
-                if slot is not None and acquired:
+                for slot in held:
                     slot.release(name)
 
         with self._lock:
@@ -233,6 +243,7 @@ class JobScheduler:
                        env: Optional[Dict[str, str]] = None,
                        timeout: Optional[float] = None,
                        device: Optional[str] = None,
+                       devices: Optional[List[str]] = None,
                        cwd: Optional[str] = None,
                        on_done: Optional[Callable[[int, str], None]] = None
                        ) -> Job:
@@ -240,7 +251,7 @@ class JobScheduler:
         group) — the multi-rank torchrun jobs go through here. ``on_done``
         gets (returncode, combined-output-tail); non-zero exit raises so the
         standard exception-to-metadata path records it."""
-        job = Job(name, device=device, timeout=timeout)
+        job = Job(name, device=device, timeout=timeout, devices=devices)
 
         def run_proc():
             full_env = dict(os.environ)
@@ -292,8 +303,9 @@ class JobScheduler:
         job.cancel_event.set()
         if job.proc is not None and job.proc.poll() is None:
             _kill_group(job.proc)
-            if job.device is not None:
-                self._device_slot(job.device).revoke()
+            for d in (job.devices or
+                      ([job.device] if job.device is not None else [])):
+                self._device_slot(d).revoke()
         if self._metadata is not None:
             try:
                 self._metadata.update_finished_flag(

@@ -108,5 +108,14 @@ def launch_distributed_train(scheduler, metadata, cfg, *, name: str,
             trainResult=result.get("result"))
         metadata.update_finished_flag(name, True)
 
+    # claim every device the rank fan-out will touch: another process job
+    # time-slicing one of these GPUs mid-train corrupts long-running kernels
+    # under mid-kernel preemption (parallel/ddp.py device_step_lock)
+    try:
+        import torch
+        n_dev = torch.cuda.device_count() if torch.cuda.is_available() else 0
+    except Exception:
+        n_dev = 0
+    devices = [f"cuda:{i}" for i in range(min(gpus, n_dev))] if n_dev else None
     return scheduler.submit_process(name, cmd, env=env, timeout=timeout,
-                                    on_done=on_done)
+                                    devices=devices, on_done=on_done)

@@ -178,3 +178,36 @@ def test_scheduler_process_job_success_output():
         on_done=lambda rc, tail: outs.append((rc, tail)))
     job.wait(timeout=60)
     assert outs and outs[0][0] == 0 and "out42" in outs[0][1]
+
+
+def test_scheduler_multi_device_job_excludes_single_device_jobs():
+    """An N-rank train claims EVERY device it spans (scheduler `devices=`):
+    a single-device job on any spanned GPU must wait until the multi-device
+    job finishes — two processes time-slicing one GPU corrupts long-running
+    kernels under mid-kernel preemption (parallel/ddp.py device_step_lock)."""
+    import sys
+    import threading
+
+    from learningorchestra_amd.executor.scheduler import JobScheduler
+    sched = JobScheduler(max_workers=4,
+                         devices=["cuda:0", "cuda:1"])
+    order = []
+    olock = threading.Lock()
+
+    big = sched.submit_process(
+        "ddp-train", [sys.executable, "-c", "import time; time.sleep(1.0)"],
+        devices=["cuda:0", "cuda:1"],
+        on_done=lambda rc, tail: order.append("big-done"))
+    time.sleep(0.3)   # let it start and take both slots
+
+    def small():
+        with olock:
+            order.append("small-ran")
+
+    j = sched.submit("single", small, device="cuda:1")
+    time.sleep(0.2)
+    # the single-device job must still be blocked on the claimed slot
+    assert "small-ran" not in order
+    big.wait(timeout=30)
+    j.wait(timeout=30)
+    assert order.index("big-done") < order.index("small-ran")
