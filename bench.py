@@ -38,9 +38,11 @@ def main() -> None:
     ap.add_argument("--rows", type=int, default=10_000_000,
                     help="gbt: total rows across all ranks (BASELINE cfg 4)")
     ap.add_argument("--no-graph", action="store_true")
-    # 0.02 keeps the fixed synthetic batch stable for 300+ step soaks;
-    # 0.05 plateaus by ~70 iterations and can diverge past ~200 (bf16)
-    ap.add_argument("--lr", type=float, default=0.02)
+    # 0.01: stable for 300+ step soaks on the fixed synthetic batch. 0.05
+    # diverges past ~200 iterations and 0.02 is borderline (atomics make
+    # the trajectory nondeterministic run to run; observed one divergence
+    # in three 300-step soaks). Throughput is lr-independent.
+    ap.add_argument("--lr", type=float, default=0.01)
     args = ap.parse_args()
 
     from learningorchestra_amd.parallel import (barrier, get_rank,

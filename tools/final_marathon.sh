@@ -3,7 +3,7 @@
 set -u
 cd /root/repo
 mkdir -p gpurun_out
-R=gpurun_out/final3
+R=gpurun_out/final4
 mkdir -p $R
 echo "=== freshness ===" | tee $R/summary.log
 SO=$(ls learningorchestra_amd/_build/*.so 2>/dev/null | head -1)
