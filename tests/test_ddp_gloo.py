@@ -179,3 +179,10 @@ def test_bench_torchrun_world4_cpu(tmp_path):
     assert doc["config"]["parallelism"] == "dp4"
     assert doc["config"]["global_batch"] == 4 * 64
     assert doc["value"] > 0
+    # full driver contract: every required key present, strict-JSON parseable
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in doc, key
+    json.loads(line, parse_constant=lambda c: (_ for _ in ()).throw(
+        ValueError(f"non-strict JSON constant {c} in bench line")))
