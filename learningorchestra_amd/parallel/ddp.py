@@ -79,24 +79,28 @@ class device_step_lock:
     def __init__(self, device_index: Optional[int] = None):
         self._active = oversubscribed()
         self._fh = None
+        self._path = None
         if self._active:
             import tempfile
             idx = (torch.cuda.current_device()
                    if device_index is None else device_index)
-            path = os.path.join(tempfile.gettempdir(), f"lo_dev{idx}.lock")
-            self._fh = open(path, "w")
+            self._path = os.path.join(tempfile.gettempdir(),
+                                      f"lo_dev{idx}.lock")
 
     def __enter__(self):
-        if self._fh is not None:
+        if self._active:
             import fcntl
+            self._fh = open(self._path, "w")
             fcntl.flock(self._fh, fcntl.LOCK_EX)
         return self
 
     def __exit__(self, *exc):
         if self._fh is not None:
-            import fcntl
-            torch.cuda.synchronize()   # drain before the peer runs
-            fcntl.flock(self._fh, fcntl.LOCK_UN)
+            try:
+                torch.cuda.synchronize()   # drain before the peer runs
+            finally:
+                self._fh.close()           # closing the fd releases the flock
+                self._fh = None
         return False
 
     def close(self) -> None:
