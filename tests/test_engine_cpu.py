@@ -253,3 +253,31 @@ def test_conv_tuple_pad_matches_autograd():
     torch.testing.assert_close(dx.float(),
                                xr.grad.permute(0, 2, 3, 1).float(),
                                atol=5e-2, rtol=5e-2)
+
+
+def test_graph_capture_failure_falls_back_to_eager(monkeypatch):
+    """The world>1 safety net for environments where hipGraph capture fails
+    (e.g. a driver/runtime combination on the 8-GPU node): step_async must
+    catch the capture error, permanently drop to eager, and keep training."""
+    import math
+
+    import torch
+
+    from learningorchestra_amd.data.synthetic import mnist_batch
+    from learningorchestra_amd.engine.trainer import Trainer, make_sgd
+    from learningorchestra_amd.models.mnist_cnn import build_mnist_cnn
+
+    m = build_mnist_cnn("cpu", seed=0)
+    tr = Trainer(m, make_sgd(m, lr=0.05), device="cpu", use_graph=False)
+    tr.use_graph = True   # force the graph branch on a CPU host
+
+    def boom(x, y):
+        raise RuntimeError("hipGraph capture unavailable")
+
+    monkeypatch.setattr(tr, "_capture", boom)
+    monkeypatch.setattr(torch.cuda, "synchronize", lambda *a, **k: None)
+    x, y = mnist_batch(64, device="cpu", dtype=torch.bfloat16, seed=0)
+    tr.step_async(x, y)                  # capture fails -> eager step ran
+    assert tr.use_graph is False and tr._graph is None
+    tr.step_async(x, y)                  # stays eager
+    assert math.isfinite(float(m.loss_sum))
