@@ -478,3 +478,39 @@ def test_observe_many_concurrent_waiters(client):
                 "slowjob", True)).start()
         results = [f.result(timeout=60) for f in futs]
     assert all(results), results.count(False)
+
+
+def test_tune_torch_gridsearch_e2e(client):
+    """The Tune verb end-to-end on the native tool: a GridSearch artifact
+    created via model/torch, fitted via tune/torch, then queried for the
+    winning candidate (reference path: tune/* through the binary executor,
+    SURVEY §2.1; native scheduler in models/tuning.py)."""
+    r = client.post(f"{PREFIX}/model/torch",
+                    json={"modelName": "gs0",
+                          "modulePath": "learningorchestra_amd.models.tuning",
+                          "class": "GridSearch",
+                          "classParameters": {
+                              "modulePath":
+                                  "learningorchestra_amd.models.tabular",
+                              "className": "LogisticRegressionClassifier",
+                              "paramGrid": {"lr": [0.05, 0.2]},
+                              "fixedParameters": {"epochs": 40},
+                              "validationSplit": 0.25}})
+    assert r.status_code == 201
+    wait_finished(client, "gs0")
+
+    # separable 1-feature data: y = x > 1.5
+    xs = "#[[float(i % 4)] for i in range(80)]"
+    ys = "#[1 if (i % 4) > 1 else 0 for i in range(80)]"
+    r = client.post(f"{PREFIX}/tune/torch",
+                    json={"name": "gs_fit", "modelName": "gs0",
+                          "parentName": "gs0", "method": "fit",
+                          "methodParameters": {"x": xs, "y": ys}})
+    assert r.status_code == 201
+    doc = wait_finished(client, "gs_fit")
+    assert doc.get("exception") in (None, ""), doc
+
+    gs = client.rt.artifacts.load("gs_fit", "tune/torch")
+    assert gs.best_params_ is not None and "lr" in gs.best_params_
+    assert gs.best_score_ >= 0.5, (gs.best_score_, gs.results_)
+    assert len(gs.results_) == 2
