@@ -514,3 +514,36 @@ def test_tune_torch_gridsearch_e2e(client):
     assert gs.best_params_ is not None and "lr" in gs.best_params_
     assert gs.best_score_ >= 0.5, (gs.best_score_, gs.results_)
     assert len(gs.results_) == 2
+
+
+def test_patch_rerun_bumps_execution_documents(client):
+    """PATCH on a train result re-executes with new parameters and appends
+    a new execution document at the next _id (reference: binary_executor
+    update_execution, server.py:74-118; docs appended at max+1)."""
+    client.post(f"{PREFIX}/model/scikitlearn",
+                json={"modelName": "m_p", "modulePath": "sklearn.linear_model",
+                      "class": "LogisticRegression",
+                      "classParameters": {"max_iter": 100}})
+    wait_finished(client, "m_p")
+    client.post(f"{PREFIX}/train/scikitlearn",
+                json={"name": "t_p", "modelName": "m_p", "parentName": "m_p",
+                      "method": "fit",
+                      "methodParameters": {"X": "#[[0.],[1.],[2.],[3.]]",
+                                           "y": "#[0,0,1,1]"}})
+    wait_finished(client, "t_p")
+    docs_before = client.rt.db["t_p"].count_documents({})
+
+    r = client.patch(f"{PREFIX}/train/scikitlearn/t_p",
+                     json={"methodParameters": {"X": "#[[0.],[1.],[2.],[5.]]",
+                                                "y": "#[0,0,1,1]"},
+                           "description": "re-fit with shifted data"})
+    assert r.status_code == 200
+    doc = wait_finished(client, "t_p")
+    assert doc.get("exception") in (None, "")
+    docs_after = client.rt.db["t_p"].count_documents({})
+    assert docs_after == docs_before + 1
+    # the newest execution doc carries the new description
+    newest = max(client.rt.db["t_p"].find({"_id": {"$gt": 0}}),
+                 key=lambda d: d["_id"])
+    assert "re-fit" in str(newest.get("description", "")) or \
+        "re-fit" in str(newest)
