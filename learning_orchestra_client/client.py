@@ -243,3 +243,31 @@ class Observe(_VerbClient):
                 return doc
             if time.time() >= deadline:
                 raise TimeoutError(f"'{name}' not finished after {timeout}s")
+
+    def watch(self, name: str, verb: str = "train", tool: str = "torch",
+              poll: float = 0.25, timeout: float = 300.0):
+        """Yield result-collection documents as they are written (the
+        reference pip client watched the result collection with Mongo
+        change streams); generator ends when the pipeline's finished flag
+        is set. ``verb``/``tool`` name the collection's poll URI."""
+        deadline = time.time() + timeout
+        seen = 1                       # _id 0 is the metadata document
+        path = f"{PREFIX}/{verb}/{tool}/{name}"
+        while True:
+            try:
+                rows = self._ctx.get(path, {
+                    "query": json.dumps({"_id": {"$gte": seen}}),
+                    "limit": 100, "skip": 0}) or []
+            except LearningOrchestraError:
+                rows = []              # collection not created yet
+            for doc in sorted((d for d in rows
+                               if isinstance(d.get("_id"), int)),
+                              key=lambda d: d["_id"]):
+                seen = doc["_id"] + 1
+                yield doc
+            meta = self.observe(name)
+            if meta and meta.get("finished"):
+                return
+            if time.time() >= deadline:
+                raise TimeoutError(f"'{name}' not finished after {timeout}s")
+            time.sleep(poll)

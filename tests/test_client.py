@@ -102,3 +102,22 @@ def test_titanic_example_runs(tmp_path):
                          text=True, timeout=300)
     assert out.returncode == 0, out.stderr[-2000:]
     assert "accuracy=" in out.stdout
+
+
+def test_observe_watch_streams_result_rows(ctx, tmp_path):
+    """Observe.watch yields result-collection rows as they appear and stops
+    at the finished flag (reference pip client: Mongo change-stream watch on
+    the result collection)."""
+    from learning_orchestra_client import Observe
+
+    p = tmp_path / "titanic.csv"
+    p.write_text(titanic_csv(rows=50))
+    DatasetCsv(ctx).insert_sync("t_watch", str(p))
+
+    TransformProjection(ctx).create("t_watch", "w_feat", ["Pclass", "Fare"])
+    rows = list(Observe(ctx).watch("w_feat", verb="transform",
+                                   tool="projection", timeout=60.0))
+    assert len(rows) == 50
+    assert all(set(r) >= {"Pclass", "Fare"} for r in rows)
+    ids = [r["_id"] for r in rows]
+    assert ids == sorted(ids) and ids[0] >= 1
