@@ -513,6 +513,9 @@ class DocumentStore:
         self._replaying = False
         self._wal_fh = None
         self._wal_fsync = wal_fsync or os.environ.get("LO_WAL_FSYNC") == "1"
+        self._wal_ops = 0
+        self._wal_autoflush = int(
+            os.environ.get("LO_WAL_AUTOFLUSH", "20000"))
         if root:
             os.makedirs(os.path.join(root, "collections"), exist_ok=True)
             for fn in os.listdir(os.path.join(root, "collections")):
@@ -541,6 +544,15 @@ class DocumentStore:
             self._wal_fh.flush()  # to the fd: survives kill -9
             if self._wal_fsync:
                 os.fsync(self._wal_fh.fileno())  # survives power loss too
+            # long-running servers: checkpoint periodically so the WAL (and
+            # replay time at next open) stays bounded. The mutation just
+            # logged is already applied in-memory, _mutlock is re-entrant,
+            # and flush snapshots + truncates under the same lock — safe to
+            # call from here.
+            self._wal_ops += 1
+            if self._wal_ops >= self._wal_autoflush:
+                self._wal_ops = 0
+                self.flush()
 
     def _replay_wal(self) -> None:
         """Re-apply mutations logged since the last snapshot (crash

@@ -211,3 +211,26 @@ def test_scheduler_multi_device_job_excludes_single_device_jobs():
     big.wait(timeout=30)
     j.wait(timeout=30)
     assert order.index("big-done") < order.index("small-ran")
+
+
+def test_wal_autoflush_bounds_wal(tmp_path, monkeypatch):
+    """A long-running server must checkpoint periodically: after the
+    LO_WAL_AUTOFLUSH-th logged mutation the store snapshots and truncates
+    the WAL, and the data survives a reopen."""
+    import os
+
+    monkeypatch.setenv("LO_WAL_AUTOFLUSH", "50")
+    from learningorchestra_amd.storage.docstore import DocumentStore
+
+    store = DocumentStore(str(tmp_path))
+    col = store["events"]
+    for i in range(120):
+        col.insert_one({"i": i})
+    wal = os.path.join(str(tmp_path), "wal.jsonl")
+    # 120 ops with threshold 50: two auto-flushes happened; the WAL holds
+    # only the tail since the last checkpoint (< threshold lines)
+    n_lines = sum(1 for _ in open(wal)) if os.path.exists(wal) else 0
+    assert n_lines < 50, n_lines
+    store.flush()
+    reopened = DocumentStore(str(tmp_path))
+    assert reopened["events"].count_documents({}) == 120
