@@ -547,3 +547,33 @@ def test_patch_rerun_bumps_execution_documents(client):
                  key=lambda d: d["_id"])
     assert "re-fit" in str(newest.get("description", "")) or \
         "re-fit" in str(newest)
+
+
+def test_train_torch_multi_rank_failure_records_exception(client):
+    """A failing N-rank torchrun job must flip finished=true WITH the
+    exception recorded (errors are data — reference parity), not hang the
+    poll contract."""
+    client.post(f"{PREFIX}/model/torch",
+                json={"modelName": "ddpbad",
+                      "modulePath": "learningorchestra_amd.models.zoo",
+                      "class": "MnistCNN",
+                      "classParameters": {"channels": [8, 8],
+                                          "fc_width": 32, "device": "cpu"}})
+    wait_finished(client, "ddpbad")
+    r = client.post(f"{PREFIX}/train/torch",
+                    json={"name": "ddpfail", "modelName": "ddpbad",
+                          "method": "fit",
+                          "methodParameters": {
+                              "gpus": 2,
+                              # wrong feature width: every rank raises
+                              "x": "#numpy.random.RandomState(0)"
+                                   ".rand(16,7).astype('float32')",
+                              "y": "#numpy.random.RandomState(1)"
+                                   ".randint(0,10,16)",
+                              "epochs": 1, "batch_size": 8}})
+    assert r.status_code == 201
+    r = client.get(f"{PREFIX}/observe/ddpfail/wait",
+                   params={"timeoutSeconds": 240})
+    doc = r.json()["result"]
+    assert doc is not None and doc.get("finished"), doc
+    assert doc.get("exception"), doc
