@@ -185,6 +185,20 @@ def create_app(runtime: Optional[Runtime] = None) -> FastAPI:
             body.get("description", ""))
         return {RESULT: _poll_uri("model", tool, name)}
 
+    # ----------------------------------------------------------------- cancel --
+    # registered BEFORE the generic POST /{verb}/{tool} route: FastAPI
+    # matches in registration order, and "cancel/{name}" would otherwise be
+    # swallowed as verb="cancel" (caught by the REST cancel test)
+    @app.post(PREFIX + "/cancel/{name}")
+    def cancel(name: str):
+        """Cancel a queued/running job (new vs the reference — its answer was
+        restarting the whole Swarm service). Process jobs (multi-GPU train)
+        are killed for real; thread jobs are cancelled cooperatively."""
+        rt.require_exists(name)
+        ok = rt.scheduler.cancel(name)
+        return {RESULT: f"cancelled {name}" if ok
+                else f"'{name}' has no active job"}
+
     # ------------------------------------------- train/tune/evaluate/predict --
     @app.post(PREFIX + "/{verb}/{tool}", status_code=201)
     def create_binary_execution(verb: str, tool: str, body: Dict[str, Any]):
@@ -281,16 +295,6 @@ def create_app(runtime: Optional[Runtime] = None) -> FastAPI:
         return JSONResponse({RESULT: doc, "timedOut": True}, status_code=200)
 
     # -------------------------------------------------------------- cancel --
-    @app.post(PREFIX + "/cancel/{name}")
-    def cancel(name: str):
-        """Cancel a queued/running job (new vs the reference — its answer was
-        restarting the whole Swarm service). Process jobs (multi-GPU train)
-        are killed for real; thread jobs are cancelled cooperatively."""
-        rt.require_exists(name)
-        ok = rt.scheduler.cancel(name)
-        return {RESULT: f"cancelled {name}" if ok
-                else f"'{name}' has no active job"}
-
     # ------------------------------------------------------------- catalog --
     @app.get(PREFIX + "/{verb}/{tool}")
     def catalog(verb: str, tool: str):

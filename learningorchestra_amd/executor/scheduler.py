@@ -146,6 +146,11 @@ class JobScheduler:
             return min(devs, key=lambda d: (loads[d],
                                             (devs.index(d) + rr) % len(devs)))
 
+    def job(self, name: str) -> Optional[Job]:
+        """Look up a submitted job by name (None if unknown/pruned)."""
+        with self._lock:
+            return self._jobs.get(name)
+
     def _device_slot(self, device: str) -> _DeviceSlot:
         with self._lock:
             if device not in self._device_slots:

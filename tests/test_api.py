@@ -577,3 +577,48 @@ def test_train_torch_multi_rank_failure_records_exception(client):
     doc = r.json()["result"]
     assert doc is not None and doc.get("finished"), doc
     assert doc.get("exception"), doc
+
+
+def test_cancel_running_multi_rank_train(client):
+    """POST /cancel/{name} kills a RUNNING N-rank train's process group and
+    records the cancellation (r2: real job control vs the reference's
+    restart-the-service answer)."""
+    client.post(f"{PREFIX}/model/torch",
+                json={"modelName": "ddpslow",
+                      "modulePath": "learningorchestra_amd.models.zoo",
+                      "class": "MnistCNN",
+                      "classParameters": {"channels": [8, 8],
+                                          "fc_width": 32, "device": "cpu"}})
+    wait_finished(client, "ddpslow")
+    r = client.post(f"{PREFIX}/train/torch",
+                    json={"name": "ddpcancel", "modelName": "ddpslow",
+                          "method": "fit",
+                          "methodParameters": {
+                              "gpus": 2,
+                              "x": "#numpy.random.RandomState(0)"
+                                   ".rand(4096,784).astype('float32')",
+                              "y": "#numpy.random.RandomState(1)"
+                                   ".randint(0,10,4096)",
+                              "epochs": 500, "batch_size": 64}})
+    assert r.status_code == 201
+    # let the torchrun tree actually start
+    deadline = time.time() + 60
+    while time.time() < deadline:
+        job = client.rt.scheduler.job("ddpcancel")
+        if job is not None and job.proc is not None \
+                and job.proc.poll() is None:
+            break
+        time.sleep(0.25)
+    else:
+        raise AssertionError("torchrun job never started")
+
+    r = client.post(f"{PREFIX}/cancel/ddpcancel")
+    assert r.status_code == 200
+    assert "cancelled" in r.json()["result"]
+    doc = wait_finished(client, "ddpcancel", timeout=60)
+    assert doc.get("exception"), doc
+    # the process tree is really gone
+    deadline = time.time() + 30
+    while time.time() < deadline and job.proc.poll() is None:
+        time.sleep(0.25)
+    assert job.proc.poll() is not None
