@@ -622,3 +622,74 @@ def test_cancel_running_multi_rank_train(client):
     while time.time() < deadline and job.proc.poll() is None:
         time.sleep(0.25)
     assert job.proc.poll() is not None
+
+
+def test_chaos_concurrent_trains_cancels_observers(client):
+    """Stress the scheduler + metadata + docstore stack: 12 concurrent
+    sklearn train jobs, observers long-polling each, random cancels racing
+    completion, catalog/metrics reads throughout. Every job must terminate
+    with finished=true (result or recorded exception) and the API must
+    never 5xx."""
+    import random
+    import threading
+
+    client.post(f"{PREFIX}/model/scikitlearn",
+                json={"modelName": "chaos_m",
+                      "modulePath": "sklearn.linear_model",
+                      "class": "LogisticRegression",
+                      "classParameters": {"max_iter": 50}})
+    wait_finished(client, "chaos_m")
+
+    names = [f"chaos_t{i}" for i in range(12)]
+    errors = []
+
+    def train(name):
+        try:
+            r = client.post(
+                f"{PREFIX}/train/scikitlearn",
+                json={"name": name, "modelName": "chaos_m",
+                      "parentName": "chaos_m", "method": "fit",
+                      "methodParameters": {
+                          "X": "#[[float(i)] for i in range(200)]",
+                          "y": "#[i % 2 for i in range(200)]"}})
+            assert r.status_code in (201, 409), r.status_code
+        except Exception as exc:  # noqa: BLE001
+            errors.append(("train", name, repr(exc)))
+
+    def observer(name):
+        try:
+            r = client.get(f"{PREFIX}/observe/{name}/wait",
+                           params={"timeoutSeconds": 60})
+            assert r.status_code == 200, r.status_code
+        except Exception as exc:  # noqa: BLE001
+            errors.append(("observe", name, repr(exc)))
+
+    def chaos_reader(stop):
+        rng = random.Random(7)
+        while not stop.is_set():
+            try:
+                assert client.get(f"{PREFIX}/metrics").status_code == 200
+                assert client.get(
+                    f"{PREFIX}/train/scikitlearn").status_code == 200
+                victim = rng.choice(names)
+                r = client.post(f"{PREFIX}/cancel/{victim}")
+                assert r.status_code in (200, 404), r.status_code
+            except Exception as exc:  # noqa: BLE001
+                errors.append(("reader", "-", repr(exc)))
+                return
+
+    threads = [threading.Thread(target=train, args=(n,)) for n in names]
+    threads += [threading.Thread(target=observer, args=(n,)) for n in names]
+    stop = threading.Event()
+    reader = threading.Thread(target=chaos_reader, args=(stop,))
+    for t in threads:
+        t.start()
+    reader.start()
+    for t in threads:
+        t.join(timeout=120)
+    stop.set()
+    reader.join(timeout=30)
+    assert not errors, errors[:5]
+    for n in names:
+        doc = client.rt.metadata.get_metadata(n)
+        assert doc is not None and doc.get("finished"), (n, doc)
