@@ -44,6 +44,18 @@ def build(verbose: bool = False):
         verbose=verbose,
         is_python_module=True,
     )
+    # stale-cache guard (bit us once: a ninja failure can leave an older
+    # cached .so importable while the sources have moved on): the linked
+    # .so must be at least as new as every source it claims to contain
+    so = os.path.join(BUILD_DIR, EXT_NAME + ".so")
+    if os.path.exists(so):
+        so_mt = os.path.getmtime(so)
+        stale = [os.path.basename(s) for s in SOURCES
+                 if os.path.getmtime(s) > so_mt + 1.0]
+        if stale:
+            raise RuntimeError(
+                f"stale kernel build: {so} predates modified sources "
+                f"{stale}; remove {BUILD_DIR} and rebuild")
     return mod
 
 
