@@ -21,6 +21,10 @@ def main() -> None:
     ctx = Context.in_process(Runtime())
     http = ctx._session  # TestClient-compatible session
 
+    # idempotent re-runs: clear any previous artifacts of the same names
+    for verb, name in (("train", "ddp_fit"), ("model", "ddp_cnn")):
+        http.delete(f"{PREFIX}/{verb}/torch/{name}")
+
     r = http.post(f"{PREFIX}/model/torch", json={
         "modelName": "ddp_cnn",
         "modulePath": "learningorchestra_amd.models.zoo",
