@@ -106,15 +106,34 @@ def main() -> None:
         x, y = imagenet_batch(batch, device=device, dtype=torch.bfloat16,
                               seed=1234 + rank)
 
+    # LO_BENCH_TRACE=1: per-50-step |w|/|g| maxima on stderr — diagnostic
+    # for the rare non-finite loss seen in 300-step soaks (PERFORMANCE.md
+    # "Open issue"); prints outside the timed region only when enabled
+    trace = os.environ.get("LO_BENCH_TRACE") == "1" and rank == 0
+
+    def _trace(tag):
+        if trace:
+            import sys
+            if use_gpu:
+                torch.cuda.synchronize()
+            a = model.arena
+            print(f"[trace {tag}] |w|max {float(a.master.abs().max()):.4e} "
+                  f"|g|max {float(a.grad.abs().max()):.4e} "
+                  f"loss_sum {float(model.loss_sum):.4e}", file=sys.stderr,
+                  flush=True)
+
     for _ in range(args.warmup):
         trainer.step_async(x, y)
+    _trace("warmup")
 
     barrier()
     if use_gpu:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
+    for i in range(args.steps):
         trainer.step_async(x, y)
+        if trace and i % 50 == 49:
+            _trace(f"step{i + 1}")
     if use_gpu:
         torch.cuda.synchronize()
     barrier()
