@@ -234,3 +234,52 @@ def test_wal_autoflush_bounds_wal(tmp_path, monkeypatch):
     store.flush()
     reopened = DocumentStore(str(tmp_path))
     assert reopened["events"].count_documents({}) == 120
+
+
+def test_docstore_concurrent_writers_and_flushes(tmp_path):
+    """8 threads hammer inserts/updates across 4 collections while another
+    thread flushes continuously: no exceptions, no lost writes, and a
+    reopen (snapshot + WAL replay) sees every document."""
+    import threading
+
+    from learningorchestra_amd.storage.docstore import DocumentStore
+
+    store = DocumentStore(str(tmp_path))
+    errors = []
+    N_PER = 200
+
+    def writer(t):
+        try:
+            col = store[f"c{t % 4}"]
+            for i in range(N_PER):
+                col.insert_one({"_id": f"w{t}_{i}", "v": i})
+                if i % 50 == 0:
+                    col.update_one({"_id": f"w{t}_{i}"},
+                                   {"$set": {"touched": True}})
+        except Exception as exc:  # noqa: BLE001
+            errors.append(repr(exc))
+
+    stop = threading.Event()
+
+    def flusher():
+        while not stop.is_set():
+            try:
+                store.flush()
+            except Exception as exc:  # noqa: BLE001
+                errors.append(repr(exc))
+                return
+
+    threads = [threading.Thread(target=writer, args=(t,)) for t in range(8)]
+    fl = threading.Thread(target=flusher)
+    fl.start()
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=120)
+    stop.set()
+    fl.join(timeout=30)
+    assert not errors, errors[:3]
+    store.flush()
+    reopened = DocumentStore(str(tmp_path))
+    total = sum(reopened[f"c{k}"].count_documents({}) for k in range(4))
+    assert total == 8 * N_PER, total
