@@ -601,8 +601,9 @@ def test_cancel_running_multi_rank_train(client):
                                    ".randint(0,10,4096)",
                               "epochs": 500, "batch_size": 64}})
     assert r.status_code == 201
-    # let the torchrun tree actually start
-    deadline = time.time() + 60
+    # let the torchrun tree actually start (generous: a loaded CI host can
+    # take a while to fork + import torch in the children)
+    deadline = time.time() + 120
     while time.time() < deadline:
         job = client.rt.scheduler.job("ddpcancel")
         if job is not None and job.proc is not None \
