@@ -659,9 +659,18 @@ def test_chaos_concurrent_trains_cancels_observers(client):
 
     def observer(name):
         try:
-            r = client.get(f"{PREFIX}/observe/{name}/wait",
-                           params={"timeoutSeconds": 60})
-            assert r.status_code == 200, r.status_code
+            # observers launch concurrently with the train POSTs: a 404 just
+            # means the metadata doc does not exist YET (correct API
+            # behavior) — retry until the create lands
+            deadline = time.time() + 60
+            while True:
+                r = client.get(f"{PREFIX}/observe/{name}/wait",
+                               params={"timeoutSeconds": 30})
+                if r.status_code == 200:
+                    break
+                assert r.status_code == 404, r.status_code
+                assert time.time() < deadline, "doc never created"
+                time.sleep(0.1)
         except Exception as exc:  # noqa: BLE001
             errors.append(("observe", name, repr(exc)))
 
